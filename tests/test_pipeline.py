@@ -1,0 +1,111 @@
+"""End-to-end pipeline tests on a synthetic assembly (BASELINE.json config 1:
+10 kb-scale synthetic draft + synthetic BAM, CPU plumbing only).
+
+The strongest consistency check needs no trained model: voting the TRUTH
+LABELS through the stitcher must reconstruct the truth sequence over the
+covered span — this exercises features (C++ pileup), label join, vote and
+stitch together.
+"""
+
+import os
+
+import numpy as np
+import pytest
+import torch
+
+from roko_amd import config as C
+from roko_amd import features as F
+from roko_amd.inference import accumulate_votes, merge_votes, stitch_contig
+from roko_amd.rkdata import RkwFile
+
+
+def small_cfg(seed=0):
+    return F.FeatureConfig(seed=seed, region_size=2000, region_overlap=300)
+
+
+def test_features_infer_end_to_end(tiny_assembly, tmp_path):
+    out = str(tmp_path / "infer.rkw")
+    n = F.run(tiny_assembly["draft_fasta"], tiny_assembly["reads_bam"], out,
+              workers=1, cfg=small_cfg(), log=lambda *a: None)
+    assert n > 50
+    f = RkwFile(out)
+    assert f.inference
+    assert f.contig_names() == ["ctg1"]
+    contig, pos, ex, lab = f.window(0)
+    assert ex.shape == (C.WINDOW_ROWS, C.WINDOW_COLS)
+    assert lab is None
+    assert ex.max() <= 11
+
+
+def test_features_train_end_to_end(tiny_assembly, tmp_path):
+    out = str(tmp_path / "train.rkw")
+    n = F.run(tiny_assembly["draft_fasta"], tiny_assembly["reads_bam"], out,
+              bam_y=tiny_assembly["truth_bam"], workers=1, cfg=small_cfg(),
+              log=lambda *a: None)
+    assert n > 50
+    f = RkwFile(out)
+    assert not f.inference
+    _, pos, ex, lab = f.group_arrays(0)
+    assert lab is not None and lab.shape[1] == C.WINDOW_COLS
+    assert lab.max() < C.NUM_CLASSES  # UNKNOWN-labelled windows dropped
+
+
+def test_features_multiprocess_matches_serial(tiny_assembly, tmp_path):
+    a, b = str(tmp_path / "a.rkw"), str(tmp_path / "b.rkw")
+    F.run(tiny_assembly["draft_fasta"], tiny_assembly["reads_bam"], a,
+          workers=1, cfg=small_cfg(), log=lambda *a: None)
+    F.run(tiny_assembly["draft_fasta"], tiny_assembly["reads_bam"], b,
+          workers=3, cfg=small_cfg(), log=lambda *a: None)
+    fa, fb = RkwFile(a), RkwFile(b)
+    assert fa.num_windows == fb.num_windows
+    for gi in range(len(fa.groups)):
+        _, pa, ea, _ = fa.group_arrays(gi)
+        _, pb, eb, _ = fb.group_arrays(gi)
+        assert np.array_equal(pa, pb)
+        assert np.array_equal(ea, eb)
+
+
+def test_label_votes_reconstruct_truth(tiny_assembly, tmp_path):
+    """Vote the labels themselves: stitched consensus == truth mid-section."""
+    out = str(tmp_path / "train.rkw")
+    F.run(tiny_assembly["draft_fasta"], tiny_assembly["reads_bam"], out,
+          bam_y=tiny_assembly["truth_bam"], workers=1, cfg=small_cfg(),
+          log=lambda *a: None)
+    f = RkwFile(out)
+    tables = []
+    for gi in range(len(f.groups)):
+        _, pos, _, lab = f.group_arrays(gi)
+        tables.append(accumulate_votes(np.asarray(pos), np.asarray(lab)))
+    keys, counts = merge_votes(tables)
+    draft = f.contig_seq("ctg1")
+    polished = stitch_contig(draft, keys, counts)
+    truth = tiny_assembly["truth"]
+    # the central chunk of the truth must appear verbatim in the polish
+    assert truth[500:2500] in polished
+
+
+def test_vote_stitch_units():
+    # two windows voting on 4 columns; window B overrides A at pos 1 by 2:1
+    pos = np.array([
+        [[0, 0], [1, 0], [1, 1], [2, 0]],
+        [[1, 0], [1, 1], [2, 0], [3, 0]],
+        [[1, 0], [1, 1], [2, 0], [3, 0]],
+    ], dtype=np.int32)
+    # A C G T = 0 1 2 3, GAP = 4
+    preds = np.array([
+        [0, 0, 4, 2],
+        [3, 4, 2, 1],
+        [3, 4, 2, 1],
+    ], dtype=np.uint8)
+    keys, counts = accumulate_votes(pos, preds)
+    draft = "AAAAAA"
+    out = stitch_contig(draft, keys, counts)
+    # majority: pos0=A, pos1=T (2 votes), (1,1)=GAP skip, pos2=G, pos3=C
+    # prefix draft[:0]="", suffix draft[4:]="AA"
+    assert out == "ATGC" + "AA"
+
+
+def test_stitch_empty_votes_returns_draft():
+    keys = np.empty(0, dtype=np.int64)
+    counts = np.empty((0, C.NUM_CLASSES), dtype=np.int64)
+    assert stitch_contig("ACGT", keys, counts) == "ACGT"
