@@ -1,0 +1,170 @@
+"""Benchmark harness (driver contract — see project brief).
+
+Measures the flagship inference step on N GPUs of one node: batches of 128
+sampled-read windows (200x90, synthetic, random-init weights) through the
+hand-written gfx950 kernel path, fused argmax, predictions copied to host —
+i.e. the per-window work of the polishing pipeline (BASELINE.json metric:
+inference bases/sec at b=128; bases = windows * 30-column stride).
+
+  python bench.py --gpus N --steps K --warmup W [--mode inference|train]
+
+Under torchrun (one rank per GPU, RCCL) ranks run independent streams of
+batches (weak scaling; inference needs no collectives — SURVEY.md §2.5) and
+the max elapsed over ranks is used.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+from roko_amd import config as C
+from roko_amd.model import RokoModel
+from roko_amd.parallel.ddp import init_distributed
+
+
+def bench_inference(args, rank, world, device):
+    torch.manual_seed(0)
+    model = RokoModel().to(device).eval()
+    if device.type == "cuda":
+        from roko_amd import ops
+        ops.require()
+        from roko_amd.ops.forward import roko_argmax
+        predict = lambda x: roko_argmax(model, x)
+    else:
+        predict = lambda x: model(x.long()).argmax(dim=2)
+
+    g = torch.Generator().manual_seed(1234 + rank)
+    nbuf = 4
+    xs = [
+        torch.randint(0, C.NUM_BASE_IDS, (args.batch, C.WINDOW_ROWS, C.WINDOW_COLS),
+                      generator=g, dtype=torch.uint8).to(device)
+        for _ in range(nbuf)
+    ]
+
+    def step(i):
+        preds = predict(xs[i % nbuf])
+        return preds.to("cpu", non_blocking=False)
+
+    for i in range(args.warmup):
+        step(i)
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    if world > 1:
+        torch.distributed.barrier()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        step(i)
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    if world > 1:
+        torch.distributed.barrier()
+    elapsed = time.perf_counter() - t0
+
+    if world > 1:
+        t = torch.tensor([elapsed], device=device if device.type == "cuda" else "cpu")
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    bases = world * args.steps * args.batch * C.WINDOW_STRIDE
+    return {
+        "metric": "inference_bases_per_sec",
+        "value": bases / elapsed,
+        "unit": "bases/s",
+        "ms_per_step": elapsed / args.steps * 1000.0,
+    }
+
+
+def bench_train(args, rank, world, device):
+    from roko_amd.ops.train import fused_train_step, train_step_available
+
+    if device.type != "cuda" or not train_step_available():
+        raise SystemExit("train bench requires the fused HIP train step on GPU")
+    torch.manual_seed(0)
+    model = RokoModel().to(device).train()
+    g = torch.Generator().manual_seed(99 + rank)
+    x = torch.randint(0, C.NUM_BASE_IDS, (args.batch, C.WINDOW_ROWS, C.WINDOW_COLS),
+                      generator=g, dtype=torch.uint8).to(device)
+    y = torch.randint(0, C.NUM_CLASSES, (args.batch, C.WINDOW_COLS),
+                      generator=g).to(device)
+
+    for _ in range(args.warmup):
+        fused_train_step(model, x, y)
+    torch.cuda.synchronize()
+    if world > 1:
+        torch.distributed.barrier()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        fused_train_step(model, x, y)
+    torch.cuda.synchronize()
+    if world > 1:
+        torch.distributed.barrier()
+    elapsed = time.perf_counter() - t0
+    if world > 1:
+        t = torch.tensor([elapsed], device=device)
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(t.item())
+    windows = world * args.steps * args.batch
+    return {
+        "metric": "train_windows_per_sec",
+        "value": windows / elapsed,
+        "unit": "windows/s",
+        "ms_per_step": elapsed / args.steps * 1000.0,
+    }
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=50)
+    p.add_argument("--warmup", type=int, default=10)
+    p.add_argument("--batch", type=int, default=C.BATCH_SIZE)
+    p.add_argument("--mode", choices=["inference", "train"], default="inference")
+    args = p.parse_args()
+
+    rank, local_rank, world = init_distributed()
+    device = (
+        torch.device("cuda", local_rank)
+        if torch.cuda.is_available()
+        else torch.device("cpu")
+    )
+
+    fn = bench_inference if args.mode == "inference" else bench_train
+    res = fn(args, rank, world, device)
+
+    if rank == 0:
+        out = {
+            "metric": res["metric"],
+            "value": res["value"],
+            "unit": res["unit"],
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": res["ms_per_step"],
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if device.type == "cuda" else "float32",
+            "data": "synthetic",
+            "config": {
+                "model": "roko bi-GRU polisher (r10-shape: 200x90 windows, "
+                         "emb50, fc 200->100->10, GRU 500/128x3 bidir, 5-class)",
+                "global_batch": args.batch * world,
+                "seq_len": C.WINDOW_COLS,
+                "parallelism": f"dp{world}",
+                "mode": args.mode,
+                "bases_per_window": C.WINDOW_STRIDE,
+            },
+        }
+        print(json.dumps(out))
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,118 @@
+"""Full-model GPU forward through the hand-written gfx950 kernels.
+
+Orchestration: the fused embed+MLP kernel produces the GRU
+input sequence; per layer, the input projections of BOTH directions and all
+90 steps are ONE hipBLASLt GEMM (plain GEMM -> library; the brief's rule) and
+the recurrence is one persistent kernel launch; the 5-class head is a fused
+GEMV(+argmax) kernel. No per-step launches, no MIOpen RNN, no eager fallback.
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+
+from .. import config as C
+
+
+def _ext():
+    from . import ext
+
+    return ext()
+
+
+def _bf16_weights(model) -> dict:
+    """Per-model cache of bf16/fp32 kernel-ready weight tensors."""
+    params = [
+        model.embedding.weight, model.fc1.weight, model.fc1.bias,
+        model.fc2.weight, model.fc2.bias, model.fc4.weight, model.fc4.bias,
+    ]
+    for l in range(C.NUM_LAYERS):
+        for suffix in ("", "_reverse"):
+            for kind in ("weight_ih", "weight_hh", "bias_ih", "bias_hh"):
+                params.append(getattr(model.gru, f"{kind}_l{l}{suffix}"))
+    ver = tuple(p._version for p in params) + (params[0].device,)
+    cache = getattr(model, "_hip_weight_cache", None)
+    if cache is not None and cache["ver"] == ver:
+        return cache
+
+    g = model.gru
+    c = {"ver": ver}
+    c["w1"] = model.fc1.weight.detach().to(torch.bfloat16).contiguous()
+    c["b1"] = model.fc1.bias.detach().float().contiguous()
+    c["w2"] = model.fc2.weight.detach().to(torch.bfloat16).contiguous()
+    c["b2"] = model.fc2.bias.detach().float().contiguous()
+    c["emb"] = model.embedding.weight.detach().to(torch.bfloat16).contiguous()
+    c["w4"] = model.fc4.weight.detach().to(torch.bfloat16).contiguous()
+    c["b4"] = model.fc4.bias.detach().float().contiguous()
+    for l in range(C.NUM_LAYERS):
+        wf = getattr(g, f"weight_ih_l{l}").detach()
+        wr = getattr(g, f"weight_ih_l{l}_reverse").detach()
+        bf = getattr(g, f"bias_ih_l{l}").detach()
+        br = getattr(g, f"bias_ih_l{l}_reverse").detach()
+        # (in, 768) so xg = x @ w_ih_t is one GEMM covering both directions
+        c[f"w_ih_t{l}"] = (
+            torch.cat([wf, wr], dim=0).to(torch.bfloat16).t().contiguous()
+        )
+        c[f"b_ih{l}"] = torch.cat([bf, br]).to(torch.bfloat16).contiguous()
+        c[f"u{l}"] = torch.stack(
+            [getattr(g, f"weight_hh_l{l}").detach(),
+             getattr(g, f"weight_hh_l{l}_reverse").detach()]
+        ).to(torch.bfloat16).contiguous()
+        c[f"bhh{l}"] = torch.stack(
+            [getattr(g, f"bias_hh_l{l}").detach(),
+             getattr(g, f"bias_hh_l{l}_reverse").detach()]
+        ).float().contiguous()
+    model._hip_weight_cache = c
+    return c
+
+
+def roko_forward(model, x: torch.Tensor) -> torch.Tensor:
+    """ids (B, 200, 90) int/uint8 on GPU -> logits (B, 90, 5) fp32."""
+    if model.training:
+        raise RuntimeError(
+            "HIP training path not wired into model.forward; use the fused "
+            "train step (roko_amd.ops.train_step) or model.eval()"
+        )
+    ext = _ext()
+    w = _bf16_weights(model)
+    B0 = x.shape[0]
+    ids = x.to(torch.uint8).contiguous()
+    pad = (-B0) % 32
+    if pad:
+        ids = torch.cat([ids, ids.new_zeros(pad, *ids.shape[1:])])
+    B = ids.shape[0]
+    T = C.WINDOW_COLS
+
+    seq = ext.embed_mlp_fwd(ids, w["w1"], w["b1"], w["w2"], w["b2"], w["emb"])
+    for l in range(C.NUM_LAYERS):
+        xg = torch.addmm(
+            w[f"b_ih{l}"], seq.reshape(T * B, -1), w[f"w_ih_t{l}"]
+        ).view(T, B, 2, 384)
+        hseq = ext.gru_layer_fwd(xg.contiguous(), w[f"u{l}"], w[f"bhh{l}"])
+        seq = hseq.view(T, B, 2 * C.HIDDEN_SIZE)
+    (logits,) = ext.head_fwd(seq, w["w4"], w["b4"], True, False)
+    return logits[:B0]
+
+
+def roko_argmax(model, x: torch.Tensor) -> torch.Tensor:
+    """ids -> per-position class predictions (B, 90) uint8, argmax fused."""
+    ext = _ext()
+    w = _bf16_weights(model)
+    B0 = x.shape[0]
+    ids = x.to(torch.uint8).contiguous()
+    pad = (-B0) % 32
+    if pad:
+        ids = torch.cat([ids, ids.new_zeros(pad, *ids.shape[1:])])
+    B = ids.shape[0]
+    T = C.WINDOW_COLS
+    seq = ext.embed_mlp_fwd(ids, w["w1"], w["b1"], w["w2"], w["b2"], w["emb"])
+    for l in range(C.NUM_LAYERS):
+        xg = torch.addmm(
+            w[f"b_ih{l}"], seq.reshape(T * B, -1), w[f"w_ih_t{l}"]
+        ).view(T, B, 2, 384)
+        hseq = ext.gru_layer_fwd(xg.contiguous(), w[f"u{l}"], w[f"bhh{l}"])
+        seq = hseq.view(T, B, 2 * C.HIDDEN_SIZE)
+    (amax,) = ext.head_fwd(seq, w["w4"], w["b4"], False, True)
+    return amax[:B0]

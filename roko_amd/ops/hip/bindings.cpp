@@ -1,0 +1,120 @@
+// Torch bindings for the gfx950 kernels (roko_amd.ops._hip_ops).
+
+#include <ATen/cuda/CUDAContext.h>
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+
+#include <cstdint>
+
+namespace rk {
+void mfma_probe(const float* a, const float* b, float* d, hipStream_t stream);
+void embed_mlp_fwd(const uint8_t* ids, const void* w1, const float* b1,
+                   const void* w2, const float* b2, const void* emb, void* out,
+                   int B, hipStream_t stream);
+void gru_layer_fwd(const void* xg, const void* u, const float* bhh, void* hseq,
+                   int T, int B, hipStream_t stream);
+void head_fwd(const void* hseq, const void* w4, const float* b4, float* logits,
+              uint8_t* amax, int T, int B, hipStream_t stream);
+}  // namespace rk
+
+namespace {
+
+hipStream_t cur_stream() {
+    return at::cuda::getCurrentCUDAStream().stream();
+}
+
+void check(const torch::Tensor& t, torch::ScalarType dt, const char* name) {
+    TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+    TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+    TORCH_CHECK(t.scalar_type() == dt, name, " has wrong dtype");
+}
+
+torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor b) {
+    check(a, torch::kFloat32, "a");
+    check(b, torch::kFloat32, "b");
+    TORCH_CHECK(a.sizes() == torch::IntArrayRef({16, 32}), "a must be (16,32)");
+    TORCH_CHECK(b.sizes() == torch::IntArrayRef({32, 16}), "b must be (32,16)");
+    auto d = torch::empty({16, 16}, a.options());
+    rk::mfma_probe(a.data_ptr<float>(), b.data_ptr<float>(), d.data_ptr<float>(),
+                   cur_stream());
+    return d;
+}
+
+// ids (B, 200, 90) u8 -> (90, B, 500) bf16
+torch::Tensor embed_mlp_fwd(torch::Tensor ids, torch::Tensor w1, torch::Tensor b1,
+                            torch::Tensor w2, torch::Tensor b2, torch::Tensor emb) {
+    check(ids, torch::kUInt8, "ids");
+    check(w1, torch::kBFloat16, "w1");
+    check(b1, torch::kFloat32, "b1");
+    check(w2, torch::kBFloat16, "w2");
+    check(b2, torch::kFloat32, "b2");
+    check(emb, torch::kBFloat16, "emb");
+    const int B = ids.size(0);
+    TORCH_CHECK(ids.size(1) == 200 && ids.size(2) == 90, "ids must be (B,200,90)");
+    TORCH_CHECK(w1.size(0) == 100 && w1.size(1) == 200, "w1 must be (100,200)");
+    TORCH_CHECK(w2.size(0) == 10 && w2.size(1) == 100, "w2 must be (10,100)");
+    TORCH_CHECK(emb.size(0) == 12 && emb.size(1) == 50, "emb must be (12,50)");
+    auto out = torch::empty({90, B, 500}, ids.options().dtype(torch::kBFloat16));
+    rk::embed_mlp_fwd(ids.data_ptr<uint8_t>(), w1.data_ptr(), b1.data_ptr<float>(),
+                      w2.data_ptr(), b2.data_ptr<float>(), emb.data_ptr(),
+                      out.data_ptr(), B, cur_stream());
+    return out;
+}
+
+// xg (T, B, 2, 384) bf16, u (2, 384, 128) bf16, bhh (2, 384) f32
+//   -> hseq (T, B, 2, 128) bf16
+torch::Tensor gru_layer_fwd(torch::Tensor xg, torch::Tensor u, torch::Tensor bhh) {
+    check(xg, torch::kBFloat16, "xg");
+    check(u, torch::kBFloat16, "u");
+    check(bhh, torch::kFloat32, "bhh");
+    const int T = xg.size(0), B = xg.size(1);
+    TORCH_CHECK(xg.size(2) == 2 && xg.size(3) == 384, "xg must be (T,B,2,384)");
+    TORCH_CHECK(u.size(0) == 2 && u.size(1) == 384 && u.size(2) == 128,
+                "u must be (2,384,128)");
+    TORCH_CHECK(bhh.size(0) == 2 && bhh.size(1) == 384, "bhh must be (2,384)");
+    TORCH_CHECK(B % 32 == 0, "batch must be a multiple of 32 (pad on host)");
+    auto hseq = torch::empty({T, B, 2, 128}, xg.options());
+    rk::gru_layer_fwd(xg.data_ptr(), u.data_ptr(), bhh.data_ptr<float>(),
+                      hseq.data_ptr(), T, B, cur_stream());
+    return hseq;
+}
+
+// hseq (T, B, 256) bf16 -> (logits (B,T,5) f32, argmax (B,T) u8) per flags
+std::vector<torch::Tensor> head_fwd(torch::Tensor hseq, torch::Tensor w4,
+                                    torch::Tensor b4, bool want_logits,
+                                    bool want_argmax) {
+    check(hseq, torch::kBFloat16, "hseq");
+    check(w4, torch::kBFloat16, "w4");
+    check(b4, torch::kFloat32, "b4");
+    const int T = hseq.size(0), B = hseq.size(1);
+    TORCH_CHECK(hseq.size(2) == 256, "hseq must be (T,B,256)");
+    TORCH_CHECK(w4.size(0) == 5 && w4.size(1) == 256, "w4 must be (5,256)");
+    torch::Tensor logits, amax;
+    float* lp = nullptr;
+    uint8_t* ap = nullptr;
+    if (want_logits) {
+        logits = torch::empty({B, T, 5}, hseq.options().dtype(torch::kFloat32));
+        lp = logits.data_ptr<float>();
+    }
+    if (want_argmax) {
+        amax = torch::empty({B, T}, hseq.options().dtype(torch::kUInt8));
+        ap = amax.data_ptr<uint8_t>();
+    }
+    rk::head_fwd(hseq.data_ptr(), w4.data_ptr(), b4.data_ptr<float>(), lp, ap,
+                 T, B, cur_stream());
+    std::vector<torch::Tensor> out;
+    if (want_logits) out.push_back(logits);
+    if (want_argmax) out.push_back(amax);
+    return out;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.doc() = "roko-mi355x CDNA4 kernels (gfx950)";
+    m.def("mfma_probe", &mfma_probe);
+    m.def("embed_mlp_fwd", &embed_mlp_fwd);
+    m.def("gru_layer_fwd", &gru_layer_fwd);
+    m.def("head_fwd", &head_fwd, py::arg("hseq"), py::arg("w4"), py::arg("b4"),
+          py::arg("want_logits") = true, py::arg("want_argmax") = false);
+}

@@ -1,0 +1,70 @@
+// Shared helpers for the CDNA4 (gfx950) kernels. HIP-only — no CUDA compat.
+//
+// MFMA note: all matrix math uses v_mfma_f32_16x16x32_bf16 (gfx950's 2xK
+// bf16 shape, cdna_hip_programming.md §3). Fragment mappings used here:
+//   A (16x32):  lane l holds A[row = l & 15][k = 8*(l >> 4) + j], j = 0..7
+//   B (32x16):  lane l holds B[k = 8*(l >> 4) + j][col = l & 15]
+//   C/D (16x16): lane l, reg i -> D[row = (l >> 4)*4 + i][col = l & 15]
+// The mappings are verified on-device by the mfma_probe kernel (test_gpu.py)
+// before anything else trusts them.
+
+#pragma once
+
+#include <hip/hip_bf16.h>
+#include <hip/hip_runtime.h>
+
+#define RK_DEV __device__ __forceinline__
+
+namespace rk {
+
+using bf16 = __hip_bfloat16;
+using f32x4 = __attribute__((__vector_size__(4 * sizeof(float)))) float;
+using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
+using short4_t = __attribute__((ext_vector_type(4))) short;
+
+RK_DEV float bf2f(bf16 v) { return __bfloat162float(v); }
+RK_DEV bf16 f2bf(float v) { return __float2bfloat16(v); }
+
+// MFMA wrapper: D = A(16x32) * B(32x16) + C, bf16 in / f32 accumulate.
+RK_DEV f32x4 mfma16x16x32(bf16x8 a, bf16x8 b, f32x4 c) {
+    return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+}
+
+// Load an A-fragment from an LDS tile of bf16 with row stride `ld` elements:
+// rows [row0, row0+16), k-columns [k0, k0+32).
+RK_DEV bf16x8 lds_load_a_frag(const bf16* tile, int row0, int k0, int ld) {
+    const int lane = threadIdx.x & 63;
+    const int row = row0 + (lane & 15);
+    const int k = k0 + 8 * (lane >> 4);
+    return *reinterpret_cast<const bf16x8*>(tile + row * ld + k);
+}
+
+// Load a B-fragment B[k][col] from an LDS tile stored K-major (i.e. tile
+// layout [col][k], row stride ld elements) — cols [col0, col0+16),
+// k in [k0, k0+32). Reading B transposed-stored keeps lanes on 16B chunks.
+RK_DEV bf16x8 lds_load_b_frag_t(const bf16* tile_t, int col0, int k0, int ld) {
+    const int lane = threadIdx.x & 63;
+    const int col = col0 + (lane & 15);
+    const int k = k0 + 8 * (lane >> 4);
+    return *reinterpret_cast<const bf16x8*>(tile_t + col * ld + k);
+}
+
+RK_DEV float sigmoidf_dev(float x) { return 1.0f / (1.0f + __expf(-x)); }
+
+// overflow-stable tanh: tanh(x) = sign(x) * (1 - e) / (1 + e), e = exp(-2|x|)
+RK_DEV float tanhf_dev(float x) {
+    float e = __expf(-2.0f * fabsf(x));
+    float t = (1.0f - e) / (1.0f + e);
+    return copysignf(t, x);
+}
+
+#define HIP_CHECK_LAST()                                                     \
+    do {                                                                     \
+        hipError_t e_ = hipGetLastError();                                   \
+        if (e_ != hipSuccess) {                                              \
+            throw std::runtime_error(std::string("HIP kernel launch: ") +    \
+                                     hipGetErrorString(e_));                 \
+        }                                                                    \
+    } while (0)
+
+}  // namespace rk
