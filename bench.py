@@ -83,12 +83,13 @@ def bench_inference(args, rank, world, device):
 
 
 def bench_train(args, rank, world, device):
-    from roko_amd.ops.train import fused_train_step, train_step_available
+    from roko_amd.ops.train import FusedAdam, fused_train_step, train_step_available
 
     if device.type != "cuda" or not train_step_available():
         raise SystemExit("train bench requires the fused HIP train step on GPU")
     torch.manual_seed(0)
     model = RokoModel().to(device).train()
+    opt = FusedAdam(list(model.parameters()), lr=C.LR)
     g = torch.Generator().manual_seed(99 + rank)
     x = torch.randint(0, C.NUM_BASE_IDS, (args.batch, C.WINDOW_ROWS, C.WINDOW_COLS),
                       generator=g, dtype=torch.uint8).to(device)
@@ -96,13 +97,13 @@ def bench_train(args, rank, world, device):
                       generator=g).to(device)
 
     for _ in range(args.warmup):
-        fused_train_step(model, x, y)
+        fused_train_step(model, x, y, opt)
     torch.cuda.synchronize()
     if world > 1:
         torch.distributed.barrier()
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        fused_train_step(model, x, y)
+        fused_train_step(model, x, y, opt)
     torch.cuda.synchronize()
     if world > 1:
         torch.distributed.barrier()

@@ -51,7 +51,13 @@ def ext():
 
 
 def model_forward(model, x: torch.Tensor) -> torch.Tensor:
-    """Full-model forward through the HIP kernels (inference/eval path)."""
+    """Full-model forward through the HIP kernels. Eval mode runs the fully
+    fused inference path; train mode runs the differentiable path (custom
+    GRU autograd + torch MLP front)."""
+    if model.training or torch.is_grad_enabled():
+        from .train import train_forward
+
+        return train_forward(model, x)
     from .forward import roko_forward
 
     return roko_forward(model, x)

@@ -70,11 +70,6 @@ def _bf16_weights(model) -> dict:
 
 def roko_forward(model, x: torch.Tensor) -> torch.Tensor:
     """ids (B, 200, 90) int/uint8 on GPU -> logits (B, 90, 5) fp32."""
-    if model.training:
-        raise RuntimeError(
-            "HIP training path not wired into model.forward; use the fused "
-            "train step (roko_amd.ops.train_step) or model.eval()"
-        )
     ext = _ext()
     w = _bf16_weights(model)
     B0 = x.shape[0]
@@ -90,7 +85,7 @@ def roko_forward(model, x: torch.Tensor) -> torch.Tensor:
         xg = torch.addmm(
             w[f"b_ih{l}"], seq.reshape(T * B, -1), w[f"w_ih_t{l}"]
         ).view(T, B, 2, 384)
-        hseq = ext.gru_layer_fwd(xg.contiguous(), w[f"u{l}"], w[f"bhh{l}"])
+        (hseq,) = ext.gru_layer_fwd(xg.contiguous(), w[f"u{l}"], w[f"bhh{l}"], False)
         seq = hseq.view(T, B, 2 * C.HIDDEN_SIZE)
     (logits,) = ext.head_fwd(seq, w["w4"], w["b4"], True, False)
     return logits[:B0]
@@ -112,7 +107,7 @@ def roko_argmax(model, x: torch.Tensor) -> torch.Tensor:
         xg = torch.addmm(
             w[f"b_ih{l}"], seq.reshape(T * B, -1), w[f"w_ih_t{l}"]
         ).view(T, B, 2, 384)
-        hseq = ext.gru_layer_fwd(xg.contiguous(), w[f"u{l}"], w[f"bhh{l}"])
+        (hseq,) = ext.gru_layer_fwd(xg.contiguous(), w[f"u{l}"], w[f"bhh{l}"], False)
         seq = hseq.view(T, B, 2 * C.HIDDEN_SIZE)
     (amax,) = ext.head_fwd(seq, w["w4"], w["b4"], False, True)
     return amax[:B0]

@@ -1,0 +1,41 @@
+// Fused Adam over ONE flat parameter buffer (reference op: torch Adam over
+// ~1.1 M params, train.py:39 — SURVEY.md §2.4 K7). The optimizer keeps all
+// parameters, gradients and moments in single flat fp32 tensors (the model's
+// param tensors are views), so the whole update is one elementwise kernel —
+// no per-tensor launch storm, one RCCL all-reduce on the flat grad in DP.
+
+#include <cstdint>
+
+#include "common.h"
+
+namespace rk {
+
+__global__ __launch_bounds__(256) void adam_step_kernel(
+    float* __restrict__ p, const float* __restrict__ g, float* __restrict__ m,
+    float* __restrict__ v, int64_t n, float lr, float beta1, float beta2,
+    float eps, float bc1, float bc2) {
+    const int64_t stride = (int64_t)gridDim.x * 256;
+    for (int64_t i = (int64_t)blockIdx.x * 256 + threadIdx.x; i < n; i += stride) {
+        const float gi = g[i];
+        const float mi = beta1 * m[i] + (1.0f - beta1) * gi;
+        const float vi = beta2 * v[i] + (1.0f - beta2) * gi * gi;
+        m[i] = mi;
+        v[i] = vi;
+        const float mhat = mi / bc1;
+        const float vhat = vi / bc2;
+        p[i] -= lr * mhat / (sqrtf(vhat) + eps);
+    }
+}
+
+void adam_step(float* p, const float* g, float* m, float* v, int64_t n,
+               float lr, float beta1, float beta2, float eps, int step,
+               hipStream_t stream) {
+    const float bc1 = 1.0f - powf(beta1, float(step));
+    const float bc2 = 1.0f - powf(beta2, float(step));
+    int blocks = int((n + 255) / 256);
+    if (blocks > 2048) blocks = 2048;
+    hipLaunchKernelGGL(adam_step_kernel, dim3(blocks), dim3(256), 0, stream, p,
+                       g, m, v, n, lr, beta1, beta2, eps, bc1, bc2);
+}
+
+}  // namespace rk

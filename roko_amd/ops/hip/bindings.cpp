@@ -12,7 +12,14 @@ void embed_mlp_fwd(const uint8_t* ids, const void* w1, const float* b1,
                    const void* w2, const float* b2, const void* emb, void* out,
                    int B, hipStream_t stream);
 void gru_layer_fwd(const void* xg, const void* u, const float* bhh, void* hseq,
-                   int T, int B, hipStream_t stream);
+                   void* cache, int T, int B, hipStream_t stream);
+void gru_layer_bwd(const void* cache, const void* hseq, const void* dhin,
+                   const void* ut, void* dg, int T, int B, hipStream_t stream);
+void ce_fwd_bwd(const float* logits, const int64_t* target, float* dlogits,
+                float* loss_sum, int64_t n, hipStream_t stream);
+void adam_step(float* p, const float* g, float* m, float* v, int64_t n,
+               float lr, float beta1, float beta2, float eps, int step,
+               hipStream_t stream);
 void head_fwd(const void* hseq, const void* w4, const float* b4, float* logits,
               uint8_t* amax, int T, int B, hipStream_t stream);
 }  // namespace rk
@@ -62,8 +69,9 @@ torch::Tensor embed_mlp_fwd(torch::Tensor ids, torch::Tensor w1, torch::Tensor b
 }
 
 // xg (T, B, 2, 384) bf16, u (2, 384, 128) bf16, bhh (2, 384) f32
-//   -> hseq (T, B, 2, 128) bf16
-torch::Tensor gru_layer_fwd(torch::Tensor xg, torch::Tensor u, torch::Tensor bhh) {
+//   -> (hseq (T, B, 2, 128) bf16 [, cache (T, B, 2, 512) bf16 when train])
+std::vector<torch::Tensor> gru_layer_fwd(torch::Tensor xg, torch::Tensor u,
+                                         torch::Tensor bhh, bool train) {
     check(xg, torch::kBFloat16, "xg");
     check(u, torch::kBFloat16, "u");
     check(bhh, torch::kFloat32, "bhh");
@@ -74,9 +82,67 @@ torch::Tensor gru_layer_fwd(torch::Tensor xg, torch::Tensor u, torch::Tensor bhh
     TORCH_CHECK(bhh.size(0) == 2 && bhh.size(1) == 384, "bhh must be (2,384)");
     TORCH_CHECK(B % 32 == 0, "batch must be a multiple of 32 (pad on host)");
     auto hseq = torch::empty({T, B, 2, 128}, xg.options());
+    torch::Tensor cache;
+    void* cp = nullptr;
+    if (train) {
+        cache = torch::empty({T, B, 2, 512}, xg.options());
+        cp = cache.data_ptr();
+    }
     rk::gru_layer_fwd(xg.data_ptr(), u.data_ptr(), bhh.data_ptr<float>(),
-                      hseq.data_ptr(), T, B, cur_stream());
-    return hseq;
+                      hseq.data_ptr(), cp, T, B, cur_stream());
+    std::vector<torch::Tensor> out{hseq};
+    if (train) out.push_back(cache);
+    return out;
+}
+
+// BPTT sequential backward: dg (T, B, 2, 512) = [dxr dxz dxn dhgn]
+torch::Tensor gru_layer_bwd(torch::Tensor cache, torch::Tensor hseq,
+                            torch::Tensor dhin, torch::Tensor ut) {
+    check(cache, torch::kBFloat16, "cache");
+    check(hseq, torch::kBFloat16, "hseq");
+    check(dhin, torch::kBFloat16, "dhin");
+    check(ut, torch::kBFloat16, "ut");
+    const int T = cache.size(0), B = cache.size(1);
+    TORCH_CHECK(cache.size(2) == 2 && cache.size(3) == 512, "cache (T,B,2,512)");
+    TORCH_CHECK(hseq.size(2) == 2 && hseq.size(3) == 128, "hseq (T,B,2,128)");
+    TORCH_CHECK(dhin.sizes() == hseq.sizes(), "dhin must match hseq");
+    TORCH_CHECK(ut.size(0) == 2 && ut.size(1) == 128 && ut.size(2) == 384,
+                "ut must be (2,128,384)");
+    auto dg = torch::empty({T, B, 2, 512}, cache.options());
+    rk::gru_layer_bwd(cache.data_ptr(), hseq.data_ptr(), dhin.data_ptr(),
+                      ut.data_ptr(), dg.data_ptr(), T, B, cur_stream());
+    return dg;
+}
+
+// fused CE: returns (loss scalar f32, dlogits (N,5) f32) for mean reduction
+std::vector<torch::Tensor> ce_fwd_bwd(torch::Tensor logits, torch::Tensor target) {
+    check(logits, torch::kFloat32, "logits");
+    TORCH_CHECK(target.is_cuda() && target.is_contiguous(), "target");
+    TORCH_CHECK(target.scalar_type() == torch::kInt64, "target must be int64");
+    TORCH_CHECK(logits.dim() == 2 && logits.size(1) == 5, "logits must be (N,5)");
+    const int64_t n = logits.size(0);
+    TORCH_CHECK(target.numel() == n, "target size mismatch");
+    auto dlogits = torch::empty_like(logits);
+    auto loss = torch::zeros({1}, logits.options());
+    rk::ce_fwd_bwd(logits.data_ptr<float>(), target.data_ptr<int64_t>(),
+                   dlogits.data_ptr<float>(), loss.data_ptr<float>(), n,
+                   cur_stream());
+    return {loss.squeeze(0) / double(n), dlogits};
+}
+
+// fused Adam on flat fp32 buffers
+void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+               torch::Tensor v, double lr, double beta1, double beta2,
+               double eps, int64_t step) {
+    check(p, torch::kFloat32, "p");
+    check(g, torch::kFloat32, "g");
+    check(m, torch::kFloat32, "m");
+    check(v, torch::kFloat32, "v");
+    const int64_t n = p.numel();
+    TORCH_CHECK(g.numel() == n && m.numel() == n && v.numel() == n, "size mismatch");
+    rk::adam_step(p.data_ptr<float>(), g.data_ptr<float>(), m.data_ptr<float>(),
+                  v.data_ptr<float>(), n, float(lr), float(beta1), float(beta2),
+                  float(eps), int(step), cur_stream());
 }
 
 // hseq (T, B, 256) bf16 -> (logits (B,T,5) f32, argmax (B,T) u8) per flags
@@ -114,7 +180,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.doc() = "roko-mi355x CDNA4 kernels (gfx950)";
     m.def("mfma_probe", &mfma_probe);
     m.def("embed_mlp_fwd", &embed_mlp_fwd);
-    m.def("gru_layer_fwd", &gru_layer_fwd);
+    m.def("gru_layer_fwd", &gru_layer_fwd, py::arg("xg"), py::arg("u"),
+          py::arg("bhh"), py::arg("train") = false);
+    m.def("gru_layer_bwd", &gru_layer_bwd);
+    m.def("ce_fwd_bwd", &ce_fwd_bwd);
+    m.def("adam_step", &adam_step);
     m.def("head_fwd", &head_fwd, py::arg("hseq"), py::arg("w4"), py::arg("b4"),
           py::arg("want_logits") = true, py::arg("want_argmax") = false);
 }

@@ -34,15 +34,18 @@ constexpr int MB = 32;       // batch rows per workgroup
 constexpr int WAVES = 8;     // 512 threads
 constexpr int HPAD = H + 8;  // LDS row padding (bank-conflict fix)
 
+template <bool TRAIN>
 __global__ __launch_bounds__(WAVES * 64) void gru_layer_fwd_kernel(
     const bf16* __restrict__ xg,   // (T, B, 2, 3H)  W_ih·x + b_ih
     const bf16* __restrict__ u,    // (2, 3H, H)     weight_hh
     const float* __restrict__ bhh, // (2, 3H)        bias_hh
     bf16* __restrict__ hseq,       // (T, B, 2, H)   output
+    bf16* __restrict__ cache,      // (T, B, 2, 4H)  [r z n hgn] or nullptr
     int T, int B) {
     __shared__ struct {
         bf16 h[MB][HPAD];            // bf16 mirror of the hidden state
         bf16 xgb[2][MB][G3];         // double-buffered step gate inputs
+        bf16 cache_st[TRAIN ? MB : 1][TRAIN ? 4 * H : 1];  // [r z n hgn] staging
     } lds;
 
     const int dir = blockIdx.y;
@@ -94,18 +97,24 @@ __global__ __launch_bounds__(WAVES * 64) void gru_layer_fwd_kernel(
     __syncthreads();
 
     // ---- T-step recurrence -------------------------------------------------
+    // Per-step staging is split issue-early / write-late (T14,
+    // cdna_hip_programming.md §5.5): the next step's xg loads issue before
+    // the MFMAs and land in LDS only after the post-combine barrier, so HBM
+    // latency hides under compute.
     int cur = 0;
     for (int ti = 0; ti < T; ++ti) {
         const int t = (dir == 0) ? ti : T - 1 - ti;
-        // stage next step's xg into the other buffer
-        if (ti + 1 < T) {
+        // issue next step's xg loads into registers
+        bf16x8 stage[3];
+        const bool has_next = (ti + 1 < T);
+        if (has_next) {
             const int tn = (dir == 0) ? ti + 1 : T - 2 - ti;
             const bf16* src = xg + (((size_t)tn * B + b0) * 2 + dir) * G3;
 #pragma unroll
             for (int p = 0; p < 3; ++p) {
                 int e = (p * WAVES * 64 + tid) * 8;
                 int row = e / G3, col = e % G3;
-                *reinterpret_cast<bf16x8*>(&lds.xgb[cur ^ 1][row][col]) =
+                stage[p] =
                     *reinterpret_cast<const bf16x8*>(src + (size_t)row * 2 * G3 + col);
             }
         }
@@ -128,6 +137,7 @@ __global__ __launch_bounds__(WAVES * 64) void gru_layer_fwd_kernel(
         }
 
         // fused gate math; updates the fp32 register hidden state
+        float rv[2][4], zv[2][4], nv[2][4], hgnv[2][4];
 #pragma unroll
         for (int mt = 0; mt < 2; ++mt) {
 #pragma unroll
@@ -136,20 +146,46 @@ __global__ __launch_bounds__(WAVES * 64) void gru_layer_fwd_kernel(
                 const float xr = bf2f(lds.xgb[cur][row][0 * H + j0 + lcol]);
                 const float xz = bf2f(lds.xgb[cur][row][1 * H + j0 + lcol]);
                 const float xn = bf2f(lds.xgb[cur][row][2 * H + j0 + lcol]);
+                const float hgn = acc[mt][2][i] + bhh_reg[2];
                 const float r = sigmoidf_dev(xr + acc[mt][0][i] + bhh_reg[0]);
                 const float z = sigmoidf_dev(xz + acc[mt][1][i] + bhh_reg[1]);
-                const float n = tanhf_dev(xn + r * (acc[mt][2][i] + bhh_reg[2]));
+                const float n = tanhf_dev(xn + r * hgn);
+                rv[mt][i] = r;
+                zv[mt][i] = z;
+                nv[mt][i] = n;
+                hgnv[mt][i] = hgn;
                 hreg[mt][i] = (1.0f - z) * n + z * hreg[mt][i];
             }
         }
 
         __syncthreads();  // all waves done reading lds.h (and prior hseq read)
+        // write-late: staged xg for step t+1 lands now, under no reader
+        if (has_next) {
+#pragma unroll
+            for (int p = 0; p < 3; ++p) {
+                int e = (p * WAVES * 64 + tid) * 8;
+                int row = e / G3, col = e % G3;
+                *reinterpret_cast<bf16x8*>(&lds.xgb[cur ^ 1][row][col]) = stage[p];
+            }
+        }
 #pragma unroll
         for (int mt = 0; mt < 2; ++mt)
 #pragma unroll
             for (int i = 0; i < 4; ++i)
                 lds.h[mt * 16 + lrow * 4 + i][j0 + lcol] = f2bf(hreg[mt][i]);
-        __syncthreads();  // new h visible
+        if constexpr (TRAIN) {
+#pragma unroll
+            for (int mt = 0; mt < 2; ++mt)
+#pragma unroll
+                for (int i = 0; i < 4; ++i) {
+                    const int row = mt * 16 + lrow * 4 + i;
+                    lds.cache_st[row][0 * H + j0 + lcol] = f2bf(rv[mt][i]);
+                    lds.cache_st[row][1 * H + j0 + lcol] = f2bf(zv[mt][i]);
+                    lds.cache_st[row][2 * H + j0 + lcol] = f2bf(nv[mt][i]);
+                    lds.cache_st[row][3 * H + j0 + lcol] = f2bf(hgnv[mt][i]);
+                }
+        }
+        __syncthreads();  // new h (+ cache staging) visible
 
         // cooperative wide store of h to hseq (coalesced 16B per lane)
         {
@@ -159,17 +195,190 @@ __global__ __launch_bounds__(WAVES * 64) void gru_layer_fwd_kernel(
             *reinterpret_cast<bf16x8*>(dst + (size_t)row * 2 * H + col) =
                 *reinterpret_cast<const bf16x8*>(&lds.h[row][col]);
         }
+        if constexpr (TRAIN) {
+            bf16* dst = cache + (((size_t)t * B + b0) * 2 + dir) * 4 * H;
+            const int row = tid / 64;           // 512 threads = 8 rows x 64
+            const int col = (tid % 64) * 8;
+#pragma unroll
+            for (int q = 0; q < 4; ++q)
+                *reinterpret_cast<bf16x8*>(dst + (size_t)(row + q * 8) * 2 * 4 * H + col) =
+                    *reinterpret_cast<const bf16x8*>(&lds.cache_st[row + q * 8][col]);
+        }
         cur ^= 1;
     }
 }
 
 void gru_layer_fwd(const void* xg, const void* u, const float* bhh, void* hseq,
-                   int T, int B, hipStream_t stream) {
+                   void* cache, int T, int B, hipStream_t stream) {
     dim3 grid(B / MB, 2);
     dim3 block(WAVES * 64);
-    hipLaunchKernelGGL(gru_layer_fwd_kernel, grid, block, 0, stream,
-                       static_cast<const bf16*>(xg), static_cast<const bf16*>(u),
-                       bhh, static_cast<bf16*>(hseq), T, B);
+    if (cache)
+        hipLaunchKernelGGL(gru_layer_fwd_kernel<true>, grid, block, 0, stream,
+                           static_cast<const bf16*>(xg), static_cast<const bf16*>(u),
+                           bhh, static_cast<bf16*>(hseq), static_cast<bf16*>(cache),
+                           T, B);
+    else
+        hipLaunchKernelGGL(gru_layer_fwd_kernel<false>, grid, block, 0, stream,
+                           static_cast<const bf16*>(xg), static_cast<const bf16*>(u),
+                           bhh, static_cast<bf16*>(hseq), nullptr, T, B);
+}
+
+// ---------------------------------------------------------------------------
+// Backward (BPTT) — the sequential part only. The kernel walks the T steps
+// in reverse, carrying dL/dh in registers and emitting per-step gate
+// gradients dg = [dxr, dxz, dxn, dhgn]; the batched weight/input gradient
+// reductions (dU = dhg^T·h_prev, dW_ih = dxg^T·x, dx = dxg·W_ih) are plain
+// GEMMs done host-side with hipBLASLt.
+//
+// Math per step (forward: r,z = sig, n = tanh(xn + r*hgn), h' = (1-z)n + zh):
+//   dn   = dh * (1-z)            dz  = dh * (h_prev - n)
+//   dan  = dn * (1 - n^2)        dxn = dan         dhgn = dan * r
+//   dr   = dan * hgn             dxr = dhgr = dr * r * (1-r)
+//   dxz  = dhgz = dz * z * (1-z)
+//   dh_prev = dh * z + dhg · U
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(WAVES * 64) void gru_layer_bwd_kernel(
+    const bf16* __restrict__ cache,  // (T, B, 2, 4H) [r z n hgn]
+    const bf16* __restrict__ hseq,   // (T, B, 2, H)
+    const bf16* __restrict__ dhin,   // (T, B, 2, H) grad wrt layer output
+    const bf16* __restrict__ ut,     // (2, H, 3H) = weight_hh^T
+    bf16* __restrict__ dg,           // (T, B, 2, 4H) out: [dxr dxz dxn dhgn]
+    int T, int B) {
+    __shared__ struct {
+        bf16 cache_st[MB][4 * H];   // staged cache[t]
+        bf16 dhin_st[MB][H];        // staged dhin[t]
+        bf16 hprev_st[MB][H];       // staged h_{t-1}
+        bf16 dhg[MB][G3 + 8];       // A-operand of the dh GEMM
+        bf16 dg_st[MB][4 * H];      // staged output tile
+    } lds;
+
+    const int dir = blockIdx.y;
+    const int b0 = blockIdx.x * MB;
+    const int tid = threadIdx.x;
+    const int wid = tid >> 6;
+    const int lane = tid & 63;
+    const int j0 = wid * 16;
+    const int lrow = lane >> 4;
+    const int lcol = lane & 15;
+
+    // B-fragments of U for dh_prev = dhg·U: B[k=j][col=khid] = Ut[khid][j]
+    bf16x8 ufrag[12];
+#pragma unroll
+    for (int kb = 0; kb < 12; ++kb) {
+        const int col = j0 + lcol;          // this wave's hidden column
+        const int j = kb * 32 + 8 * lrow;   // gate-row index
+        ufrag[kb] = *reinterpret_cast<const bf16x8*>(
+            ut + ((size_t)dir * H + col) * G3 + j);
+    }
+
+    float dhc[2][4];  // dh carry, fragment-shaped
+#pragma unroll
+    for (int mt = 0; mt < 2; ++mt)
+#pragma unroll
+        for (int i = 0; i < 4; ++i) dhc[mt][i] = 0.0f;
+
+    for (int s = 0; s < T; ++s) {
+        const int t = (dir == 0) ? T - 1 - s : s;
+        const int tp = (dir == 0) ? t - 1 : t + 1;  // h_prev timestep
+        // ---- stage cache[t], dhin[t], hprev -------------------------------
+        {
+            const bf16* src = cache + (((size_t)t * B + b0) * 2 + dir) * 4 * H;
+            const int row = tid / 64, col = (tid % 64) * 8;
+#pragma unroll
+            for (int q = 0; q < 4; ++q)
+                *reinterpret_cast<bf16x8*>(&lds.cache_st[row + q * 8][col]) =
+                    *reinterpret_cast<const bf16x8*>(
+                        src + (size_t)(row + q * 8) * 2 * 4 * H + col);
+        }
+        {
+            const bf16* src = dhin + (((size_t)t * B + b0) * 2 + dir) * H;
+            const int row = tid / 16, col = (tid % 16) * 8;
+            *reinterpret_cast<bf16x8*>(&lds.dhin_st[row][col]) =
+                *reinterpret_cast<const bf16x8*>(src + (size_t)row * 2 * H + col);
+            if (tp >= 0 && tp < T) {
+                const bf16* hp = hseq + (((size_t)tp * B + b0) * 2 + dir) * H;
+                *reinterpret_cast<bf16x8*>(&lds.hprev_st[row][col]) =
+                    *reinterpret_cast<const bf16x8*>(hp + (size_t)row * 2 * H + col);
+            } else {
+                bf16x8 zero = {};
+                *reinterpret_cast<bf16x8*>(&lds.hprev_st[row][col]) = zero;
+            }
+        }
+        __syncthreads();
+
+        // ---- gate gradients ------------------------------------------------
+        float dhp_part[2][4];
+#pragma unroll
+        for (int mt = 0; mt < 2; ++mt) {
+#pragma unroll
+            for (int i = 0; i < 4; ++i) {
+                const int row = mt * 16 + lrow * 4 + i;
+                const int j = j0 + lcol;
+                const float dh = dhc[mt][i] + bf2f(lds.dhin_st[row][j]);
+                const float r = bf2f(lds.cache_st[row][0 * H + j]);
+                const float z = bf2f(lds.cache_st[row][1 * H + j]);
+                const float n = bf2f(lds.cache_st[row][2 * H + j]);
+                const float hgn = bf2f(lds.cache_st[row][3 * H + j]);
+                const float hp = bf2f(lds.hprev_st[row][j]);
+                const float dn = dh * (1.0f - z);
+                const float dz = dh * (hp - n);
+                const float dan = dn * (1.0f - n * n);
+                const float dxn = dan;
+                const float dhgn = dan * r;
+                const float dxr = dan * hgn * r * (1.0f - r);
+                const float dxz = dz * z * (1.0f - z);
+                dhp_part[mt][i] = dh * z;
+                lds.dhg[row][0 * H + j] = f2bf(dxr);
+                lds.dhg[row][1 * H + j] = f2bf(dxz);
+                lds.dhg[row][2 * H + j] = f2bf(dhgn);
+                lds.dg_st[row][0 * H + j] = f2bf(dxr);
+                lds.dg_st[row][1 * H + j] = f2bf(dxz);
+                lds.dg_st[row][2 * H + j] = f2bf(dxn);
+                lds.dg_st[row][3 * H + j] = f2bf(dhgn);
+            }
+        }
+        __syncthreads();
+
+        // ---- dh_prev = dh*z + dhg · U  (24 MFMA per wave) -----------------
+        f32x4 acc[2];
+#pragma unroll
+        for (int mt = 0; mt < 2; ++mt) acc[mt] = f32x4{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+        for (int kb = 0; kb < 12; ++kb) {
+#pragma unroll
+            for (int mt = 0; mt < 2; ++mt) {
+                bf16x8 a = lds_load_a_frag(&lds.dhg[0][0], mt * 16, kb * 32, G3 + 8);
+                acc[mt] = mfma16x16x32(a, ufrag[kb], acc[mt]);
+            }
+        }
+#pragma unroll
+        for (int mt = 0; mt < 2; ++mt)
+#pragma unroll
+            for (int i = 0; i < 4; ++i) dhc[mt][i] = dhp_part[mt][i] + acc[mt][i];
+
+        // ---- store dg tile -------------------------------------------------
+        {
+            bf16* dst = dg + (((size_t)t * B + b0) * 2 + dir) * 4 * H;
+            const int row = tid / 64, col = (tid % 64) * 8;
+#pragma unroll
+            for (int q = 0; q < 4; ++q)
+                *reinterpret_cast<bf16x8*>(
+                    dst + (size_t)(row + q * 8) * 2 * 4 * H + col) =
+                    *reinterpret_cast<const bf16x8*>(&lds.dg_st[row + q * 8][col]);
+        }
+        __syncthreads();  // dg_st/cache_st reads done before next staging
+    }
+}
+
+void gru_layer_bwd(const void* cache, const void* hseq, const void* dhin,
+                   const void* ut, void* dg, int T, int B, hipStream_t stream) {
+    dim3 grid(B / MB, 2);
+    dim3 block(WAVES * 64);
+    hipLaunchKernelGGL(gru_layer_bwd_kernel, grid, block, 0, stream,
+                       static_cast<const bf16*>(cache), static_cast<const bf16*>(hseq),
+                       static_cast<const bf16*>(dhin), static_cast<const bf16*>(ut),
+                       static_cast<bf16*>(dg), T, B);
 }
 
 }  // namespace rk

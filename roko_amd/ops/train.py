@@ -1,0 +1,201 @@
+"""GPU training path: custom-autograd GRU (HIP fwd/bwd kernels), fused
+cross-entropy, fused flat-buffer Adam.
+
+Division of labour: the sequential recurrence (fwd + BPTT)
+runs in the persistent HIP kernels; the batched weight/input gradient
+reductions (dU = dhg^T·h_prev, dW_ih = dxg^T·x, dx = dxg·W_ih) are plain
+GEMMs on hipBLASLt via torch.matmul, exactly where the brief routes plain
+GEMMs. The embedding/MLP front (tiny: <3% of step FLOPs) stays on torch
+autograd ops so dropout semantics match the reference's training graph.
+"""
+
+from __future__ import annotations
+
+from typing import List, Optional
+
+import torch
+
+from .. import config as C
+
+
+def _ext():
+    from . import ext
+
+    return ext()
+
+
+def train_step_available() -> bool:
+    from . import available
+
+    return available() and torch.cuda.is_available()
+
+
+class GruLayerFn(torch.autograd.Function):
+    """One bidirectional GRU layer over (T, B, in) through the HIP kernels."""
+
+    @staticmethod
+    def forward(ctx, x_seq, w_ih, b_ih_all, u, bhh):
+        # x_seq (T, B, in) fp32/bf16; w_ih (768, in); b_ih_all (768,)
+        # u (2, 384, 128); bhh (2, 384)
+        ext = _ext()
+        T, B, _ = x_seq.shape
+        x_bf = x_seq.to(torch.bfloat16)
+        w_ih_bf = w_ih.detach().to(torch.bfloat16)
+        xg = torch.addmm(
+            b_ih_all.detach().to(torch.bfloat16),
+            x_bf.reshape(T * B, -1),
+            w_ih_bf.t(),
+        ).view(T, B, 2, 384).contiguous()
+        u_bf = u.detach().to(torch.bfloat16).contiguous()
+        bhh_f = bhh.detach().float().contiguous()
+        hseq, cache = ext.gru_layer_fwd(xg, u_bf, bhh_f, True)
+        ctx.save_for_backward(x_bf, w_ih_bf, u_bf, hseq, cache)
+        ctx.in_dtype = x_seq.dtype
+        return hseq.view(T, B, 2 * C.HIDDEN_SIZE)
+
+    @staticmethod
+    def backward(ctx, dout):
+        ext = _ext()
+        x_bf, w_ih_bf, u_bf, hseq, cache = ctx.saved_tensors
+        T, B, _ = x_bf.shape
+        H = C.HIDDEN_SIZE
+        dhin = dout.reshape(T, B, 2, H).to(torch.bfloat16).contiguous()
+        ut = u_bf.transpose(1, 2).contiguous()  # (2, 128, 384)
+        dg = ext.gru_layer_bwd(cache, hseq, dhin, ut)  # (T,B,2,512)
+
+        dxg = dg[..., : 3 * H]                       # (T,B,2,384)
+        dhg = torch.cat([dg[..., : 2 * H], dg[..., 3 * H :]], dim=-1)
+
+        # h_prev sequences per direction
+        zeros = hseq.new_zeros(1, B, H)
+        hp_f = torch.cat([zeros, hseq[:-1, :, 0, :]], dim=0)  # (T,B,H)
+        hp_r = torch.cat([hseq[1:, :, 1, :], zeros], dim=0)
+
+        TB = T * B
+        dhg_f = dhg[:, :, 0, :].reshape(TB, 3 * H)
+        dhg_r = dhg[:, :, 1, :].reshape(TB, 3 * H)
+        du = torch.stack(
+            [dhg_f.t().mm(hp_f.reshape(TB, H)), dhg_r.t().mm(hp_r.reshape(TB, H))]
+        ).float()
+        dbhh = torch.stack([dhg_f.sum(0), dhg_r.sum(0)]).float()
+
+        dxg_flat = dxg.reshape(TB, 2, 3 * H)
+        dxg_cat = torch.cat([dxg_flat[:, 0, :], dxg_flat[:, 1, :]], dim=1)  # (TB,768)
+        x_flat = x_bf.reshape(TB, -1)
+        dw_ih = dxg_cat.t().mm(x_flat).float()          # (768, in)
+        db_ih = dxg_cat.sum(0).float()                   # (768,)
+        dx = dxg_cat.mm(w_ih_bf).to(ctx.in_dtype).view(T, B, -1)
+
+        return dx, dw_ih, db_ih, du, dbhh
+
+
+class FusedCrossEntropy(torch.autograd.Function):
+    """Mean CE over (N, 5) logits: fwd loss and dlogits in one kernel pass."""
+
+    @staticmethod
+    def forward(ctx, logits, target):
+        ext = _ext()
+        loss, dlogits = ext.ce_fwd_bwd(
+            logits.reshape(-1, C.NUM_CLASSES).float().contiguous(),
+            target.reshape(-1).contiguous(),
+        )
+        ctx.save_for_backward(dlogits)
+        ctx.shape = logits.shape
+        return loss
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        (dlogits,) = ctx.saved_tensors
+        return (grad_out * dlogits).view(ctx.shape), None
+
+
+def fused_cross_entropy(logits: torch.Tensor, target: torch.Tensor) -> torch.Tensor:
+    """logits (B, W, 5), target (B, W) int64 -> scalar mean loss."""
+    return FusedCrossEntropy.apply(logits, target)
+
+
+class FusedAdam:
+    """Flat-buffer Adam: all params/grads/moments live in single fp32 flat
+    tensors (param tensors become views), the update is ONE kernel and the
+    DP gradient sync is ONE all-reduce on the flat grad."""
+
+    def __init__(self, params: List[torch.nn.Parameter], lr: float = C.LR,
+                 betas=(0.9, 0.999), eps: float = 1e-8):
+        self.params = [p for p in params if p.requires_grad]
+        self.lr, self.betas, self.eps = lr, betas, eps
+        self.step_count = 0
+        n = sum(p.numel() for p in self.params)
+        dev = self.params[0].device
+        self.flat_p = torch.empty(n, dtype=torch.float32, device=dev)
+        self.flat_g = torch.zeros(n, dtype=torch.float32, device=dev)
+        self.m = torch.zeros(n, dtype=torch.float32, device=dev)
+        self.v = torch.zeros(n, dtype=torch.float32, device=dev)
+        off = 0
+        for p in self.params:
+            k = p.numel()
+            self.flat_p[off : off + k].copy_(p.data.reshape(-1))
+            p.data = self.flat_p[off : off + k].view(p.shape)
+            p.grad = self.flat_g[off : off + k].view(p.shape)
+            off += k
+
+    def zero_grad(self):
+        self.flat_g.zero_()
+
+    def allreduce_grads(self):
+        import torch.distributed as dist
+
+        if dist.is_initialized() and dist.get_world_size() > 1:
+            dist.all_reduce(self.flat_g)
+            self.flat_g /= dist.get_world_size()
+
+    def step(self):
+        self.step_count += 1
+        _ext().adam_step(self.flat_p, self.flat_g, self.m, self.v, self.lr,
+                         self.betas[0], self.betas[1], self.eps, self.step_count)
+
+
+def train_forward(model, x: torch.Tensor) -> torch.Tensor:
+    """Differentiable training forward on GPU: torch ops for the MLP front
+    (keeps the reference's dropout semantics), HIP kernels for the GRU."""
+    e = model.dropout(model.embedding(x.long()))
+    e = e.permute(0, 2, 3, 1)
+    t = model.dropout(torch.relu(model.fc1(e)))
+    t = model.dropout(torch.relu(model.fc2(t)))
+    seq = t.reshape(t.shape[0], t.shape[1], -1).transpose(0, 1).contiguous()
+    # (T, B, 500)
+    g = model.gru
+    drop_p = g.dropout if model.training else 0.0
+    for l in range(C.NUM_LAYERS):
+        w_ih = torch.cat(
+            [getattr(g, f"weight_ih_l{l}"), getattr(g, f"weight_ih_l{l}_reverse")], 0
+        )
+        b_ih = torch.cat(
+            [getattr(g, f"bias_ih_l{l}"), getattr(g, f"bias_ih_l{l}_reverse")]
+        )
+        u = torch.stack(
+            [getattr(g, f"weight_hh_l{l}"), getattr(g, f"weight_hh_l{l}_reverse")]
+        )
+        bhh = torch.stack(
+            [getattr(g, f"bias_hh_l{l}"), getattr(g, f"bias_hh_l{l}_reverse")]
+        )
+        if l > 0 and drop_p > 0:
+            seq = torch.nn.functional.dropout(seq, drop_p, model.training)
+        seq = GruLayerFn.apply(seq, w_ih, b_ih, u, bhh)
+    logits = model.fc4(seq.float()).transpose(0, 1)  # (B, T, 5)
+    return logits
+
+
+def fused_train_step(model, x, y, opt: Optional[FusedAdam] = None,
+                     reducer=None) -> torch.Tensor:
+    """One full training step through the fused path; returns the loss."""
+    logits = train_forward(model, x)
+    loss = fused_cross_entropy(logits, y)
+    if opt is not None:
+        opt.zero_grad()
+    else:
+        model.zero_grad(set_to_none=False)
+    loss.backward()
+    if opt is not None:
+        opt.allreduce_grads()
+        opt.step()
+    return loss.detach()

@@ -132,9 +132,105 @@ def test_full_model_forward_matches_cpu():
 
 
 @requires_gpu
-def test_model_refuses_silent_fallback():
-    # training-mode GPU forward must not silently fall back to eager
+def test_gru_layer_bwd_vs_torch_autograd():
+    """Custom GRU autograd (HIP fwd cache + BPTT kernel + GEMM reductions)
+    vs torch.nn.GRU gradients on the same fp32 weights."""
+    from roko_amd.ops.train import GruLayerFn
+
+    torch.manual_seed(5)
+    T, B = 90, 32
+    gru = torch.nn.GRU(256, 128, num_layers=1, bidirectional=True)
+    x = (torch.randn(T, B, 256) * 0.5).cuda().requires_grad_(True)
+
+    gru_c = gru.cuda()
+    ref_out, _ = gru_c(x)
+    ref_loss = (ref_out * torch.linspace(0.5, 1.5, 256, device="cuda")).mean()
+    ref_loss.backward()
+    ref_gx = x.grad.clone()
+    ref_gw = {n: p.grad.clone() for n, p in gru_c.named_parameters()}
+
+    x.grad = None
+    gru_c.zero_grad()
+    w_ih = torch.cat([gru_c.weight_ih_l0, gru_c.weight_ih_l0_reverse], 0)
+    b_ih = torch.cat([gru_c.bias_ih_l0, gru_c.bias_ih_l0_reverse])
+    u = torch.stack([gru_c.weight_hh_l0, gru_c.weight_hh_l0_reverse])
+    bhh = torch.stack([gru_c.bias_hh_l0, gru_c.bias_hh_l0_reverse])
+    out = GruLayerFn.apply(x, w_ih, b_ih, u, bhh)
+    loss = (out * torch.linspace(0.5, 1.5, 256, device="cuda")).mean()
+    loss.backward()
+
+    # bf16 kernels vs fp32 reference: compare directions via cosine + scale
+    def close(a, b, name):
+        a, b = a.reshape(-1).float(), b.reshape(-1).float()
+        cos = torch.nn.functional.cosine_similarity(a, b, dim=0).item()
+        rel = ((a - b).norm() / (b.norm() + 1e-12)).item()
+        assert cos > 0.995, (name, cos)
+        assert rel < 0.12, (name, rel)
+
+    close(x.grad, ref_gx, "dx")
+    close(gru_c.weight_ih_l0.grad, ref_gw["weight_ih_l0"], "dw_ih")
+    close(gru_c.weight_hh_l0.grad, ref_gw["weight_hh_l0"], "dw_hh")
+    close(gru_c.weight_hh_l0_reverse.grad, ref_gw["weight_hh_l0_reverse"], "dw_hh_r")
+    close(gru_c.bias_hh_l0.grad, ref_gw["bias_hh_l0"], "dbhh")
+    close(gru_c.bias_ih_l0.grad, ref_gw["bias_ih_l0"], "dbih")
+
+
+@requires_gpu
+def test_fused_ce_vs_torch():
+    from roko_amd.ops.train import fused_cross_entropy
+
+    torch.manual_seed(6)
+    logits = (torch.randn(16, 90, 5, device="cuda") * 2).requires_grad_(True)
+    y = torch.randint(0, 5, (16, 90), device="cuda")
+    loss = fused_cross_entropy(logits, y)
+    loss.backward()
+    g1 = logits.grad.clone()
+
+    logits2 = logits.detach().clone().requires_grad_(True)
+    ref = torch.nn.functional.cross_entropy(logits2.transpose(1, 2), y)
+    ref.backward()
+    assert abs(loss.item() - ref.item()) < 1e-4
+    assert torch.allclose(g1, logits2.grad, atol=1e-6)
+
+
+@requires_gpu
+def test_fused_adam_vs_torch():
+    from roko_amd.ops.train import FusedAdam
+
+    torch.manual_seed(7)
+    p1 = torch.nn.Parameter(torch.randn(1000, device="cuda"))
+    p2 = torch.nn.Parameter(torch.randn(333, device="cuda"))
+    ref1 = torch.nn.Parameter(p1.detach().clone())
+    ref2 = torch.nn.Parameter(p2.detach().clone())
+    opt = FusedAdam([p1, p2], lr=1e-2)
+    ref_opt = torch.optim.Adam([ref1, ref2], lr=1e-2)
+    for it in range(5):
+        g1 = torch.randn_like(p1)
+        g2 = torch.randn_like(p2)
+        opt.zero_grad()
+        p1.grad.copy_(g1)
+        p2.grad.copy_(g2)
+        opt.step()
+        ref_opt.zero_grad()
+        ref1.grad = g1.clone()
+        ref2.grad = g2.clone()
+        ref_opt.step()
+    assert torch.allclose(p1.detach(), ref1.detach(), atol=1e-5)
+    assert torch.allclose(p2.detach(), ref2.detach(), atol=1e-5)
+
+
+@requires_gpu
+def test_full_train_step_loss_decreases():
+    from roko_amd.ops.train import FusedAdam, fused_train_step
+
+    torch.manual_seed(8)
     m = RokoModel().cuda().train()
-    x = torch.randint(0, 12, (2, 200, 90)).cuda()
-    with pytest.raises(RuntimeError):
-        m(x)
+    opt = FusedAdam(list(m.parameters()), lr=3e-3)
+    x = torch.randint(0, 12, (32, 200, 90), dtype=torch.uint8, device="cuda")
+    y = torch.randint(0, 5, (32, 90), device="cuda")
+    losses = [float(fused_train_step(m, x, y, opt)) for _ in range(30)]
+    assert losses[-1] < losses[0] * 0.7, losses[::10]
+    # and the generic train-mode model(x) also routes through the HIP GRU
+    logits = m(x)
+    assert logits.requires_grad
+    assert logits.shape == (32, 90, 5)
