@@ -84,7 +84,7 @@ def test_gru_layer_fwd_vs_torch():
     bhh = torch.stack([gru.bias_hh_l0, gru.bias_hh_l0_reverse])
 
     ext = ops.ext()
-    hseq = ext.gru_layer_fwd(
+    (hseq,) = ext.gru_layer_fwd(
         xg.to(torch.bfloat16).cuda().contiguous(),
         u.to(torch.bfloat16).cuda().contiguous(),
         bhh.float().cuda().contiguous(),
