@@ -58,4 +58,42 @@ def main():
         print(f"{nconc} concurrent batches: {us:8.1f} us total, {us/nconc:8.1f} us/batch")
 
 if __name__ == "__main__":
-    main()
+    import sys as _s0
+    if not (len(_s0.argv) > 1 and _s0.argv[1] == "train"):
+        main()
+
+def train_breakdown():
+    import torch
+    from roko_amd.model import RokoModel
+    from roko_amd.ops.train import FusedAdam, fused_train_step, train_forward, fused_cross_entropy
+    torch.manual_seed(0)
+    m = RokoModel().cuda().train()
+    opt = FusedAdam(list(m.parameters()))
+    x = torch.randint(0, 12, (128, 200, 90), dtype=torch.uint8, device="cuda")
+    y = torch.randint(0, 5, (128, 90), device="cuda")
+    import time
+    def timeit(fn, n=20):
+        for _ in range(5): fn()
+        torch.cuda.synchronize(); t0 = time.perf_counter()
+        for _ in range(n): fn()
+        torch.cuda.synchronize(); return (time.perf_counter()-t0)/n*1e6
+
+    print(f"train fwd            : {timeit(lambda: train_forward(m, x)):9.1f} us")
+    def fwd_loss():
+        return fused_cross_entropy(train_forward(m, x), y)
+    print(f"train fwd+loss       : {timeit(fwd_loss):9.1f} us")
+    def full():
+        fused_train_step(m, x, y, opt)
+    print(f"full step            : {timeit(full):9.1f} us")
+    # torch profiler table
+    from torch.profiler import profile, ProfilerActivity
+    with profile(activities=[ProfilerActivity.CPU, ProfilerActivity.CUDA]) as prof:
+        for _ in range(3):
+            fused_train_step(m, x, y, opt)
+        torch.cuda.synchronize()
+    print(prof.key_averages().table(sort_by="cuda_time_total", row_limit=18))
+
+if __name__ == "__main__":
+    import sys as _s
+    if len(_s.argv) > 1 and _s.argv[1] == "train":
+        train_breakdown()
