@@ -22,8 +22,8 @@ def build():
 
     build_dir = os.path.join(HERE, "_build")
     os.makedirs(build_dir, exist_ok=True)
-    names = ["bindings.cpp", "adam.hip", "ce.hip", "embed_mlp.hip", "gru.hip",
-             "head.hip", "probe.hip"]
+    names = ["bindings.cpp", "adam.hip", "ce.hip", "embed_mlp.hip", "embgrad.hip",
+             "gru.hip", "head.hip", "probe.hip"]
     sources = [os.path.join(HERE, "hip", n) for n in names]
     load(
         name="_hip_ops",

@@ -22,6 +22,8 @@ void adam_step(float* p, const float* g, float* m, float* v, int64_t n,
                hipStream_t stream);
 void head_fwd(const void* hseq, const void* w4, const float* b4, float* logits,
               uint8_t* amax, int T, int B, hipStream_t stream);
+void emb_grad(const void* dm, const uint8_t* ids, float* de, int64_t n,
+              hipStream_t stream);
 }  // namespace rk
 
 namespace {
@@ -145,6 +147,19 @@ void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                   float(eps), int(step), cur_stream());
 }
 
+// dm (N, 50) bf16 + ids (N) u8 -> dE (12, 50) f32
+torch::Tensor emb_grad(torch::Tensor dm, torch::Tensor ids) {
+    check(dm, torch::kBFloat16, "dm");
+    check(ids, torch::kUInt8, "ids");
+    TORCH_CHECK(dm.dim() == 2 && dm.size(1) == 50, "dm must be (N,50)");
+    const int64_t n = dm.size(0);
+    TORCH_CHECK(ids.numel() == n, "ids size mismatch");
+    auto de = torch::zeros({12, 50}, dm.options().dtype(torch::kFloat32));
+    rk::emb_grad(dm.data_ptr(), ids.data_ptr<uint8_t>(), de.data_ptr<float>(),
+                 n, cur_stream());
+    return de;
+}
+
 // hseq (T, B, 256) bf16 -> (logits (B,T,5) f32, argmax (B,T) u8) per flags
 std::vector<torch::Tensor> head_fwd(torch::Tensor hseq, torch::Tensor w4,
                                     torch::Tensor b4, bool want_logits,
@@ -185,6 +200,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("gru_layer_bwd", &gru_layer_bwd);
     m.def("ce_fwd_bwd", &ce_fwd_bwd);
     m.def("adam_step", &adam_step);
+    m.def("emb_grad", &emb_grad);
     m.def("head_fwd", &head_fwd, py::arg("hseq"), py::arg("w4"), py::arg("b4"),
           py::arg("want_logits") = true, py::arg("want_argmax") = false);
 }

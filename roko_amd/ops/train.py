@@ -154,29 +154,57 @@ class FusedAdam:
                          self.betas[0], self.betas[1], self.eps, self.step_count)
 
 
-def train_forward(model, x: torch.Tensor) -> torch.Tensor:
-    """Differentiable training forward on GPU: torch ops for the MLP front
-    (keeps the reference's dropout semantics), HIP kernels for the GRU.
+class EmbedGatherFn(torch.autograd.Function):
+    """Embedding gather whose backward is the fused (12, 50) LDS reduction
+    kernel instead of aten's 115M-element scatter-add (9.1 ms/step) or a
+    skinny-K Tensile GEMM (1.65 ms/step) — profiles/train_breakdown_r01."""
 
-    The embedding gather is expressed as a one-hot matmul so its backward is
-    a GEMM reduction instead of aten's scatter-add into the 12-row table
-    (measured 9.1 ms/step = 46% of the step, profiles/train_breakdown_r01);
-    the whole front runs in bf16 with fp32 master weights cast in-graph."""
-    ids = x.long()
-    B = ids.shape[0]
-    hot = torch.zeros(*ids.shape, C.NUM_BASE_IDS, dtype=torch.bfloat16,
-                      device=ids.device)
-    hot.scatter_(3, ids.unsqueeze(-1), 1.0)
-    e = hot @ model.embedding.weight.to(torch.bfloat16)  # (B, R, W, E)
+    @staticmethod
+    def forward(ctx, weight, ids_u8):
+        ctx.save_for_backward(ids_u8)
+        return torch.nn.functional.embedding(
+            ids_u8.int(), weight.to(torch.bfloat16)
+        )
+
+    @staticmethod
+    def backward(ctx, dout):
+        (ids_u8,) = ctx.saved_tensors
+        de = _ext().emb_grad(
+            dout.reshape(-1, C.EMBED_DIM).contiguous(), ids_u8.reshape(-1)
+        )
+        return de, None
+
+
+def train_forward(model, x: torch.Tensor) -> torch.Tensor:
+    """Differentiable training forward on GPU: bf16 torch GEMMs for the MLP
+    front (keeps the reference's dropout semantics), HIP kernels for the
+    embedding backward and the GRU.
+
+    The per-column read reduction is laid out as batched GEMMs over
+    (R, W*E) views — same math as the reference's fc1/fc2 over permuted
+    activations (rnn_model.py:48-54) but with zero transposed copies of the
+    115M-element activation tensors."""
+    ids = x.to(torch.uint8)
+    B, R, W = ids.shape
+    e = EmbedGatherFn.apply(model.embedding.weight, ids)  # (B, R, W, 50) bf16
     e = model.dropout(e)
-    e = e.permute(0, 2, 3, 1)  # (B, W, E, R)
-    t = model.dropout(torch.relu(
-        e @ model.fc1.weight.to(torch.bfloat16).t()
-        + model.fc1.bias.to(torch.bfloat16)))
-    t = model.dropout(torch.relu(
-        t @ model.fc2.weight.to(torch.bfloat16).t()
-        + model.fc2.bias.to(torch.bfloat16)))
-    seq = t.reshape(t.shape[0], t.shape[1], -1).transpose(0, 1).contiguous()
+    m = e.reshape(B, R, W * C.EMBED_DIM)  # (B, 200, 4500), zero-copy view
+    # t1 = relu(W1 · M): (100, 200) @ (B, 200, 4500)
+    w1 = model.fc1.weight.to(torch.bfloat16)
+    b1 = model.fc1.bias.to(torch.bfloat16)
+    t1 = torch.baddbmm(
+        b1.view(1, -1, 1), w1.unsqueeze(0).expand(B, -1, -1), m
+    )
+    t1 = model.dropout(torch.relu(t1))  # (B, 100, 4500)
+    w2 = model.fc2.weight.to(torch.bfloat16)
+    b2 = model.fc2.bias.to(torch.bfloat16)
+    t2 = torch.baddbmm(
+        b2.view(1, -1, 1), w2.unsqueeze(0).expand(B, -1, -1), t1
+    )
+    t2 = model.dropout(torch.relu(t2))  # (B, 10, 4500)
+    # (B, 10, W, E) -> (B, W, E, 10) -> (B, W, 500): out[.., e*10+j]
+    t = t2.view(B, C.FC2_OUT, W, C.EMBED_DIM).permute(0, 2, 3, 1)
+    seq = t.reshape(B, W, -1).transpose(0, 1).contiguous()
     # (T, B, 500)
     g = model.gru
     drop_p = g.dropout if model.training else 0.0
