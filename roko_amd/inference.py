@@ -135,20 +135,40 @@ def infer(
     per_contig_pos: Dict[str, List[np.ndarray]] = defaultdict(list)
     per_contig_pred: Dict[str, List[np.ndarray]] = defaultdict(list)
 
+    def account(gis, js, preds):
+        for k in range(len(gis)):
+            g, pos_arr, _, _ = rkw.group_arrays(int(gis[k]))
+            per_contig_pos[g["contig"]].append(np.asarray(pos_arr[js[k]]))
+            per_contig_pred[g["contig"]].append(preds[k])
+
     t0 = time.time()
     n_windows = 0
+    pipe = None
+    if device.type == "cuda":
+        from .ops.forward import InferencePipeline
+
+        pipe = InferencePipeline(model, batch_size, depth=4)
     with torch.no_grad():
-        for gis, js, x in dl:
-            x = x.to(device, non_blocking=True)
-            logits = model(x.long())
-            preds = logits.argmax(dim=2).to(torch.uint8).cpu().numpy()
-            gis = gis.numpy()
-            js = js.numpy()
-            n_windows += len(gis)
-            for k in range(len(gis)):
-                g, pos_arr, _, _ = rkw.group_arrays(int(gis[k]))
-                per_contig_pos[g["contig"]].append(np.asarray(pos_arr[js[k]]))
-                per_contig_pred[g["contig"]].append(preds[k])
+        if pipe is not None:
+            # pipelined: keep `depth` batches in flight; votes are harvested
+            # one pipeline-depth behind submission
+            pending: List[tuple] = []
+            for gis, js, x in dl:
+                if len(pending) >= pipe.depth:
+                    tk, pgis, pjs = pending.pop(0)
+                    account(pgis, pjs, tk().numpy())
+                ticket = pipe.submit(x, copy_out=True)
+                pending.append((ticket, gis.numpy(), js.numpy()))
+                n_windows += len(gis)
+            for tk, pgis, pjs in pending:
+                account(pgis, pjs, tk().numpy())
+        else:
+            for gis, js, x in dl:
+                x = x.to(device, non_blocking=True)
+                logits = model(x.long())
+                preds = logits.argmax(dim=2).to(torch.uint8).cpu().numpy()
+                n_windows += len(gis)
+                account(gis.numpy(), js.numpy(), preds)
     dt = time.time() - t0
     bases = n_windows * C.WINDOW_STRIDE
     log(

@@ -237,3 +237,25 @@ def test_full_train_step_loss_decreases():
     logits = m(x)
     assert logits.requires_grad
     assert logits.shape == (32, 90, 5)
+
+
+@requires_gpu
+def test_inference_pipeline_matches_argmax():
+    """Pipelined hipGraph path must give the same predictions as the plain
+    kernel path for every submitted batch (incl. a partial last batch)."""
+    from roko_amd.ops.forward import InferencePipeline, roko_argmax
+
+    torch.manual_seed(8)
+    m = RokoModel().cuda().eval()
+    pipe = InferencePipeline(m, batch=64, depth=3)
+    xs = [torch.randint(0, 12, (64, 200, 90), dtype=torch.uint8, device="cuda")
+          for _ in range(7)]
+    xs.append(torch.randint(0, 12, (29, 200, 90), dtype=torch.uint8,
+                            device="cuda"))  # partial batch
+    tickets = [pipe.submit(x, copy_out=True) for x in xs]
+    with torch.no_grad():
+        refs = [roko_argmax(m, x).cpu() for x in xs]
+    for t, ref in zip(tickets, refs):
+        got = t()
+        assert got.shape == ref.shape
+        assert (got.long() == ref.long()).all()
