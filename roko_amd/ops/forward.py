@@ -154,9 +154,29 @@ class InferencePipeline:
         return slot.submit(x, copy_out)
 
 
+class _Ticket:
+    """Handle for one submitted batch; call it to wait and get the (n, W)
+    uint8 predictions. Snapshotted automatically when its slot is reused."""
+
+    def __init__(self, slot: "_Slot", n: int):
+        self.slot = slot
+        self.n = n
+        self.data = None
+
+    def materialize(self):
+        if self.data is None:
+            self.slot.event.synchronize()
+            self.data = self.slot.host_out[: self.n].clone()
+
+    def __call__(self) -> torch.Tensor:
+        self.materialize()
+        return self.data
+
+
 class _Slot:
     def __init__(self, pipe: "InferencePipeline", batch: int, dev):
         self.pipe = pipe
+        self.pending = None
         self.stream = torch.cuda.Stream(device=dev)
         self.x = torch.zeros(
             (batch, C.WINDOW_ROWS, C.WINDOW_COLS), dtype=torch.uint8,
@@ -178,7 +198,15 @@ class _Slot:
 
     def submit(self, x: torch.Tensor, copy_out: bool):
         n = x.shape[0]
+        # preserve the previous tenant's predictions before the slot buffers
+        # are overwritten (its ticket may be called arbitrarily late)
+        if self.pending is not None:
+            self.pending.materialize()
+            self.pending = None
         self.event.synchronize()  # previous tenant fully drained
+        # the input may have been produced on another stream (usually the
+        # default one): order the slot stream behind it before copying
+        self.stream.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(self.stream):
             if x.dtype != torch.uint8:
                 x = x.to(torch.uint8)
@@ -189,13 +217,10 @@ class _Slot:
                 self.amax = self.pipe._forward_amax(self.x)
             self.host_out.copy_(self.amax, non_blocking=True)
             self.event.record(self.stream)
-
-        def ticket():
-            self.event.synchronize()
-            out = self.host_out[:n]
-            return out.clone() if copy_out else out
-
-        return ticket
+        t = _Ticket(self, n)
+        if copy_out:
+            self.pending = t
+        return t
 
 
 def roko_argmax(model, x: torch.Tensor) -> torch.Tensor:

@@ -49,6 +49,46 @@ RK_DEV bf16x8 lds_load_b_frag_t(const bf16* tile_t, int col0, int k0, int ld) {
     return *reinterpret_cast<const bf16x8*>(tile_t + col * ld + k);
 }
 
+// B-fragment B[k][col] read DIRECTLY from a row-major [k][col] tile
+// (8 scalar u16 reads per lane; use when a second transposed LDS copy of the
+// tile would blow the LDS budget — the 2-cyc scalar issues hide under MFMAs).
+RK_DEV bf16x8 lds_load_b_frag_km(const bf16* tile, int k0, int col0, int ld) {
+    const int lane = threadIdx.x & 63;
+    const int col = col0 + (lane & 15);
+    const int kb = k0 + 8 * (lane >> 4);
+    bf16x8 r;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) r[j] = tile[(size_t)(kb + j) * ld + col];
+    return r;
+}
+
+// A-fragment A[row][k] read from a TRANSPOSED [k][row] tile (8 scalar reads).
+RK_DEV bf16x8 lds_load_a_frag_t(const bf16* tile_t, int row0, int k0, int ld) {
+    const int lane = threadIdx.x & 63;
+    const int row = row0 + (lane & 15);
+    const int kb = k0 + 8 * (lane >> 4);
+    bf16x8 r;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) r[j] = tile_t[(size_t)(kb + j) * ld + row];
+    return r;
+}
+
+// Counter-based dropout hash (lowbias32): deterministic per (seed, index),
+// regenerated identically in the recompute backward — no mask tensors.
+RK_DEV uint32_t hash32(uint32_t x) {
+    x ^= x >> 16;
+    x *= 0x7feb352dU;
+    x ^= x >> 15;
+    x *= 0x846ca68bU;
+    x ^= x >> 16;
+    return x;
+}
+
+// keep iff hash < thresh; thresh = keep_prob * 2^32 (uint64 so keep=1 -> all)
+RK_DEV bool drop_keep(uint32_t seed, uint32_t idx, uint64_t thresh) {
+    return (uint64_t)hash32(idx ^ (seed * 0x9E3779B9U)) < thresh;
+}
+
 RK_DEV float sigmoidf_dev(float x) { return 1.0f / (1.0f + __expf(-x)); }
 
 // overflow-stable tanh: tanh(x) = sign(x) * (1 - e) / (1 + e), e = exp(-2|x|)

@@ -26,14 +26,19 @@ __global__ __launch_bounds__(256) void emb_grad_kernel(
     const uint8_t* __restrict__ ids,  // (N)
     float* __restrict__ de,        // (NIDS, ED) pre-zeroed
     int64_t N) {
-    __shared__ float acc[NIDS * ED];
-    for (int e = threadIdx.x; e < NIDS * ED; e += 256) acc[e] = 0.0f;
+    __shared__ float acc_lds[NIDS * ED];
+    for (int e = threadIdx.x; e < NIDS * ED; e += 256) acc_lds[e] = 0.0f;
     __syncthreads();
 
     // 256 threads = 5 rows x 50 cols (6 threads idle); each thread owns one
-    // column of 8 rows per iteration (rows rsub, rsub+5, ..., rsub+35).
+    // column of 8 rows per iteration (rows rsub, rsub+5, ..., rsub+35) and
+    // accumulates a private per-class table in 12 REGISTERS via branchless
+    // compare-select (3 VALU per class-element) — no atomics in the loop.
     const int e = threadIdx.x % ED;
     const int rsub = threadIdx.x / ED;
+    float acc[NIDS];
+#pragma unroll
+    for (int c = 0; c < NIDS; ++c) acc[c] = 0.0f;
     const int64_t iters_total =
         (N + ROWS_PER_ITER - 1) / ROWS_PER_ITER;
     const int64_t iters_per_wg = (iters_total + gridDim.x - 1) / gridDim.x;
@@ -53,12 +58,17 @@ __global__ __launch_bounds__(256) void emb_grad_kernel(
             }
 #pragma unroll
             for (int q = 0; q < 8; ++q)
-                if (v[q] != 0.0f) atomicAdd(&acc[c[q] * ED + e], v[q]);
+#pragma unroll
+                for (int cc = 0; cc < NIDS; ++cc)
+                    acc[cc] += (c[q] == cc) ? v[q] : 0.0f;
         }
     }
+#pragma unroll
+    for (int cc = 0; cc < NIDS; ++cc)
+        if (acc[cc] != 0.0f) atomicAdd(&acc_lds[cc * ED + e], acc[cc]);
     __syncthreads();
     for (int i = threadIdx.x; i < NIDS * ED; i += 256)
-        if (acc[i] != 0.0f) atomicAdd(&de[i], acc[i]);
+        if (acc_lds[i] != 0.0f) atomicAdd(&de[i], acc_lds[i]);
 }
 
 void emb_grad(const void* dm, const uint8_t* ids, float* de, int64_t n,

@@ -1,4 +1,4 @@
-// Persistent bidirectional GRU layer forward for gfx950.
+// Persistent bidirectional GRU layer forward + BPTT backward for gfx950.
 //
 // Replaces the reference's cuDNN GRU (rnn_model.py:57).
 // Design (SURVEY.md §7 hard part (a), re-thought for CDNA4):
@@ -7,14 +7,21 @@
 //     the precomputed per-step gate inputs `xg`;
 //   * ONE kernel launch runs the full T-step recurrence of a layer: grid =
 //     (B/32 batch tiles) x (2 directions); each workgroup owns 32 batch rows
-//     whose hidden state lives in fp32 registers (fragment-shaped) with a
-//     bf16 mirror in LDS as the MFMA A-operand — no per-step launches, no
-//     grid-wide sync (batch rows are independent);
+//     whose hidden state lives in fp32 registers with a bf16 mirror in LDS
+//     as the MFMA A-operand — no per-step launches, no grid-wide sync;
+//   * 8 waves per workgroup = two per SIMD: the matrix and VALU pipes are
+//     separate and arbitrate between co-resident waves, so one wave's
+//     dependent gate math overlaps its partner's MFMAs;
 //   * U = weight_hh^T fragments are loaded ONCE into registers (12 bf16x8
 //     per wave) and reused for all T steps; per-step work is 24
 //     v_mfma_f32_16x16x32_bf16 per wave + fused sigmoid/tanh gate math;
-//   * per-step xg tiles (32x384 bf16) are double-buffered through LDS so the
-//     HBM reads of step t+1 overlap the MFMAs of step t.
+//   * the hidden-state mirror is double-buffered in LDS, so each step needs
+//     exactly ONE __syncthreads(): MFMAs read h[cur] while gate math writes
+//     h[cur^1]; the barrier publishes h[cur^1] for the next step;
+//   * per-step xg tiles (32x384 bf16) are double-buffered through LDS with a
+//     prefetch distance of TWO steps (issue-early / write-late, T14): the
+//     loads for step t+2 issue before anything waits on step t+1's, hiding
+//     the full HBM/L3 latency under one whole step of compute.
 //
 // PyTorch GRU semantics (gate order r,z,n in weight rows; n-gate bias split):
 //   r = sigmoid(xr + U_r h + bhh_r)
@@ -31,21 +38,27 @@ namespace rk {
 constexpr int H = 128;       // hidden size (config.HIDDEN_SIZE)
 constexpr int G3 = 3 * H;    // gate rows
 constexpr int MB = 32;       // batch rows per workgroup
-constexpr int WAVES = 8;     // 512 threads
+constexpr int WAVES = 8;     // 512 threads, two waves per SIMD: a wave's
+                             // dependent gate VALU hides under its partner's
+                             // MFMAs (MI355X_MICROARCH.md "Two waves per
+                             // SIMD" — the pipes are separate and arbitrate)
 constexpr int HPAD = H + 8;  // LDS row padding (bank-conflict fix)
+constexpr int XCH = MB * G3 / (WAVES * 64 * 8);  // xg chunks per thread (3)
+
+using bf16x4 = __attribute__((ext_vector_type(4))) __bf16;
 
 template <bool TRAIN>
-__global__ __launch_bounds__(WAVES * 64) void gru_layer_fwd_kernel(
+__global__ __launch_bounds__(WAVES * 64, 2) void gru_layer_fwd_kernel(
     const bf16* __restrict__ xg,   // (T, B, 2, 3H)  W_ih·x + b_ih
     const bf16* __restrict__ u,    // (2, 3H, H)     weight_hh
     const float* __restrict__ bhh, // (2, 3H)        bias_hh
     bf16* __restrict__ hseq,       // (T, B, 2, H)   output
-    bf16* __restrict__ cache,      // (T, B, 2, 4H)  [r z n hgn] or nullptr
+    bf16* __restrict__ cache,      // (T, B, 2, H, 4) [r z n hgn] or nullptr
     int T, int B) {
     __shared__ struct {
-        bf16 h[MB][HPAD];            // bf16 mirror of the hidden state
+        bf16 h[2][MB][HPAD];         // double-buffered hidden-state mirror
         bf16 xgb[2][MB][G3];         // double-buffered step gate inputs
-        bf16 cache_st[TRAIN ? MB : 1][TRAIN ? 4 * H : 1];  // [r z n hgn] staging
+        bf16 cache_st[TRAIN ? 2 : 1][TRAIN ? MB : 1][TRAIN ? 4 * H : 1];
     } lds;
 
     const int dir = blockIdx.y;
@@ -58,10 +71,10 @@ __global__ __launch_bounds__(WAVES * 64) void gru_layer_fwd_kernel(
     const int lcol = lane & 15;      // fragment column
 
     // ---- load U fragments (kept in registers for all T steps) -------------
-    // B-fragment for gates = h·U^T: B[k][col] = U[gate*H + j0 + col][k]
+    // B-fragment for gates = h·U^T: B[k][col] = U[gate*H + j0 + 16ct + col][k]
     bf16x8 ufrag[3][4];
 #pragma unroll
-    for (int g = 0; g < 3; ++g) {
+    for (int g = 0; g < 3; ++g)
 #pragma unroll
         for (int kb = 0; kb < 4; ++kb) {
             const int jrow = g * H + j0 + lcol;
@@ -69,57 +82,77 @@ __global__ __launch_bounds__(WAVES * 64) void gru_layer_fwd_kernel(
             ufrag[g][kb] = *reinterpret_cast<const bf16x8*>(
                 u + (size_t)dir * G3 * H + (size_t)jrow * H + k);
         }
-    }
     float bhh_reg[3];
 #pragma unroll
     for (int g = 0; g < 3; ++g) bhh_reg[g] = bhh[dir * G3 + g * H + j0 + lcol];
 
     // ---- zero hidden state -------------------------------------------------
-    float hreg[2][4];  // fp32 master copy, fragment-shaped (rows of this wave)
+    float hreg[2][4];  // [mt][i] fp32 master copy, fragment-shaped
 #pragma unroll
     for (int mt = 0; mt < 2; ++mt)
 #pragma unroll
         for (int i = 0; i < 4; ++i) hreg[mt][i] = 0.0f;
-    for (int e = tid; e < MB * HPAD; e += WAVES * 64) lds.h[0][e] = f2bf(0.0f);
+    for (int e = tid; e < MB * HPAD; e += WAVES * 64) lds.h[0][0][e] = f2bf(0.0f);
 
-    // ---- stage xg for the first step --------------------------------------
-    const int t_first = (dir == 0) ? 0 : T - 1;
+    const int stp = (dir == 0) ? 1 : -1;
+    const int tfirst = (dir == 0) ? 0 : T - 1;
+    auto xg_src = [&](int t) {
+        return xg + (((size_t)t * B + b0) * 2 + dir) * G3;
+    };
+
+    // ---- prologue: xg[t0] -> LDS buffer 0; xg[t1] -> stage registers ------
     {
-        const bf16* src = xg + (((size_t)t_first * B + b0) * 2 + dir) * G3;
+        const bf16* src = xg_src(tfirst);
 #pragma unroll
-        for (int p = 0; p < 3; ++p) {
+        for (int p = 0; p < XCH; ++p) {
             int e = (p * WAVES * 64 + tid) * 8;
             int row = e / G3, col = e % G3;
             *reinterpret_cast<bf16x8*>(&lds.xgb[0][row][col]) =
                 *reinterpret_cast<const bf16x8*>(src + (size_t)row * 2 * G3 + col);
         }
     }
+    bf16x8 stage[XCH];
+    if (T > 1) {
+        const bf16* src = xg_src(tfirst + stp);
+#pragma unroll
+        for (int p = 0; p < XCH; ++p) {
+            int e = (p * WAVES * 64 + tid) * 8;
+            int row = e / G3, col = e % G3;
+            stage[p] =
+                *reinterpret_cast<const bf16x8*>(src + (size_t)row * 2 * G3 + col);
+        }
+    }
     __syncthreads();
 
-    // ---- T-step recurrence -------------------------------------------------
-    // Per-step staging is split issue-early / write-late (T14,
-    // cdna_hip_programming.md §5.5): the next step's xg loads issue before
-    // the MFMAs and land in LDS only after the post-combine barrier, so HBM
-    // latency hides under compute.
+    // ---- T-step recurrence, ONE barrier per step --------------------------
     int cur = 0;
     for (int ti = 0; ti < T; ++ti) {
-        const int t = (dir == 0) ? ti : T - 1 - ti;
-        // issue next step's xg loads into registers
-        bf16x8 stage[3];
-        const bool has_next = (ti + 1 < T);
-        if (has_next) {
-            const int tn = (dir == 0) ? ti + 1 : T - 2 - ti;
-            const bf16* src = xg + (((size_t)tn * B + b0) * 2 + dir) * G3;
+        const int t = tfirst + stp * ti;
+
+        // issue xg[t+2] loads FIRST (they retire next step), then write the
+        // staged xg[t+1] into the back LDS buffer (waits only on last step's
+        // loads — vmcnt counting keeps the new loads in flight)
+        bf16x8 stage2[XCH];
+        if (ti + 2 < T) {
+            const bf16* src = xg_src(t + 2 * stp);
 #pragma unroll
-            for (int p = 0; p < 3; ++p) {
+            for (int p = 0; p < XCH; ++p) {
                 int e = (p * WAVES * 64 + tid) * 8;
                 int row = e / G3, col = e % G3;
-                stage[p] =
+                stage2[p] =
                     *reinterpret_cast<const bf16x8*>(src + (size_t)row * 2 * G3 + col);
             }
         }
+        if (ti + 1 < T) {
+#pragma unroll
+            for (int p = 0; p < XCH; ++p) {
+                int e = (p * WAVES * 64 + tid) * 8;
+                int row = e / G3, col = e % G3;
+                *reinterpret_cast<bf16x8*>(&lds.xgb[cur ^ 1][row][col]) = stage[p];
+            }
+        }
 
-        // gates_h = h · U^T  (24 MFMA per wave)
+        // gates_h = h · U^T  (24 MFMA per wave, A-frags shared 3-ways)
         f32x4 acc[2][3];
 #pragma unroll
         for (int mt = 0; mt < 2; ++mt)
@@ -129,81 +162,63 @@ __global__ __launch_bounds__(WAVES * 64) void gru_layer_fwd_kernel(
         for (int kb = 0; kb < 4; ++kb) {
 #pragma unroll
             for (int mt = 0; mt < 2; ++mt) {
-                bf16x8 a = lds_load_a_frag(&lds.h[0][0], mt * 16, kb * 32, HPAD);
+                bf16x8 a = lds_load_a_frag(&lds.h[cur][0][0], mt * 16, kb * 32, HPAD);
 #pragma unroll
                 for (int g = 0; g < 3; ++g)
                     acc[mt][g] = mfma16x16x32(a, ufrag[g][kb], acc[mt][g]);
             }
         }
 
-        // fused gate math; updates the fp32 register hidden state
-        float rv[2][4], zv[2][4], nv[2][4], hgnv[2][4];
+        // fused gate math; updates the fp32 register hidden state and writes
+        // the bf16 mirror into the BACK h buffer (no reader until barrier)
 #pragma unroll
         for (int mt = 0; mt < 2; ++mt) {
 #pragma unroll
             for (int i = 0; i < 4; ++i) {
                 const int row = mt * 16 + lrow * 4 + i;
-                const float xr = bf2f(lds.xgb[cur][row][0 * H + j0 + lcol]);
-                const float xz = bf2f(lds.xgb[cur][row][1 * H + j0 + lcol]);
-                const float xn = bf2f(lds.xgb[cur][row][2 * H + j0 + lcol]);
+                const int j = j0 + lcol;
+                const float xr = bf2f(lds.xgb[cur][row][0 * H + j]);
+                const float xz = bf2f(lds.xgb[cur][row][1 * H + j]);
+                const float xn = bf2f(lds.xgb[cur][row][2 * H + j]);
                 const float hgn = acc[mt][2][i] + bhh_reg[2];
                 const float r = sigmoidf_dev(xr + acc[mt][0][i] + bhh_reg[0]);
                 const float z = sigmoidf_dev(xz + acc[mt][1][i] + bhh_reg[1]);
                 const float n = tanhf_dev(xn + r * hgn);
-                rv[mt][i] = r;
-                zv[mt][i] = z;
-                nv[mt][i] = n;
-                hgnv[mt][i] = hgn;
-                hreg[mt][i] = (1.0f - z) * n + z * hreg[mt][i];
-            }
-        }
-
-        __syncthreads();  // all waves done reading lds.h (and prior hseq read)
-        // write-late: staged xg for step t+1 lands now, under no reader
-        if (has_next) {
-#pragma unroll
-            for (int p = 0; p < 3; ++p) {
-                int e = (p * WAVES * 64 + tid) * 8;
-                int row = e / G3, col = e % G3;
-                *reinterpret_cast<bf16x8*>(&lds.xgb[cur ^ 1][row][col]) = stage[p];
-            }
-        }
-#pragma unroll
-        for (int mt = 0; mt < 2; ++mt)
-#pragma unroll
-            for (int i = 0; i < 4; ++i)
-                lds.h[mt * 16 + lrow * 4 + i][j0 + lcol] = f2bf(hreg[mt][i]);
-        if constexpr (TRAIN) {
-#pragma unroll
-            for (int mt = 0; mt < 2; ++mt)
-#pragma unroll
-                for (int i = 0; i < 4; ++i) {
-                    const int row = mt * 16 + lrow * 4 + i;
-                    lds.cache_st[row][0 * H + j0 + lcol] = f2bf(rv[mt][i]);
-                    lds.cache_st[row][1 * H + j0 + lcol] = f2bf(zv[mt][i]);
-                    lds.cache_st[row][2 * H + j0 + lcol] = f2bf(nv[mt][i]);
-                    lds.cache_st[row][3 * H + j0 + lcol] = f2bf(hgnv[mt][i]);
+                const float hnew = (1.0f - z) * n + z * hreg[mt][i];
+                hreg[mt][i] = hnew;
+                lds.h[cur ^ 1][row][j] = f2bf(hnew);
+                if constexpr (TRAIN) {
+                    bf16x4 pk = {f2bf(r), f2bf(z), f2bf(n), f2bf(hgn)};
+                    *reinterpret_cast<bf16x4*>(
+                        &lds.cache_st[cur ^ 1][row][4 * j]) = pk;
                 }
+            }
         }
-        __syncthreads();  // new h (+ cache staging) visible
+        __syncthreads();  // h[cur^1] (+ cache staging) published
 
-        // cooperative wide store of h to hseq (coalesced 16B per lane)
+        // cooperative wide store of h to hseq (coalesced 16B per lane),
+        // overlaps the next step's MFMAs as plain VMEM traffic
         {
             bf16* dst = hseq + (((size_t)t * B + b0) * 2 + dir) * H;
-            const int row = tid / 16;           // 512 threads = 32 rows x 16
-            const int col = (tid % 16) * 8;
-            *reinterpret_cast<bf16x8*>(dst + (size_t)row * 2 * H + col) =
-                *reinterpret_cast<const bf16x8*>(&lds.h[row][col]);
+            {
+                const int e = tid * 8;
+                const int row = e / H, col = e % H;
+                *reinterpret_cast<bf16x8*>(dst + (size_t)row * 2 * H + col) =
+                    *reinterpret_cast<const bf16x8*>(&lds.h[cur ^ 1][row][col]);
+            }
         }
         if constexpr (TRAIN) {
             bf16* dst = cache + (((size_t)t * B + b0) * 2 + dir) * 4 * H;
-            const int row = tid / 64;           // 512 threads = 8 rows x 64
-            const int col = (tid % 64) * 8;
 #pragma unroll
-            for (int q = 0; q < 4; ++q)
-                *reinterpret_cast<bf16x8*>(dst + (size_t)(row + q * 8) * 2 * 4 * H + col) =
-                    *reinterpret_cast<const bf16x8*>(&lds.cache_st[row + q * 8][col]);
+            for (int p = 0; p < 4; ++p) {
+                const int e = (p * WAVES * 64 + tid) * 8;
+                const int row = e / (4 * H), col = e % (4 * H);
+                *reinterpret_cast<bf16x8*>(dst + (size_t)row * 2 * 4 * H + col) =
+                    *reinterpret_cast<const bf16x8*>(&lds.cache_st[cur ^ 1][row][col]);
+            }
         }
+#pragma unroll
+        for (int p = 0; p < XCH; ++p) stage[p] = stage2[p];
         cur ^= 1;
     }
 }
@@ -238,15 +253,17 @@ void gru_layer_fwd(const void* xg, const void* u, const float* bhh, void* hseq,
 //   dh_prev = dh * z + dhg · U
 // ---------------------------------------------------------------------------
 
-__global__ __launch_bounds__(WAVES * 64) void gru_layer_bwd_kernel(
-    const bf16* __restrict__ cache,  // (T, B, 2, 4H) [r z n hgn]
+constexpr int BW_WAVES = 8;  // dhg GEMM has K=384: 24 MFMA/wave at 8 waves
+
+__global__ __launch_bounds__(BW_WAVES * 64) void gru_layer_bwd_kernel(
+    const bf16* __restrict__ cache,  // (T, B, 2, H, 4) [r z n hgn] packed
     const bf16* __restrict__ hseq,   // (T, B, 2, H)
     const bf16* __restrict__ dhin,   // (T, B, 2, H) grad wrt layer output
     const bf16* __restrict__ ut,     // (2, H, 3H) = weight_hh^T
     bf16* __restrict__ dg,           // (T, B, 2, 4H) out: [dxr dxz dxn dhgn]
     int T, int B) {
     __shared__ struct {
-        bf16 cache_st[MB][4 * H];   // staged cache[t]
+        bf16 cache_st[MB][4 * H];   // staged cache[t] (packed layout)
         bf16 dhin_st[MB][H];        // staged dhin[t]
         bf16 hprev_st[MB][H];       // staged h_{t-1}
         bf16 dhg[MB][G3 + 8];       // A-operand of the dh GEMM
@@ -316,10 +333,12 @@ __global__ __launch_bounds__(WAVES * 64) void gru_layer_bwd_kernel(
                 const int row = mt * 16 + lrow * 4 + i;
                 const int j = j0 + lcol;
                 const float dh = dhc[mt][i] + bf2f(lds.dhin_st[row][j]);
-                const float r = bf2f(lds.cache_st[row][0 * H + j]);
-                const float z = bf2f(lds.cache_st[row][1 * H + j]);
-                const float n = bf2f(lds.cache_st[row][2 * H + j]);
-                const float hgn = bf2f(lds.cache_st[row][3 * H + j]);
+                const bf16x4 pk = *reinterpret_cast<const bf16x4*>(
+                    &lds.cache_st[row][4 * j]);
+                const float r = bf2f(pk[0]);
+                const float z = bf2f(pk[1]);
+                const float n = bf2f(pk[2]);
+                const float hgn = bf2f(pk[3]);
                 const float hp = bf2f(lds.hprev_st[row][j]);
                 const float dn = dh * (1.0f - z);
                 const float dz = dh * (hp - n);
@@ -374,7 +393,7 @@ __global__ __launch_bounds__(WAVES * 64) void gru_layer_bwd_kernel(
 void gru_layer_bwd(const void* cache, const void* hseq, const void* dhin,
                    const void* ut, void* dg, int T, int B, hipStream_t stream) {
     dim3 grid(B / MB, 2);
-    dim3 block(WAVES * 64);
+    dim3 block(BW_WAVES * 64);
     hipLaunchKernelGGL(gru_layer_bwd_kernel, grid, block, 0, stream,
                        static_cast<const bf16*>(cache), static_cast<const bf16*>(hseq),
                        static_cast<const bf16*>(dhin), static_cast<const bf16*>(ut),
