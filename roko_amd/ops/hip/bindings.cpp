@@ -24,6 +24,14 @@ void head_fwd(const void* hseq, const void* w4, const float* b4, float* logits,
               uint8_t* amax, int T, int B, hipStream_t stream);
 void emb_grad(const void* dm, const uint8_t* ids, float* de, int64_t n,
               hipStream_t stream);
+void front_fwd(const uint8_t* ids, const void* w1, const float* b1,
+               const void* w2, const float* b2, const void* emb, void* out,
+               int B, uint32_t seed, float keep, hipStream_t stream);
+void front_bwd(const uint8_t* ids, const void* dseq, const void* w1,
+               const float* b1, const void* w2, const float* b2,
+               const void* emb, float* dw1, float* db1, float* dw2, float* db2,
+               float* de, int B, uint32_t seed, float keep,
+               hipStream_t stream);
 }  // namespace rk
 
 namespace {
@@ -189,6 +197,56 @@ std::vector<torch::Tensor> head_fwd(torch::Tensor hseq, torch::Tensor w4,
     return out;
 }
 
+// fused train front fwd: ids -> (W, B, 500) bf16 GRU input sequence
+torch::Tensor front_fwd(torch::Tensor ids, torch::Tensor w1, torch::Tensor b1,
+                        torch::Tensor w2, torch::Tensor b2, torch::Tensor emb,
+                        int64_t seed, double keep) {
+    check(ids, torch::kUInt8, "ids");
+    check(w1, torch::kBFloat16, "w1");
+    check(b1, torch::kFloat32, "b1");
+    check(w2, torch::kBFloat16, "w2");
+    check(b2, torch::kFloat32, "b2");
+    check(emb, torch::kBFloat16, "emb");
+    const int B = ids.size(0);
+    TORCH_CHECK(ids.size(1) == 200 && ids.size(2) == 90, "ids must be (B,200,90)");
+    auto out = torch::empty({90, B, 500}, ids.options().dtype(torch::kBFloat16));
+    rk::front_fwd(ids.data_ptr<uint8_t>(), w1.data_ptr(), b1.data_ptr<float>(),
+                  w2.data_ptr(), b2.data_ptr<float>(), emb.data_ptr(),
+                  out.data_ptr(), B, (uint32_t)seed, (float)keep, cur_stream());
+    return out;
+}
+
+// fused train front bwd: -> (de, dw1, db1, dw2, db2) fp32
+std::vector<torch::Tensor> front_bwd(torch::Tensor ids, torch::Tensor dseq,
+                                     torch::Tensor w1, torch::Tensor b1,
+                                     torch::Tensor w2, torch::Tensor b2,
+                                     torch::Tensor emb, int64_t seed,
+                                     double keep) {
+    check(ids, torch::kUInt8, "ids");
+    check(dseq, torch::kBFloat16, "dseq");
+    check(w1, torch::kBFloat16, "w1");
+    check(b1, torch::kFloat32, "b1");
+    check(w2, torch::kBFloat16, "w2");
+    check(b2, torch::kFloat32, "b2");
+    check(emb, torch::kBFloat16, "emb");
+    const int B = ids.size(0);
+    TORCH_CHECK(dseq.size(0) == 90 && dseq.size(1) == B && dseq.size(2) == 500,
+                "dseq must be (90,B,500)");
+    auto opt = b1.options();
+    auto dw1 = torch::zeros({100, 200}, opt);
+    auto db1 = torch::zeros({100}, opt);
+    auto dw2 = torch::zeros({10, 100}, opt);
+    auto db2 = torch::zeros({10}, opt);
+    auto de = torch::zeros({12, 50}, opt);
+    rk::front_bwd(ids.data_ptr<uint8_t>(), dseq.data_ptr(), w1.data_ptr(),
+                  b1.data_ptr<float>(), w2.data_ptr(), b2.data_ptr<float>(),
+                  emb.data_ptr(), dw1.data_ptr<float>(), db1.data_ptr<float>(),
+                  dw2.data_ptr<float>(), db2.data_ptr<float>(),
+                  de.data_ptr<float>(), B, (uint32_t)seed, (float)keep,
+                  cur_stream());
+    return {de, dw1, db1, dw2, db2};
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -201,6 +259,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("ce_fwd_bwd", &ce_fwd_bwd);
     m.def("adam_step", &adam_step);
     m.def("emb_grad", &emb_grad);
+    m.def("front_fwd", &front_fwd);
+    m.def("front_bwd", &front_bwd);
     m.def("head_fwd", &head_fwd, py::arg("hseq"), py::arg("w4"), py::arg("b4"),
           py::arg("want_logits") = true, py::arg("want_argmax") = false);
 }

@@ -154,6 +154,38 @@ class FusedAdam:
                          self.betas[0], self.betas[1], self.eps, self.step_count)
 
 
+class FrontFn(torch.autograd.Function):
+    """Fused embedding+MLP training front (fwd + recompute backward in two
+    HIP kernels — front_train.hip). Dropout masks are counter-based hashes
+    of (seed, index); the backward regenerates them, so no mask tensors and
+    no activation round-trips through HBM."""
+
+    @staticmethod
+    def forward(ctx, emb_w, w1, b1, w2, b2, ids_u8, p_drop, training):
+        ext = _ext()
+        keep = 1.0 - (p_drop if training else 0.0)
+        seed = int(torch.randint(0, 2**31 - 1, (1,)).item())
+        w1b = w1.detach().to(torch.bfloat16).contiguous()
+        b1f = b1.detach().float().contiguous()
+        w2b = w2.detach().to(torch.bfloat16).contiguous()
+        b2f = b2.detach().float().contiguous()
+        embb = emb_w.detach().to(torch.bfloat16).contiguous()
+        out = ext.front_fwd(ids_u8, w1b, b1f, w2b, b2f, embb, seed, keep)
+        ctx.save_for_backward(ids_u8, w1b, b1f, w2b, b2f, embb)
+        ctx.seed, ctx.keep = seed, keep
+        return out  # (90, B, 500) bf16
+
+    @staticmethod
+    def backward(ctx, dseq):
+        ext = _ext()
+        ids_u8, w1b, b1f, w2b, b2f, embb = ctx.saved_tensors
+        de, dw1, db1, dw2, db2 = ext.front_bwd(
+            ids_u8, dseq.to(torch.bfloat16).contiguous(), w1b, b1f, w2b, b2f,
+            embb, ctx.seed, ctx.keep,
+        )
+        return de, dw1, db1, dw2, db2, None, None, None
+
+
 class EmbedGatherFn(torch.autograd.Function):
     """Embedding gather whose backward is the fused (12, 50) LDS reduction
     kernel instead of aten's 115M-element scatter-add (9.1 ms/step) or a
@@ -180,32 +212,17 @@ def train_forward(model, x: torch.Tensor) -> torch.Tensor:
     front (keeps the reference's dropout semantics), HIP kernels for the
     embedding backward and the GRU.
 
-    The per-column read reduction is laid out as batched GEMMs over
-    (R, W*E) views — same math as the reference's fc1/fc2 over permuted
-    activations (rnn_model.py:48-54) but with zero transposed copies of the
-    115M-element activation tensors."""
+    The whole front (embedding gather, three dropouts, fc1/fc2 GEMMs, relu,
+    and the permute into GRU layout) is TWO kernel launches via FrontFn —
+    the reference's semantics (rnn_model.py:47-56) with hash-based dropout
+    masks instead of materialised ones."""
     ids = x.to(torch.uint8)
-    B, R, W = ids.shape
-    e = EmbedGatherFn.apply(model.embedding.weight, ids)  # (B, R, W, 50) bf16
-    e = model.dropout(e)
-    m = e.reshape(B, R, W * C.EMBED_DIM)  # (B, 200, 4500), zero-copy view
-    # t1 = relu(W1 · M): (100, 200) @ (B, 200, 4500)
-    w1 = model.fc1.weight.to(torch.bfloat16)
-    b1 = model.fc1.bias.to(torch.bfloat16)
-    t1 = torch.baddbmm(
-        b1.view(1, -1, 1), w1.unsqueeze(0).expand(B, -1, -1), m
+    seq = FrontFn.apply(
+        model.embedding.weight, model.fc1.weight, model.fc1.bias,
+        model.fc2.weight, model.fc2.bias, ids, float(model.dropout.p),
+        model.training,
     )
-    t1 = model.dropout(torch.relu(t1))  # (B, 100, 4500)
-    w2 = model.fc2.weight.to(torch.bfloat16)
-    b2 = model.fc2.bias.to(torch.bfloat16)
-    t2 = torch.baddbmm(
-        b2.view(1, -1, 1), w2.unsqueeze(0).expand(B, -1, -1), t1
-    )
-    t2 = model.dropout(torch.relu(t2))  # (B, 10, 4500)
-    # (B, 10, W, E) -> (B, W, E, 10) -> (B, W, 500): out[.., e*10+j]
-    t = t2.view(B, C.FC2_OUT, W, C.EMBED_DIM).permute(0, 2, 3, 1)
-    seq = t.reshape(B, W, -1).transpose(0, 1).contiguous()
-    # (T, B, 500)
+    # (T, B, 500) bf16
     g = model.gru
     drop_p = g.dropout if model.training else 0.0
     for l in range(C.NUM_LAYERS):
