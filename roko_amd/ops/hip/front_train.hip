@@ -6,25 +6,32 @@
 // cost ~3 ms/step at b=128 in dropout/reduce/scatter/GEMM glue
 // (profiles/train_r01_kernel_stats.txt). Design:
 //
-//   * one workgroup (8 waves) per window; all weights staged in LDS once;
-//     the 90 columns are processed sequentially, each as two small MFMA
-//     GEMMs (t1 = relu(W1·m + b1), t2 = relu(W2·t1 + b2)) over LDS tiles;
-//   * dropout masks are COUNTER-BASED HASHES (common.h drop_keep) of
-//     (seed, element index) — nothing is materialised; the backward kernel
-//     regenerates the embedding mask from the same seed and recovers the
-//     relu/dropout deriatives of t1/t2 from the recomputed activations'
-//     signs (post > 0 <=> kept AND pre > 0);
+//   * a workgroup (8 waves) owns a SLICE of one window's 90 columns
+//     (CSPLIT workgroups per window — grid B x CSPLIT fills the 256 CUs);
+//     all weights are staged in LDS once; each column is a chain of small
+//     MFMA GEMMs over LDS tiles;
+//   * dropout masks are COUNTER-BASED HASHES (common.h hash32) of
+//     (seed, element index) — nothing is materialised. One 32-bit hash
+//     yields TWO 16-bit keep decisions for an element pair, halving the
+//     hash VALU. The backward regenerates the embedding mask from the same
+//     seed and recovers the relu/dropout derivatives of t1/t2 from the
+//     recomputed activations' signs (post > 0 <=> kept AND pre > 0);
 //   * the backward RECOMPUTES m and t1 per column (MFMA is ~100x cheaper
 //     than round-tripping the 173 MB of activations through HBM), carries
-//     dW1/dW2/db1/db2 in REGISTER fragment accumulators across all 90
-//     columns, accumulates the embedding gradient in a (12,50) LDS table,
-//     and commits everything with one atomic pass per workgroup at the end;
+//     dW1/dW2 in REGISTER fragment accumulators across its columns, reduces
+//     db1/db2 with lane shuffles into small LDS tables, accumulates the
+//     embedding gradient in a (12,50) LDS table, and commits everything
+//     with one atomic pass per workgroup at the end;
+//   * EVERY MFMA operand is a 16-byte ds_read_b128 fragment read — tiles
+//     are written in the layouts their consumers need. The LDS array is
+//     per-CU, and scalar u16 fragment reads cost 16x more array cycles per
+//     byte: the first version of this kernel was LDS-array-bound at 39.5%
+//     SQ_LDS_IDX_ACTIVE (profiles/pmc_front_r01.txt). The one exception
+//     (dm's W1 operand) reuses the resident row-major w1t via 8-scalar
+//     reads, where a second 50 KB transposed copy would not fit; phase-dead
+//     tiles share storage (unions) to stay under 160 KB, with the aliased
+//     pad regions re-zeroed each column;
 //   * dx is never needed: the input is integer base ids.
-//
-// Weight-layout notes: B-operand fragments are read either from transposed
-// LDS tiles (ds_read_b128, lds_load_b_frag_t) or DIRECTLY from row-major
-// tiles with 8 scalar reads (lds_load_b_frag_km) where a second transposed
-// copy would not fit the 160 KB LDS budget.
 
 #include <cstdint>
 
@@ -44,6 +51,28 @@ constexpr int KP = 224;        // R padded to 7 k-steps of 32
 constexpr int KP_LD = KP + 8;  // +8 bf16 padding against bank conflicts
 constexpr int MP = 112;        // F1 padded to 7 m-tiles
 constexpr int EP = 64;         // E padded to 4 n-tiles
+constexpr int CSPLIT = 2;      // workgroups per window (column slices)
+
+// dropout-mask hashes: one 32-bit hash covers an index PAIR; bit-halves give
+// the two 16-bit keep decisions. thresh16 = keep * 65536.
+RK_DEV uint32_t mask_hash(uint32_t seed, uint32_t pair_key) {
+    return hash32(pair_key ^ (seed * 0x9E3779B9U));
+}
+RK_DEV bool keep_half(uint32_t h, int which, uint32_t thresh16) {
+    return ((h >> (16 * which)) & 0xFFFFu) < thresh16;
+}
+// embedding-mask pair key: pairs over consecutive reads (r0 = 2q, r0+1)
+RK_DEV uint32_t ekey(int b, int w, int q, int e) {
+    return ((uint32_t)(b * W + w) * (R / 2) + q) * E + e;
+}
+// t1-mask pair key: pairs over consecutive fc1 rows (f0 = 2p, f0+1)
+RK_DEV uint32_t t1key(int b, int w, int p, int e) {
+    return ((uint32_t)(b * W + w) * (MP / 2) + p) * E + e;
+}
+// t2 mask: single elements (tiny count), half 0 of its hash
+RK_DEV uint32_t t2key(int b, int w, int j, int e) {
+    return ((uint32_t)(b * W + w) * F2 + j) * E + e;
+}
 
 // ---------------------------------------------------------------------------
 // forward
@@ -59,14 +88,14 @@ __global__ __launch_bounds__(512, 2) void front_fwd_kernel(
     bf16* __restrict__ out,           // (W, B, OUT)
     int B, uint32_t seed, float keep) {
     __shared__ struct {
-        bf16 w1t[MP][KP_LD];     // zero-padded W1 [f][r]
-        bf16 m_t[EP][KP_LD];     // masked embedding tile [e][r]
-        bf16 t1_t[EP][136];      // t1 post-activation [e][f]
-        bf16 w2_lds[16][136];    // zero-padded W2 [j][f]
+        bf16 w1t[MP][KP_LD];     // zero-padded W1 [f][r] (G1 A-operand)
+        bf16 m_t[EP][KP_LD];     // masked embedding tile [e][r] (G1 B)
+        bf16 t1_t[EP][136];      // t1 post-activation [e][f] (G3 B)
+        bf16 w2_lds[16][136];    // zero-padded W2 [j][f] (G3 A)
         bf16 emb_s[12][E];
         bf16 t2st[OUT + 12];
-        float b1s[MP];
-        float b2s[16];
+        bf16 b1s[MP];
+        bf16 b2s[16];
         uint8_t col_ids[R];
     } lds;
 
@@ -76,36 +105,43 @@ __global__ __launch_bounds__(512, 2) void front_fwd_kernel(
     const int lane = tid & 63;
     const int lrow = lane >> 4;
     const int lcol = lane & 15;
-    const uint64_t thresh = (uint64_t)((double)keep * 4294967296.0);
+    const uint32_t thresh16 = (uint32_t)(keep * 65536.0f);
     const float inv_keep = 1.0f / keep;
+    const int w_begin = blockIdx.y * (W / CSPLIT);
+    const int w_end = (blockIdx.y + 1 == CSPLIT) ? W : w_begin + W / CSPLIT;
 
     // ---- one-time staging -------------------------------------------------
     for (int e = tid; e < MP * KP_LD; e += 512) (&lds.w1t[0][0])[e] = f2bf(0.f);
     for (int e = tid; e < EP * KP_LD; e += 512) (&lds.m_t[0][0])[e] = f2bf(0.f);
     for (int e = tid; e < EP * 136; e += 512) (&lds.t1_t[0][0])[e] = f2bf(0.f);
     for (int e = tid; e < 16 * 136; e += 512) (&lds.w2_lds[0][0])[e] = f2bf(0.f);
-    for (int e = tid; e < MP; e += 512) lds.b1s[e] = 0.f;
-    for (int e = tid; e < 16; e += 512) lds.b2s[e] = 0.f;
+    for (int e = tid; e < MP; e += 512) lds.b1s[e] = f2bf(0.f);
+    for (int e = tid; e < 16; e += 512) lds.b2s[e] = f2bf(0.f);
     __syncthreads();
     for (int e = tid; e < F1 * R; e += 512) lds.w1t[e / R][e % R] = w1[e];
     for (int e = tid; e < 12 * E; e += 512) lds.emb_s[e / E][e % E] = emb[e];
     for (int e = tid; e < F2 * F1; e += 512) lds.w2_lds[e / F1][e % F1] = w2[e];
-    for (int e = tid; e < F1; e += 512) lds.b1s[e] = b1[e];
-    for (int e = tid; e < F2; e += 512) lds.b2s[e] = b2[e];
+    for (int e = tid; e < F1; e += 512) lds.b1s[e] = f2bf(b1[e]);
+    for (int e = tid; e < F2; e += 512) lds.b2s[e] = f2bf(b2[e]);
     __syncthreads();
 
-    for (int w = 0; w < W; ++w) {
+    for (int w = w_begin; w < w_end; ++w) {
         // ---- stage this column's read ids --------------------------------
         if (tid < R) lds.col_ids[tid] = ids[((size_t)b * R + tid) * W + w];
         __syncthreads();
-        // ---- masked embedding tile m[r][e] stored [e][r] ------------------
-        for (int i = tid; i < R * E; i += 512) {
-            const int r = i / E, e = i % E;
-            const uint32_t idx = ((uint32_t)(b * R + r) * W + w) * E + e;
-            float v = 0.f;
-            if (drop_keep(seed, idx, thresh))
-                v = bf2f(lds.emb_s[lds.col_ids[r]][e]) * inv_keep;
-            lds.m_t[e][r] = f2bf(v);
+        // ---- masked embedding tile, one hash per read PAIR ----------------
+        for (int p = tid; p < E * (R / 2); p += 512) {
+            const int e = p / (R / 2), q = p % (R / 2);
+            const int r0 = 2 * q;
+            const uint32_t h = mask_hash(seed, ekey(b, w, q, e));
+            float v0 = 0.f, v1 = 0.f;
+            if (keep_half(h, 0, thresh16))
+                v0 = bf2f(lds.emb_s[lds.col_ids[r0]][e]) * inv_keep;
+            if (keep_half(h, 1, thresh16))
+                v1 = bf2f(lds.emb_s[lds.col_ids[r0 + 1]][e]) * inv_keep;
+            bf16 pk[2] = {f2bf(v0), f2bf(v1)};
+            *reinterpret_cast<uint32_t*>(&lds.m_t[e][r0]) =
+                *reinterpret_cast<const uint32_t*>(pk);
         }
         __syncthreads();
 
@@ -122,15 +158,22 @@ __global__ __launch_bounds__(512, 2) void front_fwd_kernel(
                     bf16x8 bb = lds_load_b_frag_t(&lds.m_t[0][0], nt * 16, kb * 32, KP_LD);
                     acc = mfma16x16x32(a, bb, acc);
                 }
+                const int e = nt * 16 + lcol;
 #pragma unroll
-                for (int i = 0; i < 4; ++i) {
+                for (int i = 0; i < 4; i += 2) {
                     const int f = mt * 16 + lrow * 4 + i;
-                    const int e = nt * 16 + lcol;
-                    float v = fmaxf(acc[i] + lds.b1s[f], 0.f);
-                    const uint32_t idx = ((uint32_t)(b * W + w) * F1 + f) * E + e;
-                    const bool live = (f < F1) && (e < E) &&
-                                      drop_keep(seed ^ 0x51u, idx, thresh);
-                    lds.t1_t[e][f] = f2bf(live ? v * inv_keep : 0.f);
+                    const uint32_t h =
+                        mask_hash(seed ^ 0x51u, t1key(b, w, f >> 1, e));
+                    bf16 pk[2];
+#pragma unroll
+                    for (int uu = 0; uu < 2; ++uu) {
+                        float v = fmaxf(acc[i + uu] + bf2f(lds.b1s[f + uu]), 0.f);
+                        const bool live = (f + uu < F1) && (e < E) &&
+                                          keep_half(h, uu, thresh16);
+                        pk[uu] = f2bf(live ? v * inv_keep : 0.f);
+                    }
+                    *reinterpret_cast<uint32_t*>(&lds.t1_t[e][f]) =
+                        *reinterpret_cast<const uint32_t*>(pk);
                 }
             }
         }
@@ -151,9 +194,10 @@ __global__ __launch_bounds__(512, 2) void front_fwd_kernel(
                 const int j = lrow * 4 + i;
                 const int e = nt * 16 + lcol;
                 if (j < F2 && e < E) {
-                    float v = fmaxf(acc[i] + lds.b2s[j], 0.f);
-                    const uint32_t idx = ((uint32_t)(b * W + w) * F2 + j) * E + e;
-                    const bool live = drop_keep(seed ^ 0x52u, idx, thresh);
+                    float v = fmaxf(acc[i] + bf2f(lds.b2s[j]), 0.f);
+                    const uint32_t h =
+                        mask_hash(seed ^ 0x52u, t2key(b, w, j, e));
+                    const bool live = keep_half(h, 0, thresh16);
                     lds.t2st[e * F2 + j] = f2bf(live ? v * inv_keep : 0.f);
                 }
             }
@@ -191,19 +235,35 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
     float* __restrict__ de,   // (12, E)
     int B, uint32_t seed, float keep) {
     __shared__ struct {
-        bf16 w1t[128][KP_LD];    // zero-padded W1 [f][r] (128 rows: K reads)
-        bf16 m_t[EP][KP_LD];     // [e][r]
-        bf16 t1_t[EP][136];      // [e][f] recomputed t1 post
-        bf16 w2_lds[16][136];    // [j][f]
-        bf16 w2t_t[MP][40];      // [f][j] = W2^T zero-padded
-        bf16 dt2_je[32][72];     // [j][e] dt2 pre-activation grads
-        bf16 dt1_fe[128][72];    // [f][e] dt1 pre-activation grads
-        bf16 dseq_st[OUT + 12];
+        bf16 w1t[MP][KP_LD];  // [f][r]: G1 A-operand; dm B via 8-scalar reads
+                              // (dm's K pad rows 112..127 read past this
+                              // array into the union below — safe: the
+                              // matching A-operand columns are zero)
+        union {
+            bf16 m_t[EP][KP_LD];  // [e][r]: G1 B-operand (dead after G1)
+            struct {              // live from the G3 phase on
+                bf16 dt2_je[16][72];   // [j][e]: dW2 A-operand
+                bf16 dt2_ej[EP][40];   // [e][j]: dt1 B-operand
+                bf16 dt1_ef[EP][136];  // [e][f]: dm A-operand
+            } g;
+        } u1;
+        union {
+            bf16 t1_t[EP][136];   // [e][f]: G3 B-operand (dead after G3)
+            bf16 dt1_fe[MP][72];  // [f][e]: dW1 A-operand
+        } u2;
+        bf16 m_rt[R + 8][72];    // [r][e]: dW1 B-operand
+        bf16 t1_fe[MP][72];      // [f][e]: dW2 B-operand + relu/drop mask
+        bf16 w2_lds[12][136];    // [j][f]: G3 A-operand. Only 12 rows: the
+                                 // A-fragment's row 12..15 reads run past
+                                 // the array (garbage), feeding accumulator
+                                 // rows j >= 10 that every epilogue ignores
+        bf16 w2t_t[MP][40];      // [f][j]: dt1 A-operand
+        bf16 dseq_st[OUT];
         bf16 emb_s[12][E];
-        float b1s[MP];
-        float b2s[16];
+        bf16 b1s[MP];
+        bf16 b2s[16];
         float de_lds[12][E];
-        float db1_lds[MP];
+        float db1_lds[F1];
         float db2_lds[16];
         uint8_t col_ids[R];
     } lds;
@@ -214,22 +274,24 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
     const int lane = tid & 63;
     const int lrow = lane >> 4;
     const int lcol = lane & 15;
-    const uint64_t thresh = (uint64_t)((double)keep * 4294967296.0);
+    const uint32_t thresh16 = (uint32_t)(keep * 65536.0f);
     const float inv_keep = 1.0f / keep;
+    const int w_begin = blockIdx.y * (W / CSPLIT);
+    const int w_end = (blockIdx.y + 1 == CSPLIT) ? W : w_begin + W / CSPLIT;
 
     // ---- one-time staging + zero ------------------------------------------
-    for (int e = tid; e < 128 * KP_LD; e += 512) (&lds.w1t[0][0])[e] = f2bf(0.f);
-    for (int e = tid; e < EP * KP_LD; e += 512) (&lds.m_t[0][0])[e] = f2bf(0.f);
-    for (int e = tid; e < EP * 136; e += 512) (&lds.t1_t[0][0])[e] = f2bf(0.f);
-    for (int e = tid; e < 16 * 136; e += 512) (&lds.w2_lds[0][0])[e] = f2bf(0.f);
+    for (int e = tid; e < MP * KP_LD; e += 512) (&lds.w1t[0][0])[e] = f2bf(0.f);
+    for (int e = tid; e < EP * KP_LD; e += 512) (&lds.u1.m_t[0][0])[e] = f2bf(0.f);
+    for (int e = tid; e < EP * 136; e += 512) (&lds.u2.t1_t[0][0])[e] = f2bf(0.f);
+    for (int e = tid; e < (R + 8) * 72; e += 512) (&lds.m_rt[0][0])[e] = f2bf(0.f);
+    for (int e = tid; e < MP * 72; e += 512) (&lds.t1_fe[0][0])[e] = f2bf(0.f);
+    for (int e = tid; e < 12 * 136; e += 512) (&lds.w2_lds[0][0])[e] = f2bf(0.f);
     for (int e = tid; e < MP * 40; e += 512) (&lds.w2t_t[0][0])[e] = f2bf(0.f);
-    for (int e = tid; e < 32 * 72; e += 512) (&lds.dt2_je[0][0])[e] = f2bf(0.f);
-    for (int e = tid; e < 128 * 72; e += 512) (&lds.dt1_fe[0][0])[e] = f2bf(0.f);
     for (int e = tid; e < 12 * E; e += 512) (&lds.de_lds[0][0])[e] = 0.f;
-    for (int e = tid; e < MP; e += 512) lds.db1_lds[e] = 0.f;
+    for (int e = tid; e < F1; e += 512) lds.db1_lds[e] = 0.f;
     for (int e = tid; e < 16; e += 512) lds.db2_lds[e] = 0.f;
-    for (int e = tid; e < MP; e += 512) lds.b1s[e] = 0.f;
-    for (int e = tid; e < 16; e += 512) lds.b2s[e] = 0.f;
+    for (int e = tid; e < MP; e += 512) lds.b1s[e] = f2bf(0.f);
+    for (int e = tid; e < 16; e += 512) lds.b2s[e] = f2bf(0.f);
     __syncthreads();
     for (int e = tid; e < F1 * R; e += 512) lds.w1t[e / R][e % R] = w1[e];
     for (int e = tid; e < 12 * E; e += 512) lds.emb_s[e / E][e % E] = emb[e];
@@ -237,34 +299,53 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
         lds.w2_lds[e / F1][e % F1] = w2[e];
         lds.w2t_t[e % F1][e / F1] = w2[e];
     }
-    for (int e = tid; e < F1; e += 512) lds.b1s[e] = b1[e];
-    for (int e = tid; e < F2; e += 512) lds.b2s[e] = b2[e];
-    __syncthreads();
+    for (int e = tid; e < F1; e += 512) lds.b1s[e] = f2bf(b1[e]);
+    for (int e = tid; e < F2; e += 512) lds.b2s[e] = f2bf(b2[e]);
 
-    // register accumulators carried across all 90 columns
+    // register accumulators carried across this slice's columns
     f32x4 dw1acc[12];
 #pragma unroll
     for (int s = 0; s < 12; ++s) dw1acc[s] = f32x4{0.f, 0.f, 0.f, 0.f};
     f32x4 dw2acc = {0.f, 0.f, 0.f, 0.f};
+    __syncthreads();
 
-    for (int w = 0; w < W; ++w) {
-        // ---- stage ids column + dseq column ------------------------------
+    for (int w = w_begin; w < w_end; ++w) {
+        // ---- stage ids + dseq; rebuild m in BOTH layouts; re-zero the pad
+        // regions of the aliased tiles (trashed by last column's grads) ----
         if (tid < R) lds.col_ids[tid] = ids[((size_t)b * R + tid) * W + w];
         if (tid < OUT)
             lds.dseq_st[tid] = dseq[((size_t)w * B + b) * OUT + tid];
+        for (int z = tid; z < EP * 3; z += 512) {   // m_t K-pad r in [200,224)
+            const int e = z / 3, c = 200 + (z % 3) * 8;
+            *reinterpret_cast<bf16x8*>(&lds.u1.m_t[e][c]) = bf16x8{};
+        }
+        for (int z = tid; z < 14 * 25; z += 512) {  // m_t rows e in [50,64)
+            const int e = 50 + z / 25, c = (z % 25) * 8;
+            *reinterpret_cast<bf16x8*>(&lds.u1.m_t[e][c]) = bf16x8{};
+        }
+        for (int z = tid; z < EP * 3; z += 512) {   // t1_t K-pad f in [112,136)
+            const int e = z / 3, c = 112 + (z % 3) * 8;
+            *reinterpret_cast<bf16x8*>(&lds.u2.t1_t[e][c]) = bf16x8{};
+        }
         __syncthreads();
-        // ---- recompute masked embedding tile ------------------------------
-        for (int i = tid; i < R * E; i += 512) {
-            const int r = i / E, e = i % E;
-            const uint32_t idx = ((uint32_t)(b * R + r) * W + w) * E + e;
-            float v = 0.f;
-            if (drop_keep(seed, idx, thresh))
-                v = bf2f(lds.emb_s[lds.col_ids[r]][e]) * inv_keep;
-            lds.m_t[e][r] = f2bf(v);
+        for (int p = tid; p < E * (R / 2); p += 512) {
+            const int e = p / (R / 2), q = p % (R / 2);
+            const int r0 = 2 * q;
+            const uint32_t h = mask_hash(seed, ekey(b, w, q, e));
+            float v0 = 0.f, v1 = 0.f;
+            if (keep_half(h, 0, thresh16))
+                v0 = bf2f(lds.emb_s[lds.col_ids[r0]][e]) * inv_keep;
+            if (keep_half(h, 1, thresh16))
+                v1 = bf2f(lds.emb_s[lds.col_ids[r0 + 1]][e]) * inv_keep;
+            bf16 pk[2] = {f2bf(v0), f2bf(v1)};
+            *reinterpret_cast<uint32_t*>(&lds.u1.m_t[e][r0]) =
+                *reinterpret_cast<const uint32_t*>(pk);
+            lds.m_rt[r0][e] = pk[0];
+            lds.m_rt[r0 + 1][e] = pk[1];
         }
         __syncthreads();
 
-        // ---- recompute t1 (same code path as forward => same bf16 bits) ---
+        // ---- G1 recompute: t1 (same code path as forward => same bits) ----
 #pragma clang loop unroll(disable)
         for (int s = 0; s < 4; ++s) {
             const int tile = wid + s * 8;
@@ -274,124 +355,147 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
 #pragma unroll
                 for (int kb = 0; kb < 7; ++kb) {
                     bf16x8 a = lds_load_a_frag(&lds.w1t[0][0], mt * 16, kb * 32, KP_LD);
-                    bf16x8 bb = lds_load_b_frag_t(&lds.m_t[0][0], nt * 16, kb * 32, KP_LD);
+                    bf16x8 bb = lds_load_b_frag_t(&lds.u1.m_t[0][0], nt * 16, kb * 32, KP_LD);
                     acc = mfma16x16x32(a, bb, acc);
                 }
+                const int e = nt * 16 + lcol;
 #pragma unroll
-                for (int i = 0; i < 4; ++i) {
+                for (int i = 0; i < 4; i += 2) {
                     const int f = mt * 16 + lrow * 4 + i;
-                    const int e = nt * 16 + lcol;
-                    float v = fmaxf(acc[i] + lds.b1s[f], 0.f);
-                    const uint32_t idx = ((uint32_t)(b * W + w) * F1 + f) * E + e;
-                    const bool live = (f < F1) && (e < E) &&
-                                      drop_keep(seed ^ 0x51u, idx, thresh);
-                    lds.t1_t[e][f] = f2bf(live ? v * inv_keep : 0.f);
+                    const uint32_t h =
+                        mask_hash(seed ^ 0x51u, t1key(b, w, f >> 1, e));
+                    bf16 pk[2];
+#pragma unroll
+                    for (int uu = 0; uu < 2; ++uu) {
+                        float v = fmaxf(acc[i + uu] + bf2f(lds.b1s[f + uu]), 0.f);
+                        const bool live = (f + uu < F1) && (e < E) &&
+                                          keep_half(h, uu, thresh16);
+                        pk[uu] = f2bf(live ? v * inv_keep : 0.f);
+                    }
+                    *reinterpret_cast<uint32_t*>(&lds.u2.t1_t[e][f]) =
+                        *reinterpret_cast<const uint32_t*>(pk);
+                    lds.t1_fe[f][e] = pk[0];
+                    lds.t1_fe[f + 1][e] = pk[1];
                 }
             }
         }
         __syncthreads();
 
-        // ---- recompute t2pre; dt2 = dseq ⊙ drop2' ⊙ relu2' ---------------
+        // ---- G3 recompute -> dt2 in both layouts (m_t region dies here) ---
         if (wid < 4) {
             const int nt = wid;
             f32x4 acc = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
             for (int kb = 0; kb < 4; ++kb) {
                 bf16x8 a = lds_load_a_frag(&lds.w2_lds[0][0], 0, kb * 32, 136);
-                bf16x8 bb = lds_load_b_frag_t(&lds.t1_t[0][0], nt * 16, kb * 32, 136);
+                bf16x8 bb = lds_load_b_frag_t(&lds.u2.t1_t[0][0], nt * 16, kb * 32, 136);
                 acc = mfma16x16x32(a, bb, acc);
             }
 #pragma unroll
             for (int i = 0; i < 4; ++i) {
                 const int j = lrow * 4 + i;
                 const int e = nt * 16 + lcol;
+                float g = 0.f;
                 if (j < F2 && e < E) {
-                    const float t2pre = acc[i] + lds.b2s[j];
-                    const uint32_t idx = ((uint32_t)(b * W + w) * F2 + j) * E + e;
-                    const bool live = (t2pre > 0.f) &&
-                                      drop_keep(seed ^ 0x52u, idx, thresh);
-                    const float g = live
-                        ? bf2f(lds.dseq_st[e * F2 + j]) * inv_keep : 0.f;
-                    lds.dt2_je[j][e] = f2bf(g);
-                    float red = g;
-#pragma unroll
-                    for (int moff = 1; moff < 16; moff <<= 1)
-                        red += __shfl_xor(red, moff, 16);
-                    if (lcol == 0) atomicAdd(&lds.db2_lds[j], red);
+                    const float t2pre = acc[i] + bf2f(lds.b2s[j]);
+                    const uint32_t h =
+                        mask_hash(seed ^ 0x52u, t2key(b, w, j, e));
+                    if (t2pre > 0.f && keep_half(h, 0, thresh16))
+                        g = bf2f(lds.dseq_st[e * F2 + j]) * inv_keep;
                 }
+                lds.u1.g.dt2_je[j][e] = f2bf(g);
+                lds.u1.g.dt2_ej[e][j] = f2bf(g);
+                float red = g;
+#pragma unroll
+                for (int moff = 1; moff < 16; moff <<= 1)
+                    red += __shfl_xor(red, moff, 16);
+                if (lcol == 0 && j < F2) atomicAdd(&lds.db2_lds[j], red);
+            }
+        } else {
+            // waves 4..7: zero the j in [16,32) pad of dt2_ej (read by the
+            // dt1 GEMM's K loop; aliased memory holds stale m values)
+            for (int z = tid - 256; z < EP * 2; z += 256) {
+                const int e = z >> 1, c = 16 + (z & 1) * 8;
+                *reinterpret_cast<bf16x8*>(&lds.u1.g.dt2_ej[e][c]) = bf16x8{};
             }
         }
         __syncthreads();
 
-        // ---- dt1 = W2^T · dt2, through relu1'/drop1' ----------------------
+        // ---- dt1 = W2^T · dt2, through relu1'/drop1' (both layouts) -------
 #pragma clang loop unroll(disable)
         for (int s = 0; s < 4; ++s) {
             const int tile = wid + s * 8;
             if (tile < 28) {
                 const int mt = tile >> 2, nt = tile & 3;
                 bf16x8 a = lds_load_a_frag(&lds.w2t_t[0][0], mt * 16, 0, 40);
-                bf16x8 bb = lds_load_b_frag_km(&lds.dt2_je[0][0], 0, nt * 16, 72);
+                bf16x8 bb = lds_load_b_frag_t(&lds.u1.g.dt2_ej[0][0], nt * 16, 0, 40);
                 f32x4 acc = mfma16x16x32(a, bb, f32x4{0.f, 0.f, 0.f, 0.f});
+                const int e = nt * 16 + lcol;
 #pragma unroll
                 for (int i = 0; i < 4; ++i) {
                     const int f = mt * 16 + lrow * 4 + i;
-                    const int e = nt * 16 + lcol;
                     // post > 0 <=> kept AND pre > 0 (chain rule collapses)
-                    const float t1post = bf2f(lds.t1_t[e][f]);
+                    const float t1post = bf2f(lds.t1_fe[f][e]);
                     const float g = (t1post > 0.f) ? acc[i] * inv_keep : 0.f;
-                    lds.dt1_fe[f][e] = f2bf(g);
+                    lds.u2.dt1_fe[f][e] = f2bf(g);
+                    lds.u1.g.dt1_ef[e][f] = f2bf(g);
                     float red = g;
 #pragma unroll
                     for (int moff = 1; moff < 16; moff <<= 1)
                         red += __shfl_xor(red, moff, 16);
-                    if (lcol == 0) atomicAdd(&lds.db1_lds[f], red);
+                    if (lcol == 0 && f < F1) atomicAdd(&lds.db1_lds[f], red);
                 }
             }
+        }
+        // zero dt1_ef's K-pad cols f in [112,136) (read by dm's A loop)
+        for (int z = tid; z < EP * 3; z += 512) {
+            const int e = z / 3, c = 112 + (z % 3) * 8;
+            *reinterpret_cast<bf16x8*>(&lds.u1.g.dt1_ef[e][c]) = bf16x8{};
         }
         __syncthreads();
 
         // ---- read-only phase: dW2, dW1 accumulate; dm -> de ---------------
-        if (wid < 7) {  // dW2 += dt2 · t1^T   (A (j,e), B (e,f))
+        if (wid < 7) {  // dW2 += dt2 · t1^T   (A (j,e) b128, B (e,f) b128)
             const int nt = wid;
 #pragma unroll
             for (int kb = 0; kb < 2; ++kb) {
-                bf16x8 a = lds_load_a_frag(&lds.dt2_je[0][0], 0, kb * 32, 72);
-                bf16x8 bb = lds_load_b_frag_km(&lds.t1_t[0][0], kb * 32, nt * 16, 136);
+                bf16x8 a = lds_load_a_frag(&lds.u1.g.dt2_je[0][0], 0, kb * 32, 72);
+                bf16x8 bb = lds_load_b_frag_t(&lds.t1_fe[0][0], nt * 16, kb * 32, 72);
                 dw2acc = mfma16x16x32(a, bb, dw2acc);
             }
         }
 #pragma unroll
-        for (int s = 0; s < 12; ++s) {  // dW1 += dt1 · m^T  (A (f,e), B (e,r))
+        for (int s = 0; s < 12; ++s) {  // dW1 += dt1 · m^T (A (f,e), B (e,r))
             const int tile = wid + s * 8;
             if (tile < 91) {
                 const int mt = tile / 13, nt = tile % 13;
 #pragma unroll
                 for (int kb = 0; kb < 2; ++kb) {
-                    bf16x8 a = lds_load_a_frag(&lds.dt1_fe[0][0], mt * 16, kb * 32, 72);
-                    bf16x8 bb = lds_load_b_frag_km(&lds.m_t[0][0], kb * 32, nt * 16, KP_LD);
+                    bf16x8 a = lds_load_a_frag(&lds.u2.dt1_fe[0][0], mt * 16, kb * 32, 72);
+                    bf16x8 bb = lds_load_b_frag_t(&lds.m_rt[0][0], nt * 16, kb * 32, 72);
                     dw1acc[s] = mfma16x16x32(a, bb, dw1acc[s]);
                 }
             }
         }
 #pragma clang loop unroll(disable)
-        for (int s = 0; s < 7; ++s) {  // dm = W1^T · dt1  (A (r,f), B (f,e))
-            const int tile = wid + s * 8;
+        for (int s = 0; s < 7; ++s) {  // dm^T = dt1^T · W1 (A (e,f) b128,
+            const int tile = wid + s * 8;  //       B (f,r) 8-scalar from w1t)
             if (tile < 52) {
-                const int mt = tile >> 2, nt = tile & 3;
+                const int emt = tile & 3, rnt = tile >> 2;
                 f32x4 acc = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
                 for (int kb = 0; kb < 4; ++kb) {
-                    bf16x8 a = lds_load_a_frag_t(&lds.w1t[0][0], mt * 16, kb * 32, KP_LD);
-                    bf16x8 bb = lds_load_b_frag_km(&lds.dt1_fe[0][0], kb * 32, nt * 16, 72);
+                    bf16x8 a = lds_load_a_frag(&lds.u1.g.dt1_ef[0][0], emt * 16, kb * 32, 136);
+                    bf16x8 bb = lds_load_b_frag_km(&lds.w1t[0][0], kb * 32, rnt * 16, KP_LD);
                     acc = mfma16x16x32(a, bb, acc);
                 }
 #pragma unroll
                 for (int i = 0; i < 4; ++i) {
-                    const int r = mt * 16 + lrow * 4 + i;
-                    const int e = nt * 16 + lcol;
+                    const int e = emt * 16 + lrow * 4 + i;
+                    const int r = rnt * 16 + lcol;
                     if (r < R && e < E) {
-                        const uint32_t idx = ((uint32_t)(b * R + r) * W + w) * E + e;
-                        if (drop_keep(seed, idx, thresh))
+                        const uint32_t h = mask_hash(seed, ekey(b, w, r >> 1, e));
+                        if (keep_half(h, r & 1, thresh16))
                             atomicAdd(&lds.de_lds[lds.col_ids[r]][e],
                                       acc[i] * inv_keep);
                     }
@@ -439,7 +543,8 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
 void front_fwd(const uint8_t* ids, const void* w1, const float* b1,
                const void* w2, const float* b2, const void* emb, void* out,
                int B, uint32_t seed, float keep, hipStream_t stream) {
-    hipLaunchKernelGGL(front::front_fwd_kernel, dim3(B), dim3(512), 0, stream,
+    hipLaunchKernelGGL(front::front_fwd_kernel,
+                       dim3(B, front::CSPLIT), dim3(512), 0, stream,
                        ids, static_cast<const bf16*>(w1), b1,
                        static_cast<const bf16*>(w2), b2,
                        static_cast<const bf16*>(emb), static_cast<bf16*>(out),
@@ -451,7 +556,8 @@ void front_bwd(const uint8_t* ids, const void* dseq, const void* w1,
                const void* emb, float* dw1, float* db1, float* dw2, float* db2,
                float* de, int B, uint32_t seed, float keep,
                hipStream_t stream) {
-    hipLaunchKernelGGL(front::front_bwd_kernel, dim3(B), dim3(512), 0, stream,
+    hipLaunchKernelGGL(front::front_bwd_kernel,
+                       dim3(B, front::CSPLIT), dim3(512), 0, stream,
                        ids, static_cast<const bf16*>(dseq),
                        static_cast<const bf16*>(w1), b1,
                        static_cast<const bf16*>(w2), b2,
