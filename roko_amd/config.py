@@ -120,6 +120,8 @@ class TrainConfig:
     workers: int = 0
     in_memory: bool = False
     seed: int = 0
+    #: resume from <out>/train_state.pt when present
+    resume: bool = False
     #: gradient-bucket size (bytes) for the RCCL all-reduce overlap
     bucket_bytes: int = 2 << 20
 

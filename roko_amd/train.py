@@ -58,7 +58,11 @@ class EarlyStopper:
 
 class CheckpointManager:
     """Keep the best-by-score reference-format state_dicts
-    (reference: train.py:82-84 — ignite ModelCheckpoint semantics)."""
+    (reference: train.py:82-84 — ignite ModelCheckpoint semantics), plus a
+    full resume sidecar (``train_state.pt``: optimizer moments, epoch, RNG,
+    early-stop state) — the reference cannot resume at all (SURVEY.md §5.4);
+    the ``.pth`` weight files stay pure state_dicts for cross-framework
+    compatibility."""
 
     def __init__(self, out_dir: str, keep: int = 2):
         self.out_dir = out_dir
@@ -78,6 +82,40 @@ class CheckpointManager:
             if os.path.exists(drop):
                 os.remove(drop)
         return path
+
+    @property
+    def state_path(self) -> str:
+        return os.path.join(self.out_dir, "train_state.pt")
+
+    def save_state(self, model, opt, epoch: int, stopper: "EarlyStopper") -> None:
+        torch.save(
+            {
+                "model": model.state_dict(),
+                "opt": opt.state_dict(),
+                "epoch": epoch,
+                "stopper": {"best": stopper.best, "bad": stopper.bad},
+                "rng": torch.get_rng_state(),
+                "cuda_rng": (
+                    torch.cuda.get_rng_state()
+                    if torch.cuda.is_available() else None
+                ),
+                "saved": self.saved,
+            },
+            self.state_path,
+        )
+
+    def load_state(self, model, opt, stopper: "EarlyStopper") -> int:
+        """Restore a previous run; returns the next epoch to run (1-based)."""
+        st = torch.load(self.state_path, map_location="cpu", weights_only=False)
+        model.load_state_dict(st["model"])
+        opt.load_state_dict(st["opt"])
+        stopper.best = st["stopper"]["best"]
+        stopper.bad = st["stopper"]["bad"]
+        torch.set_rng_state(st["rng"])
+        if st.get("cuda_rng") is not None and torch.cuda.is_available():
+            torch.cuda.set_rng_state(st["cuda_rng"])
+        self.saved = [tuple(x) for x in st.get("saved", [])]
+        return int(st["epoch"]) + 1
 
 
 def evaluate(model, loader, device) -> tuple[float, float]:
@@ -143,11 +181,17 @@ def train(
 
     stopper = EarlyStopper(cfg.patience)
     ckpt = CheckpointManager(out_dir)
+    start_epoch = 1
+    if cfg.resume and os.path.exists(ckpt.state_path):
+        start_epoch = ckpt.load_state(model, opt, stopper)
+        model = model.to(device)
+        if rank == 0:
+            log(f"resumed from {ckpt.state_path} at epoch {start_epoch}")
     model.train()
 
     step = 0
     history = []
-    for epoch in range(1, cfg.epochs + 1):
+    for epoch in range(start_epoch, cfg.epochs + 1):
         if sampler is not None:
             sampler.set_epoch(epoch)
         t0 = time.time()
@@ -184,6 +228,7 @@ def train(
                 f"val_acc {acc:.4f} ({wps:.0f} windows/s, {dt:.1f}s)"
             )
             ckpt.save(model, epoch, score if not math.isnan(score) else 0.0)
+            ckpt.save_state(model, opt, epoch, stopper)
         if max_steps is not None and step >= max_steps:
             break
         if val_dl is not None and stopper.step(score):
@@ -207,10 +252,12 @@ def main(argv=None):
     p.add_argument("--lr", type=float, default=C.LR)
     p.add_argument("--patience", type=int, default=C.PATIENCE)
     p.add_argument("--seed", type=int, default=0)
+    p.add_argument("--resume", action="store_true",
+                   help="resume from <out>/train_state.pt if present")
     a = p.parse_args(argv)
     cfg = TrainConfig(
         batch_size=a.b, epochs=a.epochs, lr=a.lr, patience=a.patience,
-        workers=a.t, in_memory=a.memory, seed=a.seed,
+        workers=a.t, in_memory=a.memory, seed=a.seed, resume=a.resume,
     )
     train(a.train, a.out, val_path=a.val, cfg=cfg)
 

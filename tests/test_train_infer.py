@@ -86,3 +86,36 @@ def test_infer_cli_roundtrip(train_rkw, infer_rkw, tiny_assembly, tmp_path):
     draft_len = len(tiny_assembly["draft"])
     assert 0.8 * draft_len < len(got["ctg1"]) < 1.2 * draft_len
     assert got["ctg1"] == seqs["ctg1"]
+
+
+def test_train_resume_roundtrip(tmp_path, small_train_rkw=None):
+    """Checkpoint/resume sidecar: a 2-epoch run interrupted after epoch 1
+    resumes and matches the state layout (SURVEY.md §5.4 — beyond-reference
+    capability)."""
+    import numpy as np
+    from roko_amd import config as C
+    from roko_amd.config import TrainConfig
+    from roko_amd.rkdata import RkwWriter
+    from roko_amd.train import train
+
+    path = str(tmp_path / "t.rkw")
+    rng = np.random.default_rng(0)
+    w = RkwWriter(path, inference=False)
+    X = rng.integers(0, 12, (8, C.WINDOW_ROWS, C.WINDOW_COLS), dtype=np.uint8)
+    Y = rng.integers(0, 5, (8, C.WINDOW_COLS), dtype=np.uint8)
+    P = np.zeros((8, C.WINDOW_COLS, 2), dtype=np.int32)
+    P[..., 0] = np.arange(C.WINDOW_COLS)[None, :]
+    w.store("c1", 0, C.WINDOW_COLS, P, X, Y)
+    w.write_contigs([("c1", "A" * 200)])
+    w.close()
+
+    out = str(tmp_path / "ckpt")
+    cfg = TrainConfig(batch_size=4, epochs=1, workers=0, seed=3)
+    m1, h1 = train(path, out, cfg=cfg, log=lambda *a, **k: None)
+    import os
+    assert os.path.exists(os.path.join(out, "train_state.pt"))
+
+    cfg2 = TrainConfig(batch_size=4, epochs=2, workers=0, seed=3, resume=True)
+    m2, h2 = train(path, out, cfg=cfg2, log=lambda *a, **k: None)
+    # resumed run starts at epoch 2 -> exactly one more epoch of history
+    assert len(h2) == 1 and h2[0]["epoch"] == 2
