@@ -35,6 +35,7 @@ from .io.fasta import write_fasta
 from .model import RokoModel
 from .parallel.ddp import init_distributed
 from .rkdata import RkwFile
+from .utils.metrics import Meter, trace_range
 
 #: vote tables: contig -> (keys int64 (K,), counts int64 (K, NUM_CLASSES))
 VoteTable = Dict[str, Tuple[np.ndarray, np.ndarray]]
@@ -143,6 +144,7 @@ def infer(
 
     t0 = time.time()
     n_windows = 0
+    meter = Meter("inference", rank=rank)
     pipe = None
     if device.type == "cuda":
         from .ops.forward import InferencePipeline
@@ -157,9 +159,11 @@ def infer(
                 if len(pending) >= pipe.depth:
                     tk, pgis, pjs = pending.pop(0)
                     account(pgis, pjs, tk().numpy())
-                ticket = pipe.submit(x, copy_out=True)
+                with trace_range("submit"):
+                    ticket = pipe.submit(x, copy_out=True)
                 pending.append((ticket, gis.numpy(), js.numpy()))
                 n_windows += len(gis)
+                meter.add(windows=len(gis), bases=len(gis) * C.WINDOW_STRIDE)
             for tk, pgis, pjs in pending:
                 account(pgis, pjs, tk().numpy())
         else:
@@ -168,7 +172,9 @@ def infer(
                 logits = model(x.long())
                 preds = logits.argmax(dim=2).to(torch.uint8).cpu().numpy()
                 n_windows += len(gis)
+                meter.add(windows=len(gis), bases=len(gis) * C.WINDOW_STRIDE)
                 account(gis.numpy(), js.numpy(), preds)
+    meter.close()
     dt = time.time() - t0
     bases = n_windows * C.WINDOW_STRIDE
     log(

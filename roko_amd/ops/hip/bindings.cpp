@@ -31,7 +31,7 @@ void front_bwd(const uint8_t* ids, const void* dseq, const void* w1,
                const float* b1, const void* w2, const float* b2,
                const void* emb, float* dw1, float* db1, float* dw2, float* db2,
                float* de, int B, uint32_t seed, float keep,
-               hipStream_t stream);
+               hipStream_t stream, uint32_t phase_mask);
 }  // namespace rk
 
 namespace {
@@ -221,7 +221,7 @@ std::vector<torch::Tensor> front_bwd(torch::Tensor ids, torch::Tensor dseq,
                                      torch::Tensor w1, torch::Tensor b1,
                                      torch::Tensor w2, torch::Tensor b2,
                                      torch::Tensor emb, int64_t seed,
-                                     double keep) {
+                                     double keep, int64_t phase_mask) {
     check(ids, torch::kUInt8, "ids");
     check(dseq, torch::kBFloat16, "dseq");
     check(w1, torch::kBFloat16, "w1");
@@ -243,7 +243,7 @@ std::vector<torch::Tensor> front_bwd(torch::Tensor ids, torch::Tensor dseq,
                   emb.data_ptr(), dw1.data_ptr<float>(), db1.data_ptr<float>(),
                   dw2.data_ptr<float>(), db2.data_ptr<float>(),
                   de.data_ptr<float>(), B, (uint32_t)seed, (float)keep,
-                  cur_stream());
+                  cur_stream(), (uint32_t)phase_mask);
     return {de, dw1, db1, dw2, db2};
 }
 
@@ -260,7 +260,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("adam_step", &adam_step);
     m.def("emb_grad", &emb_grad);
     m.def("front_fwd", &front_fwd);
-    m.def("front_bwd", &front_bwd);
+    m.def("front_bwd", &front_bwd, py::arg("ids"), py::arg("dseq"),
+          py::arg("w1"), py::arg("b1"), py::arg("w2"), py::arg("b2"),
+          py::arg("emb"), py::arg("seed"), py::arg("keep"),
+          py::arg("phase_mask") = 0x1F);
     m.def("head_fwd", &head_fwd, py::arg("hseq"), py::arg("w4"), py::arg("b4"),
           py::arg("want_logits") = true, py::arg("want_argmax") = false);
 }

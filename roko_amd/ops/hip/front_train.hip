@@ -233,7 +233,9 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
     float* __restrict__ dw2,  // (F2, F1)
     float* __restrict__ db2,  // (F2)
     float* __restrict__ de,   // (12, E)
-    int B, uint32_t seed, float keep) {
+    int B, uint32_t seed, float keep, uint32_t phase_mask) {
+    // phase_mask: timing-experiment switch (default 0x1F = all phases).
+    // bit0 G1 recompute, bit1 G3+dt2, bit2 dt1, bit3 dW1/dW2, bit4 dm/de.
     __shared__ struct {
         bf16 w1t[MP][KP_LD];  // [f][r]: G1 A-operand; dm B via 8-scalar reads
                               // (dm's K pad rows 112..127 read past this
@@ -347,7 +349,7 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
 
         // ---- G1 recompute: t1 (same code path as forward => same bits) ----
 #pragma clang loop unroll(disable)
-        for (int s = 0; s < 4; ++s) {
+        for (int s = 0; s < ((phase_mask & 1u) ? 4 : 0); ++s) {
             const int tile = wid + s * 8;
             if (tile < 28) {
                 const int mt = tile >> 2, nt = tile & 3;
@@ -382,7 +384,7 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
         __syncthreads();
 
         // ---- G3 recompute -> dt2 in both layouts (m_t region dies here) ---
-        if (wid < 4) {
+        if (wid < 4 && (phase_mask & 2u)) {
             const int nt = wid;
             f32x4 acc = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
@@ -411,7 +413,7 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
                     red += __shfl_xor(red, moff, 16);
                 if (lcol == 0 && j < F2) atomicAdd(&lds.db2_lds[j], red);
             }
-        } else {
+        } else if (wid >= 4 && (phase_mask & 2u)) {
             // waves 4..7: zero the j in [16,32) pad of dt2_ej (read by the
             // dt1 GEMM's K loop; aliased memory holds stale m values)
             for (int z = tid - 256; z < EP * 2; z += 256) {
@@ -423,7 +425,7 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
 
         // ---- dt1 = W2^T · dt2, through relu1'/drop1' (both layouts) -------
 #pragma clang loop unroll(disable)
-        for (int s = 0; s < 4; ++s) {
+        for (int s = 0; s < ((phase_mask & 4u) ? 4 : 0); ++s) {
             const int tile = wid + s * 8;
             if (tile < 28) {
                 const int mt = tile >> 2, nt = tile & 3;
@@ -455,7 +457,7 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
         __syncthreads();
 
         // ---- read-only phase: dW2, dW1 accumulate; dm -> de ---------------
-        if (wid < 7) {  // dW2 += dt2 · t1^T   (A (j,e) b128, B (e,f) b128)
+        if (wid < 7 && (phase_mask & 8u)) {  // dW2 += dt2 · t1^T
             const int nt = wid;
 #pragma unroll
             for (int kb = 0; kb < 2; ++kb) {
@@ -467,7 +469,7 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
 #pragma unroll
         for (int s = 0; s < 12; ++s) {  // dW1 += dt1 · m^T (A (f,e), B (e,r))
             const int tile = wid + s * 8;
-            if (tile < 91) {
+            if (tile < 91 && (phase_mask & 8u)) {
                 const int mt = tile / 13, nt = tile % 13;
 #pragma unroll
                 for (int kb = 0; kb < 2; ++kb) {
@@ -478,7 +480,7 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
             }
         }
 #pragma clang loop unroll(disable)
-        for (int s = 0; s < 7; ++s) {  // dm^T = dt1^T · W1 (A (e,f) b128,
+        for (int s = 0; s < ((phase_mask & 16u) ? 7 : 0); ++s) {  // dm^T
             const int tile = wid + s * 8;  //       B (f,r) 8-scalar from w1t)
             if (tile < 52) {
                 const int emt = tile & 3, rnt = tile >> 2;
@@ -555,14 +557,14 @@ void front_bwd(const uint8_t* ids, const void* dseq, const void* w1,
                const float* b1, const void* w2, const float* b2,
                const void* emb, float* dw1, float* db1, float* dw2, float* db2,
                float* de, int B, uint32_t seed, float keep,
-               hipStream_t stream) {
+               hipStream_t stream, uint32_t phase_mask) {
     hipLaunchKernelGGL(front::front_bwd_kernel,
                        dim3(B, front::CSPLIT), dim3(512), 0, stream,
                        ids, static_cast<const bf16*>(dseq),
                        static_cast<const bf16*>(w1), b1,
                        static_cast<const bf16*>(w2), b2,
                        static_cast<const bf16*>(emb), dw1, db1, dw2, db2, de,
-                       B, seed, keep);
+                       B, seed, keep, phase_mask);
 }
 
 }  // namespace rk

@@ -111,35 +111,37 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gru_layer_fwd_kernel(
                 *reinterpret_cast<const bf16x8*>(src + (size_t)row * 2 * G3 + col);
         }
     }
-    bf16x8 stage[XCH];
+    bf16x8 stageA[XCH], stageB[XCH];
     if (T > 1) {
         const bf16* src = xg_src(tfirst + stp);
 #pragma unroll
         for (int p = 0; p < XCH; ++p) {
             int e = (p * WAVES * 64 + tid) * 8;
             int row = e / G3, col = e % G3;
-            stage[p] =
+            stageA[p] =
                 *reinterpret_cast<const bf16x8*>(src + (size_t)row * 2 * G3 + col);
         }
     }
     __syncthreads();
 
     // ---- T-step recurrence, ONE barrier per step --------------------------
-    int cur = 0;
-    for (int ti = 0; ti < T; ++ti) {
+    // The loop is unrolled by TWO with alternating stage register sets so no
+    // register copy ever waits on the loads issued in the SAME step: the xg
+    // write into LDS only touches registers loaded a FULL step earlier (the
+    // v3 kernel's stage rotation copy forced an end-of-step vmcnt(0) on
+    // loads ~200 cycles old — profiles/pmc_gru_r01.txt's 36% SQ_WAIT_ANY).
+    auto body = [&](int ti, int curp, bf16x8 (&st_wr)[XCH],
+                    bf16x8 (&st_ld)[XCH]) {
         const int t = tfirst + stp * ti;
-
         // issue xg[t+2] loads FIRST (they retire next step), then write the
-        // staged xg[t+1] into the back LDS buffer (waits only on last step's
-        // loads — vmcnt counting keeps the new loads in flight)
-        bf16x8 stage2[XCH];
+        // staged xg[t+1] (loaded last step) into the back LDS buffer
         if (ti + 2 < T) {
             const bf16* src = xg_src(t + 2 * stp);
 #pragma unroll
             for (int p = 0; p < XCH; ++p) {
                 int e = (p * WAVES * 64 + tid) * 8;
                 int row = e / G3, col = e % G3;
-                stage2[p] =
+                st_ld[p] =
                     *reinterpret_cast<const bf16x8*>(src + (size_t)row * 2 * G3 + col);
             }
         }
@@ -148,7 +150,7 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gru_layer_fwd_kernel(
             for (int p = 0; p < XCH; ++p) {
                 int e = (p * WAVES * 64 + tid) * 8;
                 int row = e / G3, col = e % G3;
-                *reinterpret_cast<bf16x8*>(&lds.xgb[cur ^ 1][row][col]) = stage[p];
+                *reinterpret_cast<bf16x8*>(&lds.xgb[curp ^ 1][row][col]) = st_wr[p];
             }
         }
 
@@ -162,7 +164,7 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gru_layer_fwd_kernel(
         for (int kb = 0; kb < 4; ++kb) {
 #pragma unroll
             for (int mt = 0; mt < 2; ++mt) {
-                bf16x8 a = lds_load_a_frag(&lds.h[cur][0][0], mt * 16, kb * 32, HPAD);
+                bf16x8 a = lds_load_a_frag(&lds.h[curp][0][0], mt * 16, kb * 32, HPAD);
 #pragma unroll
                 for (int g = 0; g < 3; ++g)
                     acc[mt][g] = mfma16x16x32(a, ufrag[g][kb], acc[mt][g]);
@@ -177,35 +179,33 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gru_layer_fwd_kernel(
             for (int i = 0; i < 4; ++i) {
                 const int row = mt * 16 + lrow * 4 + i;
                 const int j = j0 + lcol;
-                const float xr = bf2f(lds.xgb[cur][row][0 * H + j]);
-                const float xz = bf2f(lds.xgb[cur][row][1 * H + j]);
-                const float xn = bf2f(lds.xgb[cur][row][2 * H + j]);
+                const float xr = bf2f(lds.xgb[curp][row][0 * H + j]);
+                const float xz = bf2f(lds.xgb[curp][row][1 * H + j]);
+                const float xn = bf2f(lds.xgb[curp][row][2 * H + j]);
                 const float hgn = acc[mt][2][i] + bhh_reg[2];
                 const float r = sigmoidf_dev(xr + acc[mt][0][i] + bhh_reg[0]);
                 const float z = sigmoidf_dev(xz + acc[mt][1][i] + bhh_reg[1]);
                 const float n = tanhf_dev(xn + r * hgn);
                 const float hnew = (1.0f - z) * n + z * hreg[mt][i];
                 hreg[mt][i] = hnew;
-                lds.h[cur ^ 1][row][j] = f2bf(hnew);
+                lds.h[curp ^ 1][row][j] = f2bf(hnew);
                 if constexpr (TRAIN) {
                     bf16x4 pk = {f2bf(r), f2bf(z), f2bf(n), f2bf(hgn)};
                     *reinterpret_cast<bf16x4*>(
-                        &lds.cache_st[cur ^ 1][row][4 * j]) = pk;
+                        &lds.cache_st[curp ^ 1][row][4 * j]) = pk;
                 }
             }
         }
-        __syncthreads();  // h[cur^1] (+ cache staging) published
+        __syncthreads();  // h[curp^1] (+ cache staging) published
 
         // cooperative wide store of h to hseq (coalesced 16B per lane),
         // overlaps the next step's MFMAs as plain VMEM traffic
         {
             bf16* dst = hseq + (((size_t)t * B + b0) * 2 + dir) * H;
-            {
-                const int e = tid * 8;
-                const int row = e / H, col = e % H;
-                *reinterpret_cast<bf16x8*>(dst + (size_t)row * 2 * H + col) =
-                    *reinterpret_cast<const bf16x8*>(&lds.h[cur ^ 1][row][col]);
-            }
+            const int e = tid * 8;
+            const int row = e / H, col = e % H;
+            *reinterpret_cast<bf16x8*>(dst + (size_t)row * 2 * H + col) =
+                *reinterpret_cast<const bf16x8*>(&lds.h[curp ^ 1][row][col]);
         }
         if constexpr (TRAIN) {
             bf16* dst = cache + (((size_t)t * B + b0) * 2 + dir) * 4 * H;
@@ -214,13 +214,17 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gru_layer_fwd_kernel(
                 const int e = (p * WAVES * 64 + tid) * 8;
                 const int row = e / (4 * H), col = e % (4 * H);
                 *reinterpret_cast<bf16x8*>(dst + (size_t)row * 2 * 4 * H + col) =
-                    *reinterpret_cast<const bf16x8*>(&lds.cache_st[cur ^ 1][row][col]);
+                    *reinterpret_cast<const bf16x8*>(&lds.cache_st[curp ^ 1][row][col]);
             }
         }
-#pragma unroll
-        for (int p = 0; p < XCH; ++p) stage[p] = stage2[p];
-        cur ^= 1;
+    };
+
+    int ti = 0;
+    for (; ti + 2 <= T; ti += 2) {
+        body(ti, 0, stageA, stageB);
+        body(ti + 1, 1, stageB, stageA);
     }
+    if (ti < T) body(ti, 0, stageA, stageB);
 }
 
 void gru_layer_fwd(const void* xg, const void* u, const float* bhh, void* hseq,

@@ -35,6 +35,7 @@ from .config import TrainConfig
 from .datasets import InMemoryTrainDataset, TrainDataset
 from .model import RokoModel
 from .parallel.ddp import GradReducer, init_distributed
+from .utils.metrics import Meter
 
 
 class EarlyStopper:
@@ -191,6 +192,7 @@ def train(
 
     step = 0
     history = []
+    meter = Meter("train", rank=rank)
     for epoch in range(start_epoch, cfg.epochs + 1):
         if sampler is not None:
             sampler.set_epoch(epoch)
@@ -208,6 +210,7 @@ def train(
             run_loss += loss.item()
             n_batches += 1
             step += 1
+            meter.add(windows=len(x))
             if max_steps is not None and step >= max_steps:
                 break
         dt = time.time() - t0
@@ -236,6 +239,7 @@ def train(
                 log(f"early stop at epoch {epoch} (no val-acc gain in {cfg.patience})")
             break
 
+    meter.close()
     reducer.remove()
     return model, history
 
