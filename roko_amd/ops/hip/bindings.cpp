@@ -30,8 +30,10 @@ void front_fwd(const uint8_t* ids, const void* w1, const float* b1,
 void front_bwd(const uint8_t* ids, const void* dseq, const void* w1,
                const float* b1, const void* w2, const float* b2,
                const void* emb, float* dw1, float* db1, float* dw2, float* db2,
-               float* de, int B, uint32_t seed, float keep,
+               void* dt1g, int B, uint32_t seed, float keep,
                hipStream_t stream, uint32_t phase_mask);
+void front_de(const uint8_t* ids, const void* dt1g, const void* w1, float* de,
+              int B, uint32_t seed, float keep, hipStream_t stream);
 }  // namespace rk
 
 namespace {
@@ -238,12 +240,18 @@ std::vector<torch::Tensor> front_bwd(torch::Tensor ids, torch::Tensor dseq,
     auto dw2 = torch::zeros({10, 100}, opt);
     auto db2 = torch::zeros({10}, opt);
     auto de = torch::zeros({12, 50}, opt);
+    // dt1 workspace feeds the standalone embedding-grad kernel
+    auto dt1g = torch::empty({B, 90, 112, 64}, dseq.options());
     rk::front_bwd(ids.data_ptr<uint8_t>(), dseq.data_ptr(), w1.data_ptr(),
                   b1.data_ptr<float>(), w2.data_ptr(), b2.data_ptr<float>(),
                   emb.data_ptr(), dw1.data_ptr<float>(), db1.data_ptr<float>(),
                   dw2.data_ptr<float>(), db2.data_ptr<float>(),
-                  de.data_ptr<float>(), B, (uint32_t)seed, (float)keep,
+                  dt1g.data_ptr(), B, (uint32_t)seed, (float)keep,
                   cur_stream(), (uint32_t)phase_mask);
+    if (phase_mask == 0x1F)
+        rk::front_de(ids.data_ptr<uint8_t>(), dt1g.data_ptr(), w1.data_ptr(),
+                     de.data_ptr<float>(), B, (uint32_t)seed, (float)keep,
+                     cur_stream());
     return {de, dw1, db1, dw2, db2};
 }
 

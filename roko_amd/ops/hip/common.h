@@ -20,6 +20,7 @@ namespace rk {
 using bf16 = __hip_bfloat16;
 using f32x4 = __attribute__((__vector_size__(4 * sizeof(float)))) float;
 using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
+using bf16x4 = __attribute__((ext_vector_type(4))) __bf16;
 using short4_t = __attribute__((ext_vector_type(4))) short;
 
 RK_DEV float bf2f(bf16 v) { return __bfloat162float(v); }

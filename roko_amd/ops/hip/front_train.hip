@@ -232,7 +232,12 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
     float* __restrict__ db1,  // (F1)
     float* __restrict__ dw2,  // (F2, F1)
     float* __restrict__ db2,  // (F2)
-    float* __restrict__ de,   // (12, E)
+    bf16* __restrict__ dt1g,  // (B, W, MP, EP) dt1 pre-activation grads —
+                              // consumed by front_de_kernel (dm/de split off
+                              // into its own kernel where a transposed W1
+                              // tile fits LDS; scalar W1 reads made the dm
+                              // phase HALF this kernel's time,
+                              // profiles/front_bwd_phases_r01)
     int B, uint32_t seed, float keep, uint32_t phase_mask) {
     // phase_mask: timing-experiment switch (default 0x1F = all phases).
     // bit0 G1 recompute, bit1 G3+dt2, bit2 dt1, bit3 dW1/dW2, bit4 dm/de.
@@ -245,8 +250,7 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
             bf16 m_t[EP][KP_LD];  // [e][r]: G1 B-operand (dead after G1)
             struct {              // live from the G3 phase on
                 bf16 dt2_je[16][72];   // [j][e]: dW2 A-operand
-                bf16 dt2_ej[EP][40];   // [e][j]: dt1 B-operand
-                bf16 dt1_ef[EP][136];  // [e][f]: dm A-operand
+                bf16 dt2_ej[EP][40];   // [e][j]: dt1 A-operand
             } g;
         } u1;
         union {
@@ -259,12 +263,11 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
                                  // A-fragment's row 12..15 reads run past
                                  // the array (garbage), feeding accumulator
                                  // rows j >= 10 that every epilogue ignores
-        bf16 w2t_t[MP][40];      // [f][j]: dt1 A-operand
+        bf16 w2t_t[MP][40];      // [f][j]: dt1^T B-operand
         bf16 dseq_st[OUT];
         bf16 emb_s[12][E];
         bf16 b1s[MP];
         bf16 b2s[16];
-        float de_lds[12][E];
         float db1_lds[F1];
         float db2_lds[16];
         uint8_t col_ids[R];
@@ -289,7 +292,6 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
     for (int e = tid; e < MP * 72; e += 512) (&lds.t1_fe[0][0])[e] = f2bf(0.f);
     for (int e = tid; e < 12 * 136; e += 512) (&lds.w2_lds[0][0])[e] = f2bf(0.f);
     for (int e = tid; e < MP * 40; e += 512) (&lds.w2t_t[0][0])[e] = f2bf(0.f);
-    for (int e = tid; e < 12 * E; e += 512) (&lds.de_lds[0][0])[e] = 0.f;
     for (int e = tid; e < F1; e += 512) lds.db1_lds[e] = 0.f;
     for (int e = tid; e < 16; e += 512) lds.db2_lds[e] = 0.f;
     for (int e = tid; e < MP; e += 512) lds.b1s[e] = f2bf(0.f);
@@ -423,40 +425,44 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
         }
         __syncthreads();
 
-        // ---- dt1 = W2^T · dt2, through relu1'/drop1' (both layouts) -------
+        // ---- dt1^T = dt2^T · W2, through relu1'/drop1' --------------------
+        // (rows = e so each lane holds 4 CONSECUTIVE e values: the global
+        // dt1g store packs them into one 8-byte write)
 #pragma clang loop unroll(disable)
         for (int s = 0; s < ((phase_mask & 4u) ? 4 : 0); ++s) {
             const int tile = wid + s * 8;
             if (tile < 28) {
-                const int mt = tile >> 2, nt = tile & 3;
-                bf16x8 a = lds_load_a_frag(&lds.w2t_t[0][0], mt * 16, 0, 40);
-                bf16x8 bb = lds_load_b_frag_t(&lds.u1.g.dt2_ej[0][0], nt * 16, 0, 40);
+                const int emt = tile & 3, fnt = tile >> 2;
+                bf16x8 a = lds_load_a_frag(&lds.u1.g.dt2_ej[0][0], emt * 16, 0, 40);
+                bf16x8 bb = lds_load_b_frag_t(&lds.w2t_t[0][0], fnt * 16, 0, 40);
                 f32x4 acc = mfma16x16x32(a, bb, f32x4{0.f, 0.f, 0.f, 0.f});
-                const int e = nt * 16 + lcol;
+                const int f = fnt * 16 + lcol;
+                bf16x4 pk;
+                float fsum = 0.f;
 #pragma unroll
                 for (int i = 0; i < 4; ++i) {
-                    const int f = mt * 16 + lrow * 4 + i;
+                    const int e = emt * 16 + lrow * 4 + i;
                     // post > 0 <=> kept AND pre > 0 (chain rule collapses)
                     const float t1post = bf2f(lds.t1_fe[f][e]);
                     const float g = (t1post > 0.f) ? acc[i] * inv_keep : 0.f;
                     lds.u2.dt1_fe[f][e] = f2bf(g);
-                    lds.u1.g.dt1_ef[e][f] = f2bf(g);
-                    float red = g;
-#pragma unroll
-                    for (int moff = 1; moff < 16; moff <<= 1)
-                        red += __shfl_xor(red, moff, 16);
-                    if (lcol == 0 && f < F1) atomicAdd(&lds.db1_lds[f], red);
+                    pk[i] = f2bf(g);
+                    fsum += g;
                 }
+                {
+                    bf16* dst = dt1g +
+                        (((size_t)b * W + w) * MP + f) * EP + emt * 16 + lrow * 4;
+                    *reinterpret_cast<bf16x4*>(dst) = pk;
+                }
+                // db1[f]: sum this lane's 4 e's, then fold the 4 lrow groups
+                fsum += __shfl_xor(fsum, 16);
+                fsum += __shfl_xor(fsum, 32);
+                if (lrow == 0 && f < F1) atomicAdd(&lds.db1_lds[f], fsum);
             }
-        }
-        // zero dt1_ef's K-pad cols f in [112,136) (read by dm's A loop)
-        for (int z = tid; z < EP * 3; z += 512) {
-            const int e = z / 3, c = 112 + (z % 3) * 8;
-            *reinterpret_cast<bf16x8*>(&lds.u1.g.dt1_ef[e][c]) = bf16x8{};
         }
         __syncthreads();
 
-        // ---- read-only phase: dW2, dW1 accumulate; dm -> de ---------------
+        // ---- read-only phase: dW2 and dW1 fragment accumulation -----------
         if (wid < 7 && (phase_mask & 8u)) {  // dW2 += dt2 · t1^T
             const int nt = wid;
 #pragma unroll
@@ -476,31 +482,6 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
                     bf16x8 a = lds_load_a_frag(&lds.u2.dt1_fe[0][0], mt * 16, kb * 32, 72);
                     bf16x8 bb = lds_load_b_frag_t(&lds.m_rt[0][0], nt * 16, kb * 32, 72);
                     dw1acc[s] = mfma16x16x32(a, bb, dw1acc[s]);
-                }
-            }
-        }
-#pragma clang loop unroll(disable)
-        for (int s = 0; s < ((phase_mask & 16u) ? 7 : 0); ++s) {  // dm^T
-            const int tile = wid + s * 8;  //       B (f,r) 8-scalar from w1t)
-            if (tile < 52) {
-                const int emt = tile & 3, rnt = tile >> 2;
-                f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-#pragma unroll
-                for (int kb = 0; kb < 4; ++kb) {
-                    bf16x8 a = lds_load_a_frag(&lds.u1.g.dt1_ef[0][0], emt * 16, kb * 32, 136);
-                    bf16x8 bb = lds_load_b_frag_km(&lds.w1t[0][0], kb * 32, rnt * 16, KP_LD);
-                    acc = mfma16x16x32(a, bb, acc);
-                }
-#pragma unroll
-                for (int i = 0; i < 4; ++i) {
-                    const int e = emt * 16 + lrow * 4 + i;
-                    const int r = rnt * 16 + lcol;
-                    if (r < R && e < E) {
-                        const uint32_t h = mask_hash(seed, ekey(b, w, r >> 1, e));
-                        if (keep_half(h, r & 1, thresh16))
-                            atomicAdd(&lds.de_lds[lds.col_ids[r]][e],
-                                      acc[i] * inv_keep);
-                    }
                 }
             }
         }
@@ -535,6 +516,98 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
         if (lds.db1_lds[i] != 0.f) atomicAdd(&db1[i], lds.db1_lds[i]);
     for (int i = tid; i < F2; i += 512)
         if (lds.db2_lds[i] != 0.f) atomicAdd(&db2[i], lds.db2_lds[i]);
+}
+
+// ---------------------------------------------------------------------------
+// embedding gradient (dm -> de), standalone
+// ---------------------------------------------------------------------------
+// de[c,e] = sum over (b,w,r) of [ids==c] * maskE(r,e) * dm[b,w,r,e], with
+// dm = W1^T · dt1 recomputed per column from the dt1g workspace. Kept out of
+// front_bwd_kernel because the dm GEMM needs W1 in [r][f] layout — both W1
+// layouts plus the backward's tiles exceed 160 KB LDS, and reading the
+// resident [f][r] copy with scalar fragment loads measured as HALF the
+// backward kernel (profiles/front_bwd_phases_r01). Here the transposed W1
+// tile has LDS to itself and every operand is a b128 fragment read.
+
+__global__ __launch_bounds__(512, 2) void front_de_kernel(
+    const uint8_t* __restrict__ ids,   // (B, R, W)
+    const bf16* __restrict__ dt1g,     // (B, W, MP, EP)
+    const bf16* __restrict__ w1,       // (F1, R)
+    float* __restrict__ de,            // (12, E) pre-zeroed
+    int B, uint32_t seed, float keep) {
+    __shared__ struct {
+        bf16 w1_rt[R + 8][136];   // [r][f] = W1^T, zero-padded (A-operand)
+        bf16 dt1_ef[EP][136];     // [e][f] staged column of dt1g (B-operand)
+        float de_lds[12][E];
+        uint8_t col_ids[R];
+    } lds;
+
+    const int b = blockIdx.x;
+    const int tid = threadIdx.x;
+    const int wid = tid >> 6;
+    const int lane = tid & 63;
+    const int lrow = lane >> 4;
+    const int lcol = lane & 15;
+    const uint32_t thresh16 = (uint32_t)(keep * 65536.0f);
+    const float inv_keep = 1.0f / keep;
+    const int w_begin = blockIdx.y * (W / CSPLIT);
+    const int w_end = (blockIdx.y + 1 == CSPLIT) ? W : w_begin + W / CSPLIT;
+
+    for (int e = tid; e < (R + 8) * 136; e += 512)
+        (&lds.w1_rt[0][0])[e] = f2bf(0.f);
+    for (int e = tid; e < EP * 136; e += 512)
+        (&lds.dt1_ef[0][0])[e] = f2bf(0.f);
+    for (int e = tid; e < 12 * E; e += 512) (&lds.de_lds[0][0])[e] = 0.f;
+    __syncthreads();
+    for (int e = tid; e < F1 * R; e += 512)
+        lds.w1_rt[e % R][e / R] = w1[e];
+    __syncthreads();
+
+    for (int w = w_begin; w < w_end; ++w) {
+        if (tid < R) lds.col_ids[tid] = ids[((size_t)b * R + tid) * W + w];
+        // stage dt1[b, w] (MP, EP) -> [e][f]: vector global reads, scalar
+        // transposed LDS writes
+        {
+            const bf16* src = dt1g + ((size_t)b * W + w) * MP * EP;
+            for (int p = tid; p < MP * EP / 8; p += 512) {
+                const int f = p / (EP / 8), e0 = (p % (EP / 8)) * 8;
+                bf16 v[8];
+                *reinterpret_cast<bf16x8*>(v) = *reinterpret_cast<const bf16x8*>(
+                    src + (size_t)f * EP + e0);
+#pragma unroll
+                for (int q = 0; q < 8; ++q) lds.dt1_ef[e0 + q][f] = v[q];
+            }
+        }
+        __syncthreads();
+
+        // dm[r, e] = W1^T · dt1 — 13x4 tiles over 8 waves, all b128
+#pragma clang loop unroll(disable)
+        for (int s = 0; s < 7; ++s) {
+            const int tile = wid + s * 8;
+            if (tile < 52) {
+                const int rmt = tile >> 2, ent = tile & 3;
+                f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+                for (int kb = 0; kb < 4; ++kb) {
+                    bf16x8 a = lds_load_a_frag(&lds.w1_rt[0][0], rmt * 16, kb * 32, 136);
+                    bf16x8 bb = lds_load_b_frag_t(&lds.dt1_ef[0][0], ent * 16, kb * 32, 136);
+                    acc = mfma16x16x32(a, bb, acc);
+                }
+#pragma unroll
+                for (int i = 0; i < 4; ++i) {
+                    const int r = rmt * 16 + lrow * 4 + i;
+                    const int e = ent * 16 + lcol;
+                    if (r < R && e < E) {
+                        const uint32_t h = mask_hash(seed, ekey(b, w, r >> 1, e));
+                        if (keep_half(h, r & 1, thresh16))
+                            atomicAdd(&lds.de_lds[lds.col_ids[r]][e],
+                                      acc[i] * inv_keep);
+                    }
+                }
+            }
+        }
+        __syncthreads();
+    }
     for (int i = tid; i < 12 * E; i += 512)
         if ((&lds.de_lds[0][0])[i] != 0.f)
             atomicAdd(&de[i], (&lds.de_lds[0][0])[i]);
@@ -553,18 +626,26 @@ void front_fwd(const uint8_t* ids, const void* w1, const float* b1,
                        B, seed, keep);
 }
 
+void front_de(const uint8_t* ids, const void* dt1g, const void* w1, float* de,
+              int B, uint32_t seed, float keep, hipStream_t stream) {
+    hipLaunchKernelGGL(front::front_de_kernel,
+                       dim3(B, front::CSPLIT), dim3(512), 0, stream,
+                       ids, static_cast<const bf16*>(dt1g),
+                       static_cast<const bf16*>(w1), de, B, seed, keep);
+}
+
 void front_bwd(const uint8_t* ids, const void* dseq, const void* w1,
                const float* b1, const void* w2, const float* b2,
                const void* emb, float* dw1, float* db1, float* dw2, float* db2,
-               float* de, int B, uint32_t seed, float keep,
+               void* dt1g, int B, uint32_t seed, float keep,
                hipStream_t stream, uint32_t phase_mask) {
     hipLaunchKernelGGL(front::front_bwd_kernel,
                        dim3(B, front::CSPLIT), dim3(512), 0, stream,
                        ids, static_cast<const bf16*>(dseq),
                        static_cast<const bf16*>(w1), b1,
                        static_cast<const bf16*>(w2), b2,
-                       static_cast<const bf16*>(emb), dw1, db1, dw2, db2, de,
-                       B, seed, keep, phase_mask);
+                       static_cast<const bf16*>(emb), dw1, db1, dw2, db2,
+                       static_cast<bf16*>(dt1g), B, seed, keep, phase_mask);
 }
 
 }  // namespace rk

@@ -45,8 +45,6 @@ constexpr int WAVES = 8;     // 512 threads, two waves per SIMD: a wave's
 constexpr int HPAD = H + 8;  // LDS row padding (bank-conflict fix)
 constexpr int XCH = MB * G3 / (WAVES * 64 * 8);  // xg chunks per thread (3)
 
-using bf16x4 = __attribute__((ext_vector_type(4))) __bf16;
-
 template <bool TRAIN>
 __global__ __launch_bounds__(WAVES * 64, 2) void gru_layer_fwd_kernel(
     const bf16* __restrict__ xg,   // (T, B, 2, 3H)  W_ih·x + b_ih
