@@ -534,7 +534,8 @@ __global__ __launch_bounds__(512, 2) void front_de_kernel(
     const bf16* __restrict__ dt1g,     // (B, W, MP, EP)
     const bf16* __restrict__ w1,       // (F1, R)
     float* __restrict__ de,            // (12, E) pre-zeroed
-    int B, uint32_t seed, float keep) {
+    int B, uint32_t seed, float keep,
+    unsigned long long* __restrict__ timing) {  // optional (4): stage/gemm/epi/total cycles
     __shared__ struct {
         bf16 w1_rt[R + 8][136];   // [r][f] = W1^T, zero-padded (A-operand)
         bf16 dt1_ef[EP][136];     // [e][f] staged column of dt1g (B-operand)
@@ -563,7 +564,10 @@ __global__ __launch_bounds__(512, 2) void front_de_kernel(
         lds.w1_rt[e % R][e / R] = w1[e];
     __syncthreads();
 
+    unsigned long long tacc[3] = {0, 0, 0};
+    unsigned long long tw0 = (timing && tid == 0) ? __builtin_amdgcn_s_memtime() : 0;
     for (int w = w_begin; w < w_end; ++w) {
+        unsigned long long tp0 = (timing && tid == 0) ? __builtin_amdgcn_s_memtime() : 0;
         if (tid < R) lds.col_ids[tid] = ids[((size_t)b * R + tid) * W + w];
         // stage dt1[b, w] (MP, EP) -> [e][f]: vector global reads, scalar
         // transposed LDS writes
@@ -579,6 +583,11 @@ __global__ __launch_bounds__(512, 2) void front_de_kernel(
             }
         }
         __syncthreads();
+        if (timing && tid == 0) {
+            unsigned long long t1 = __builtin_amdgcn_s_memtime();
+            tacc[0] += t1 - tp0;
+            tp0 = t1;
+        }
 
         // dm[r, e] = W1^T · dt1 — 13x4 tiles over 8 waves, all b128
 #pragma clang loop unroll(disable)
@@ -607,6 +616,12 @@ __global__ __launch_bounds__(512, 2) void front_de_kernel(
             }
         }
         __syncthreads();
+        if (timing && tid == 0) tacc[1] += __builtin_amdgcn_s_memtime() - tp0;
+    }
+    if (timing && tid == 0) {
+        atomicAdd(&timing[0], tacc[0]);
+        atomicAdd(&timing[1], tacc[1]);
+        atomicAdd(&timing[3], __builtin_amdgcn_s_memtime() - tw0);
     }
     for (int i = tid; i < 12 * E; i += 512)
         if ((&lds.de_lds[0][0])[i] != 0.f)
@@ -627,11 +642,12 @@ void front_fwd(const uint8_t* ids, const void* w1, const float* b1,
 }
 
 void front_de(const uint8_t* ids, const void* dt1g, const void* w1, float* de,
-              int B, uint32_t seed, float keep, hipStream_t stream) {
+              int B, uint32_t seed, float keep, hipStream_t stream,
+              unsigned long long* timing) {
     hipLaunchKernelGGL(front::front_de_kernel,
                        dim3(B, front::CSPLIT), dim3(512), 0, stream,
                        ids, static_cast<const bf16*>(dt1g),
-                       static_cast<const bf16*>(w1), de, B, seed, keep);
+                       static_cast<const bf16*>(w1), de, B, seed, keep, timing);
 }
 
 void front_bwd(const uint8_t* ids, const void* dseq, const void* w1,

@@ -256,8 +256,9 @@ void gru_layer_fwd(const void* xg, const void* u, const float* bhh, void* hseq,
 // ---------------------------------------------------------------------------
 
 constexpr int BW_WAVES = 8;  // dhg GEMM has K=384: 24 MFMA/wave at 8 waves
+constexpr int BWCH_C = 4;    // cache chunks per thread (32x512 / 512 / 8)
 
-__global__ __launch_bounds__(BW_WAVES * 64) void gru_layer_bwd_kernel(
+__global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
     const bf16* __restrict__ cache,  // (T, B, 2, H, 4) [r z n hgn] packed
     const bf16* __restrict__ hseq,   // (T, B, 2, H)
     const bf16* __restrict__ dhin,   // (T, B, 2, H) grad wrt layer output
@@ -265,11 +266,11 @@ __global__ __launch_bounds__(BW_WAVES * 64) void gru_layer_bwd_kernel(
     bf16* __restrict__ dg,           // (T, B, 2, 4H) out: [dxr dxz dxn dhgn]
     int T, int B) {
     __shared__ struct {
-        bf16 cache_st[MB][4 * H];   // staged cache[t] (packed layout)
-        bf16 dhin_st[MB][H];        // staged dhin[t]
-        bf16 hprev_st[MB][H];       // staged h_{t-1}
-        bf16 dhg[MB][G3 + 8];       // A-operand of the dh GEMM
-        bf16 dg_st[MB][4 * H];      // staged output tile
+        bf16 cache_st[2][MB][4 * H];  // double-buffered staged cache[t]
+        bf16 dhin_st[2][MB][H];       // double-buffered staged dhin[t]
+        bf16 hprev_st[2][MB][H];      // double-buffered staged h_{t-1}
+        bf16 dhg[MB][G3 + 8];         // A-operand of the dh GEMM
+        bf16 dg_st[MB][4 * H];        // staged output tile
     } lds;
 
     const int dir = blockIdx.y;
@@ -297,36 +298,58 @@ __global__ __launch_bounds__(BW_WAVES * 64) void gru_layer_bwd_kernel(
 #pragma unroll
         for (int i = 0; i < 4; ++i) dhc[mt][i] = 0.0f;
 
-    for (int s = 0; s < T; ++s) {
-        const int t = (dir == 0) ? T - 1 - s : s;
-        const int tp = (dir == 0) ? t - 1 : t + 1;  // h_prev timestep
-        // ---- stage cache[t], dhin[t], hprev -------------------------------
-        {
-            const bf16* src = cache + (((size_t)t * B + b0) * 2 + dir) * 4 * H;
-            const int row = tid / 64, col = (tid % 64) * 8;
-#pragma unroll
-            for (int q = 0; q < 4; ++q)
-                *reinterpret_cast<bf16x8*>(&lds.cache_st[row + q * 8][col]) =
-                    *reinterpret_cast<const bf16x8*>(
-                        src + (size_t)(row + q * 8) * 2 * 4 * H + col);
-        }
-        {
-            const bf16* src = dhin + (((size_t)t * B + b0) * 2 + dir) * H;
-            const int row = tid / 16, col = (tid % 16) * 8;
-            *reinterpret_cast<bf16x8*>(&lds.dhin_st[row][col]) =
-                *reinterpret_cast<const bf16x8*>(src + (size_t)row * 2 * H + col);
-            if (tp >= 0 && tp < T) {
-                const bf16* hp = hseq + (((size_t)tp * B + b0) * 2 + dir) * H;
-                *reinterpret_cast<bf16x8*>(&lds.hprev_st[row][col]) =
-                    *reinterpret_cast<const bf16x8*>(hp + (size_t)row * 2 * H + col);
-            } else {
-                bf16x8 zero = {};
-                *reinterpret_cast<bf16x8*>(&lds.hprev_st[row][col]) = zero;
-            }
-        }
-        __syncthreads();
+    const int stp = (dir == 0) ? -1 : 1;     // BPTT walks t backwards
+    const int t0 = (dir == 0) ? T - 1 : 0;
+    auto t_of = [&](int sidx) { return t0 + stp * sidx; };
 
-        // ---- gate gradients ------------------------------------------------
+    // register-staged loads for one step: cache (4 chunks) + dhin + hprev
+    auto issue_loads = [&](int sidx, bf16x8 (&rc)[BWCH_C], bf16x8& rdh,
+                           bf16x8& rhp) {
+        const int t = t_of(sidx);
+        const int tp = (dir == 0) ? t - 1 : t + 1;
+        const bf16* src = cache + (((size_t)t * B + b0) * 2 + dir) * 4 * H;
+        const int row4 = tid / 64, col4 = (tid % 64) * 8;
+#pragma unroll
+        for (int q = 0; q < BWCH_C; ++q)
+            rc[q] = *reinterpret_cast<const bf16x8*>(
+                src + (size_t)(row4 + q * 8) * 2 * 4 * H + col4);
+        const int row = tid / 16, col = (tid % 16) * 8;
+        rdh = *reinterpret_cast<const bf16x8*>(
+            dhin + (((size_t)t * B + b0) * 2 + dir) * H + (size_t)row * 2 * H + col);
+        if (tp >= 0 && tp < T)
+            rhp = *reinterpret_cast<const bf16x8*>(
+                hseq + (((size_t)tp * B + b0) * 2 + dir) * H + (size_t)row * 2 * H + col);
+        else
+            rhp = bf16x8{};
+    };
+    auto write_stage = [&](int buf, bf16x8 (&rc)[BWCH_C], bf16x8& rdh,
+                           bf16x8& rhp) {
+        const int row4 = tid / 64, col4 = (tid % 64) * 8;
+#pragma unroll
+        for (int q = 0; q < BWCH_C; ++q)
+            *reinterpret_cast<bf16x8*>(&lds.cache_st[buf][row4 + q * 8][col4]) = rc[q];
+        const int row = tid / 16, col = (tid % 16) * 8;
+        *reinterpret_cast<bf16x8*>(&lds.dhin_st[buf][row][col]) = rdh;
+        *reinterpret_cast<bf16x8*>(&lds.hprev_st[buf][row][col]) = rhp;
+    };
+
+    bf16x8 rcA[BWCH_C], rcB[BWCH_C];
+    bf16x8 rdhA, rhpA, rdhB, rhpB;
+    // prologue: step 0 straight to LDS buffer 0; step 1 to registers A
+    issue_loads(0, rcA, rdhA, rhpA);
+    write_stage(0, rcA, rdhA, rhpA);
+    if (T > 1) issue_loads(1, rcA, rdhA, rhpA);
+    __syncthreads();
+
+    auto body = [&](int sidx, int curp, bf16x8 (&rc_wr)[BWCH_C], bf16x8& rdh_wr,
+                    bf16x8& rhp_wr, bf16x8 (&rc_ld)[BWCH_C], bf16x8& rdh_ld,
+                    bf16x8& rhp_ld) {
+        const int t = t_of(sidx);
+        // issue step s+2 loads FIRST, then write step s+1's staged registers
+        if (sidx + 2 < T) issue_loads(sidx + 2, rc_ld, rdh_ld, rhp_ld);
+        if (sidx + 1 < T) write_stage(curp ^ 1, rc_wr, rdh_wr, rhp_wr);
+
+        // ---- gate gradients from stage[curp] ------------------------------
         float dhp_part[2][4];
 #pragma unroll
         for (int mt = 0; mt < 2; ++mt) {
@@ -334,14 +357,14 @@ __global__ __launch_bounds__(BW_WAVES * 64) void gru_layer_bwd_kernel(
             for (int i = 0; i < 4; ++i) {
                 const int row = mt * 16 + lrow * 4 + i;
                 const int j = j0 + lcol;
-                const float dh = dhc[mt][i] + bf2f(lds.dhin_st[row][j]);
+                const float dh = dhc[mt][i] + bf2f(lds.dhin_st[curp][row][j]);
                 const bf16x4 pk = *reinterpret_cast<const bf16x4*>(
-                    &lds.cache_st[row][4 * j]);
+                    &lds.cache_st[curp][row][4 * j]);
                 const float r = bf2f(pk[0]);
                 const float z = bf2f(pk[1]);
                 const float n = bf2f(pk[2]);
                 const float hgn = bf2f(pk[3]);
-                const float hp = bf2f(lds.hprev_st[row][j]);
+                const float hp = bf2f(lds.hprev_st[curp][row][j]);
                 const float dn = dh * (1.0f - z);
                 const float dz = dh * (hp - n);
                 const float dan = dn * (1.0f - n * n);
@@ -359,7 +382,7 @@ __global__ __launch_bounds__(BW_WAVES * 64) void gru_layer_bwd_kernel(
                 lds.dg_st[row][3 * H + j] = f2bf(dhgn);
             }
         }
-        __syncthreads();
+        __syncthreads();  // dhg/dg_st and stage[curp^1] published
 
         // ---- dh_prev = dh*z + dhg · U  (24 MFMA per wave) -----------------
         f32x4 acc[2];
@@ -388,8 +411,15 @@ __global__ __launch_bounds__(BW_WAVES * 64) void gru_layer_bwd_kernel(
                     dst + (size_t)(row + q * 8) * 2 * 4 * H + col) =
                     *reinterpret_cast<const bf16x8*>(&lds.dg_st[row + q * 8][col]);
         }
-        __syncthreads();  // dg_st/cache_st reads done before next staging
+        __syncthreads();  // dhg/dg_st/stage[curp] reads done before reuse
+    };
+
+    int sidx = 0;
+    for (; sidx + 2 <= T; sidx += 2) {
+        body(sidx, 0, rcA, rdhA, rhpA, rcB, rdhB, rhpB);
+        body(sidx + 1, 1, rcB, rdhB, rhpB, rcA, rdhA, rhpA);
     }
+    if (sidx < T) body(sidx, 0, rcA, rdhA, rhpA, rcB, rdhB, rhpB);
 }
 
 void gru_layer_bwd(const void* cache, const void* hseq, const void* dhin,
