@@ -59,15 +59,29 @@ def _features_for_region(bam, contig, start, end, cfg: FeatureConfig):
 
 
 def generate_infer(args):
-    """Inference worker: windows only (reference: features.py:97-110)."""
+    """Inference worker: windows only (reference: features.py:97-110).
+    Failures are returned, not raised: one bad region must not kill the
+    whole run (the reference dies on any worker exception).
+    """
     bam_x, contig, start, end, cfg = args
-    positions, examples = _features_for_region(bam_x, contig, start, end, cfg)
+    try:
+        positions, examples = _features_for_region(bam_x, contig, start, end, cfg)
+    except Exception as e:  # noqa: BLE001 — deliberate region-level fence
+        return ("__error__", f"{contig}:{start}-{end}", repr(e))
     return contig, start, end, positions, examples, None
 
 
 def generate_train(args):
     """Training worker: windows joined with truth labels
-    (reference: features.py:37-94)."""
+    (reference: features.py:37-94). Failures are returned, not raised."""
+    try:
+        return _generate_train(args)
+    except Exception as e:  # noqa: BLE001 — deliberate region-level fence
+        _, _, contig, start, end, _ = args
+        return ("__error__", f"{contig}:{start}-{end}", repr(e))
+
+
+def _generate_train(args):
     bam_x, bam_y, contig, start, end, cfg = args
     from .labels import filter_aligns, get_aligns, get_pos_and_labels
 
@@ -162,9 +176,15 @@ def run(
     with RkwWriter(out_path, inference=inference) as writer:
         writer.write_contigs(refs)
 
+        n_errors = 0
+
         def consume(result):
-            nonlocal n_windows
+            nonlocal n_windows, n_errors
             if result is None:
+                return
+            if result[0] == "__error__":
+                n_errors += 1
+                log(f"WARNING: region {result[1]} failed: {result[2]} (skipped)")
                 return
             contig, start, end, positions, examples, labs = result
             writer.store(contig, start, end, positions, examples, labs)
@@ -179,7 +199,8 @@ def run(
                     consume(result)
     dt = time.time() - t0
     log(f"wrote {n_windows} windows to {out_path} in {dt:.1f}s "
-        f"({n_windows / max(dt, 1e-9):.0f} windows/s)")
+        f"({n_windows / max(dt, 1e-9):.0f} windows/s)"
+        + (f"; {n_errors} regions FAILED and were skipped" if n_errors else ""))
     return n_windows
 
 

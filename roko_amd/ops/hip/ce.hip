@@ -43,7 +43,7 @@ __global__ __launch_bounds__(256) void ce_fwd_bwd_kernel(
     // wave-level sum, one atomic per wave
 #pragma unroll
     for (int off = 32; off > 0; off >>= 1) loss += __shfl_down(loss, off, 64);
-    if ((threadIdx.x & 63) == 0) atomicAdd(loss_sum, loss);
+    if ((threadIdx.x & 63) == 0) agent_atomic_add(loss_sum, loss);
 }
 
 void ce_fwd_bwd(const float* logits, const int64_t* target, float* dlogits,

@@ -90,6 +90,18 @@ RK_DEV bool drop_keep(uint32_t seed, uint32_t idx, uint64_t thresh) {
     return (uint64_t)hash32(idx ^ (seed * 0x9E3779B9U)) < thresh;
 }
 
+// Relaxed atomic adds: plain atomicAdd defaults to seq_cst and hipcc fences
+// EVERY ds_add_f32 behind an s_waitcnt lgkmcnt(0) — a full LDS-pipe drain
+// per atomic (measured as the dominant cost of the de reduction). We only
+// need atomicity, never ordering: results are read after kernel end (global)
+// or after a __syncthreads() (LDS).
+RK_DEV void lds_atomic_add(float* p, float v) {
+    __hip_atomic_fetch_add(p, v, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_WORKGROUP);
+}
+RK_DEV void agent_atomic_add(float* p, float v) {
+    __hip_atomic_fetch_add(p, v, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+}
+
 RK_DEV float sigmoidf_dev(float x) { return 1.0f / (1.0f + __expf(-x)); }
 
 // overflow-stable tanh: tanh(x) = sign(x) * (1 - e) / (1 + e), e = exp(-2|x|)

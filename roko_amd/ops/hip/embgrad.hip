@@ -65,10 +65,10 @@ __global__ __launch_bounds__(256) void emb_grad_kernel(
     }
 #pragma unroll
     for (int cc = 0; cc < NIDS; ++cc)
-        if (acc[cc] != 0.0f) atomicAdd(&acc_lds[cc * ED + e], acc[cc]);
+        if (acc[cc] != 0.0f) lds_atomic_add(&acc_lds[cc * ED + e], acc[cc]);
     __syncthreads();
     for (int i = threadIdx.x; i < NIDS * ED; i += 256)
-        if (acc_lds[i] != 0.0f) atomicAdd(&de[i], acc_lds[i]);
+        if (acc_lds[i] != 0.0f) agent_atomic_add(&de[i], acc_lds[i]);
 }
 
 void emb_grad(const void* dm, const uint8_t* ids, float* de, int64_t n,

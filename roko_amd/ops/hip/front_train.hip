@@ -413,7 +413,7 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
 #pragma unroll
                 for (int moff = 1; moff < 16; moff <<= 1)
                     red += __shfl_xor(red, moff, 16);
-                if (lcol == 0 && j < F2) atomicAdd(&lds.db2_lds[j], red);
+                if (lcol == 0 && j < F2) lds_atomic_add(&lds.db2_lds[j], red);
             }
         } else if (wid >= 4 && (phase_mask & 2u)) {
             // waves 4..7: zero the j in [16,32) pad of dt2_ej (read by the
@@ -457,7 +457,7 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
                 // db1[f]: sum this lane's 4 e's, then fold the 4 lrow groups
                 fsum += __shfl_xor(fsum, 16);
                 fsum += __shfl_xor(fsum, 32);
-                if (lrow == 0 && f < F1) atomicAdd(&lds.db1_lds[f], fsum);
+                if (lrow == 0 && f < F1) lds_atomic_add(&lds.db1_lds[f], fsum);
             }
         }
         __syncthreads();
@@ -498,7 +498,7 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
             for (int i = 0; i < 4; ++i) {
                 const int f = mt * 16 + lrow * 4 + i;
                 const int r = nt * 16 + lcol;
-                if (f < F1 && r < R) atomicAdd(&dw1[f * R + r], dw1acc[s][i]);
+                if (f < F1 && r < R) agent_atomic_add(&dw1[f * R + r], dw1acc[s][i]);
             }
         }
     }
@@ -508,14 +508,14 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
         for (int i = 0; i < 4; ++i) {
             const int j = lrow * 4 + i;
             const int f = nt * 16 + lcol;
-            if (j < F2 && f < F1) atomicAdd(&dw2[j * F1 + f], dw2acc[i]);
+            if (j < F2 && f < F1) agent_atomic_add(&dw2[j * F1 + f], dw2acc[i]);
         }
     }
     __syncthreads();
     for (int i = tid; i < F1; i += 512)
-        if (lds.db1_lds[i] != 0.f) atomicAdd(&db1[i], lds.db1_lds[i]);
+        if (lds.db1_lds[i] != 0.f) agent_atomic_add(&db1[i], lds.db1_lds[i]);
     for (int i = tid; i < F2; i += 512)
-        if (lds.db2_lds[i] != 0.f) atomicAdd(&db2[i], lds.db2_lds[i]);
+        if (lds.db2_lds[i] != 0.f) agent_atomic_add(&db2[i], lds.db2_lds[i]);
 }
 
 // ---------------------------------------------------------------------------
@@ -609,8 +609,8 @@ __global__ __launch_bounds__(512, 2) void front_de_kernel(
                     if (r < R && e < E) {
                         const uint32_t h = mask_hash(seed, ekey(b, w, r >> 1, e));
                         if (keep_half(h, r & 1, thresh16))
-                            atomicAdd(&lds.de_lds[lds.col_ids[r]][e],
-                                      acc[i] * inv_keep);
+                            lds_atomic_add(&lds.de_lds[lds.col_ids[r]][e],
+                                           acc[i] * inv_keep);
                     }
                 }
             }
@@ -625,7 +625,7 @@ __global__ __launch_bounds__(512, 2) void front_de_kernel(
     }
     for (int i = tid; i < 12 * E; i += 512)
         if ((&lds.de_lds[0][0])[i] != 0.f)
-            atomicAdd(&de[i], (&lds.de_lds[0][0])[i]);
+            agent_atomic_add(&de[i], (&lds.de_lds[0][0])[i]);
 }
 
 }  // namespace front
