@@ -80,7 +80,7 @@ def roko_forward(model, x: torch.Tensor) -> torch.Tensor:
     B = ids.shape[0]
     T = C.WINDOW_COLS
 
-    seq = ext.embed_mlp_fwd(ids, w["w1"], w["b1"], w["w2"], w["b2"], w["emb"])
+    seq = ext.front_fwd(ids, w["w1"], w["b1"], w["w2"], w["b2"], w["emb"], 0, 1.0)
     for l in range(C.NUM_LAYERS):
         xg = torch.addmm(
             w[f"b_ih{l}"], seq.reshape(T * B, -1), w[f"w_ih_t{l}"]
@@ -131,8 +131,8 @@ class InferencePipeline:
         w = self.w
         B = ids_u8.shape[0]
         T = C.WINDOW_COLS
-        seq = ext.embed_mlp_fwd(ids_u8, w["w1"], w["b1"], w["w2"], w["b2"],
-                                w["emb"])
+        seq = ext.front_fwd(ids_u8, w["w1"], w["b1"], w["w2"], w["b2"],
+                            w["emb"], 0, 1.0)
         for l in range(C.NUM_LAYERS):
             xg = torch.addmm(
                 w[f"b_ih{l}"], seq.reshape(T * B, -1), w[f"w_ih_t{l}"]
@@ -234,7 +234,7 @@ def roko_argmax(model, x: torch.Tensor) -> torch.Tensor:
         ids = torch.cat([ids, ids.new_zeros(pad, *ids.shape[1:])])
     B = ids.shape[0]
     T = C.WINDOW_COLS
-    seq = ext.embed_mlp_fwd(ids, w["w1"], w["b1"], w["w2"], w["b2"], w["emb"])
+    seq = ext.front_fwd(ids, w["w1"], w["b1"], w["w2"], w["b2"], w["emb"], 0, 1.0)
     for l in range(C.NUM_LAYERS):
         xg = torch.addmm(
             w[f"b_ih{l}"], seq.reshape(T * B, -1), w[f"w_ih_t{l}"]

@@ -107,6 +107,7 @@ __global__ __launch_bounds__(512, 2) void front_fwd_kernel(
     const int lcol = lane & 15;
     const uint32_t thresh16 = (uint32_t)(keep * 65536.0f);
     const float inv_keep = 1.0f / keep;
+    const bool no_drop = keep >= 1.0f;  // eval/inference: skip every hash
     const int w_begin = blockIdx.y * (W / CSPLIT);
     const int w_end = (blockIdx.y + 1 == CSPLIT) ? W : w_begin + W / CSPLIT;
 
@@ -130,18 +131,76 @@ __global__ __launch_bounds__(512, 2) void front_fwd_kernel(
         if (tid < R) lds.col_ids[tid] = ids[((size_t)b * R + tid) * W + w];
         __syncthreads();
         // ---- masked embedding tile, one hash per read PAIR ----------------
-        for (int p = tid; p < E * (R / 2); p += 512) {
-            const int e = p / (R / 2), q = p % (R / 2);
-            const int r0 = 2 * q;
-            const uint32_t h = mask_hash(seed, ekey(b, w, q, e));
-            float v0 = 0.f, v1 = 0.f;
-            if (keep_half(h, 0, thresh16))
-                v0 = bf2f(lds.emb_s[lds.col_ids[r0]][e]) * inv_keep;
-            if (keep_half(h, 1, thresh16))
-                v1 = bf2f(lds.emb_s[lds.col_ids[r0 + 1]][e]) * inv_keep;
-            bf16 pk[2] = {f2bf(v0), f2bf(v1)};
-            *reinterpret_cast<uint32_t*>(&lds.m_t[e][r0]) =
-                *reinterpret_cast<const uint32_t*>(pk);
+        if (no_drop) {
+            uint8_t bid0[10], bid1[10];
+            bf16 bev0[10], bev1[10];
+#pragma unroll
+            for (int k = 0; k < 10; ++k) {
+                const int p = tid + k * 512;
+                if (p < E * (R / 2)) {
+                    const int q = p % (R / 2);
+                    bid0[k] = lds.col_ids[2 * q];
+                    bid1[k] = lds.col_ids[2 * q + 1];
+                }
+            }
+#pragma unroll
+            for (int k = 0; k < 10; ++k) {
+                const int p = tid + k * 512;
+                if (p < E * (R / 2)) {
+                    const int e = p / (R / 2);
+                    bev0[k] = lds.emb_s[bid0[k]][e];
+                    bev1[k] = lds.emb_s[bid1[k]][e];
+                }
+            }
+#pragma unroll
+            for (int k = 0; k < 10; ++k) {
+                const int p = tid + k * 512;
+                if (p < E * (R / 2)) {
+                    const int e = p / (R / 2), q = p % (R / 2);
+                    bf16 pk[2] = {bev0[k], bev1[k]};
+                    *reinterpret_cast<uint32_t*>(&lds.m_t[e][2 * q]) =
+                        *reinterpret_cast<const uint32_t*>(pk);
+                }
+            }
+        } else {
+            // three passes so the id reads, the embedding reads and the
+            // stores each issue as one batch — the naive fused loop was a
+            // col_ids -> emb -> store dependent LDS chain per pair
+            uint8_t bid0[10], bid1[10];
+            bf16 bev0[10], bev1[10];
+#pragma unroll
+            for (int k = 0; k < 10; ++k) {
+                const int p = tid + k * 512;
+                if (p < E * (R / 2)) {
+                    const int q = p % (R / 2);
+                    bid0[k] = lds.col_ids[2 * q];
+                    bid1[k] = lds.col_ids[2 * q + 1];
+                }
+            }
+#pragma unroll
+            for (int k = 0; k < 10; ++k) {
+                const int p = tid + k * 512;
+                if (p < E * (R / 2)) {
+                    const int e = p / (R / 2);
+                    bev0[k] = lds.emb_s[bid0[k]][e];
+                    bev1[k] = lds.emb_s[bid1[k]][e];
+                }
+            }
+#pragma unroll
+            for (int k = 0; k < 10; ++k) {
+                const int p = tid + k * 512;
+                if (p < E * (R / 2)) {
+                    const int e = p / (R / 2), q = p % (R / 2);
+                    const uint32_t h = mask_hash(seed, ekey(b, w, q, e));
+                    float v0 = keep_half(h, 0, thresh16)
+                                   ? bf2f(bev0[k]) * inv_keep : 0.f;
+                    float v1 = keep_half(h, 1, thresh16)
+                                   ? bf2f(bev1[k]) * inv_keep : 0.f;
+                    bf16 pk[2] = {f2bf(v0), f2bf(v1)};
+                    *reinterpret_cast<uint32_t*>(&lds.m_t[e][2 * q]) =
+                        *reinterpret_cast<const uint32_t*>(pk);
+                }
+            }
         }
         __syncthreads();
 
@@ -162,15 +221,15 @@ __global__ __launch_bounds__(512, 2) void front_fwd_kernel(
 #pragma unroll
                 for (int i = 0; i < 4; i += 2) {
                     const int f = mt * 16 + lrow * 4 + i;
-                    const uint32_t h =
-                        mask_hash(seed ^ 0x51u, t1key(b, w, f >> 1, e));
+                    const uint32_t h = no_drop ? 0u
+                        : mask_hash(seed ^ 0x51u, t1key(b, w, f >> 1, e));
                     bf16 pk[2];
 #pragma unroll
                     for (int uu = 0; uu < 2; ++uu) {
                         float v = fmaxf(acc[i + uu] + bf2f(lds.b1s[f + uu]), 0.f);
                         const bool live = (f + uu < F1) && (e < E) &&
-                                          keep_half(h, uu, thresh16);
-                        pk[uu] = f2bf(live ? v * inv_keep : 0.f);
+                                          (no_drop || keep_half(h, uu, thresh16));
+                        pk[uu] = f2bf(live ? (no_drop ? v : v * inv_keep) : 0.f);
                     }
                     *reinterpret_cast<uint32_t*>(&lds.t1_t[e][f]) =
                         *reinterpret_cast<const uint32_t*>(pk);
@@ -195,10 +254,11 @@ __global__ __launch_bounds__(512, 2) void front_fwd_kernel(
                 const int e = nt * 16 + lcol;
                 if (j < F2 && e < E) {
                     float v = fmaxf(acc[i] + bf2f(lds.b2s[j]), 0.f);
-                    const uint32_t h =
-                        mask_hash(seed ^ 0x52u, t2key(b, w, j, e));
-                    const bool live = keep_half(h, 0, thresh16);
-                    lds.t2st[e * F2 + j] = f2bf(live ? v * inv_keep : 0.f);
+                    const bool live = no_drop ||
+                        keep_half(mask_hash(seed ^ 0x52u, t2key(b, w, j, e)),
+                                  0, thresh16);
+                    lds.t2st[e * F2 + j] =
+                        f2bf(live ? (no_drop ? v : v * inv_keep) : 0.f);
                 }
             }
         }
@@ -332,20 +392,45 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
             *reinterpret_cast<bf16x8*>(&lds.u2.t1_t[e][c]) = bf16x8{};
         }
         __syncthreads();
-        for (int p = tid; p < E * (R / 2); p += 512) {
-            const int e = p / (R / 2), q = p % (R / 2);
-            const int r0 = 2 * q;
-            const uint32_t h = mask_hash(seed, ekey(b, w, q, e));
-            float v0 = 0.f, v1 = 0.f;
-            if (keep_half(h, 0, thresh16))
-                v0 = bf2f(lds.emb_s[lds.col_ids[r0]][e]) * inv_keep;
-            if (keep_half(h, 1, thresh16))
-                v1 = bf2f(lds.emb_s[lds.col_ids[r0 + 1]][e]) * inv_keep;
-            bf16 pk[2] = {f2bf(v0), f2bf(v1)};
-            *reinterpret_cast<uint32_t*>(&lds.u1.m_t[e][r0]) =
-                *reinterpret_cast<const uint32_t*>(pk);
-            lds.m_rt[r0][e] = pk[0];
-            lds.m_rt[r0 + 1][e] = pk[1];
+        {
+            // batched 3-pass build (see the forward kernel's comment)
+            uint8_t bid0[10], bid1[10];
+            bf16 bev0[10], bev1[10];
+#pragma unroll
+            for (int k = 0; k < 10; ++k) {
+                const int p = tid + k * 512;
+                if (p < E * (R / 2)) {
+                    const int q = p % (R / 2);
+                    bid0[k] = lds.col_ids[2 * q];
+                    bid1[k] = lds.col_ids[2 * q + 1];
+                }
+            }
+#pragma unroll
+            for (int k = 0; k < 10; ++k) {
+                const int p = tid + k * 512;
+                if (p < E * (R / 2)) {
+                    const int e = p / (R / 2);
+                    bev0[k] = lds.emb_s[bid0[k]][e];
+                    bev1[k] = lds.emb_s[bid1[k]][e];
+                }
+            }
+#pragma unroll
+            for (int k = 0; k < 10; ++k) {
+                const int p = tid + k * 512;
+                if (p < E * (R / 2)) {
+                    const int e = p / (R / 2), q = p % (R / 2);
+                    const uint32_t h = mask_hash(seed, ekey(b, w, q, e));
+                    float v0 = keep_half(h, 0, thresh16)
+                                   ? bf2f(bev0[k]) * inv_keep : 0.f;
+                    float v1 = keep_half(h, 1, thresh16)
+                                   ? bf2f(bev1[k]) * inv_keep : 0.f;
+                    bf16 pk[2] = {f2bf(v0), f2bf(v1)};
+                    *reinterpret_cast<uint32_t*>(&lds.u1.m_t[e][2 * q]) =
+                        *reinterpret_cast<const uint32_t*>(pk);
+                    lds.m_rt[2 * q][e] = pk[0];
+                    lds.m_rt[2 * q + 1][e] = pk[1];
+                }
+            }
         }
         __syncthreads();
 
@@ -437,18 +522,21 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
                 bf16x8 bb = lds_load_b_frag_t(&lds.w2t_t[0][0], fnt * 16, 0, 40);
                 f32x4 acc = mfma16x16x32(a, bb, f32x4{0.f, 0.f, 0.f, 0.f});
                 const int f = fnt * 16 + lcol;
+                const int e0 = emt * 16 + lrow * 4;
+                // one 8-byte read for the lane's 4 consecutive t1post values
+                const bf16x4 t1p4 = *reinterpret_cast<const bf16x4*>(
+                    &lds.t1_fe[f][e0]);
                 bf16x4 pk;
                 float fsum = 0.f;
 #pragma unroll
                 for (int i = 0; i < 4; ++i) {
-                    const int e = emt * 16 + lrow * 4 + i;
                     // post > 0 <=> kept AND pre > 0 (chain rule collapses)
-                    const float t1post = bf2f(lds.t1_fe[f][e]);
-                    const float g = (t1post > 0.f) ? acc[i] * inv_keep : 0.f;
-                    lds.u2.dt1_fe[f][e] = f2bf(g);
+                    const float g =
+                        (bf2f(t1p4[i]) > 0.f) ? acc[i] * inv_keep : 0.f;
                     pk[i] = f2bf(g);
                     fsum += g;
                 }
+                *reinterpret_cast<bf16x4*>(&lds.u2.dt1_fe[f][e0]) = pk;
                 {
                     bf16* dst = dt1g +
                         (((size_t)b * W + w) * MP + f) * EP + emt * 16 + lrow * 4;
@@ -589,27 +677,33 @@ __global__ __launch_bounds__(512, 2) void front_de_kernel(
             tp0 = t1;
         }
 
-        // dm[r, e] = W1^T · dt1 — 13x4 tiles over 8 waves, all b128
+        // dm^T[e, r] = dt1^T · W1 — 13x4 tiles over 8 waves, all b128.
+        // The lane's read id is loaded ONCE per tile, FIRST, so its LDS
+        // latency hides under the fragment loads: a per-element
+        // col_ids read fed the atomic's ADDRESS and serialized the whole
+        // epilogue into read -> lgkmcnt(0) -> ds_add chains (~150 cycles
+        // per element, the dominant cost of v1 of this kernel).
 #pragma clang loop unroll(disable)
         for (int s = 0; s < 7; ++s) {
             const int tile = wid + s * 8;
             if (tile < 52) {
-                const int rmt = tile >> 2, ent = tile & 3;
+                const int emt = tile & 3, rnt = tile >> 2;
+                const int r = rnt * 16 + lcol;
+                const int myid = lds.col_ids[r < R ? r : 0];
                 f32x4 acc = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
                 for (int kb = 0; kb < 4; ++kb) {
-                    bf16x8 a = lds_load_a_frag(&lds.w1_rt[0][0], rmt * 16, kb * 32, 136);
-                    bf16x8 bb = lds_load_b_frag_t(&lds.dt1_ef[0][0], ent * 16, kb * 32, 136);
+                    bf16x8 a = lds_load_a_frag(&lds.dt1_ef[0][0], emt * 16, kb * 32, 136);
+                    bf16x8 bb = lds_load_b_frag_t(&lds.w1_rt[0][0], rnt * 16, kb * 32, 136);
                     acc = mfma16x16x32(a, bb, acc);
                 }
 #pragma unroll
                 for (int i = 0; i < 4; ++i) {
-                    const int r = rmt * 16 + lrow * 4 + i;
-                    const int e = ent * 16 + lcol;
+                    const int e = emt * 16 + lrow * 4 + i;
                     if (r < R && e < E) {
                         const uint32_t h = mask_hash(seed, ekey(b, w, r >> 1, e));
                         if (keep_half(h, r & 1, thresh16))
-                            lds_atomic_add(&lds.de_lds[lds.col_ids[r]][e],
+                            lds_atomic_add(&lds.de_lds[myid][e],
                                            acc[i] * inv_keep);
                     }
                 }
