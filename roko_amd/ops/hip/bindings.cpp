@@ -34,7 +34,7 @@ void front_bwd(const uint8_t* ids, const void* dseq, const void* w1,
                hipStream_t stream, uint32_t phase_mask);
 void front_de(const uint8_t* ids, const void* dt1g, const void* w1, float* de,
               int B, uint32_t seed, float keep, hipStream_t stream,
-              unsigned long long* timing);
+              unsigned long long* timing, uint32_t dbg);
 }  // namespace rk
 
 namespace {
@@ -252,14 +252,14 @@ std::vector<torch::Tensor> front_bwd(torch::Tensor ids, torch::Tensor dseq,
     if (phase_mask == 0x1F)
         rk::front_de(ids.data_ptr<uint8_t>(), dt1g.data_ptr(), w1.data_ptr(),
                      de.data_ptr<float>(), B, (uint32_t)seed, (float)keep,
-                     cur_stream(), nullptr);
+                     cur_stream(), nullptr, 0);
     return {de, dw1, db1, dw2, db2};
 }
 
 // standalone de kernel with optional cycle-timing output (4 uint64)
 std::vector<torch::Tensor> front_de_timed(torch::Tensor ids, torch::Tensor dt1g,
                                           torch::Tensor w1, int64_t seed,
-                                          double keep) {
+                                          double keep, int64_t dbg) {
     check(ids, torch::kUInt8, "ids");
     check(dt1g, torch::kBFloat16, "dt1g");
     check(w1, torch::kBFloat16, "w1");
@@ -269,7 +269,8 @@ std::vector<torch::Tensor> front_de_timed(torch::Tensor ids, torch::Tensor dt1g,
     rk::front_de(ids.data_ptr<uint8_t>(), dt1g.data_ptr(), w1.data_ptr(),
                  de.data_ptr<float>(), B, (uint32_t)seed, (float)keep,
                  cur_stream(),
-                 reinterpret_cast<unsigned long long*>(tim.data_ptr<int64_t>()));
+                 reinterpret_cast<unsigned long long*>(tim.data_ptr<int64_t>()),
+                 (uint32_t)dbg);
     return {de, tim};
 }
 
@@ -286,7 +287,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("adam_step", &adam_step);
     m.def("emb_grad", &emb_grad);
     m.def("front_fwd", &front_fwd);
-    m.def("front_de_timed", &front_de_timed);
+    m.def("front_de_timed", &front_de_timed, py::arg("ids"), py::arg("dt1g"),
+          py::arg("w1"), py::arg("seed"), py::arg("keep"), py::arg("dbg") = 0);
     m.def("front_bwd", &front_bwd, py::arg("ids"), py::arg("dseq"),
           py::arg("w1"), py::arg("b1"), py::arg("w2"), py::arg("b2"),
           py::arg("emb"), py::arg("seed"), py::arg("keep"),

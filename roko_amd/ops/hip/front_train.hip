@@ -623,11 +623,13 @@ __global__ __launch_bounds__(512, 2) void front_de_kernel(
     const bf16* __restrict__ w1,       // (F1, R)
     float* __restrict__ de,            // (12, E) pre-zeroed
     int B, uint32_t seed, float keep,
-    unsigned long long* __restrict__ timing) {  // optional (4): stage/gemm/epi/total cycles
+    unsigned long long* __restrict__ timing,  // optional (4) cycle counters
+    uint32_t dbg) {  // timing-bisection: 1 skip atomics, 2 skip hash, 4 skip epi
     __shared__ struct {
-        bf16 w1_rt[R + 8][136];   // [r][f] = W1^T, zero-padded (A-operand)
-        bf16 dt1_ef[EP][136];     // [e][f] staged column of dt1g (B-operand)
-        float de_lds[12][E];
+        bf16 w1_rt[R + 8][136];   // [r][f] = W1^T, zero-padded
+        bf16 dt1_ef[EP][136];     // [e][f] staged column of dt1g
+        bf16 dmm_er[EP][KP_LD];   // [e][r] masked dm (de-GEMM B-operand)
+        bf16 hot_t[16][KP_LD];    // [c][r] one-hot of col_ids (de-GEMM A)
         uint8_t col_ids[R];
     } lds;
 
@@ -646,7 +648,8 @@ __global__ __launch_bounds__(512, 2) void front_de_kernel(
         (&lds.w1_rt[0][0])[e] = f2bf(0.f);
     for (int e = tid; e < EP * 136; e += 512)
         (&lds.dt1_ef[0][0])[e] = f2bf(0.f);
-    for (int e = tid; e < 12 * E; e += 512) (&lds.de_lds[0][0])[e] = 0.f;
+    for (int e = tid; e < EP * KP_LD; e += 512)
+        (&lds.dmm_er[0][0])[e] = f2bf(0.f);
     __syncthreads();
     for (int e = tid; e < F1 * R; e += 512)
         lds.w1_rt[e % R][e / R] = w1[e];
@@ -654,9 +657,13 @@ __global__ __launch_bounds__(512, 2) void front_de_kernel(
 
     unsigned long long tacc[3] = {0, 0, 0};
     unsigned long long tw0 = (timing && tid == 0) ? __builtin_amdgcn_s_memtime() : 0;
+    float sink = 0.f;
+    f32x4 de_acc = {0.f, 0.f, 0.f, 0.f};  // waves 0..3: de tile fragments
     for (int w = w_begin; w < w_end; ++w) {
         unsigned long long tp0 = (timing && tid == 0) ? __builtin_amdgcn_s_memtime() : 0;
         if (tid < R) lds.col_ids[tid] = ids[((size_t)b * R + tid) * W + w];
+        for (int z = tid; z < 16 * KP_LD / 8; z += 512)
+            *reinterpret_cast<bf16x8*>(&lds.hot_t[0][0] + z * 8) = bf16x8{};
         // stage dt1[b, w] (MP, EP) -> [e][f]: vector global reads, scalar
         // transposed LDS writes
         {
@@ -671,6 +678,7 @@ __global__ __launch_bounds__(512, 2) void front_de_kernel(
             }
         }
         __syncthreads();
+        if (tid < R) lds.hot_t[lds.col_ids[tid]][tid] = f2bf(1.0f);
         if (timing && tid == 0) {
             unsigned long long t1 = __builtin_amdgcn_s_memtime();
             tacc[0] += t1 - tp0;
@@ -697,29 +705,57 @@ __global__ __launch_bounds__(512, 2) void front_de_kernel(
                     bf16x8 bb = lds_load_b_frag_t(&lds.w1_rt[0][0], rnt * 16, kb * 32, 136);
                     acc = mfma16x16x32(a, bb, acc);
                 }
+                (void)myid;
+                if (dbg & 4u) {  // timing: epilogue stripped, sink the acc
+                    sink += acc[0] + acc[1] + acc[2] + acc[3];
+                } else {
 #pragma unroll
-                for (int i = 0; i < 4; ++i) {
-                    const int e = emt * 16 + lrow * 4 + i;
-                    if (r < R && e < E) {
-                        const uint32_t h = mask_hash(seed, ekey(b, w, r >> 1, e));
-                        if (keep_half(h, r & 1, thresh16))
-                            lds_atomic_add(&lds.de_lds[myid][e],
-                                           acc[i] * inv_keep);
+                    for (int i = 0; i < 4; ++i) {
+                        const int e = emt * 16 + lrow * 4 + i;
+                        float v = 0.f;
+                        if (r < R && e < E) {
+                            const uint32_t h = (dbg & 2u) ? (uint32_t)e
+                                : mask_hash(seed, ekey(b, w, r >> 1, e));
+                            if (keep_half(h, r & 1, thresh16))
+                                v = acc[i] * inv_keep;
+                        }
+                        lds.dmm_er[e][r] = f2bf(v);
                     }
                 }
             }
         }
         __syncthreads();
+        // de += Hot^T · dmm — ONE 16x64xK GEMM per column instead of ~13k
+        // LDS atomics (the atomics measured ~20k cycles/column); the de
+        // fragments live in registers across all columns
+        if (wid < 4 && !(dbg & 4u)) {
+            const int nt = wid;
+#pragma unroll
+            for (int kb = 0; kb < 7; ++kb) {
+                bf16x8 a = lds_load_a_frag(&lds.hot_t[0][0], 0, kb * 32, KP_LD);
+                bf16x8 bb = lds_load_b_frag_t(&lds.dmm_er[0][0], nt * 16, kb * 32, KP_LD);
+                de_acc = mfma16x16x32(a, bb, de_acc);
+            }
+        }
+        __syncthreads();
         if (timing && tid == 0) tacc[1] += __builtin_amdgcn_s_memtime() - tp0;
+    }
+    if (wid < 4) {
+        const int nt = wid;
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+            const int c = lrow * 4 + i;
+            const int e = nt * 16 + lcol;
+            if (c < 12 && e < E && de_acc[i] != 0.f)
+                agent_atomic_add(&de[c * E + e], de_acc[i]);
+        }
     }
     if (timing && tid == 0) {
         atomicAdd(&timing[0], tacc[0]);
         atomicAdd(&timing[1], tacc[1]);
         atomicAdd(&timing[3], __builtin_amdgcn_s_memtime() - tw0);
     }
-    for (int i = tid; i < 12 * E; i += 512)
-        if ((&lds.de_lds[0][0])[i] != 0.f)
-            agent_atomic_add(&de[i], (&lds.de_lds[0][0])[i]);
+    if (dbg && sink == 12345.678f) de[0] = sink;  // keep the sink alive
 }
 
 }  // namespace front
@@ -737,11 +773,12 @@ void front_fwd(const uint8_t* ids, const void* w1, const float* b1,
 
 void front_de(const uint8_t* ids, const void* dt1g, const void* w1, float* de,
               int B, uint32_t seed, float keep, hipStream_t stream,
-              unsigned long long* timing) {
+              unsigned long long* timing, uint32_t dbg) {
     hipLaunchKernelGGL(front::front_de_kernel,
                        dim3(B, front::CSPLIT), dim3(512), 0, stream,
                        ids, static_cast<const bf16*>(dt1g),
-                       static_cast<const bf16*>(w1), de, B, seed, keep, timing);
+                       static_cast<const bf16*>(w1), de, B, seed, keep, timing,
+                       dbg);
 }
 
 void front_bwd(const uint8_t* ids, const void* dseq, const void* w1,
