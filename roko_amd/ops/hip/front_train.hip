@@ -328,8 +328,6 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
         bf16 emb_s[12][E];
         bf16 b1s[MP];
         bf16 b2s[16];
-        float db1_lds[F1];
-        float db2_lds[16];
         uint8_t col_ids[R];
     } lds;
 
@@ -352,8 +350,6 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
     for (int e = tid; e < MP * 72; e += 512) (&lds.t1_fe[0][0])[e] = f2bf(0.f);
     for (int e = tid; e < 12 * 136; e += 512) (&lds.w2_lds[0][0])[e] = f2bf(0.f);
     for (int e = tid; e < MP * 40; e += 512) (&lds.w2t_t[0][0])[e] = f2bf(0.f);
-    for (int e = tid; e < F1; e += 512) lds.db1_lds[e] = 0.f;
-    for (int e = tid; e < 16; e += 512) lds.db2_lds[e] = 0.f;
     for (int e = tid; e < MP; e += 512) lds.b1s[e] = f2bf(0.f);
     for (int e = tid; e < 16; e += 512) lds.b2s[e] = f2bf(0.f);
     __syncthreads();
@@ -371,6 +367,8 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
 #pragma unroll
     for (int s = 0; s < 12; ++s) dw1acc[s] = f32x4{0.f, 0.f, 0.f, 0.f};
     f32x4 dw2acc = {0.f, 0.f, 0.f, 0.f};
+    float db1acc[4] = {0.f, 0.f, 0.f, 0.f};
+    float db2acc[4] = {0.f, 0.f, 0.f, 0.f};
     __syncthreads();
 
     for (int w = w_begin; w < w_end; ++w) {
@@ -494,11 +492,7 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
                 }
                 lds.u1.g.dt2_je[j][e] = f2bf(g);
                 lds.u1.g.dt2_ej[e][j] = f2bf(g);
-                float red = g;
-#pragma unroll
-                for (int moff = 1; moff < 16; moff <<= 1)
-                    red += __shfl_xor(red, moff, 16);
-                if (lcol == 0 && j < F2) lds_atomic_add(&lds.db2_lds[j], red);
+                db2acc[i] += g;  // per-lane partial; reduced once at the end
             }
         } else if (wid >= 4 && (phase_mask & 2u)) {
             // waves 4..7: zero the j in [16,32) pad of dt2_ej (read by the
@@ -542,10 +536,7 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
                         (((size_t)b * W + w) * MP + f) * EP + emt * 16 + lrow * 4;
                     *reinterpret_cast<bf16x4*>(dst) = pk;
                 }
-                // db1[f]: sum this lane's 4 e's, then fold the 4 lrow groups
-                fsum += __shfl_xor(fsum, 16);
-                fsum += __shfl_xor(fsum, 32);
-                if (lrow == 0 && f < F1) lds_atomic_add(&lds.db1_lds[f], fsum);
+                db1acc[s] += fsum;  // per-lane partial; reduced at the end
             }
         }
         __syncthreads();
@@ -599,11 +590,28 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
             if (j < F2 && f < F1) agent_atomic_add(&dw2[j * F1 + f], dw2acc[i]);
         }
     }
-    __syncthreads();
-    for (int i = tid; i < F1; i += 512)
-        if (lds.db1_lds[i] != 0.f) agent_atomic_add(&db1[i], lds.db1_lds[i]);
-    for (int i = tid; i < F2; i += 512)
-        if (lds.db2_lds[i] != 0.f) agent_atomic_add(&db2[i], lds.db2_lds[i]);
+#pragma unroll
+    for (int s = 0; s < 4; ++s) {  // db1: fold lrow groups, lane per f commits
+        const int tile = wid + s * 8;
+        if (tile < 28) {
+            const int f = (tile >> 2) * 16 + lcol;
+            float red = db1acc[s];
+            red += __shfl_xor(red, 16);
+            red += __shfl_xor(red, 32);
+            if (lrow == 0 && f < F1) agent_atomic_add(&db1[f], red);
+        }
+    }
+    if (wid < 4) {
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {  // db2: fold the 16 lcol lanes
+            const int j = lrow * 4 + i;
+            float red = db2acc[i];
+#pragma unroll
+            for (int moff = 1; moff < 16; moff <<= 1)
+                red += __shfl_xor(red, moff, 16);
+            if (lcol == 0 && j < F2) agent_atomic_add(&db2[j], red);
+        }
+    }
 }
 
 // ---------------------------------------------------------------------------
