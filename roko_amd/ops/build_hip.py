@@ -22,7 +22,7 @@ def build():
 
     build_dir = os.path.join(HERE, "_build")
     os.makedirs(build_dir, exist_ok=True)
-    names = ["bindings.cpp", "adam.hip", "ce.hip", "embed_mlp.hip", "embgrad.hip", "front_train.hip",
+    names = ["bindings.cpp", "adam.hip", "ce.hip", "embed_mlp.hip", "embgrad.hip", "front_train.hip", "gemm.hip",
              "gru.hip", "head.hip", "probe.hip"]
     sources = [os.path.join(HERE, "hip", n) for n in names]
     load(

@@ -14,7 +14,8 @@ void embed_mlp_fwd(const uint8_t* ids, const void* w1, const float* b1,
 void gru_layer_fwd(const void* xg, const void* u, const float* bhh, void* hseq,
                    void* cache, int T, int B, hipStream_t stream);
 void gru_layer_bwd(const void* cache, const void* hseq, const void* dhin,
-                   const void* ut, void* dg, int T, int B, hipStream_t stream);
+                   const void* ut, void* dxg, void* dhg, int T, int B,
+                   hipStream_t stream);
 void ce_fwd_bwd(const float* logits, const int64_t* target, float* dlogits,
                 float* loss_sum, int64_t n, hipStream_t stream);
 void adam_step(float* p, const float* g, float* m, float* v, int64_t n,
@@ -32,6 +33,8 @@ void front_bwd(const uint8_t* ids, const void* dseq, const void* w1,
                const void* emb, float* dw1, float* db1, float* dw2, float* db2,
                void* dt1g, int B, uint32_t seed, float keep,
                hipStream_t stream, uint32_t phase_mask);
+void gemm_bias(const void* A, const void* B, const float* bias, void* C,
+               int M, int N, int K, hipStream_t stream);
 void front_de(const uint8_t* ids, const void* dt1g, const void* w1, float* de,
               int B, uint32_t seed, float keep, hipStream_t stream,
               unsigned long long* timing, uint32_t dbg);
@@ -108,9 +111,10 @@ std::vector<torch::Tensor> gru_layer_fwd(torch::Tensor xg, torch::Tensor u,
     return out;
 }
 
-// BPTT sequential backward: dg (T, B, 2, 512) = [dxr dxz dxn dhgn]
-torch::Tensor gru_layer_bwd(torch::Tensor cache, torch::Tensor hseq,
-                            torch::Tensor dhin, torch::Tensor ut) {
+// BPTT sequential backward -> dxg (T,B,2,384) [dxr dxz dxn] and
+// dhg (2,T,B,384) [dxr dxz dhgn] in GEMM-ready layouts
+std::vector<torch::Tensor> gru_layer_bwd(torch::Tensor cache, torch::Tensor hseq,
+                                         torch::Tensor dhin, torch::Tensor ut) {
     check(cache, torch::kBFloat16, "cache");
     check(hseq, torch::kBFloat16, "hseq");
     check(dhin, torch::kBFloat16, "dhin");
@@ -121,10 +125,12 @@ torch::Tensor gru_layer_bwd(torch::Tensor cache, torch::Tensor hseq,
     TORCH_CHECK(dhin.sizes() == hseq.sizes(), "dhin must match hseq");
     TORCH_CHECK(ut.size(0) == 2 && ut.size(1) == 128 && ut.size(2) == 384,
                 "ut must be (2,128,384)");
-    auto dg = torch::empty({T, B, 2, 512}, cache.options());
+    auto dxg = torch::empty({T, B, 2, 384}, cache.options());
+    auto dhg = torch::empty({2, T, B, 384}, cache.options());
     rk::gru_layer_bwd(cache.data_ptr(), hseq.data_ptr(), dhin.data_ptr(),
-                      ut.data_ptr(), dg.data_ptr(), T, B, cur_stream());
-    return dg;
+                      ut.data_ptr(), dxg.data_ptr(), dhg.data_ptr(), T, B,
+                      cur_stream());
+    return {dxg, dhg};
 }
 
 // fused CE: returns (loss scalar f32, dlogits (N,5) f32) for mean reduction
@@ -274,6 +280,25 @@ std::vector<torch::Tensor> front_de_timed(torch::Tensor ids, torch::Tensor dt1g,
     return {de, tim};
 }
 
+// C (M,N) bf16 = A (M,K) bf16 · B (K,N) bf16 [+ bias f32]
+torch::Tensor gemm_bias(torch::Tensor A, torch::Tensor B,
+                        c10::optional<torch::Tensor> bias) {
+    check(A, torch::kBFloat16, "A");
+    check(B, torch::kBFloat16, "B");
+    const int M = A.size(0), K = A.size(1), N = B.size(1);
+    TORCH_CHECK(B.size(0) == K, "inner dims mismatch");
+    const float* bp = nullptr;
+    if (bias.has_value()) {
+        check(*bias, torch::kFloat32, "bias");
+        TORCH_CHECK(bias->numel() == N, "bias size");
+        bp = bias->data_ptr<float>();
+    }
+    auto Cout = torch::empty({M, N}, A.options());
+    rk::gemm_bias(A.data_ptr(), B.data_ptr(), bp, Cout.data_ptr(), M, N, K,
+                  cur_stream());
+    return Cout;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -287,6 +312,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("adam_step", &adam_step);
     m.def("emb_grad", &emb_grad);
     m.def("front_fwd", &front_fwd);
+    m.def("gemm_bias", &gemm_bias, py::arg("A"), py::arg("B"),
+          py::arg("bias") = c10::nullopt);
     m.def("front_de_timed", &front_de_timed, py::arg("ids"), py::arg("dt1g"),
           py::arg("w1"), py::arg("seed"), py::arg("keep"), py::arg("dbg") = 0);
     m.def("front_bwd", &front_bwd, py::arg("ids"), py::arg("dseq"),

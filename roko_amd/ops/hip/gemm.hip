@@ -1,0 +1,162 @@
+// bf16 GEMM + bias for the GRU input-projection shape family
+// (M ~ 11520 = T*B, N <= 768, K <= 768) on gfx950.
+//
+// hipBLASLt's heuristic picks ~110 TF/s tiles for these shapes (xg GEMM
+// measured 81 us — profiles/train_r01_kernel_stats.txt); the MFMA floor is
+// ~10x lower. Canonical CDNA GEMM structure (cdna_hip_programming.md §5):
+// 256x64 C-tiles, K staged in 32-wide slices through double-buffered LDS,
+// A-fragments ds_read_b128 from row-major [row][k] images, B transposed at
+// staging time into [col][k] images, bias fused into the epilogue, output
+// staged through LDS for 16-byte coalesced stores.
+//
+//   C (M, N) bf16 = A (M, K) bf16 · B (K, N) bf16 + bias (N) f32
+//
+// M/N/K need not be multiples of the tile sizes (tails are masked).
+
+#include <cstdint>
+
+#include "common.h"
+
+namespace rk {
+namespace gemm {
+
+constexpr int BM = 256;  // C-tile rows per workgroup
+constexpr int BN = 64;   // C-tile cols per workgroup
+constexpr int BK = 32;   // K slice per stage
+constexpr int LDA = BK + 8;   // LDS row strides (bank-conflict pad)
+constexpr int WAVES = 8;
+
+__global__ __launch_bounds__(WAVES * 64, 2) void gemm_bias_kernel(
+    const bf16* __restrict__ A,   // (M, K) row-major
+    const bf16* __restrict__ B,   // (K, N) row-major
+    const float* __restrict__ bias,  // (N) or nullptr
+    bf16* __restrict__ C,         // (M, N) row-major
+    int M, int N, int K) {
+    __shared__ struct {
+        bf16 a[2][BM][LDA];   // [row][k]
+        bf16 bt[2][BN][LDA];  // [col][k] (transposed at staging)
+    } lds;
+
+    const int m0 = blockIdx.x * BM;
+    const int n0 = blockIdx.y * BN;
+    const int tid = threadIdx.x;
+    const int wid = tid >> 6;
+    const int lane = tid & 63;
+    const int lrow = lane >> 4;
+    const int lcol = lane & 15;
+
+    // wave tiling: 8 waves x (2 m-subtiles x 4 n-subtiles)? -> each wave owns
+    // two 16-row strips across two 16-col strips: tiles (mt, nt) with
+    // mt = wid * 2 + {0,1} over 16 m-subtiles, nt = {0..3}
+    // accumulators: 2 mt x 4 nt = 8 fragments
+    f32x4 acc[2][4];
+#pragma unroll
+    for (int a_ = 0; a_ < 2; ++a_)
+#pragma unroll
+        for (int b_ = 0; b_ < 4; ++b_) acc[a_][b_] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+    const int kq = (K + BK - 1) / BK;
+
+    // cooperative staging: A tile BM x BK (8192 elems = 16/thread = 2 x b128),
+    // B tile BK x BN (2048 elems = 4/thread, transposed scalar writes)
+    auto stage = [&](int buf, int k0) {
+        const bool full_k = (k0 + BK <= K);
+#pragma unroll
+        for (int p = 0; p < 2; ++p) {
+            const int e = (p * 512 + tid) * 8;
+            const int r = e / BK, kk = e % BK;
+            const int row = m0 + r;
+            bf16 v[8];
+            if (full_k && row < M) {
+                *reinterpret_cast<bf16x8*>(v) =
+                    *reinterpret_cast<const bf16x8*>(A + (size_t)row * K + k0 + kk);
+            } else {
+#pragma unroll
+                for (int q = 0; q < 8; ++q)
+                    v[q] = (row < M && k0 + kk + q < K)
+                               ? A[(size_t)row * K + k0 + kk + q] : f2bf(0.f);
+            }
+            *reinterpret_cast<bf16x8*>(&lds.a[buf][r][kk]) =
+                *reinterpret_cast<const bf16x8*>(v);
+        }
+#pragma unroll
+        for (int p = 0; p < 4; ++p) {
+            const int e = p * 512 + tid;
+            const int kk = e / BN, c = e % BN;
+            bf16 v = f2bf(0.f);
+            if (k0 + kk < K && n0 + c < N) v = B[(size_t)(k0 + kk) * N + n0 + c];
+            lds.bt[buf][c][kk] = v;
+        }
+    };
+
+    stage(0, 0);
+    __syncthreads();
+
+    for (int q = 0; q < kq; ++q) {
+        if (q + 1 < kq) stage((q + 1) & 1, (q + 1) * BK);
+        bf16x8 bfr[4];
+#pragma unroll
+        for (int b_ = 0; b_ < 4; ++b_)
+            bfr[b_] = lds_load_b_frag_t(&lds.bt[q & 1][0][0], b_ * 16, 0, LDA);
+#pragma unroll
+        for (int a_ = 0; a_ < 2; ++a_) {
+            const int mt = wid * 2 + a_;
+            bf16x8 af = lds_load_a_frag(&lds.a[q & 1][0][0], mt * 16, 0, LDA);
+#pragma unroll
+            for (int b_ = 0; b_ < 4; ++b_)
+                acc[a_][b_] = mfma16x16x32(af, bfr[b_], acc[a_][b_]);
+        }
+        __syncthreads();
+    }
+
+    // epilogue: bias + staged coalesced store (reuse the A buffers)
+    bf16* cst = &lds.a[0][0][0];  // BM x BN staging, stride BN
+#pragma unroll
+    for (int a_ = 0; a_ < 2; ++a_) {
+        const int mt = wid * 2 + a_;
+#pragma unroll
+        for (int b_ = 0; b_ < 4; ++b_) {
+            const int c = b_ * 16 + lcol;
+            const float bv = bias ? bias[min(n0 + c, N - 1)] : 0.f;
+#pragma unroll
+            for (int i = 0; i < 4; ++i) {
+                const int r = mt * 16 + lrow * 4 + i;
+                cst[r * BN + c] = f2bf(acc[a_][b_][i] + bv);
+            }
+        }
+    }
+    __syncthreads();
+    {
+        const int n_full = (n0 + BN <= N);
+#pragma unroll
+        for (int p = 0; p < 4; ++p) {
+            const int e = (p * 512 + tid) * 8;
+            const int r = e / BN, c = e % BN;
+            const int row = m0 + r;
+            if (row < M) {
+                if (n_full) {
+                    *reinterpret_cast<bf16x8*>(C + (size_t)row * N + n0 + c) =
+                        *reinterpret_cast<const bf16x8*>(&cst[r * BN + c]);
+                } else {
+#pragma unroll
+                    for (int u = 0; u < 8; ++u)
+                        if (n0 + c + u < N)
+                            C[(size_t)row * N + n0 + c + u] = cst[r * BN + c + u];
+                }
+            }
+        }
+    }
+}
+
+}  // namespace gemm
+
+void gemm_bias(const void* A, const void* B, const float* bias, void* C,
+               int M, int N, int K, hipStream_t stream) {
+    dim3 grid((M + gemm::BM - 1) / gemm::BM, (N + gemm::BN - 1) / gemm::BN);
+    hipLaunchKernelGGL(gemm::gemm_bias_kernel, grid, dim3(gemm::WAVES * 64), 0,
+                       stream, static_cast<const bf16*>(A),
+                       static_cast<const bf16*>(B), bias, static_cast<bf16*>(C),
+                       M, N, K);
+}
+
+}  // namespace rk

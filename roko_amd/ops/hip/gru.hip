@@ -263,7 +263,11 @@ __global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
     const bf16* __restrict__ hseq,   // (T, B, 2, H)
     const bf16* __restrict__ dhin,   // (T, B, 2, H) grad wrt layer output
     const bf16* __restrict__ ut,     // (2, H, 3H) = weight_hh^T
-    bf16* __restrict__ dg,           // (T, B, 2, 4H) out: [dxr dxz dxn dhgn]
+    bf16* __restrict__ dxg,          // (T, B, 2, 3H) out: [dxr dxz dxn] —
+                                     // (T*B, 768) view feeds dW_ih/dx GEMMs
+    bf16* __restrict__ dhg,          // (2, T, B, 3H) out: [dxr dxz dhgn]
+                                     // dir-major so each dir slice is a
+                                     // contiguous (T*B, 384) GEMM operand
     int T, int B) {
     __shared__ struct {
         bf16 cache_st[2][MB][4 * H];  // double-buffered staged cache[t]
@@ -402,14 +406,32 @@ __global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
             for (int i = 0; i < 4; ++i) dhc[mt][i] = dhp_part[mt][i] + acc[mt][i];
 
         // ---- store dg tile -------------------------------------------------
+        // ---- store dxg + dhg tiles (GEMM-ready layouts, no host cats) -----
         {
-            bf16* dst = dg + (((size_t)t * B + b0) * 2 + dir) * 4 * H;
+            bf16* dst = dxg + (((size_t)t * B + b0) * 2 + dir) * G3;
+            const int row = tid / 64, col = (tid % 64) * 8;
+            if (col < G3) {
+#pragma unroll
+                for (int q = 0; q < 4; ++q)
+                    *reinterpret_cast<bf16x8*>(
+                        dst + (size_t)(row + q * 8) * 2 * G3 + col) =
+                        *reinterpret_cast<const bf16x8*>(&lds.dg_st[row + q * 8][col]);
+            }
+        }
+        {
+            // dhg = [dxr dxz dhgn]: columns [0, 2H) then the dhgn block
+            bf16* dst = dhg + (((size_t)dir * T + t) * B + b0) * G3;
             const int row = tid / 64, col = (tid % 64) * 8;
 #pragma unroll
-            for (int q = 0; q < 4; ++q)
-                *reinterpret_cast<bf16x8*>(
-                    dst + (size_t)(row + q * 8) * 2 * 4 * H + col) =
-                    *reinterpret_cast<const bf16x8*>(&lds.dg_st[row + q * 8][col]);
+            for (int q = 0; q < 4; ++q) {
+                const int r2 = row + q * 8;
+                if (col < 2 * H)
+                    *reinterpret_cast<bf16x8*>(dst + (size_t)r2 * G3 + col) =
+                        *reinterpret_cast<const bf16x8*>(&lds.dg_st[r2][col]);
+                else if (col < 3 * H)
+                    *reinterpret_cast<bf16x8*>(dst + (size_t)r2 * G3 + col) =
+                        *reinterpret_cast<const bf16x8*>(&lds.dg_st[r2][col + H]);
+            }
         }
         __syncthreads();  // dhg/dg_st/stage[curp] reads done before reuse
     };
@@ -423,13 +445,14 @@ __global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
 }
 
 void gru_layer_bwd(const void* cache, const void* hseq, const void* dhin,
-                   const void* ut, void* dg, int T, int B, hipStream_t stream) {
+                   const void* ut, void* dxg, void* dhg, int T, int B,
+                   hipStream_t stream) {
     dim3 grid(B / MB, 2);
     dim3 block(BW_WAVES * 64);
     hipLaunchKernelGGL(gru_layer_bwd_kernel, grid, block, 0, stream,
                        static_cast<const bf16*>(cache), static_cast<const bf16*>(hseq),
                        static_cast<const bf16*>(dhin), static_cast<const bf16*>(ut),
-                       static_cast<bf16*>(dg), T, B);
+                       static_cast<bf16*>(dxg), static_cast<bf16*>(dhg), T, B);
 }
 
 }  // namespace rk

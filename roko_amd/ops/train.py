@@ -41,11 +41,12 @@ class GruLayerFn(torch.autograd.Function):
         T, B, _ = x_seq.shape
         x_bf = x_seq.to(torch.bfloat16)
         w_ih_bf = w_ih.detach().to(torch.bfloat16)
-        xg = torch.addmm(
-            b_ih_all.detach().to(torch.bfloat16),
-            x_bf.reshape(T * B, -1),
-            w_ih_bf.t(),
-        ).view(T, B, 2, 384).contiguous()
+        # custom MFMA GEMM (hipBLASLt picks ~110 TF/s tiles at these shapes)
+        xg = ext.gemm_bias(
+            x_bf.reshape(T * B, -1).contiguous(),
+            w_ih_bf.t().contiguous(),
+            b_ih_all.detach().float().contiguous(),
+        ).view(T, B, 2, 384)
         u_bf = u.detach().to(torch.bfloat16).contiguous()
         bhh_f = bhh.detach().float().contiguous()
         hseq, cache = ext.gru_layer_fwd(xg, u_bf, bhh_f, True)
@@ -61,10 +62,9 @@ class GruLayerFn(torch.autograd.Function):
         H = C.HIDDEN_SIZE
         dhin = dout.reshape(T, B, 2, H).to(torch.bfloat16).contiguous()
         ut = u_bf.transpose(1, 2).contiguous()  # (2, 128, 384)
-        dg = ext.gru_layer_bwd(cache, hseq, dhin, ut)  # (T,B,2,512)
-
-        dxg = dg[..., : 3 * H]                       # (T,B,2,384)
-        dhg = torch.cat([dg[..., : 2 * H], dg[..., 3 * H :]], dim=-1)
+        # kernel emits GEMM-ready layouts: dxg (T,B,2,384) -> (TB,768) view;
+        # dhg (2,T,B,384) -> contiguous per-direction (TB,384) slices
+        dxg, dhg = ext.gru_layer_bwd(cache, hseq, dhin, ut)
 
         # h_prev sequences per direction
         zeros = hseq.new_zeros(1, B, H)
@@ -72,14 +72,13 @@ class GruLayerFn(torch.autograd.Function):
         hp_r = torch.cat([hseq[1:, :, 1, :], zeros], dim=0)
 
         TB = T * B
-        dhg_f = dhg[:, :, 0, :].reshape(TB, 3 * H)
-        dhg_r = dhg[:, :, 1, :].reshape(TB, 3 * H)
+        dhg_f = dhg[0].reshape(TB, 3 * H)
+        dhg_r = dhg[1].reshape(TB, 3 * H)
         du = torch.stack(
             [dhg_f.t().mm(hp_f.reshape(TB, H)), dhg_r.t().mm(hp_r.reshape(TB, H))]
         ).float()
 
-        dxg_flat = dxg.reshape(TB, 2, 3 * H)
-        dxg_cat = torch.cat([dxg_flat[:, 0, :], dxg_flat[:, 1, :]], dim=1)  # (TB,768)
+        dxg_cat = dxg.reshape(TB, 2 * 3 * H)  # zero-copy (TB, 768)
         x_flat = x_bf.reshape(TB, -1)
         dw_ih = dxg_cat.t().mm(x_flat).float()          # (768, in)
         # column sums as one hipBLASLt GEMV each — aten's strided bf16
@@ -89,7 +88,8 @@ class GruLayerFn(torch.autograd.Function):
             [ones.mm(dhg_f).squeeze(0), ones.mm(dhg_r).squeeze(0)]
         ).float()
         db_ih = ones.mm(dxg_cat).squeeze(0).float()      # (768,)
-        dx = dxg_cat.mm(w_ih_bf).to(ctx.in_dtype).view(T, B, -1)
+        dx = ext.gemm_bias(dxg_cat.contiguous(), w_ih_bf.contiguous(), None) \
+            .to(ctx.in_dtype).view(T, B, -1)
 
         return dx, dw_ih, db_ih, du, dbhh
 
