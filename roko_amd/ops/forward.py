@@ -55,7 +55,7 @@ def _bf16_weights(model) -> dict:
         c[f"w_ih_t{l}"] = (
             torch.cat([wf, wr], dim=0).to(torch.bfloat16).t().contiguous()
         )
-        c[f"b_ih{l}"] = torch.cat([bf, br]).float().contiguous()
+        c[f"b_ih{l}"] = torch.cat([bf, br]).to(torch.bfloat16).contiguous()
         c[f"u{l}"] = torch.stack(
             [getattr(g, f"weight_hh_l{l}").detach(),
              getattr(g, f"weight_hh_l{l}_reverse").detach()]
@@ -82,10 +82,10 @@ def roko_forward(model, x: torch.Tensor) -> torch.Tensor:
 
     seq = ext.front_fwd(ids, w["w1"], w["b1"], w["w2"], w["b2"], w["emb"], 0, 1.0)
     for l in range(C.NUM_LAYERS):
-        xg = ext.gemm_bias(
-            seq.reshape(T * B, -1), w[f"w_ih_t{l}"], w[f"b_ih{l}"]
+        xg = torch.addmm(
+            w[f"b_ih{l}"], seq.reshape(T * B, -1), w[f"w_ih_t{l}"]
         ).view(T, B, 2, 384)
-        (hseq,) = ext.gru_layer_fwd(xg, w[f"u{l}"], w[f"bhh{l}"], False)
+        (hseq,) = ext.gru_layer_fwd(xg.contiguous(), w[f"u{l}"], w[f"bhh{l}"], False)
         seq = hseq.view(T, B, 2 * C.HIDDEN_SIZE)
     (logits,) = ext.head_fwd(seq, w["w4"], w["b4"], True, False)
     return logits[:B0]
@@ -134,10 +134,12 @@ class InferencePipeline:
         seq = ext.front_fwd(ids_u8, w["w1"], w["b1"], w["w2"], w["b2"],
                             w["emb"], 0, 1.0)
         for l in range(C.NUM_LAYERS):
-            xg = ext.gemm_bias(
-                seq.reshape(T * B, -1), w[f"w_ih_t{l}"], w[f"b_ih{l}"]
+            xg = torch.addmm(
+                w[f"b_ih{l}"], seq.reshape(T * B, -1), w[f"w_ih_t{l}"]
             ).view(T, B, 2, 384)
-            (hseq,) = ext.gru_layer_fwd(xg, w[f"u{l}"], w[f"bhh{l}"], False)
+            (hseq,) = ext.gru_layer_fwd(
+                xg.contiguous(), w[f"u{l}"], w[f"bhh{l}"], False
+            )
             seq = hseq.view(T, B, 2 * C.HIDDEN_SIZE)
         (amax,) = ext.head_fwd(seq, w["w4"], w["b4"], False, True)
         return amax
@@ -234,10 +236,10 @@ def roko_argmax(model, x: torch.Tensor) -> torch.Tensor:
     T = C.WINDOW_COLS
     seq = ext.front_fwd(ids, w["w1"], w["b1"], w["w2"], w["b2"], w["emb"], 0, 1.0)
     for l in range(C.NUM_LAYERS):
-        xg = ext.gemm_bias(
-            seq.reshape(T * B, -1), w[f"w_ih_t{l}"], w[f"b_ih{l}"]
+        xg = torch.addmm(
+            w[f"b_ih{l}"], seq.reshape(T * B, -1), w[f"w_ih_t{l}"]
         ).view(T, B, 2, 384)
-        (hseq,) = ext.gru_layer_fwd(xg, w[f"u{l}"], w[f"bhh{l}"], False)
+        (hseq,) = ext.gru_layer_fwd(xg.contiguous(), w[f"u{l}"], w[f"bhh{l}"], False)
         seq = hseq.view(T, B, 2 * C.HIDDEN_SIZE)
     (amax,) = ext.head_fwd(seq, w["w4"], w["b4"], False, True)
     return amax[:B0]

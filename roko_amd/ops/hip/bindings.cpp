@@ -35,6 +35,8 @@ void front_bwd(const uint8_t* ids, const void* dseq, const void* w1,
                hipStream_t stream, uint32_t phase_mask);
 void gemm_bias(const void* A, const void* B, const float* bias, void* C,
                int M, int N, int K, hipStream_t stream);
+void atb_splitk(const void* A, const void* B, float* C, int M, int N, int K,
+                hipStream_t stream);
 void front_de(const uint8_t* ids, const void* dt1g, const void* w1, float* de,
               int B, uint32_t seed, float keep, hipStream_t stream,
               unsigned long long* timing, uint32_t dbg);
@@ -299,6 +301,18 @@ torch::Tensor gemm_bias(torch::Tensor A, torch::Tensor B,
     return Cout;
 }
 
+// C (M,N) f32 = A (K,M)^T · B (K,N), split-K with atomic accumulation
+torch::Tensor atb_splitk(torch::Tensor A, torch::Tensor B) {
+    check(A, torch::kBFloat16, "A");
+    check(B, torch::kBFloat16, "B");
+    const int K = A.size(0), M = A.size(1), N = B.size(1);
+    TORCH_CHECK(B.size(0) == K, "inner dims mismatch");
+    auto Cout = torch::zeros({M, N}, A.options().dtype(torch::kFloat32));
+    rk::atb_splitk(A.data_ptr(), B.data_ptr(), Cout.data_ptr<float>(), M, N, K,
+                   cur_stream());
+    return Cout;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -312,6 +326,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("adam_step", &adam_step);
     m.def("emb_grad", &emb_grad);
     m.def("front_fwd", &front_fwd);
+    m.def("atb_splitk", &atb_splitk);
     m.def("gemm_bias", &gemm_bias, py::arg("A"), py::arg("B"),
           py::arg("bias") = c10::nullopt);
     m.def("front_de_timed", &front_de_timed, py::arg("ids"), py::arg("dt1g"),

@@ -159,4 +159,133 @@ void gemm_bias(const void* A, const void* B, const float* bias, void* C,
                        M, N, K);
 }
 
+// ---------------------------------------------------------------------------
+// split-K  C (M, N) f32  +=  A (K, M)^T · B (K, N),  bf16 inputs
+// ---------------------------------------------------------------------------
+// The GRU weight-gradient reductions (dU = dhg^T·h_prev at (384,128) and
+// dW_ih = dxg^T·x at (768,500), both K = T*B ~ 11520) map to transpose-A
+// GEMMs that hipBLASLt runs WITHOUT split-K — 6 workgroups on a 256-CU chip,
+// 81 us each (profiles/train_r01_latest). Here K is sliced across the grid
+// (one 64-row stage per iteration, both operands transposed into LDS at
+// staging time) and every workgroup commits its partial tile with relaxed
+// agent-scope f32 atomics.
+
+namespace gemmatb {
+
+constexpr int BM = 128;  // C rows (= A cols) per workgroup
+constexpr int BN = 128;  // C cols per workgroup
+constexpr int BK = 64;   // K slice per stage
+constexpr int KSLICE = 256;  // K rows per workgroup (KSLICE/BK stages)
+constexpr int LD = BK + 8;
+constexpr int WAVES = 8;
+
+__global__ __launch_bounds__(WAVES * 64, 2) void atb_splitk_kernel(
+    const bf16* __restrict__ A,  // (K, M)
+    const bf16* __restrict__ B,  // (K, N)
+    float* __restrict__ C,       // (M, N) pre-zeroed, atomic-accumulated
+    int M, int N, int K) {
+    __shared__ struct {
+        bf16 at[2][BM][LD];  // [m][k]
+        bf16 bt[2][BN][LD];  // [n][k]
+    } lds;
+
+    const int k_begin = blockIdx.x * KSLICE;
+    const int m0 = blockIdx.y * BM;
+    const int n0 = blockIdx.z * BN;
+    const int tid = threadIdx.x;
+    const int wid = tid >> 6;
+    const int lane = tid & 63;
+    const int lrow = lane >> 4;
+    const int lcol = lane & 15;
+
+    // 8x8 = 64 sub-tiles over 8 waves: wave owns one 16-row strip (mt = wid)
+    // across all 8 col strips
+    f32x4 acc[8];
+#pragma unroll
+    for (int i = 0; i < 8; ++i) acc[i] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+    // stage one BK-slice: A rows [k0, k0+BK) cols [m0, m0+BM) transposed to
+    // [m][k]; same for B. 128*64 elems each = 16/thread.
+    auto stage = [&](int buf, int k0) {
+#pragma unroll
+        for (int p = 0; p < 2; ++p) {
+            const int e = (p * 512 + tid) * 8;   // element in (BK, BM) slab
+            const int kk = e / BM, m = e % BM;   // 8 consecutive m per thread
+            bf16 v[8];
+            const int krow = k0 + kk;
+            if (krow < K && m0 + m + 7 < M) {
+                *reinterpret_cast<bf16x8*>(v) = *reinterpret_cast<const bf16x8*>(
+                    A + (size_t)krow * M + m0 + m);
+            } else {
+#pragma unroll
+                for (int q = 0; q < 8; ++q)
+                    v[q] = (krow < K && m0 + m + q < M)
+                               ? A[(size_t)krow * M + m0 + m + q] : f2bf(0.f);
+            }
+#pragma unroll
+            for (int q = 0; q < 8; ++q) lds.at[buf][m + q][kk] = v[q];
+        }
+#pragma unroll
+        for (int p = 0; p < 2; ++p) {
+            const int e = (p * 512 + tid) * 8;
+            const int kk = e / BN, n = e % BN;
+            bf16 v[8];
+            const int krow = k0 + kk;
+            if (krow < K && n0 + n + 7 < N) {
+                *reinterpret_cast<bf16x8*>(v) = *reinterpret_cast<const bf16x8*>(
+                    B + (size_t)krow * N + n0 + n);
+            } else {
+#pragma unroll
+                for (int q = 0; q < 8; ++q)
+                    v[q] = (krow < K && n0 + n + q < N)
+                               ? B[(size_t)krow * N + n0 + n + q] : f2bf(0.f);
+            }
+#pragma unroll
+            for (int q = 0; q < 8; ++q) lds.bt[buf][n + q][kk] = v[q];
+        }
+    };
+
+    stage(0, k_begin);
+    __syncthreads();
+    const int stages = KSLICE / BK;
+    for (int s = 0; s < stages; ++s) {
+        if (s + 1 < stages) stage((s + 1) & 1, k_begin + (s + 1) * BK);
+#pragma unroll
+        for (int kb = 0; kb < BK / 32; ++kb) {
+            bf16x8 af = lds_load_a_frag(&lds.at[s & 1][0][0], wid * 16, kb * 32, LD);
+#pragma unroll
+            for (int b_ = 0; b_ < 8; ++b_) {
+                bf16x8 bf_ = lds_load_b_frag_t(&lds.bt[s & 1][0][0], b_ * 16, kb * 32, LD);
+                acc[b_] = mfma16x16x32(af, bf_, acc[b_]);
+            }
+        }
+        __syncthreads();
+    }
+
+    // commit the partial tile (relaxed agent atomics; C pre-zeroed)
+#pragma unroll
+    for (int b_ = 0; b_ < 8; ++b_) {
+        const int n = n0 + b_ * 16 + lcol;
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+            const int m = m0 + wid * 16 + lrow * 4 + i;
+            if (m < M && n < N && acc[b_][i] != 0.f)
+                agent_atomic_add(&C[(size_t)m * N + n], acc[b_][i]);
+        }
+    }
+}
+
+}  // namespace gemmatb
+
+void atb_splitk(const void* A, const void* B, float* C, int M, int N, int K,
+                hipStream_t stream) {
+    dim3 grid((K + gemmatb::KSLICE - 1) / gemmatb::KSLICE,
+              (M + gemmatb::BM - 1) / gemmatb::BM,
+              (N + gemmatb::BN - 1) / gemmatb::BN);
+    hipLaunchKernelGGL(gemmatb::atb_splitk_kernel, grid,
+                       dim3(gemmatb::WAVES * 64), 0, stream,
+                       static_cast<const bf16*>(A), static_cast<const bf16*>(B),
+                       C, M, N, K);
+}
+
 }  // namespace rk

@@ -41,12 +41,11 @@ class GruLayerFn(torch.autograd.Function):
         T, B, _ = x_seq.shape
         x_bf = x_seq.to(torch.bfloat16)
         w_ih_bf = w_ih.detach().to(torch.bfloat16)
-        # custom MFMA GEMM (hipBLASLt picks ~110 TF/s tiles at these shapes)
-        xg = ext.gemm_bias(
-            x_bf.reshape(T * B, -1).contiguous(),
-            w_ih_bf.t().contiguous(),
-            b_ih_all.detach().float().contiguous(),
-        ).view(T, B, 2, 384)
+        xg = torch.addmm(
+            b_ih_all.detach().to(torch.bfloat16),
+            x_bf.reshape(T * B, -1),
+            w_ih_bf.t(),
+        ).view(T, B, 2, 384).contiguous()
         u_bf = u.detach().to(torch.bfloat16).contiguous()
         bhh_f = bhh.detach().float().contiguous()
         hseq, cache = ext.gru_layer_fwd(xg, u_bf, bhh_f, True)
@@ -74,13 +73,16 @@ class GruLayerFn(torch.autograd.Function):
         TB = T * B
         dhg_f = dhg[0].reshape(TB, 3 * H)
         dhg_r = dhg[1].reshape(TB, 3 * H)
+        # custom split-K A^T·B kernels: hipBLASLt schedules these K=11520
+        # transpose-A reductions on 6 workgroups (81 us each)
         du = torch.stack(
-            [dhg_f.t().mm(hp_f.reshape(TB, H)), dhg_r.t().mm(hp_r.reshape(TB, H))]
-        ).float()
+            [ext.atb_splitk(dhg_f, hp_f.reshape(TB, H).contiguous()),
+             ext.atb_splitk(dhg_r, hp_r.reshape(TB, H).contiguous())]
+        )
 
         dxg_cat = dxg.reshape(TB, 2 * 3 * H)  # zero-copy (TB, 768)
         x_flat = x_bf.reshape(TB, -1)
-        dw_ih = dxg_cat.t().mm(x_flat).float()          # (768, in)
+        dw_ih = ext.atb_splitk(dxg_cat, x_flat.contiguous())  # (768, in) f32
         # column sums as one hipBLASLt GEMV each — aten's strided bf16
         # .sum(0) over (TB, 768) was ~100 us/call (profiles/train_r01)
         ones = x_bf.new_ones(1, TB)
@@ -88,8 +90,7 @@ class GruLayerFn(torch.autograd.Function):
             [ones.mm(dhg_f).squeeze(0), ones.mm(dhg_r).squeeze(0)]
         ).float()
         db_ih = ones.mm(dxg_cat).squeeze(0).float()      # (768,)
-        dx = ext.gemm_bias(dxg_cat.contiguous(), w_ih_bf.contiguous(), None) \
-            .to(ctx.in_dtype).view(T, B, -1)
+        dx = dxg_cat.mm(w_ih_bf).to(ctx.in_dtype).view(T, B, -1)
 
         return dx, dw_ih, db_ih, du, dbhh
 

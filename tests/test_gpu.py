@@ -334,3 +334,18 @@ def test_gemm_bias_vs_addmm():
         ref2 = A.float() @ B.float()
         rel2 = (got2 - ref2).norm() / ref2.norm()
         assert rel2.item() < 2e-2, (M, N, K, rel2.item())
+
+
+@requires_gpu
+def test_atb_splitk_vs_mm():
+    """Split-K A^T·B kernel vs hipBLASLt for the weight-grad shapes."""
+    ext = ops.ext()
+    torch.manual_seed(12)
+    for K, M, N in [(11520, 384, 128), (11520, 768, 500), (2880, 768, 256),
+                    (1000, 130, 70)]:
+        A = (torch.randn(K, M, device="cuda") * 0.2).to(torch.bfloat16)
+        B = (torch.randn(K, N, device="cuda") * 0.2).to(torch.bfloat16)
+        got = ext.atb_splitk(A, B)
+        ref = A.float().t() @ B.float()
+        rel = (got - ref).norm() / ref.norm()
+        assert rel.item() < 2e-2, (K, M, N, rel.item())
