@@ -12,7 +12,7 @@ void embed_mlp_fwd(const uint8_t* ids, const void* w1, const float* b1,
                    const void* w2, const float* b2, const void* emb, void* out,
                    int B, hipStream_t stream);
 void gru_layer_fwd(const void* xg, const void* u, const float* bhh, void* hseq,
-                   void* cache, int T, int B, hipStream_t stream);
+                   void* cache, int T, int B, hipStream_t stream, uint32_t dbg);
 void gru_layer_bwd(const void* cache, const void* hseq, const void* dhin,
                    const void* ut, void* dxg, void* dhg, int T, int B,
                    hipStream_t stream);
@@ -89,7 +89,8 @@ torch::Tensor embed_mlp_fwd(torch::Tensor ids, torch::Tensor w1, torch::Tensor b
 // xg (T, B, 2, 384) bf16, u (2, 384, 128) bf16, bhh (2, 384) f32
 //   -> (hseq (T, B, 2, 128) bf16 [, cache (T, B, 2, 512) bf16 when train])
 std::vector<torch::Tensor> gru_layer_fwd(torch::Tensor xg, torch::Tensor u,
-                                         torch::Tensor bhh, bool train) {
+                                         torch::Tensor bhh, bool train,
+                                         int64_t dbg) {
     check(xg, torch::kBFloat16, "xg");
     check(u, torch::kBFloat16, "u");
     check(bhh, torch::kFloat32, "bhh");
@@ -107,7 +108,7 @@ std::vector<torch::Tensor> gru_layer_fwd(torch::Tensor xg, torch::Tensor u,
         cp = cache.data_ptr();
     }
     rk::gru_layer_fwd(xg.data_ptr(), u.data_ptr(), bhh.data_ptr<float>(),
-                      hseq.data_ptr(), cp, T, B, cur_stream());
+                      hseq.data_ptr(), cp, T, B, cur_stream(), (uint32_t)dbg);
     std::vector<torch::Tensor> out{hseq};
     if (train) out.push_back(cache);
     return out;
@@ -320,7 +321,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("mfma_probe", &mfma_probe);
     m.def("embed_mlp_fwd", &embed_mlp_fwd);
     m.def("gru_layer_fwd", &gru_layer_fwd, py::arg("xg"), py::arg("u"),
-          py::arg("bhh"), py::arg("train") = false);
+          py::arg("bhh"), py::arg("train") = false, py::arg("dbg") = 0);
     m.def("gru_layer_bwd", &gru_layer_bwd);
     m.def("ce_fwd_bwd", &ce_fwd_bwd);
     m.def("adam_step", &adam_step);

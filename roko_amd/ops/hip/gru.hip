@@ -52,7 +52,9 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gru_layer_fwd_kernel(
     const float* __restrict__ bhh, // (2, 3H)        bias_hh
     bf16* __restrict__ hseq,       // (T, B, 2, H)   output
     bf16* __restrict__ cache,      // (T, B, 2, H, 4) [r z n hgn] or nullptr
-    int T, int B) {
+    int T, int B,
+    uint32_t dbg) {  // timing bisection: 1 no hseq store, 2 no cache store,
+                     // 4 no gate VALU, 8 no xg staging, 16 no MFMA
     __shared__ struct {
         bf16 h[2][MB][HPAD];         // double-buffered hidden-state mirror
         bf16 xgb[2][MB][G3];         // double-buffered step gate inputs
@@ -133,7 +135,7 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gru_layer_fwd_kernel(
         const int t = tfirst + stp * ti;
         // issue xg[t+2] loads FIRST (they retire next step), then write the
         // staged xg[t+1] (loaded last step) into the back LDS buffer
-        if (ti + 2 < T) {
+        if (ti + 2 < T && !(dbg & 8u)) {
             const bf16* src = xg_src(t + 2 * stp);
 #pragma unroll
             for (int p = 0; p < XCH; ++p) {
@@ -143,7 +145,7 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gru_layer_fwd_kernel(
                     *reinterpret_cast<const bf16x8*>(src + (size_t)row * 2 * G3 + col);
             }
         }
-        if (ti + 1 < T) {
+        if (ti + 1 < T && !(dbg & 8u)) {
 #pragma unroll
             for (int p = 0; p < XCH; ++p) {
                 int e = (p * WAVES * 64 + tid) * 8;
@@ -158,14 +160,16 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gru_layer_fwd_kernel(
         for (int mt = 0; mt < 2; ++mt)
 #pragma unroll
             for (int g = 0; g < 3; ++g) acc[mt][g] = f32x4{0.f, 0.f, 0.f, 0.f};
+        if (!(dbg & 16u)) {
 #pragma unroll
-        for (int kb = 0; kb < 4; ++kb) {
+            for (int kb = 0; kb < 4; ++kb) {
 #pragma unroll
-            for (int mt = 0; mt < 2; ++mt) {
-                bf16x8 a = lds_load_a_frag(&lds.h[curp][0][0], mt * 16, kb * 32, HPAD);
+                for (int mt = 0; mt < 2; ++mt) {
+                    bf16x8 a = lds_load_a_frag(&lds.h[curp][0][0], mt * 16, kb * 32, HPAD);
 #pragma unroll
-                for (int g = 0; g < 3; ++g)
-                    acc[mt][g] = mfma16x16x32(a, ufrag[g][kb], acc[mt][g]);
+                    for (int g = 0; g < 3; ++g)
+                        acc[mt][g] = mfma16x16x32(a, ufrag[g][kb], acc[mt][g]);
+                }
             }
         }
 
@@ -177,14 +181,20 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gru_layer_fwd_kernel(
             for (int i = 0; i < 4; ++i) {
                 const int row = mt * 16 + lrow * 4 + i;
                 const int j = j0 + lcol;
-                const float xr = bf2f(lds.xgb[curp][row][0 * H + j]);
-                const float xz = bf2f(lds.xgb[curp][row][1 * H + j]);
-                const float xn = bf2f(lds.xgb[curp][row][2 * H + j]);
-                const float hgn = acc[mt][2][i] + bhh_reg[2];
-                const float r = sigmoidf_dev(xr + acc[mt][0][i] + bhh_reg[0]);
-                const float z = sigmoidf_dev(xz + acc[mt][1][i] + bhh_reg[1]);
-                const float n = tanhf_dev(xn + r * hgn);
-                const float hnew = (1.0f - z) * n + z * hreg[mt][i];
+                float hnew, r, z, n, hgn;
+                if (dbg & 4u) {  // timing: gate math stripped
+                    hnew = acc[mt][0][i] + acc[mt][1][i] + acc[mt][2][i];
+                    r = z = n = hgn = hnew;
+                } else {
+                    const float xr = bf2f(lds.xgb[curp][row][0 * H + j]);
+                    const float xz = bf2f(lds.xgb[curp][row][1 * H + j]);
+                    const float xn = bf2f(lds.xgb[curp][row][2 * H + j]);
+                    hgn = acc[mt][2][i] + bhh_reg[2];
+                    r = sigmoidf_dev(xr + acc[mt][0][i] + bhh_reg[0]);
+                    z = sigmoidf_dev(xz + acc[mt][1][i] + bhh_reg[1]);
+                    n = tanhf_dev(xn + r * hgn);
+                    hnew = (1.0f - z) * n + z * hreg[mt][i];
+                }
                 hreg[mt][i] = hnew;
                 lds.h[curp ^ 1][row][j] = f2bf(hnew);
                 if constexpr (TRAIN) {
@@ -198,7 +208,7 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gru_layer_fwd_kernel(
 
         // cooperative wide store of h to hseq (coalesced 16B per lane),
         // overlaps the next step's MFMAs as plain VMEM traffic
-        {
+        if (!(dbg & 1u)) {
             bf16* dst = hseq + (((size_t)t * B + b0) * 2 + dir) * H;
             const int e = tid * 8;
             const int row = e / H, col = e % H;
@@ -206,13 +216,15 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gru_layer_fwd_kernel(
                 *reinterpret_cast<const bf16x8*>(&lds.h[curp ^ 1][row][col]);
         }
         if constexpr (TRAIN) {
-            bf16* dst = cache + (((size_t)t * B + b0) * 2 + dir) * 4 * H;
+            if (!(dbg & 2u)) {
+                bf16* dst = cache + (((size_t)t * B + b0) * 2 + dir) * 4 * H;
 #pragma unroll
-            for (int p = 0; p < 4; ++p) {
-                const int e = (p * WAVES * 64 + tid) * 8;
-                const int row = e / (4 * H), col = e % (4 * H);
-                *reinterpret_cast<bf16x8*>(dst + (size_t)row * 2 * 4 * H + col) =
-                    *reinterpret_cast<const bf16x8*>(&lds.cache_st[curp ^ 1][row][col]);
+                for (int p = 0; p < 4; ++p) {
+                    const int e = (p * WAVES * 64 + tid) * 8;
+                    const int row = e / (4 * H), col = e % (4 * H);
+                    *reinterpret_cast<bf16x8*>(dst + (size_t)row * 2 * 4 * H + col) =
+                        *reinterpret_cast<const bf16x8*>(&lds.cache_st[curp ^ 1][row][col]);
+                }
             }
         }
     };
@@ -226,18 +238,19 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gru_layer_fwd_kernel(
 }
 
 void gru_layer_fwd(const void* xg, const void* u, const float* bhh, void* hseq,
-                   void* cache, int T, int B, hipStream_t stream) {
+                   void* cache, int T, int B, hipStream_t stream,
+                   uint32_t dbg) {
     dim3 grid(B / MB, 2);
     dim3 block(WAVES * 64);
     if (cache)
         hipLaunchKernelGGL(gru_layer_fwd_kernel<true>, grid, block, 0, stream,
                            static_cast<const bf16*>(xg), static_cast<const bf16*>(u),
                            bhh, static_cast<bf16*>(hseq), static_cast<bf16*>(cache),
-                           T, B);
+                           T, B, dbg);
     else
         hipLaunchKernelGGL(gru_layer_fwd_kernel<false>, grid, block, 0, stream,
                            static_cast<const bf16*>(xg), static_cast<const bf16*>(u),
-                           bhh, static_cast<bf16*>(hseq), nullptr, T, B);
+                           bhh, static_cast<bf16*>(hseq), nullptr, T, B, dbg);
 }
 
 // ---------------------------------------------------------------------------
