@@ -35,8 +35,9 @@ void front_bwd(const uint8_t* ids, const void* dseq, const void* w1,
                hipStream_t stream, uint32_t phase_mask);
 void gemm_bias(const void* A, const void* B, const float* bias, void* C,
                int M, int N, int K, hipStream_t stream);
-void atb_splitk(const void* A, const void* B, float* C, int M, int N, int K,
-                hipStream_t stream);
+int atb_splitk_nslices(int K);
+void atb_splitk(const void* A, const void* B, float* ws, float* C, int M,
+                int N, int K, hipStream_t stream);
 void front_de(const uint8_t* ids, const void* dt1g, const void* w1, float* de,
               int B, uint32_t seed, float keep, hipStream_t stream,
               unsigned long long* timing, uint32_t dbg);
@@ -308,9 +309,11 @@ torch::Tensor atb_splitk(torch::Tensor A, torch::Tensor B) {
     check(B, torch::kBFloat16, "B");
     const int K = A.size(0), M = A.size(1), N = B.size(1);
     TORCH_CHECK(B.size(0) == K, "inner dims mismatch");
-    auto Cout = torch::zeros({M, N}, A.options().dtype(torch::kFloat32));
-    rk::atb_splitk(A.data_ptr(), B.data_ptr(), Cout.data_ptr<float>(), M, N, K,
-                   cur_stream());
+    auto Cout = torch::empty({M, N}, A.options().dtype(torch::kFloat32));
+    auto ws = torch::empty({rk::atb_splitk_nslices(K), M, N},
+                           Cout.options());
+    rk::atb_splitk(A.data_ptr(), B.data_ptr(), ws.data_ptr<float>(),
+                   Cout.data_ptr<float>(), M, N, K, cur_stream());
     return Cout;
 }
 

@@ -154,23 +154,6 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gru_layer_fwd_kernel(
             }
         }
 
-        // prefetch this step's 24 xg gate inputs FIRST: their LDS latency
-        // hides under the MFMA issue below (the fused read-inside-gate-math
-        // version serialized into per-element read->exp chains — the gate
-        // phase measured 1.6 us of the 2.9 us step)
-        bf16 xgv[3][2][4];
-        if (!(dbg & 4u)) {
-#pragma unroll
-            for (int mt = 0; mt < 2; ++mt)
-#pragma unroll
-                for (int i = 0; i < 4; ++i) {
-                    const int row = mt * 16 + lrow * 4 + i;
-                    const int j = j0 + lcol;
-#pragma unroll
-                    for (int g = 0; g < 3; ++g)
-                        xgv[g][mt][i] = lds.xgb[curp][row][g * H + j];
-                }
-        }
         // gates_h = h · U^T  (24 MFMA per wave, A-frags shared 3-ways)
         f32x4 acc[2][3];
 #pragma unroll
@@ -203,9 +186,9 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gru_layer_fwd_kernel(
                     hnew = acc[mt][0][i] + acc[mt][1][i] + acc[mt][2][i];
                     r = z = n = hgn = hnew;
                 } else {
-                    const float xr = bf2f(xgv[0][mt][i]);
-                    const float xz = bf2f(xgv[1][mt][i]);
-                    const float xn = bf2f(xgv[2][mt][i]);
+                    const float xr = bf2f(lds.xgb[curp][row][0 * H + j]);
+                    const float xz = bf2f(lds.xgb[curp][row][1 * H + j]);
+                    const float xn = bf2f(lds.xgb[curp][row][2 * H + j]);
                     hgn = acc[mt][2][i] + bhh_reg[2];
                     r = sigmoidf_dev(xr + acc[mt][0][i] + bhh_reg[0]);
                     z = sigmoidf_dev(xz + acc[mt][1][i] + bhh_reg[1]);
@@ -383,21 +366,7 @@ __global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
         if (sidx + 2 < T) issue_loads(sidx + 2, rc_ld, rdh_ld, rhp_ld);
         if (sidx + 1 < T) write_stage(curp ^ 1, rc_wr, rdh_wr, rhp_wr);
 
-        // ---- gate gradients from stage[curp] (inputs batch-read first so
-        // the LDS latencies overlap instead of chaining per element) -------
-        bf16 dhv[2][4], hpv[2][4];
-        bf16x4 pkv[2][4];
-#pragma unroll
-        for (int mt = 0; mt < 2; ++mt)
-#pragma unroll
-            for (int i = 0; i < 4; ++i) {
-                const int row = mt * 16 + lrow * 4 + i;
-                const int j = j0 + lcol;
-                dhv[mt][i] = lds.dhin_st[curp][row][j];
-                hpv[mt][i] = lds.hprev_st[curp][row][j];
-                pkv[mt][i] = *reinterpret_cast<const bf16x4*>(
-                    &lds.cache_st[curp][row][4 * j]);
-            }
+        // ---- gate gradients from stage[curp] ------------------------------
         float dhp_part[2][4];
 #pragma unroll
         for (int mt = 0; mt < 2; ++mt) {
@@ -405,13 +374,14 @@ __global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
             for (int i = 0; i < 4; ++i) {
                 const int row = mt * 16 + lrow * 4 + i;
                 const int j = j0 + lcol;
-                const float dh = dhc[mt][i] + bf2f(dhv[mt][i]);
-                const bf16x4 pk = pkv[mt][i];
+                const float dh = dhc[mt][i] + bf2f(lds.dhin_st[curp][row][j]);
+                const bf16x4 pk = *reinterpret_cast<const bf16x4*>(
+                    &lds.cache_st[curp][row][4 * j]);
                 const float r = bf2f(pk[0]);
                 const float z = bf2f(pk[1]);
                 const float n = bf2f(pk[2]);
                 const float hgn = bf2f(pk[3]);
-                const float hp = bf2f(hpv[mt][i]);
+                const float hp = bf2f(lds.hprev_st[curp][row][j]);
                 const float dn = dh * (1.0f - z);
                 const float dz = dh * (hp - n);
                 const float dan = dn * (1.0f - n * n);
