@@ -21,6 +21,11 @@ void ce_fwd_bwd(const float* logits, const int64_t* target, float* dlogits,
 void adam_step(float* p, const float* g, float* m, float* v, int64_t n,
                float lr, float beta1, float beta2, float eps, int step,
                hipStream_t stream);
+void adam_mt(const int64_t* table, int n_params, float* p, float* m, float* v,
+             float lr, float beta1, float beta2, float eps, int step,
+             hipStream_t stream);
+void grad_gather(const int64_t* table, int n_params, float* flat_g,
+                 hipStream_t stream);
 void head_fwd(const void* hseq, const void* w4, const float* b4, float* logits,
               uint8_t* amax, int T, int B, hipStream_t stream);
 void emb_grad(const void* dm, const uint8_t* ids, float* de, int64_t n,
@@ -166,6 +171,23 @@ void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
     rk::adam_step(p.data_ptr<float>(), g.data_ptr<float>(), m.data_ptr<float>(),
                   v.data_ptr<float>(), n, float(lr), float(beta1), float(beta2),
                   float(eps), int(step), cur_stream());
+}
+
+// multi-tensor Adam: table rows [grad_ptr, flat_offset, numel] (GPU int64)
+void adam_mt(torch::Tensor table, int64_t n_params, torch::Tensor p,
+             torch::Tensor m, torch::Tensor v, double lr, double beta1,
+             double beta2, double eps, int64_t step) {
+    TORCH_CHECK(table.is_cuda() && table.scalar_type() == torch::kInt64);
+    rk::adam_mt(table.data_ptr<int64_t>(), (int)n_params, p.data_ptr<float>(),
+                m.data_ptr<float>(), v.data_ptr<float>(), (float)lr,
+                (float)beta1, (float)beta2, (float)eps, (int)step,
+                cur_stream());
+}
+
+void grad_gather(torch::Tensor table, int64_t n_params, torch::Tensor flat_g) {
+    TORCH_CHECK(table.is_cuda() && table.scalar_type() == torch::kInt64);
+    rk::grad_gather(table.data_ptr<int64_t>(), (int)n_params,
+                    flat_g.data_ptr<float>(), cur_stream());
 }
 
 // dm (N, 50) bf16 + ids (N) u8 -> dE (12, 50) f32
@@ -328,6 +350,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("gru_layer_bwd", &gru_layer_bwd);
     m.def("ce_fwd_bwd", &ce_fwd_bwd);
     m.def("adam_step", &adam_step);
+    m.def("adam_mt", &adam_mt);
+    m.def("grad_gather", &grad_gather);
     m.def("emb_grad", &emb_grad);
     m.def("front_fwd", &front_fwd);
     m.def("atb_splitk", &atb_splitk);

@@ -121,9 +121,13 @@ def fused_cross_entropy(logits: torch.Tensor, target: torch.Tensor) -> torch.Ten
 
 
 class FusedAdam:
-    """Flat-buffer Adam: all params/grads/moments live in single fp32 flat
-    tensors (param tensors become views), the update is ONE kernel and the
-    DP gradient sync is ONE all-reduce on the flat grad."""
+    """Flat-buffer Adam with grad=None accumulation semantics: parameters and
+    moments live in flat fp32 tensors (param tensors are views); gradients
+    stay per-tensor (autograd ASSIGNS them — no AccumulateGrad add kernel per
+    parameter, which cost ~123 us/step as 26 tiny launches) and the update is
+    one multi-tensor kernel driven by a [grad_ptr, offset, numel] table. In
+    DP a single gather kernel packs the grads into one flat tensor for ONE
+    all-reduce."""
 
     def __init__(self, params: List[torch.nn.Parameter], lr: float = C.LR,
                  betas=(0.9, 0.999), eps: float = 1e-8):
@@ -136,28 +140,54 @@ class FusedAdam:
         self.flat_g = torch.zeros(n, dtype=torch.float32, device=dev)
         self.m = torch.zeros(n, dtype=torch.float32, device=dev)
         self.v = torch.zeros(n, dtype=torch.float32, device=dev)
+        self.offs = []
         off = 0
         for p in self.params:
             k = p.numel()
             self.flat_p[off : off + k].copy_(p.data.reshape(-1))
             p.data = self.flat_p[off : off + k].view(p.shape)
-            p.grad = self.flat_g[off : off + k].view(p.shape)
+            self.offs.append(off)
             off += k
 
     def zero_grad(self):
-        self.flat_g.zero_()
+        for p in self.params:
+            p.grad = None
+
+    def _table(self):
+        rows = []
+        for p, off in zip(self.params, self.offs):
+            g = p.grad
+            assert g is not None, "param missing grad"
+            if g.dtype != torch.float32 or not g.is_contiguous():
+                g = g.float().contiguous()
+                p.grad = g
+            rows.append((g.data_ptr(), off, g.numel()))
+        cpu = torch.tensor(rows, dtype=torch.int64)
+        return cpu.to(self.flat_p.device, non_blocking=True)
 
     def allreduce_grads(self):
         import torch.distributed as dist
 
         if dist.is_initialized() and dist.get_world_size() > 1:
+            tab = self._table()
+            _ext().grad_gather(tab, len(self.params), self.flat_g)
             dist.all_reduce(self.flat_g)
             self.flat_g /= dist.get_world_size()
+            self._synced = True
+        else:
+            self._synced = False
 
     def step(self):
         self.step_count += 1
-        _ext().adam_step(self.flat_p, self.flat_g, self.m, self.v, self.lr,
-                         self.betas[0], self.betas[1], self.eps, self.step_count)
+        if getattr(self, "_synced", False):
+            _ext().adam_step(self.flat_p, self.flat_g, self.m, self.v, self.lr,
+                             self.betas[0], self.betas[1], self.eps,
+                             self.step_count)
+        else:
+            tab = self._table()
+            _ext().adam_mt(tab, len(self.params), self.flat_p, self.m, self.v,
+                           self.lr, self.betas[0], self.betas[1], self.eps,
+                           self.step_count)
 
 
 class FrontFn(torch.autograd.Function):
