@@ -207,9 +207,9 @@ def test_fused_adam_vs_torch():
     for it in range(5):
         g1 = torch.randn_like(p1)
         g2 = torch.randn_like(p2)
-        opt.zero_grad()
-        p1.grad.copy_(g1)
-        p2.grad.copy_(g2)
+        opt.zero_grad()  # grad=None semantics: assign fresh tensors
+        p1.grad = g1.clone()
+        p2.grad = g2.clone()
         opt.step()
         ref_opt.zero_grad()
         ref1.grad = g1.clone()
