@@ -14,8 +14,8 @@ void embed_mlp_fwd(const uint8_t* ids, const void* w1, const float* b1,
 void gru_layer_fwd(const void* xg, const void* u, const float* bhh, void* hseq,
                    void* cache, int T, int B, hipStream_t stream, uint32_t dbg);
 void gru_layer_bwd(const void* cache, const void* hseq, const void* dhin,
-                   const void* ut, void* dxg, void* dhg, int T, int B,
-                   hipStream_t stream);
+                   const void* ut, void* dxg, void* dhg, float* db_ih,
+                   float* dbhh, int T, int B, hipStream_t stream);
 void ce_fwd_bwd(const float* logits, const int64_t* target, float* dlogits,
                 float* loss_sum, int64_t n, hipStream_t stream);
 void adam_step(float* p, const float* g, float* m, float* v, int64_t n,
@@ -136,10 +136,14 @@ std::vector<torch::Tensor> gru_layer_bwd(torch::Tensor cache, torch::Tensor hseq
                 "ut must be (2,128,384)");
     auto dxg = torch::empty({T, B, 2, 384}, cache.options());
     auto dhg = torch::empty({2, T, B, 384}, cache.options());
+    auto fopt = cache.options().dtype(torch::kFloat32);
+    auto db_ih = torch::zeros({2, 384}, fopt);
+    auto dbhh = torch::zeros({2, 384}, fopt);
     rk::gru_layer_bwd(cache.data_ptr(), hseq.data_ptr(), dhin.data_ptr(),
-                      ut.data_ptr(), dxg.data_ptr(), dhg.data_ptr(), T, B,
+                      ut.data_ptr(), dxg.data_ptr(), dhg.data_ptr(),
+                      db_ih.data_ptr<float>(), dbhh.data_ptr<float>(), T, B,
                       cur_stream());
-    return {dxg, dhg};
+    return {dxg, dhg, db_ih, dbhh};
 }
 
 // fused CE: returns (loss scalar f32, dlogits (N,5) f32) for mean reduction

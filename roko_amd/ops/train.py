@@ -63,7 +63,7 @@ class GruLayerFn(torch.autograd.Function):
         ut = u_bf.transpose(1, 2).contiguous()  # (2, 128, 384)
         # kernel emits GEMM-ready layouts: dxg (T,B,2,384) -> (TB,768) view;
         # dhg (2,T,B,384) -> contiguous per-direction (TB,384) slices
-        dxg, dhg = ext.gru_layer_bwd(cache, hseq, dhin, ut)
+        dxg, dhg, db2_ih, dbhh2 = ext.gru_layer_bwd(cache, hseq, dhin, ut)
 
         # h_prev sequences per direction
         zeros = hseq.new_zeros(1, B, H)
@@ -83,13 +83,9 @@ class GruLayerFn(torch.autograd.Function):
         dxg_cat = dxg.reshape(TB, 2 * 3 * H)  # zero-copy (TB, 768)
         x_flat = x_bf.reshape(TB, -1)
         dw_ih = ext.atb_splitk(dxg_cat, x_flat.contiguous())  # (768, in) f32
-        # column sums as one hipBLASLt GEMV each — aten's strided bf16
-        # .sum(0) over (TB, 768) was ~100 us/call (profiles/train_r01)
-        ones = x_bf.new_ones(1, TB)
-        dbhh = torch.stack(
-            [ones.mm(dhg_f).squeeze(0), ones.mm(dhg_r).squeeze(0)]
-        ).float()
-        db_ih = ones.mm(dxg_cat).squeeze(0).float()      # (768,)
+        # bias grads come straight from the BPTT kernel's per-lane partials
+        db_ih = db2_ih.reshape(-1)
+        dbhh = dbhh2
         dx = dxg_cat.mm(w_ih_bf).to(ctx.in_dtype).view(T, B, -1)
 
         return dx, dw_ih, db_ih, du, dbhh
