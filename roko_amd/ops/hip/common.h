@@ -102,6 +102,18 @@ RK_DEV void agent_atomic_add(float* p, float v) {
     __hip_atomic_fetch_add(p, v, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
 }
 
+// async global->LDS DMA, 16 B per lane (gfx950 global_load_lds_dwordx4).
+// LDS destination is WAVE-UNIFORM base + lane*16 — the caller's LDS layout
+// must be lane-linear; the global source address is per-lane.
+RK_DEV void glds_b128(const void* g, void* l) {
+#pragma clang diagnostic push
+#pragma clang diagnostic ignored "-Wold-style-cast"
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) uint32_t*)(g),
+        (__attribute__((address_space(3))) uint32_t*)(l), 16, 0, 0);
+#pragma clang diagnostic pop
+}
+
 RK_DEV float sigmoidf_dev(float x) { return 1.0f / (1.0f + __expf(-x)); }
 
 // overflow-stable tanh: tanh(x) = sign(x) * (1 - e) / (1 + e), e = exp(-2|x|)

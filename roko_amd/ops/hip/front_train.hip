@@ -86,7 +86,8 @@ __global__ __launch_bounds__(512, 2) void front_fwd_kernel(
     const float* __restrict__ b2,     // (F2)
     const bf16* __restrict__ emb,     // (12, E)
     bf16* __restrict__ out,           // (W, B, OUT)
-    int B, uint32_t seed, float keep) {
+    int B, uint32_t seed, float keep,
+    const uint32_t* __restrict__ seed_ptr) {  // overrides `seed` if non-null
     __shared__ struct {
         bf16 w1t[MP][KP_LD];     // zero-padded W1 [f][r] (G1 A-operand)
         bf16 m_t[EP][KP_LD];     // masked embedding tile [e][r] (G1 B)
@@ -105,6 +106,7 @@ __global__ __launch_bounds__(512, 2) void front_fwd_kernel(
     const int lane = tid & 63;
     const int lrow = lane >> 4;
     const int lcol = lane & 15;
+    if (seed_ptr) seed = *seed_ptr;
     const uint32_t thresh16 = (uint32_t)(keep * 65536.0f);
     const float inv_keep = 1.0f / keep;
     const bool no_drop = keep >= 1.0f;  // eval/inference: skip every hash
@@ -298,7 +300,8 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
                               // tile fits LDS; scalar W1 reads made the dm
                               // phase HALF this kernel's time,
                               // profiles/front_bwd_phases_r01)
-    int B, uint32_t seed, float keep, uint32_t phase_mask) {
+    int B, uint32_t seed, float keep, uint32_t phase_mask,
+    const uint32_t* __restrict__ seed_ptr) {
     // phase_mask: timing-experiment switch (default 0x1F = all phases).
     // bit0 G1 recompute, bit1 G3+dt2, bit2 dt1, bit3 dW1/dW2, bit4 dm/de.
     __shared__ struct {
@@ -337,6 +340,7 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
     const int lane = tid & 63;
     const int lrow = lane >> 4;
     const int lcol = lane & 15;
+    if (seed_ptr) seed = *seed_ptr;
     const uint32_t thresh16 = (uint32_t)(keep * 65536.0f);
     const float inv_keep = 1.0f / keep;
     const int w_begin = blockIdx.y * (W / CSPLIT);
@@ -632,7 +636,8 @@ __global__ __launch_bounds__(512, 2) void front_de_kernel(
     float* __restrict__ de,            // (12, E) pre-zeroed
     int B, uint32_t seed, float keep,
     unsigned long long* __restrict__ timing,  // optional (4) cycle counters
-    uint32_t dbg) {  // timing-bisection: 1 skip atomics, 2 skip hash, 4 skip epi
+    uint32_t dbg,
+    const uint32_t* __restrict__ seed_ptr) {  // timing-bisection: 1 skip atomics, 2 skip hash, 4 skip epi
     __shared__ struct {
         bf16 w1_rt[R + 8][136];   // [r][f] = W1^T, zero-padded
         bf16 dt1_ef[EP][136];     // [e][f] staged column of dt1g
@@ -647,6 +652,7 @@ __global__ __launch_bounds__(512, 2) void front_de_kernel(
     const int lane = tid & 63;
     const int lrow = lane >> 4;
     const int lcol = lane & 15;
+    if (seed_ptr) seed = *seed_ptr;
     const uint32_t thresh16 = (uint32_t)(keep * 65536.0f);
     const float inv_keep = 1.0f / keep;
     const int w_begin = blockIdx.y * (W / CSPLIT);
@@ -770,37 +776,41 @@ __global__ __launch_bounds__(512, 2) void front_de_kernel(
 
 void front_fwd(const uint8_t* ids, const void* w1, const float* b1,
                const void* w2, const float* b2, const void* emb, void* out,
-               int B, uint32_t seed, float keep, hipStream_t stream) {
+               int B, uint32_t seed, float keep, hipStream_t stream,
+               const uint32_t* seed_ptr) {
     hipLaunchKernelGGL(front::front_fwd_kernel,
                        dim3(B, front::CSPLIT), dim3(512), 0, stream,
                        ids, static_cast<const bf16*>(w1), b1,
                        static_cast<const bf16*>(w2), b2,
                        static_cast<const bf16*>(emb), static_cast<bf16*>(out),
-                       B, seed, keep);
+                       B, seed, keep, seed_ptr);
 }
 
 void front_de(const uint8_t* ids, const void* dt1g, const void* w1, float* de,
               int B, uint32_t seed, float keep, hipStream_t stream,
-              unsigned long long* timing, uint32_t dbg) {
+              unsigned long long* timing, uint32_t dbg,
+              const uint32_t* seed_ptr) {
     hipLaunchKernelGGL(front::front_de_kernel,
                        dim3(B, front::CSPLIT), dim3(512), 0, stream,
                        ids, static_cast<const bf16*>(dt1g),
                        static_cast<const bf16*>(w1), de, B, seed, keep, timing,
-                       dbg);
+                       dbg, seed_ptr);
 }
 
 void front_bwd(const uint8_t* ids, const void* dseq, const void* w1,
                const float* b1, const void* w2, const float* b2,
                const void* emb, float* dw1, float* db1, float* dw2, float* db2,
                void* dt1g, int B, uint32_t seed, float keep,
-               hipStream_t stream, uint32_t phase_mask) {
+               hipStream_t stream, uint32_t phase_mask,
+               const uint32_t* seed_ptr) {
     hipLaunchKernelGGL(front::front_bwd_kernel,
                        dim3(B, front::CSPLIT), dim3(512), 0, stream,
                        ids, static_cast<const bf16*>(dseq),
                        static_cast<const bf16*>(w1), b1,
                        static_cast<const bf16*>(w2), b2,
                        static_cast<const bf16*>(emb), dw1, db1, dw2, db2,
-                       static_cast<bf16*>(dt1g), B, seed, keep, phase_mask);
+                       static_cast<bf16*>(dt1g), B, seed, keep, phase_mask,
+                       seed_ptr);
 }
 
 }  // namespace rk

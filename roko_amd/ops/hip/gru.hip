@@ -301,15 +301,11 @@ __global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
     const int lrow = lane >> 4;
     const int lcol = lane & 15;
 
-    // B-fragments of U for dh_prev = dhg·U: B[k=j][col=khid] = Ut[khid][j]
-    bf16x8 ufrag[12];
-#pragma unroll
-    for (int kb = 0; kb < 12; ++kb) {
-        const int col = j0 + lcol;          // this wave's hidden column
-        const int j = kb * 32 + 8 * lrow;   // gate-row index
-        ufrag[kb] = *reinterpret_cast<const bf16x8*>(
-            ut + ((size_t)dir * H + col) * G3 + j);
-    }
+    // U fragments are re-read from L2 every step instead of living in 48
+    // registers: together with the staging/bias-sum registers the persistent
+    // copy spilled to scratch inside the 90-step loop
+    const bf16* u_base =
+        ut + ((size_t)dir * H + j0 + lcol) * G3 + 8 * lrow;
 
     float dhc[2][4];  // dh carry, fragment-shaped
 #pragma unroll
@@ -324,52 +320,44 @@ __global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
     const int t0 = (dir == 0) ? T - 1 : 0;
     auto t_of = [&](int sidx) { return t0 + stp * sidx; };
 
-    // register-staged loads for one step: cache (4 chunks) + dhin + hprev
-    auto issue_loads = [&](int sidx, bf16x8 (&rc)[BWCH_C], bf16x8& rdh,
-                           bf16x8& rhp) {
+    // staging via direct global->LDS DMA: no staging registers at all (the
+    // register pipeline + bias-sum accumulators spilled to scratch). The
+    // implicit vmcnt(0) drain at the next __syncthreads() is the completion
+    // guarantee — still more than a full HBM latency of cover.
+    auto glds_stage = [&](int buf, int sidx) {
         const int t = t_of(sidx);
         const int tp = (dir == 0) ? t - 1 : t + 1;
-        const bf16* src = cache + (((size_t)t * B + b0) * 2 + dir) * 4 * H;
-        const int row4 = tid / 64, col4 = (tid % 64) * 8;
+        {
+            const bf16* src = cache + (((size_t)t * B + b0) * 2 + dir) * 4 * H;
+            const int row4 = tid / 64, col4 = (tid % 64) * 8;
 #pragma unroll
-        for (int q = 0; q < BWCH_C; ++q)
-            rc[q] = *reinterpret_cast<const bf16x8*>(
-                src + (size_t)(row4 + q * 8) * 2 * 4 * H + col4);
-        const int row = tid / 16, col = (tid % 16) * 8;
-        rdh = *reinterpret_cast<const bf16x8*>(
-            dhin + (((size_t)t * B + b0) * 2 + dir) * H + (size_t)row * 2 * H + col);
-        if (tp >= 0 && tp < T)
-            rhp = *reinterpret_cast<const bf16x8*>(
-                hseq + (((size_t)tp * B + b0) * 2 + dir) * H + (size_t)row * 2 * H + col);
-        else
-            rhp = bf16x8{};
-    };
-    auto write_stage = [&](int buf, bf16x8 (&rc)[BWCH_C], bf16x8& rdh,
-                           bf16x8& rhp) {
-        const int row4 = tid / 64, col4 = (tid % 64) * 8;
-#pragma unroll
-        for (int q = 0; q < BWCH_C; ++q)
-            *reinterpret_cast<bf16x8*>(&lds.cache_st[buf][row4 + q * 8][col4]) = rc[q];
-        const int row = tid / 16, col = (tid % 16) * 8;
-        *reinterpret_cast<bf16x8*>(&lds.dhin_st[buf][row][col]) = rdh;
-        *reinterpret_cast<bf16x8*>(&lds.hprev_st[buf][row][col]) = rhp;
+            for (int q = 0; q < BWCH_C; ++q)
+                glds_b128(src + (size_t)(row4 + q * 8) * 2 * 4 * H + col4,
+                          &lds.cache_st[buf][wid + q * 8][0]);
+        }
+        {
+            const int row = tid / 16, col = (tid % 16) * 8;
+            glds_b128(dhin + (((size_t)t * B + b0) * 2 + dir) * H +
+                          (size_t)row * 2 * H + col,
+                      &lds.dhin_st[buf][wid * 4][0]);
+            if (tp >= 0 && tp < T) {
+                glds_b128(hseq + (((size_t)tp * B + b0) * 2 + dir) * H +
+                              (size_t)row * 2 * H + col,
+                          &lds.hprev_st[buf][wid * 4][0]);
+            } else {
+                *reinterpret_cast<bf16x8*>(
+                    &lds.hprev_st[buf][row][col]) = bf16x8{};
+            }
+        }
     };
 
-    bf16x8 rcA[BWCH_C], rcB[BWCH_C];
-    bf16x8 rdhA, rhpA, rdhB, rhpB;
-    // prologue: step 0 straight to LDS buffer 0; step 1 to registers A
-    issue_loads(0, rcA, rdhA, rhpA);
-    write_stage(0, rcA, rdhA, rhpA);
-    if (T > 1) issue_loads(1, rcA, rdhA, rhpA);
+    glds_stage(0, 0);
     __syncthreads();
 
-    auto body = [&](int sidx, int curp, bf16x8 (&rc_wr)[BWCH_C], bf16x8& rdh_wr,
-                    bf16x8& rhp_wr, bf16x8 (&rc_ld)[BWCH_C], bf16x8& rdh_ld,
-                    bf16x8& rhp_ld) {
+    auto body = [&](int sidx, int curp) {
         const int t = t_of(sidx);
-        // issue step s+2 loads FIRST, then write step s+1's staged registers
-        if (sidx + 2 < T) issue_loads(sidx + 2, rc_ld, rdh_ld, rhp_ld);
-        if (sidx + 1 < T) write_stage(curp ^ 1, rc_wr, rdh_wr, rhp_wr);
+        // issue next step's LDS-DMA into the back buffers
+        if (sidx + 1 < T) glds_stage(curp ^ 1, sidx + 1);
 
         // ---- gate gradients from stage[curp] ------------------------------
         float dhp_part[2][4];
@@ -414,12 +402,17 @@ __global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
         f32x4 acc[2];
 #pragma unroll
         for (int mt = 0; mt < 2; ++mt) acc[mt] = f32x4{0.f, 0.f, 0.f, 0.f};
+        // opaque pointer copy so LICM cannot hoist the 12 loads back into
+        // 48 persistent registers
+        const bf16* ub = u_base;
+        asm volatile("" : "+v"(ub));
 #pragma unroll
         for (int kb = 0; kb < 12; ++kb) {
+            const bf16x8 uf = *reinterpret_cast<const bf16x8*>(ub + kb * 32);
 #pragma unroll
             for (int mt = 0; mt < 2; ++mt) {
                 bf16x8 a = lds_load_a_frag(&lds.dhg[0][0], mt * 16, kb * 32, G3 + 8);
-                acc[mt] = mfma16x16x32(a, ufrag[kb], acc[mt]);
+                acc[mt] = mfma16x16x32(a, uf, acc[mt]);
             }
         }
 #pragma unroll
@@ -458,12 +451,7 @@ __global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
         __syncthreads();  // dhg/dg_st/stage[curp] reads done before reuse
     };
 
-    int sidx = 0;
-    for (; sidx + 2 <= T; sidx += 2) {
-        body(sidx, 0, rcA, rdhA, rhpA, rcB, rdhB, rhpB);
-        body(sidx + 1, 1, rcB, rdhB, rhpB, rcA, rdhA, rhpA);
-    }
-    if (sidx < T) body(sidx, 0, rcA, rdhA, rhpA, rcB, rdhB, rhpB);
+    for (int sidx = 0; sidx < T; ++sidx) body(sidx, sidx & 1);
 
     // fold the 4 lrow groups (lanes 16 apart share j), then lane-per-j
     // commits the six bias-grad sums
