@@ -301,11 +301,18 @@ __global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
     const int lrow = lane >> 4;
     const int lcol = lane & 15;
 
-    // U fragments are re-read from L2 every step instead of living in 48
-    // registers: together with the staging/bias-sum registers the persistent
-    // copy spilled to scratch inside the 90-step loop
-    const bf16* u_base =
-        ut + ((size_t)dir * H + j0 + lcol) * G3 + 8 * lrow;
+    // B-fragments of U for dh_prev = dhg·U, persistent in registers: an
+    // in-loop global re-read forces a vmcnt(0) at its first use that drains
+    // the in-flight LDS-DMA staging every step (glds+ordinary-load caveat,
+    // cdna_hip_programming.md §5)
+    bf16x8 ufrag[12];
+#pragma unroll
+    for (int kb = 0; kb < 12; ++kb) {
+        const int col = j0 + lcol;
+        const int j = kb * 32 + 8 * lrow;
+        ufrag[kb] = *reinterpret_cast<const bf16x8*>(
+            ut + ((size_t)dir * H + col) * G3 + j);
+    }
 
     float dhc[2][4];  // dh carry, fragment-shaped
 #pragma unroll
@@ -402,17 +409,12 @@ __global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
         f32x4 acc[2];
 #pragma unroll
         for (int mt = 0; mt < 2; ++mt) acc[mt] = f32x4{0.f, 0.f, 0.f, 0.f};
-        // opaque pointer copy so LICM cannot hoist the 12 loads back into
-        // 48 persistent registers
-        const bf16* ub = u_base;
-        asm volatile("" : "+v"(ub));
 #pragma unroll
         for (int kb = 0; kb < 12; ++kb) {
-            const bf16x8 uf = *reinterpret_cast<const bf16x8*>(ub + kb * 32);
 #pragma unroll
             for (int mt = 0; mt < 2; ++mt) {
                 bf16x8 a = lds_load_a_frag(&lds.dhg[0][0], mt * 16, kb * 32, G3 + 8);
-                acc[mt] = mfma16x16x32(a, uf, acc[mt]);
+                acc[mt] = mfma16x16x32(a, ufrag[kb], acc[mt]);
             }
         }
 #pragma unroll
