@@ -89,7 +89,8 @@ def bench_inference(args, rank, world, device):
 
 
 def bench_train(args, rank, world, device):
-    from roko_amd.ops.train import FusedAdam, fused_train_step, train_step_available
+    from roko_amd.ops.train import (FusedAdam, GraphedTrainStep,
+                                    fused_train_step, train_step_available)
 
     if device.type != "cuda" or not train_step_available():
         raise SystemExit("train bench requires the fused HIP train step on GPU")
@@ -101,6 +102,18 @@ def bench_train(args, rank, world, device):
                       generator=g, dtype=torch.uint8).to(device)
     y = torch.randint(0, C.NUM_CLASSES, (args.batch, C.WINDOW_COLS),
                       generator=g).to(device)
+
+    stepper = None
+    if os.environ.get("ROKO_NO_GRAPH") != "1":
+        try:
+            stepper = GraphedTrainStep(model, opt, args.batch, world)
+        except Exception as e:  # noqa: BLE001 — capture support is optional
+            if rank == 0:
+                print(f"hipGraph train capture unavailable ({e!r}); "
+                      "falling back to eager steps", file=sys.stderr)
+    if stepper is not None:
+        def fused_train_step(model_, x_, y_, opt_):  # noqa: F811 shadow
+            return stepper(x_, y_)
 
     for _ in range(args.warmup):
         fused_train_step(model, x, y, opt)

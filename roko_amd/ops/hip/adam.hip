@@ -13,7 +13,12 @@ namespace rk {
 __global__ __launch_bounds__(256) void adam_step_kernel(
     float* __restrict__ p, const float* __restrict__ g, float* __restrict__ m,
     float* __restrict__ v, int64_t n, float lr, float beta1, float beta2,
-    float eps, float bc1, float bc2) {
+    float eps, float bc1, float bc2, const int* __restrict__ step_ptr) {
+    if (step_ptr) {  // hipGraph path: device-resident step counter
+        const float st = (float)*step_ptr;
+        bc1 = 1.0f - __powf(beta1, st);
+        bc2 = 1.0f - __powf(beta2, st);
+    }
     const int64_t stride = (int64_t)gridDim.x * 256;
     for (int64_t i = (int64_t)blockIdx.x * 256 + threadIdx.x; i < n; i += stride) {
         const float gi = g[i];
@@ -29,13 +34,13 @@ __global__ __launch_bounds__(256) void adam_step_kernel(
 
 void adam_step(float* p, const float* g, float* m, float* v, int64_t n,
                float lr, float beta1, float beta2, float eps, int step,
-               hipStream_t stream) {
+               hipStream_t stream, const int* step_ptr) {
     const float bc1 = 1.0f - powf(beta1, float(step));
     const float bc2 = 1.0f - powf(beta2, float(step));
     int blocks = int((n + 255) / 256);
     if (blocks > 2048) blocks = 2048;
     hipLaunchKernelGGL(adam_step_kernel, dim3(blocks), dim3(256), 0, stream, p,
-                       g, m, v, n, lr, beta1, beta2, eps, bc1, bc2);
+                       g, m, v, n, lr, beta1, beta2, eps, bc1, bc2, step_ptr);
 }
 
 // ---------------------------------------------------------------------------
@@ -48,9 +53,17 @@ void adam_step(float* p, const float* g, float* m, float* v, int64_t n,
 __global__ __launch_bounds__(256) void adam_mt_kernel(
     const int64_t* __restrict__ table, int n_params, float* __restrict__ p,
     float* __restrict__ m, float* __restrict__ v, float lr, float beta1,
-    float beta2, float eps, float bc1, float bc2) {
+    float beta2, float eps, float bc1, float bc2,
+    const int* __restrict__ step_ptr) {  // hipGraph path: step counter lives
+                                         // on device, bias corrections are
+                                         // computed in-kernel
     const int pi = blockIdx.y;
     if (pi >= n_params) return;
+    if (step_ptr) {
+        const float st = (float)*step_ptr;
+        bc1 = 1.0f - __powf(beta1, st);
+        bc2 = 1.0f - __powf(beta2, st);
+    }
     const float* g = reinterpret_cast<const float*>(table[pi * 3 + 0]);
     const int64_t off = table[pi * 3 + 1];
     const int64_t n = table[pi * 3 + 2];
@@ -84,12 +97,12 @@ __global__ __launch_bounds__(256) void grad_gather_kernel(
 
 void adam_mt(const int64_t* table, int n_params, float* p, float* m, float* v,
              float lr, float beta1, float beta2, float eps, int step,
-             hipStream_t stream) {
+             hipStream_t stream, const int* step_ptr) {
     const float bc1 = 1.0f - powf(beta1, float(step));
     const float bc2 = 1.0f - powf(beta2, float(step));
     hipLaunchKernelGGL(adam_mt_kernel, dim3(32, n_params), dim3(256), 0,
                        stream, table, n_params, p, m, v, lr, beta1, beta2, eps,
-                       bc1, bc2);
+                       bc1, bc2, step_ptr);
 }
 
 void grad_gather(const int64_t* table, int n_params, float* flat_g,

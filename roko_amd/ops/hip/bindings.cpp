@@ -14,16 +14,16 @@ void embed_mlp_fwd(const uint8_t* ids, const void* w1, const float* b1,
 void gru_layer_fwd(const void* xg, const void* u, const float* bhh, void* hseq,
                    void* cache, int T, int B, hipStream_t stream, uint32_t dbg);
 void gru_layer_bwd(const void* cache, const void* hseq, const void* dhin,
-                   const void* ut, void* dxg, void* dhg, float* db_ih,
-                   float* dbhh, int T, int B, hipStream_t stream);
+                   const void* ut, void* dxg, void* dhg, int T, int B,
+                   hipStream_t stream);
 void ce_fwd_bwd(const float* logits, const int64_t* target, float* dlogits,
                 float* loss_sum, int64_t n, hipStream_t stream);
 void adam_step(float* p, const float* g, float* m, float* v, int64_t n,
                float lr, float beta1, float beta2, float eps, int step,
-               hipStream_t stream);
+               hipStream_t stream, const int* step_ptr);
 void adam_mt(const int64_t* table, int n_params, float* p, float* m, float* v,
              float lr, float beta1, float beta2, float eps, int step,
-             hipStream_t stream);
+             hipStream_t stream, const int* step_ptr);
 void grad_gather(const int64_t* table, int n_params, float* flat_g,
                  hipStream_t stream);
 void head_fwd(const void* hseq, const void* w4, const float* b4, float* logits,
@@ -32,12 +32,14 @@ void emb_grad(const void* dm, const uint8_t* ids, float* de, int64_t n,
               hipStream_t stream);
 void front_fwd(const uint8_t* ids, const void* w1, const float* b1,
                const void* w2, const float* b2, const void* emb, void* out,
-               int B, uint32_t seed, float keep, hipStream_t stream);
+               int B, uint32_t seed, float keep, hipStream_t stream,
+               const uint32_t* seed_ptr);
 void front_bwd(const uint8_t* ids, const void* dseq, const void* w1,
                const float* b1, const void* w2, const float* b2,
                const void* emb, float* dw1, float* db1, float* dw2, float* db2,
                void* dt1g, int B, uint32_t seed, float keep,
-               hipStream_t stream, uint32_t phase_mask);
+               hipStream_t stream, uint32_t phase_mask,
+               const uint32_t* seed_ptr);
 void gemm_bias(const void* A, const void* B, const float* bias, void* C,
                int M, int N, int K, hipStream_t stream);
 int atb_splitk_nslices(int K);
@@ -45,7 +47,8 @@ void atb_splitk(const void* A, const void* B, float* ws, float* C, int M,
                 int N, int K, hipStream_t stream);
 void front_de(const uint8_t* ids, const void* dt1g, const void* w1, float* de,
               int B, uint32_t seed, float keep, hipStream_t stream,
-              unsigned long long* timing, uint32_t dbg);
+              unsigned long long* timing, uint32_t dbg,
+              const uint32_t* seed_ptr);
 }  // namespace rk
 
 namespace {
@@ -136,14 +139,10 @@ std::vector<torch::Tensor> gru_layer_bwd(torch::Tensor cache, torch::Tensor hseq
                 "ut must be (2,128,384)");
     auto dxg = torch::empty({T, B, 2, 384}, cache.options());
     auto dhg = torch::empty({2, T, B, 384}, cache.options());
-    auto fopt = cache.options().dtype(torch::kFloat32);
-    auto db_ih = torch::zeros({2, 384}, fopt);
-    auto dbhh = torch::zeros({2, 384}, fopt);
     rk::gru_layer_bwd(cache.data_ptr(), hseq.data_ptr(), dhin.data_ptr(),
-                      ut.data_ptr(), dxg.data_ptr(), dhg.data_ptr(),
-                      db_ih.data_ptr<float>(), dbhh.data_ptr<float>(), T, B,
+                      ut.data_ptr(), dxg.data_ptr(), dhg.data_ptr(), T, B,
                       cur_stream());
-    return {dxg, dhg, db_ih, dbhh};
+    return {dxg, dhg};
 }
 
 // fused CE: returns (loss scalar f32, dlogits (N,5) f32) for mean reduction
@@ -165,27 +164,41 @@ std::vector<torch::Tensor> ce_fwd_bwd(torch::Tensor logits, torch::Tensor target
 // fused Adam on flat fp32 buffers
 void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                torch::Tensor v, double lr, double beta1, double beta2,
-               double eps, int64_t step) {
+               double eps, int64_t step,
+               c10::optional<torch::Tensor> step_buf) {
     check(p, torch::kFloat32, "p");
     check(g, torch::kFloat32, "g");
     check(m, torch::kFloat32, "m");
     check(v, torch::kFloat32, "v");
     const int64_t n = p.numel();
     TORCH_CHECK(g.numel() == n && m.numel() == n && v.numel() == n, "size mismatch");
+    const int* sp = nullptr;
+    if (step_buf.has_value()) {
+        TORCH_CHECK(step_buf->is_cuda() &&
+                    step_buf->scalar_type() == torch::kInt32);
+        sp = step_buf->data_ptr<int>();
+    }
     rk::adam_step(p.data_ptr<float>(), g.data_ptr<float>(), m.data_ptr<float>(),
                   v.data_ptr<float>(), n, float(lr), float(beta1), float(beta2),
-                  float(eps), int(step), cur_stream());
+                  float(eps), int(step), cur_stream(), sp);
 }
 
 // multi-tensor Adam: table rows [grad_ptr, flat_offset, numel] (GPU int64)
 void adam_mt(torch::Tensor table, int64_t n_params, torch::Tensor p,
              torch::Tensor m, torch::Tensor v, double lr, double beta1,
-             double beta2, double eps, int64_t step) {
+             double beta2, double eps, int64_t step,
+             c10::optional<torch::Tensor> step_buf) {
     TORCH_CHECK(table.is_cuda() && table.scalar_type() == torch::kInt64);
+    const int* sp = nullptr;
+    if (step_buf.has_value()) {
+        TORCH_CHECK(step_buf->is_cuda() &&
+                    step_buf->scalar_type() == torch::kInt32);
+        sp = step_buf->data_ptr<int>();
+    }
     rk::adam_mt(table.data_ptr<int64_t>(), (int)n_params, p.data_ptr<float>(),
                 m.data_ptr<float>(), v.data_ptr<float>(), (float)lr,
                 (float)beta1, (float)beta2, (float)eps, (int)step,
-                cur_stream());
+                cur_stream(), sp);
 }
 
 void grad_gather(torch::Tensor table, int64_t n_params, torch::Tensor flat_g) {
@@ -237,9 +250,17 @@ std::vector<torch::Tensor> head_fwd(torch::Tensor hseq, torch::Tensor w4,
 }
 
 // fused train front fwd: ids -> (W, B, 500) bf16 GRU input sequence
+const uint32_t* seed_ptr_of(const c10::optional<torch::Tensor>& s) {
+    if (!s.has_value()) return nullptr;
+    TORCH_CHECK(s->is_cuda() && s->scalar_type() == torch::kInt32,
+                "seed_buf must be a cuda int32 tensor");
+    return reinterpret_cast<const uint32_t*>(s->data_ptr<int>());
+}
+
 torch::Tensor front_fwd(torch::Tensor ids, torch::Tensor w1, torch::Tensor b1,
                         torch::Tensor w2, torch::Tensor b2, torch::Tensor emb,
-                        int64_t seed, double keep) {
+                        int64_t seed, double keep,
+                        c10::optional<torch::Tensor> seed_buf) {
     check(ids, torch::kUInt8, "ids");
     check(w1, torch::kBFloat16, "w1");
     check(b1, torch::kFloat32, "b1");
@@ -251,7 +272,8 @@ torch::Tensor front_fwd(torch::Tensor ids, torch::Tensor w1, torch::Tensor b1,
     auto out = torch::empty({90, B, 500}, ids.options().dtype(torch::kBFloat16));
     rk::front_fwd(ids.data_ptr<uint8_t>(), w1.data_ptr(), b1.data_ptr<float>(),
                   w2.data_ptr(), b2.data_ptr<float>(), emb.data_ptr(),
-                  out.data_ptr(), B, (uint32_t)seed, (float)keep, cur_stream());
+                  out.data_ptr(), B, (uint32_t)seed, (float)keep, cur_stream(),
+                  seed_ptr_of(seed_buf));
     return out;
 }
 
@@ -260,7 +282,8 @@ std::vector<torch::Tensor> front_bwd(torch::Tensor ids, torch::Tensor dseq,
                                      torch::Tensor w1, torch::Tensor b1,
                                      torch::Tensor w2, torch::Tensor b2,
                                      torch::Tensor emb, int64_t seed,
-                                     double keep, int64_t phase_mask) {
+                                     double keep, int64_t phase_mask,
+                                     c10::optional<torch::Tensor> seed_buf) {
     check(ids, torch::kUInt8, "ids");
     check(dseq, torch::kBFloat16, "dseq");
     check(w1, torch::kBFloat16, "w1");
@@ -284,11 +307,11 @@ std::vector<torch::Tensor> front_bwd(torch::Tensor ids, torch::Tensor dseq,
                   emb.data_ptr(), dw1.data_ptr<float>(), db1.data_ptr<float>(),
                   dw2.data_ptr<float>(), db2.data_ptr<float>(),
                   dt1g.data_ptr(), B, (uint32_t)seed, (float)keep,
-                  cur_stream(), (uint32_t)phase_mask);
+                  cur_stream(), (uint32_t)phase_mask, seed_ptr_of(seed_buf));
     if (phase_mask == 0x1F)
         rk::front_de(ids.data_ptr<uint8_t>(), dt1g.data_ptr(), w1.data_ptr(),
                      de.data_ptr<float>(), B, (uint32_t)seed, (float)keep,
-                     cur_stream(), nullptr, 0);
+                     cur_stream(), nullptr, 0, seed_ptr_of(seed_buf));
     return {de, dw1, db1, dw2, db2};
 }
 
@@ -306,7 +329,7 @@ std::vector<torch::Tensor> front_de_timed(torch::Tensor ids, torch::Tensor dt1g,
                  de.data_ptr<float>(), B, (uint32_t)seed, (float)keep,
                  cur_stream(),
                  reinterpret_cast<unsigned long long*>(tim.data_ptr<int64_t>()),
-                 (uint32_t)dbg);
+                 (uint32_t)dbg, nullptr);
     return {de, tim};
 }
 
@@ -353,11 +376,20 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("bhh"), py::arg("train") = false, py::arg("dbg") = 0);
     m.def("gru_layer_bwd", &gru_layer_bwd);
     m.def("ce_fwd_bwd", &ce_fwd_bwd);
-    m.def("adam_step", &adam_step);
-    m.def("adam_mt", &adam_mt);
+    m.def("adam_step", &adam_step, py::arg("p"), py::arg("g"),
+          py::arg("m"), py::arg("v"), py::arg("lr"), py::arg("beta1"),
+          py::arg("beta2"), py::arg("eps"), py::arg("step"),
+          py::arg("step_buf") = c10::nullopt);
+    m.def("adam_mt", &adam_mt, py::arg("table"), py::arg("n_params"),
+          py::arg("p"), py::arg("m"), py::arg("v"), py::arg("lr"),
+          py::arg("beta1"), py::arg("beta2"), py::arg("eps"),
+          py::arg("step"), py::arg("step_buf") = c10::nullopt);
     m.def("grad_gather", &grad_gather);
     m.def("emb_grad", &emb_grad);
-    m.def("front_fwd", &front_fwd);
+    m.def("front_fwd", &front_fwd, py::arg("ids"), py::arg("w1"),
+          py::arg("b1"), py::arg("w2"), py::arg("b2"), py::arg("emb"),
+          py::arg("seed"), py::arg("keep"),
+          py::arg("seed_buf") = c10::nullopt);
     m.def("atb_splitk", &atb_splitk);
     m.def("gemm_bias", &gemm_bias, py::arg("A"), py::arg("B"),
           py::arg("bias") = c10::nullopt);
@@ -366,7 +398,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("front_bwd", &front_bwd, py::arg("ids"), py::arg("dseq"),
           py::arg("w1"), py::arg("b1"), py::arg("w2"), py::arg("b2"),
           py::arg("emb"), py::arg("seed"), py::arg("keep"),
-          py::arg("phase_mask") = 0x1F);
+          py::arg("phase_mask") = 0x1F, py::arg("seed_buf") = c10::nullopt);
     m.def("head_fwd", &head_fwd, py::arg("hseq"), py::arg("w4"), py::arg("b4"),
           py::arg("want_logits") = true, py::arg("want_argmax") = false);
 }

@@ -281,8 +281,6 @@ __global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
     bf16* __restrict__ dhg,          // (2, T, B, 3H) out: [dxr dxz dhgn]
                                      // dir-major so each dir slice is a
                                      // contiguous (T*B, 384) GEMM operand
-    float* __restrict__ db_ih,       // (2, 3H) pre-zeroed: column sums of dxg
-    float* __restrict__ dbhh,        // (2, 3H) pre-zeroed: column sums of dhg
     int T, int B) {
     __shared__ struct {
         bf16 cache_st[2][MB][4 * H];  // double-buffered staged cache[t]
@@ -301,15 +299,12 @@ __global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
     const int lrow = lane >> 4;
     const int lcol = lane & 15;
 
-    // B-fragments of U for dh_prev = dhg·U, persistent in registers: an
-    // in-loop global re-read forces a vmcnt(0) at its first use that drains
-    // the in-flight LDS-DMA staging every step (glds+ordinary-load caveat,
-    // cdna_hip_programming.md §5)
+    // B-fragments of U for dh_prev = dhg·U: B[k=j][col=khid] = Ut[khid][j]
     bf16x8 ufrag[12];
 #pragma unroll
     for (int kb = 0; kb < 12; ++kb) {
-        const int col = j0 + lcol;
-        const int j = kb * 32 + 8 * lrow;
+        const int col = j0 + lcol;          // this wave's hidden column
+        const int j = kb * 32 + 8 * lrow;   // gate-row index
         ufrag[kb] = *reinterpret_cast<const bf16x8*>(
             ut + ((size_t)dir * H + col) * G3 + j);
     }
@@ -319,52 +314,57 @@ __global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
     for (int mt = 0; mt < 2; ++mt)
 #pragma unroll
         for (int i = 0; i < 4; ++i) dhc[mt][i] = 0.0f;
-    // per-lane bias-grad partials for hidden column j (replaces three
-    // hipBLASLt ones-GEMVs per layer, ~22 us each)
-    float sxr = 0.f, sxz = 0.f, sxn = 0.f, shgn = 0.f;
 
     const int stp = (dir == 0) ? -1 : 1;     // BPTT walks t backwards
     const int t0 = (dir == 0) ? T - 1 : 0;
     auto t_of = [&](int sidx) { return t0 + stp * sidx; };
 
-    // staging via direct global->LDS DMA: no staging registers at all (the
-    // register pipeline + bias-sum accumulators spilled to scratch). The
-    // implicit vmcnt(0) drain at the next __syncthreads() is the completion
-    // guarantee — still more than a full HBM latency of cover.
-    auto glds_stage = [&](int buf, int sidx) {
+    // register-staged loads for one step: cache (4 chunks) + dhin + hprev
+    auto issue_loads = [&](int sidx, bf16x8 (&rc)[BWCH_C], bf16x8& rdh,
+                           bf16x8& rhp) {
         const int t = t_of(sidx);
         const int tp = (dir == 0) ? t - 1 : t + 1;
-        {
-            const bf16* src = cache + (((size_t)t * B + b0) * 2 + dir) * 4 * H;
-            const int row4 = tid / 64, col4 = (tid % 64) * 8;
+        const bf16* src = cache + (((size_t)t * B + b0) * 2 + dir) * 4 * H;
+        const int row4 = tid / 64, col4 = (tid % 64) * 8;
 #pragma unroll
-            for (int q = 0; q < BWCH_C; ++q)
-                glds_b128(src + (size_t)(row4 + q * 8) * 2 * 4 * H + col4,
-                          &lds.cache_st[buf][wid + q * 8][0]);
-        }
-        {
-            const int row = tid / 16, col = (tid % 16) * 8;
-            glds_b128(dhin + (((size_t)t * B + b0) * 2 + dir) * H +
-                          (size_t)row * 2 * H + col,
-                      &lds.dhin_st[buf][wid * 4][0]);
-            if (tp >= 0 && tp < T) {
-                glds_b128(hseq + (((size_t)tp * B + b0) * 2 + dir) * H +
-                              (size_t)row * 2 * H + col,
-                          &lds.hprev_st[buf][wid * 4][0]);
-            } else {
-                *reinterpret_cast<bf16x8*>(
-                    &lds.hprev_st[buf][row][col]) = bf16x8{};
-            }
-        }
+        for (int q = 0; q < BWCH_C; ++q)
+            rc[q] = *reinterpret_cast<const bf16x8*>(
+                src + (size_t)(row4 + q * 8) * 2 * 4 * H + col4);
+        const int row = tid / 16, col = (tid % 16) * 8;
+        rdh = *reinterpret_cast<const bf16x8*>(
+            dhin + (((size_t)t * B + b0) * 2 + dir) * H + (size_t)row * 2 * H + col);
+        if (tp >= 0 && tp < T)
+            rhp = *reinterpret_cast<const bf16x8*>(
+                hseq + (((size_t)tp * B + b0) * 2 + dir) * H + (size_t)row * 2 * H + col);
+        else
+            rhp = bf16x8{};
+    };
+    auto write_stage = [&](int buf, bf16x8 (&rc)[BWCH_C], bf16x8& rdh,
+                           bf16x8& rhp) {
+        const int row4 = tid / 64, col4 = (tid % 64) * 8;
+#pragma unroll
+        for (int q = 0; q < BWCH_C; ++q)
+            *reinterpret_cast<bf16x8*>(&lds.cache_st[buf][row4 + q * 8][col4]) = rc[q];
+        const int row = tid / 16, col = (tid % 16) * 8;
+        *reinterpret_cast<bf16x8*>(&lds.dhin_st[buf][row][col]) = rdh;
+        *reinterpret_cast<bf16x8*>(&lds.hprev_st[buf][row][col]) = rhp;
     };
 
-    glds_stage(0, 0);
+    bf16x8 rcA[BWCH_C], rcB[BWCH_C];
+    bf16x8 rdhA, rhpA, rdhB, rhpB;
+    // prologue: step 0 straight to LDS buffer 0; step 1 to registers A
+    issue_loads(0, rcA, rdhA, rhpA);
+    write_stage(0, rcA, rdhA, rhpA);
+    if (T > 1) issue_loads(1, rcA, rdhA, rhpA);
     __syncthreads();
 
-    auto body = [&](int sidx, int curp) {
+    auto body = [&](int sidx, int curp, bf16x8 (&rc_wr)[BWCH_C], bf16x8& rdh_wr,
+                    bf16x8& rhp_wr, bf16x8 (&rc_ld)[BWCH_C], bf16x8& rdh_ld,
+                    bf16x8& rhp_ld) {
         const int t = t_of(sidx);
-        // issue next step's LDS-DMA into the back buffers
-        if (sidx + 1 < T) glds_stage(curp ^ 1, sidx + 1);
+        // issue step s+2 loads FIRST, then write step s+1's staged registers
+        if (sidx + 2 < T) issue_loads(sidx + 2, rc_ld, rdh_ld, rhp_ld);
+        if (sidx + 1 < T) write_stage(curp ^ 1, rc_wr, rdh_wr, rhp_wr);
 
         // ---- gate gradients from stage[curp] ------------------------------
         float dhp_part[2][4];
@@ -390,10 +390,6 @@ __global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
                 const float dxr = dan * hgn * r * (1.0f - r);
                 const float dxz = dz * z * (1.0f - z);
                 dhp_part[mt][i] = dh * z;
-                sxr += dxr;
-                sxz += dxz;
-                sxn += dxn;
-                shgn += dhgn;
                 lds.dhg[row][0 * H + j] = f2bf(dxr);
                 lds.dhg[row][1 * H + j] = f2bf(dxz);
                 lds.dhg[row][2 * H + j] = f2bf(dhgn);
@@ -453,35 +449,23 @@ __global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
         __syncthreads();  // dhg/dg_st/stage[curp] reads done before reuse
     };
 
-    for (int sidx = 0; sidx < T; ++sidx) body(sidx, sidx & 1);
-
-    // fold the 4 lrow groups (lanes 16 apart share j), then lane-per-j
-    // commits the six bias-grad sums
-    sxr += __shfl_xor(sxr, 16); sxr += __shfl_xor(sxr, 32);
-    sxz += __shfl_xor(sxz, 16); sxz += __shfl_xor(sxz, 32);
-    sxn += __shfl_xor(sxn, 16); sxn += __shfl_xor(sxn, 32);
-    shgn += __shfl_xor(shgn, 16); shgn += __shfl_xor(shgn, 32);
-    if (lrow == 0) {
-        const int j = j0 + lcol;
-        agent_atomic_add(&db_ih[dir * G3 + 0 * H + j], sxr);
-        agent_atomic_add(&db_ih[dir * G3 + 1 * H + j], sxz);
-        agent_atomic_add(&db_ih[dir * G3 + 2 * H + j], sxn);
-        agent_atomic_add(&dbhh[dir * G3 + 0 * H + j], sxr);
-        agent_atomic_add(&dbhh[dir * G3 + 1 * H + j], sxz);
-        agent_atomic_add(&dbhh[dir * G3 + 2 * H + j], shgn);
+    int sidx = 0;
+    for (; sidx + 2 <= T; sidx += 2) {
+        body(sidx, 0, rcA, rdhA, rhpA, rcB, rdhB, rhpB);
+        body(sidx + 1, 1, rcB, rdhB, rhpB, rcA, rdhA, rhpA);
     }
+    if (sidx < T) body(sidx, 0, rcA, rdhA, rhpA, rcB, rdhB, rhpB);
 }
 
 void gru_layer_bwd(const void* cache, const void* hseq, const void* dhin,
-                   const void* ut, void* dxg, void* dhg, float* db_ih,
-                   float* dbhh, int T, int B, hipStream_t stream) {
+                   const void* ut, void* dxg, void* dhg, int T, int B,
+                   hipStream_t stream) {
     dim3 grid(B / MB, 2);
     dim3 block(BW_WAVES * 64);
     hipLaunchKernelGGL(gru_layer_bwd_kernel, grid, block, 0, stream,
                        static_cast<const bf16*>(cache), static_cast<const bf16*>(hseq),
                        static_cast<const bf16*>(dhin), static_cast<const bf16*>(ut),
-                       static_cast<bf16*>(dxg), static_cast<bf16*>(dhg), db_ih,
-                       dbhh, T, B);
+                       static_cast<bf16*>(dxg), static_cast<bf16*>(dhg), T, B);
 }
 
 }  // namespace rk

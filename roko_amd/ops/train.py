@@ -63,7 +63,7 @@ class GruLayerFn(torch.autograd.Function):
         ut = u_bf.transpose(1, 2).contiguous()  # (2, 128, 384)
         # kernel emits GEMM-ready layouts: dxg (T,B,2,384) -> (TB,768) view;
         # dhg (2,T,B,384) -> contiguous per-direction (TB,384) slices
-        dxg, dhg, db2_ih, dbhh2 = ext.gru_layer_bwd(cache, hseq, dhin, ut)
+        dxg, dhg = ext.gru_layer_bwd(cache, hseq, dhin, ut)
 
         # h_prev sequences per direction
         zeros = hseq.new_zeros(1, B, H)
@@ -83,9 +83,12 @@ class GruLayerFn(torch.autograd.Function):
         dxg_cat = dxg.reshape(TB, 2 * 3 * H)  # zero-copy (TB, 768)
         x_flat = x_bf.reshape(TB, -1)
         dw_ih = ext.atb_splitk(dxg_cat, x_flat.contiguous())  # (768, in) f32
-        # bias grads come straight from the BPTT kernel's per-lane partials
-        db_ih = db2_ih.reshape(-1)
-        dbhh = dbhh2
+        # column sums as one hipBLASLt GEMV each
+        ones = x_bf.new_ones(1, TB)
+        dbhh = torch.stack(
+            [ones.mm(dhg_f).squeeze(0), ones.mm(dhg_r).squeeze(0)]
+        ).float()
+        db_ih = ones.mm(dxg_cat).squeeze(0).float()      # (768,)
         dx = dxg_cat.mm(w_ih_bf).to(ctx.in_dtype).view(T, B, -1)
 
         return dx, dw_ih, db_ih, du, dbhh
@@ -149,6 +152,13 @@ class FusedAdam:
         for p in self.params:
             p.grad = None
 
+    def bind_flat_grads(self):
+        """Captured mode: gradients become views of the flat buffer so the
+        graph records accumulation into STABLE addresses; zeroing is one
+        fill on flat_g."""
+        for p, off in zip(self.params, self.offs):
+            p.grad = self.flat_g[off : off + p.numel()].view(p.shape)
+
     def _table(self):
         rows = []
         for p, off in zip(self.params, self.offs):
@@ -193,18 +203,24 @@ class FrontFn(torch.autograd.Function):
     no activation round-trips through HBM."""
 
     @staticmethod
-    def forward(ctx, emb_w, w1, b1, w2, b2, ids_u8, p_drop, training):
+    def forward(ctx, emb_w, w1, b1, w2, b2, ids_u8, p_drop, training,
+                seed_buf=None):
         ext = _ext()
         keep = 1.0 - (p_drop if training else 0.0)
-        seed = int(torch.randint(0, 2**31 - 1, (1,)).item())
+        # hipGraph capture: the seed lives in a device buffer advanced by a
+        # captured add, so every replay gets fresh dropout masks with no
+        # host RNG (a host randint would freeze the mask into the graph)
+        seed = 0 if seed_buf is not None else int(
+            torch.randint(0, 2**31 - 1, (1,)).item())
         w1b = w1.detach().to(torch.bfloat16).contiguous()
         b1f = b1.detach().float().contiguous()
         w2b = w2.detach().to(torch.bfloat16).contiguous()
         b2f = b2.detach().float().contiguous()
         embb = emb_w.detach().to(torch.bfloat16).contiguous()
-        out = ext.front_fwd(ids_u8, w1b, b1f, w2b, b2f, embb, seed, keep)
+        out = ext.front_fwd(ids_u8, w1b, b1f, w2b, b2f, embb, seed, keep,
+                            seed_buf)
         ctx.save_for_backward(ids_u8, w1b, b1f, w2b, b2f, embb)
-        ctx.seed, ctx.keep = seed, keep
+        ctx.seed, ctx.keep, ctx.seed_buf = seed, keep, seed_buf
         return out  # (90, B, 500) bf16
 
     @staticmethod
@@ -213,9 +229,9 @@ class FrontFn(torch.autograd.Function):
         ids_u8, w1b, b1f, w2b, b2f, embb = ctx.saved_tensors
         de, dw1, db1, dw2, db2 = ext.front_bwd(
             ids_u8, dseq.to(torch.bfloat16).contiguous(), w1b, b1f, w2b, b2f,
-            embb, ctx.seed, ctx.keep,
+            embb, ctx.seed, ctx.keep, 0x1F, ctx.seed_buf,
         )
-        return de, dw1, db1, dw2, db2, None, None, None
+        return de, dw1, db1, dw2, db2, None, None, None, None
 
 
 class EmbedGatherFn(torch.autograd.Function):
@@ -239,7 +255,7 @@ class EmbedGatherFn(torch.autograd.Function):
         return de, None
 
 
-def train_forward(model, x: torch.Tensor) -> torch.Tensor:
+def train_forward(model, x: torch.Tensor, seed_buf=None) -> torch.Tensor:
     """Differentiable training forward on GPU: bf16 torch GEMMs for the MLP
     front (keeps the reference's dropout semantics), HIP kernels for the
     embedding backward and the GRU.
@@ -252,7 +268,7 @@ def train_forward(model, x: torch.Tensor) -> torch.Tensor:
     seq = FrontFn.apply(
         model.embedding.weight, model.fc1.weight, model.fc1.bias,
         model.fc2.weight, model.fc2.bias, ids, float(model.dropout.p),
-        model.training,
+        model.training, seed_buf,
     )
     # (T, B, 500) bf16
     g = model.gru
@@ -291,3 +307,62 @@ def fused_train_step(model, x, y, opt: Optional[FusedAdam] = None,
         opt.allreduce_grads()
         opt.step()
     return loss.detach()
+
+
+class GraphedTrainStep:
+    """Whole-iteration hipGraph capture: dropout-seed bump, forward, fused
+    cross-entropy, backward, (DP all-reduce,) fused Adam — one graph replay
+    per step instead of ~60 kernel launches. Gradients are flat-buffer views
+    (stable addresses across replays); the dropout seed and Adam step live
+    in device buffers advanced by captured adds.
+
+    Falls back cleanly: construct inside try/except and use
+    fused_train_step when capture raises.
+    """
+
+    def __init__(self, model, opt: FusedAdam, batch: int, world: int = 1):
+        import torch.distributed as dist
+
+        dev = next(model.parameters()).device
+        self.x = torch.zeros((batch, C.WINDOW_ROWS, C.WINDOW_COLS),
+                             dtype=torch.uint8, device=dev)
+        self.y = torch.zeros((batch, C.WINDOW_COLS), dtype=torch.int64,
+                             device=dev)
+        self.seed_buf = torch.randint(0, 2**31 - 1, (1,), dtype=torch.int32,
+                                      device=dev)
+        self.step_buf = torch.zeros(1, dtype=torch.int32, device=dev)
+        self.opt = opt
+        self.world = world
+        opt.bind_flat_grads()
+
+        def one_step():
+            self.seed_buf.add_(747796405)  # odd constant: full-period walk
+            self.step_buf.add_(1)
+            opt.flat_g.zero_()
+            logits = train_forward(model, self.x, seed_buf=self.seed_buf)
+            loss = fused_cross_entropy(logits, self.y)
+            loss.backward()
+            if world > 1:
+                dist.all_reduce(opt.flat_g)
+                opt.flat_g /= world
+            _ext().adam_step(opt.flat_p, opt.flat_g, opt.m, opt.v, opt.lr,
+                             opt.betas[0], opt.betas[1], opt.eps, 0,
+                             self.step_buf)
+            return loss
+
+        # warmup on a side stream (torch full-network capture recipe)
+        s = torch.cuda.Stream(device=dev)
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(3):
+                one_step()
+        torch.cuda.current_stream().wait_stream(s)
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph):
+            self.loss = one_step()
+
+    def __call__(self, x, y):
+        self.x.copy_(x, non_blocking=True)
+        self.y.copy_(y, non_blocking=True)
+        self.graph.replay()
+        return self.loss
