@@ -357,9 +357,22 @@ class GraphedTrainStep:
             for _ in range(3):
                 one_step()
         torch.cuda.current_stream().wait_stream(s)
-        self.graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.graph):
-            self.loss = one_step()
+        # a GC run during capture can free non-pool GPU memory (dead graphs,
+        # pools from earlier work) -> hipFree inside capture -> abort in a
+        # destructor. Reclaim garbage now and hold GC for the capture.
+        import gc
+
+        torch.cuda.synchronize()
+        gc.collect()
+        gc_was_enabled = gc.isenabled()
+        gc.disable()
+        try:
+            self.graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(self.graph):
+                self.loss = one_step()
+        finally:
+            if gc_was_enabled:
+                gc.enable()
 
     def __call__(self, x, y):
         self.x.copy_(x, non_blocking=True)
