@@ -143,8 +143,8 @@ def bench_train(args, rank, world, device):
 def main():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
-    p.add_argument("--steps", type=int, default=50)
-    p.add_argument("--warmup", type=int, default=10)
+    p.add_argument("--steps", type=int, default=400)
+    p.add_argument("--warmup", type=int, default=50)
     p.add_argument("--batch", type=int, default=C.BATCH_SIZE)
     p.add_argument("--mode", choices=["inference", "train"], default="inference")
     p.add_argument("--depth", type=int, default=4,
