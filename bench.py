@@ -104,7 +104,10 @@ def bench_train(args, rank, world, device):
                       generator=g).to(device)
 
     stepper = None
-    if os.environ.get("ROKO_NO_GRAPH") != "1":
+    # whole-iteration hipGraph capture measured ~equal to the eager fused
+    # path at b=128 (flat-view grad accumulation trades the launch savings);
+    # opt-in until it wins
+    if os.environ.get("ROKO_GRAPH") == "1":
         try:
             stepper = GraphedTrainStep(model, opt, args.batch, world)
         except Exception as e:  # noqa: BLE001 — capture support is optional

@@ -365,5 +365,6 @@ def test_graphed_train_step_learns():
     y = torch.randint(0, 5, (32, 90), device="cuda")
     losses = [float(step(x, y)) for _ in range(80)]
     assert losses[-1] < losses[0] * 0.7, (losses[0], losses[-1])
-    # device counters advanced once per replay (+ 3 warmup + 1 capture)
-    assert int(step.step_buf.item()) == 80 + 4
+    # device counters advance once per replay + 3 warmup runs (the capture
+    # itself records without executing)
+    assert int(step.step_buf.item()) == 80 + 3
