@@ -130,7 +130,8 @@ def infer(
     rkw = RkwFile(data_path)
     my_groups = [gi for gi in range(len(rkw.groups)) if gi % world == rank]
     ds = InferenceDataset(data_path, groups=my_groups)
-    dl = DataLoader(ds, batch_size=batch_size, num_workers=workers)
+    dl = DataLoader(ds, batch_size=batch_size, num_workers=workers,
+                    pin_memory=torch.cuda.is_available())
 
     # per-contig accumulation of (positions, preds) then one vectorised vote
     per_contig_pos: Dict[str, List[np.ndarray]] = defaultdict(list)
