@@ -150,7 +150,7 @@ def main():
     p.add_argument("--warmup", type=int, default=50)
     p.add_argument("--batch", type=int, default=C.BATCH_SIZE)
     p.add_argument("--mode", choices=["inference", "train"], default="inference")
-    p.add_argument("--depth", type=int, default=16,
+    p.add_argument("--depth", type=int, default=32,
                    help="in-flight batches / HIP streams (inference mode)")
     args = p.parse_args()
 

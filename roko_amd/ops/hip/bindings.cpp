@@ -45,8 +45,8 @@ void gemm_bias(const void* A, const void* B, const float* bias, void* C,
 int atb_splitk_nslices(int K);
 void atb_splitk(const void* A, const void* B, float* ws, float* C, int M,
                 int N, int K, hipStream_t stream);
-void front_de(const uint8_t* ids, const void* dt1g, const void* w1, float* de,
-              int B, uint32_t seed, float keep, hipStream_t stream,
+void front_de(const uint8_t* ids, const void* dt1g, const void* w1t_g,
+              float* de, int B, uint32_t seed, float keep, hipStream_t stream,
               unsigned long long* timing, uint32_t dbg,
               const uint32_t* seed_ptr);
 }  // namespace rk
@@ -308,10 +308,14 @@ std::vector<torch::Tensor> front_bwd(torch::Tensor ids, torch::Tensor dseq,
                   dw2.data_ptr<float>(), db2.data_ptr<float>(),
                   dt1g.data_ptr(), B, (uint32_t)seed, (float)keep,
                   cur_stream(), (uint32_t)phase_mask, seed_ptr_of(seed_buf));
-    if (phase_mask == 0x1F)
-        rk::front_de(ids.data_ptr<uint8_t>(), dt1g.data_ptr(), w1.data_ptr(),
+    if (phase_mask == 0x1F) {
+        // zero-padded W1^T image for the de kernel's L2-read A-fragments
+        auto w1t_g = torch::zeros({208, 128}, w1.options());
+        w1t_g.slice(0, 0, 200).slice(1, 0, 100).copy_(w1.t());
+        rk::front_de(ids.data_ptr<uint8_t>(), dt1g.data_ptr(), w1t_g.data_ptr(),
                      de.data_ptr<float>(), B, (uint32_t)seed, (float)keep,
                      cur_stream(), nullptr, 0, seed_ptr_of(seed_buf));
+    }
     return {de, dw1, db1, dw2, db2};
 }
 
@@ -325,7 +329,9 @@ std::vector<torch::Tensor> front_de_timed(torch::Tensor ids, torch::Tensor dt1g,
     const int B = ids.size(0);
     auto de = torch::zeros({12, 50}, w1.options().dtype(torch::kFloat32));
     auto tim = torch::zeros({4}, w1.options().dtype(torch::kInt64));
-    rk::front_de(ids.data_ptr<uint8_t>(), dt1g.data_ptr(), w1.data_ptr(),
+    auto w1t_g = torch::zeros({208, 128}, w1.options());
+    w1t_g.slice(0, 0, 200).slice(1, 0, 100).copy_(w1.t());
+    rk::front_de(ids.data_ptr<uint8_t>(), dt1g.data_ptr(), w1t_g.data_ptr(),
                  de.data_ptr<float>(), B, (uint32_t)seed, (float)keep,
                  cur_stream(),
                  reinterpret_cast<unsigned long long*>(tim.data_ptr<int64_t>()),

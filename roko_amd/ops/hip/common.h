@@ -74,6 +74,15 @@ RK_DEV bf16x8 lds_load_a_frag_t(const bf16* tile_t, int row0, int k0, int ld) {
     return r;
 }
 
+// A-fragment loaded straight from a row-major GLOBAL tile (L2-resident
+// weights): 16 B per lane, same mapping as lds_load_a_frag.
+RK_DEV bf16x8 global_load_a_frag(const bf16* tile, int row0, int k0, int ld) {
+    const int lane = threadIdx.x & 63;
+    const int row = row0 + (lane & 15);
+    const int k = k0 + 8 * (lane >> 4);
+    return *reinterpret_cast<const bf16x8*>(tile + (size_t)row * ld + k);
+}
+
 // Counter-based dropout hash (lowbias32): deterministic per (seed, index),
 // regenerated identically in the recompute backward — no mask tensors.
 RK_DEV uint32_t hash32(uint32_t x) {

@@ -632,14 +632,17 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
 __global__ __launch_bounds__(512, 2) void front_de_kernel(
     const uint8_t* __restrict__ ids,   // (B, R, W)
     const bf16* __restrict__ dt1g,     // (B, W, MP, EP)
-    const bf16* __restrict__ w1,       // (F1, R)
+    const bf16* __restrict__ w1t_g,    // (R+8, 128) = W1^T zero-padded —
+                                       // A-fragments read straight from L2
+                                       // (53 KB shared by every CU) so LDS
+                                       // stays under 80 KB and TWO
+                                       // workgroups share each CU
     float* __restrict__ de,            // (12, E) pre-zeroed
     int B, uint32_t seed, float keep,
     unsigned long long* __restrict__ timing,  // optional (4) cycle counters
-    uint32_t dbg,
-    const uint32_t* __restrict__ seed_ptr) {  // timing-bisection: 1 skip atomics, 2 skip hash, 4 skip epi
+    uint32_t dbg,  // timing-bisection: 1 skip atomics, 2 skip hash, 4 skip epi
+    const uint32_t* __restrict__ seed_ptr) {
     __shared__ struct {
-        bf16 w1_rt[R + 8][136];   // [r][f] = W1^T, zero-padded
         bf16 dt1_ef[EP][136];     // [e][f] staged column of dt1g
         bf16 dmm_er[EP][KP_LD];   // [e][r] masked dm (de-GEMM B-operand)
         bf16 hot_t[16][KP_LD];    // [c][r] one-hot of col_ids (de-GEMM A)
@@ -658,15 +661,10 @@ __global__ __launch_bounds__(512, 2) void front_de_kernel(
     const int w_begin = blockIdx.y * (W / CSPLIT);
     const int w_end = (blockIdx.y + 1 == CSPLIT) ? W : w_begin + W / CSPLIT;
 
-    for (int e = tid; e < (R + 8) * 136; e += 512)
-        (&lds.w1_rt[0][0])[e] = f2bf(0.f);
     for (int e = tid; e < EP * 136; e += 512)
         (&lds.dt1_ef[0][0])[e] = f2bf(0.f);
     for (int e = tid; e < EP * KP_LD; e += 512)
         (&lds.dmm_er[0][0])[e] = f2bf(0.f);
-    __syncthreads();
-    for (int e = tid; e < F1 * R; e += 512)
-        lds.w1_rt[e % R][e / R] = w1[e];
     __syncthreads();
 
     unsigned long long tacc[3] = {0, 0, 0};
@@ -716,7 +714,7 @@ __global__ __launch_bounds__(512, 2) void front_de_kernel(
 #pragma unroll
                 for (int kb = 0; kb < 4; ++kb) {
                     bf16x8 a = lds_load_a_frag(&lds.dt1_ef[0][0], emt * 16, kb * 32, 136);
-                    bf16x8 bb = lds_load_b_frag_t(&lds.w1_rt[0][0], rnt * 16, kb * 32, 136);
+                    bf16x8 bb = global_load_a_frag(w1t_g, rnt * 16, kb * 32, 128);
                     acc = mfma16x16x32(a, bb, acc);
                 }
                 (void)myid;
@@ -786,15 +784,15 @@ void front_fwd(const uint8_t* ids, const void* w1, const float* b1,
                        B, seed, keep, seed_ptr);
 }
 
-void front_de(const uint8_t* ids, const void* dt1g, const void* w1, float* de,
-              int B, uint32_t seed, float keep, hipStream_t stream,
+void front_de(const uint8_t* ids, const void* dt1g, const void* w1t_g,
+              float* de, int B, uint32_t seed, float keep, hipStream_t stream,
               unsigned long long* timing, uint32_t dbg,
               const uint32_t* seed_ptr) {
     hipLaunchKernelGGL(front::front_de_kernel,
                        dim3(B, front::CSPLIT), dim3(512), 0, stream,
                        ids, static_cast<const bf16*>(dt1g),
-                       static_cast<const bf16*>(w1), de, B, seed, keep, timing,
-                       dbg, seed_ptr);
+                       static_cast<const bf16*>(w1t_g), de, B, seed, keep,
+                       timing, dbg, seed_ptr);
 }
 
 void front_bwd(const uint8_t* ids, const void* dseq, const void* w1,
