@@ -117,6 +117,13 @@ def bench_train(args, rank, world, device):
     if stepper is not None:
         def fused_train_step(model_, x_, y_, opt_):  # noqa: F811 shadow
             return stepper(x_, y_)
+    elif os.environ.get("ROKO_DUAL") != "0":
+        from roko_amd.ops.train import dual_stream_train_step
+        dual_streams = (torch.cuda.Stream(device=device),
+                        torch.cuda.Stream(device=device))
+
+        def fused_train_step(model_, x_, y_, opt_):  # noqa: F811 shadow
+            return dual_stream_train_step(model_, x_, y_, opt_, dual_streams)
 
     for _ in range(args.warmup):
         fused_train_step(model, x, y, opt)

@@ -152,6 +152,11 @@ class FusedAdam:
         for p in self.params:
             p.grad = None
 
+    def _flat_g2(self):
+        if not hasattr(self, "_g2"):
+            self._g2 = torch.empty_like(self.flat_g)
+        return self._g2
+
     def bind_flat_grads(self):
         """Captured mode: gradients become views of the flat buffer so the
         graph records accumulation into STABLE addresses; zeroing is one
@@ -379,3 +384,78 @@ class GraphedTrainStep:
         self.y.copy_(y, non_blocking=True)
         self.graph.replay()
         return self.loss
+
+
+def dual_stream_train_step(model, x, y, opt: FusedAdam, streams=None):
+    """One optimizer step over batch (x, y) as TWO half-batches on two HIP
+    streams. The GRU recurrence is latency-bound at ~3% chip occupancy
+    (8 workgroups), so one half's sequential GRU chain overlaps the other
+    half's front/GEMM work almost for free; halving the batch does not
+    change the GRU kernels' duration (fewer workgroups, same 90 steps).
+
+    Race-free gradient handling without replicas: after the first
+    backward's launches are enqueued, its grad TENSORS are harvested and
+    p.grad reset to None, so the second backward ASSIGNS fresh tensors —
+    the two halves' grads live in different buffers and are merged with two
+    gather kernels into the flat buffer ((gA + gB) / 2: each half-loss is a
+    mean over its half).
+    """
+    import torch.distributed as dist
+
+    dev = x.device
+    if streams is None:
+        streams = (torch.cuda.Stream(device=dev), torch.cuda.Stream(device=dev))
+    s0, s1 = streams
+    B = x.shape[0]
+    h = B // 2
+    cur = torch.cuda.current_stream()
+    s0.wait_stream(cur)
+    s1.wait_stream(cur)
+
+    opt.zero_grad()
+    with torch.cuda.stream(s0):
+        logits0 = train_forward(model, x[:h])
+        loss0 = fused_cross_entropy(logits0, y[:h])
+        loss0.backward()
+    grads_a = []
+    for p in opt.params:
+        grads_a.append(p.grad)
+        p.grad = None
+    with torch.cuda.stream(s1):
+        logits1 = train_forward(model, x[h:])
+        loss1 = fused_cross_entropy(logits1, y[h:])
+        loss1.backward()
+
+    cur.wait_stream(s0)
+    cur.wait_stream(s1)
+    # merge: flat_g = (gA + gB) / 2, then (all-reduce and) Adam
+    ext = _ext()
+
+    def table_of(grads):
+        rows = []
+        for g, off in zip(grads, opt.offs):
+            assert g is not None
+            if g.dtype != torch.float32 or not g.is_contiguous():
+                g = g.float().contiguous()
+            rows.append((g.data_ptr(), off, g.numel()))
+        return torch.tensor(rows, dtype=torch.int64).to(dev, non_blocking=True)
+
+    tab_a = table_of(grads_a)
+    grads_b = [p.grad for p in opt.params]
+    tab_b = table_of(grads_b)
+    for g in grads_a:
+        g.record_stream(cur)
+    for g in grads_b:
+        g.record_stream(cur)
+    ext.grad_gather(tab_a, len(opt.params), opt.flat_g)
+    # second gather adds? grad_gather overwrites — accumulate via temp
+    ext.grad_gather(tab_b, len(opt.params), opt._flat_g2())
+    opt.flat_g.add_(opt._flat_g2()).mul_(0.5)
+    if dist.is_initialized() and dist.get_world_size() > 1:
+        dist.all_reduce(opt.flat_g)
+        opt.flat_g /= dist.get_world_size()
+    opt.step_count += 1
+    ext.adam_step(opt.flat_p, opt.flat_g, opt.m, opt.v, opt.lr, opt.betas[0],
+                  opt.betas[1], opt.eps, opt.step_count)
+    # keep the harvested tensors alive until the merge kernels ran
+    return (loss0.detach() + loss1.detach()) * 0.5

@@ -368,3 +368,41 @@ def test_graphed_train_step_learns():
     # device counters advance once per replay + 3 warmup runs (the capture
     # itself records without executing)
     assert int(step.step_buf.item()) == 80 + 3
+
+
+@requires_gpu
+def test_dual_stream_step_matches_single():
+    """Two half-batches on two streams must produce the same update as one
+    full-batch step (eval-mode dropout off for exact comparison)."""
+    from roko_amd.ops.train import (FusedAdam, dual_stream_train_step,
+                                    fused_train_step)
+
+    torch.manual_seed(14)
+    m1 = RokoModel().cuda().eval()   # dropout off
+    m2 = RokoModel().cuda().eval()
+    m2.load_state_dict(m1.state_dict())
+    o1 = FusedAdam(list(m1.parameters()), lr=1e-3)
+    o2 = FusedAdam(list(m2.parameters()), lr=1e-3)
+    x = torch.randint(0, 12, (64, 200, 90), dtype=torch.uint8, device="cuda")
+    y = torch.randint(0, 5, (64, 90), device="cuda")
+
+    l1 = fused_train_step(m1, x, y, o1)
+    l2 = dual_stream_train_step(m2, x, y, o2)
+    assert abs(float(l1) - float(l2)) < 2e-3, (float(l1), float(l2))
+    for (n1, p1), (n2, p2) in zip(m1.named_parameters(), m2.named_parameters()):
+        a, b = p1.detach().reshape(-1), p2.detach().reshape(-1)
+        rel = ((a - b).norm() / (a.norm() + 1e-9)).item()
+        assert rel < 5e-3, (n1, rel)
+
+
+@requires_gpu
+def test_dual_stream_step_learns():
+    from roko_amd.ops.train import FusedAdam, dual_stream_train_step
+
+    torch.manual_seed(15)
+    m = RokoModel().cuda().train()
+    opt = FusedAdam(list(m.parameters()), lr=3e-3)
+    x = torch.randint(0, 12, (64, 200, 90), dtype=torch.uint8, device="cuda")
+    y = torch.randint(0, 5, (64, 90), device="cuda")
+    losses = [float(dual_stream_train_step(m, x, y, opt)) for _ in range(80)]
+    assert losses[-1] < losses[0] * 0.7, (losses[0], losses[-1])
