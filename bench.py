@@ -117,7 +117,10 @@ def bench_train(args, rank, world, device):
     if stepper is not None:
         def fused_train_step(model_, x_, y_, opt_):  # noqa: F811 shadow
             return stepper(x_, y_)
-    elif os.environ.get("ROKO_DUAL") != "0":
+    elif os.environ.get("ROKO_DUAL") == "1":
+        # measured SLOWER than the single-stream fused step (27.6k vs 33.2k
+        # windows/s): halving the batch doubles the launch count and the
+        # host becomes the critical path before the GRU-overlap win lands
         from roko_amd.ops.train import dual_stream_train_step
         dual_streams = (torch.cuda.Stream(device=device),
                         torch.cuda.Stream(device=device))
