@@ -104,12 +104,12 @@ def bench_train(args, rank, world, device):
                       generator=g).to(device)
 
     stepper = None
-    # whole-iteration hipGraph capture measured ~equal to the eager fused
-    # path at b=128 (flat-view grad accumulation trades the launch savings);
-    # opt-in until it wins
-    if os.environ.get("ROKO_GRAPH") == "1":
+    mode = os.environ.get("ROKO_TRAIN_STEP", "dualgraph")
+    if mode in ("dualgraph", "graph"):
         try:
-            stepper = GraphedTrainStep(model, opt, args.batch, world)
+            from roko_amd.ops.train import GraphedDualTrainStep
+            cls = GraphedDualTrainStep if mode == "dualgraph" else GraphedTrainStep
+            stepper = cls(model, opt, args.batch, world)
         except Exception as e:  # noqa: BLE001 — capture support is optional
             if rank == 0:
                 print(f"hipGraph train capture unavailable ({e!r}); "
@@ -117,7 +117,7 @@ def bench_train(args, rank, world, device):
     if stepper is not None:
         def fused_train_step(model_, x_, y_, opt_):  # noqa: F811 shadow
             return stepper(x_, y_)
-    elif os.environ.get("ROKO_DUAL") == "1":
+    elif mode == "dual":
         # measured SLOWER than the single-stream fused step (27.6k vs 33.2k
         # windows/s): halving the batch doubles the launch count and the
         # host becomes the critical path before the GRU-overlap win lands

@@ -459,3 +459,105 @@ def dual_stream_train_step(model, x, y, opt: FusedAdam, streams=None):
                   opt.betas[1], opt.eps, opt.step_count)
     # keep the harvested tensors alive until the merge kernels ran
     return (loss0.detach() + loss1.detach()) * 0.5
+
+
+class GraphedDualTrainStep:
+    """Dual-stream half-batch step captured as ONE hipGraph: the plain
+    dual-stream variant wins GRU-latency overlap but loses to doubled host
+    launch cost; capture removes the host from the loop entirely. Per-half
+    device seed buffers keep dropout fresh per replay; the per-half grad
+    tables are filled after capture (capture records, never executes)."""
+
+    def __init__(self, model, opt: FusedAdam, batch: int, world: int = 1):
+        import torch.distributed as dist
+
+        dev = next(model.parameters()).device
+        h = batch // 2
+        self.x = torch.zeros((batch, C.WINDOW_ROWS, C.WINDOW_COLS),
+                             dtype=torch.uint8, device=dev)
+        self.y = torch.zeros((batch, C.WINDOW_COLS), dtype=torch.int64,
+                             device=dev)
+        self.seed0 = torch.randint(0, 2**31 - 1, (1,), dtype=torch.int32,
+                                   device=dev)
+        self.seed1 = torch.randint(0, 2**31 - 1, (1,), dtype=torch.int32,
+                                   device=dev)
+        self.step_buf = torch.zeros(1, dtype=torch.int32, device=dev)
+        P = len(opt.params)
+        self.tab_a = torch.zeros((P, 3), dtype=torch.int64, device=dev)
+        self.tab_b = torch.zeros((P, 3), dtype=torch.int64, device=dev)
+        self.opt = opt
+        s0 = torch.cuda.Stream(device=dev)
+        s1 = torch.cuda.Stream(device=dev)
+        ext = _ext()
+
+        def host_table(grads):
+            rows = [(g.data_ptr(), off, g.numel())
+                    for g, off in zip(grads, opt.offs)]
+            return torch.tensor(rows, dtype=torch.int64).to(dev)
+
+        def one_step(static_tabs: bool):
+            self.seed0.add_(747796405)
+            self.seed1.add_(931541387)
+            self.step_buf.add_(1)
+            cur = torch.cuda.current_stream()
+            s0.wait_stream(cur)
+            s1.wait_stream(cur)
+            opt.zero_grad()
+            with torch.cuda.stream(s0):
+                logits0 = train_forward(model, self.x[:h],
+                                        seed_buf=self.seed0)
+                loss0 = fused_cross_entropy(logits0, self.y[:h])
+                loss0.backward()
+            grads_a = [p.grad for p in opt.params]
+            for p in opt.params:
+                p.grad = None
+            with torch.cuda.stream(s1):
+                logits1 = train_forward(model, self.x[h:],
+                                        seed_buf=self.seed1)
+                loss1 = fused_cross_entropy(logits1, self.y[h:])
+                loss1.backward()
+            grads_b = [p.grad for p in opt.params]
+            cur.wait_stream(s0)
+            cur.wait_stream(s1)
+            ta = self.tab_a if static_tabs else host_table(grads_a)
+            tb = self.tab_b if static_tabs else host_table(grads_b)
+            ext.grad_gather(ta, len(opt.params), opt.flat_g)
+            ext.grad_gather(tb, len(opt.params), opt._flat_g2())
+            opt.flat_g.add_(opt._flat_g2()).mul_(0.5)
+            if world > 1:
+                dist.all_reduce(opt.flat_g)
+                opt.flat_g /= world
+            ext.adam_step(opt.flat_p, opt.flat_g, opt.m, opt.v, opt.lr,
+                          opt.betas[0], opt.betas[1], opt.eps, 0,
+                          self.step_buf)
+            return (loss0 + loss1) * 0.5, grads_a, grads_b
+
+        side = torch.cuda.Stream(device=dev)
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(3):
+                one_step(static_tabs=False)
+        torch.cuda.current_stream().wait_stream(side)
+
+        import gc
+
+        torch.cuda.synchronize()
+        gc.collect()
+        gc_on = gc.isenabled()
+        gc.disable()
+        try:
+            self.graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(self.graph):
+                self.loss, self._ga, self._gb = one_step(static_tabs=True)
+        finally:
+            if gc_on:
+                gc.enable()
+        # now the grad addresses are final: fill the recorded tables once
+        self.tab_a.copy_(host_table(self._ga))
+        self.tab_b.copy_(host_table(self._gb))
+
+    def __call__(self, x, y):
+        self.x.copy_(x, non_blocking=True)
+        self.y.copy_(y, non_blocking=True)
+        self.graph.replay()
+        return self.loss
