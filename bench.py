@@ -104,7 +104,10 @@ def bench_train(args, rank, world, device):
                       generator=g).to(device)
 
     stepper = None
-    mode = os.environ.get("ROKO_TRAIN_STEP", "dualgraph")
+    # default is the proven eager fused step: the captured variants abort
+    # (uncatchable SIGABRT) on some capture-illegal operation inside the
+    # multi-stream graph — keep them strictly opt-in experiments
+    mode = os.environ.get("ROKO_TRAIN_STEP", "eager")
     if mode in ("dualgraph", "graph"):
         try:
             from roko_amd.ops.train import GraphedDualTrainStep
