@@ -1,12 +1,18 @@
-"""GPU training path: custom-autograd GRU (HIP fwd/bwd kernels), fused
-cross-entropy, fused flat-buffer Adam.
+"""GPU training path: fused front (FrontFn), custom-autograd GRU (HIP
+fwd/BPTT kernels), fused cross-entropy, fused multi-tensor Adam.
 
-Division of labour: the sequential recurrence (fwd + BPTT)
-runs in the persistent HIP kernels; the batched weight/input gradient
-reductions (dU = dhg^T·h_prev, dW_ih = dxg^T·x, dx = dxg·W_ih) are plain
-GEMMs on hipBLASLt via torch.matmul, exactly where the brief routes plain
-GEMMs. The embedding/MLP front (tiny: <3% of step FLOPs) stays on torch
-autograd ops so dropout semantics match the reference's training graph.
+Division of labour: the embedding/MLP front and the
+sequential GRU recurrence run in hand-written HIP kernels; the weight-grad
+reductions go to the split-K AtB kernel (dU, dW_ih) or hipBLASLt (xg, dx —
+shapes where its tiles win). Dropout everywhere is counter-hash based so
+the recompute backward regenerates masks without tensors.
+
+Experimental steppers kept for the record: GraphedTrainStep (whole-step
+hipGraph — measured parity with eager), dual_stream_train_step (half-batch
+streams — host-launch-bound), GraphedDualTrainStep (their combination —
+aborts in capture: the per-parameter AccumulateGrad node is SHARED by both
+halves' graphs and its cross-stream gradient hand-off is capture-illegal).
+The production step is fused_train_step.
 """
 
 from __future__ import annotations
@@ -462,11 +468,14 @@ def dual_stream_train_step(model, x, y, opt: FusedAdam, streams=None):
 
 
 class GraphedDualTrainStep:
-    """Dual-stream half-batch step captured as ONE hipGraph: the plain
-    dual-stream variant wins GRU-latency overlap but loses to doubled host
-    launch cost; capture removes the host from the loop entirely. Per-half
-    device seed buffers keep dropout fresh per replay; the per-half grad
-    tables are filled after capture (capture records, never executes)."""
+    """EXPERIMENTAL — currently aborts during capture: both halves share
+    each parameter's AccumulateGrad node, whose cross-stream gradient
+    hand-off is illegal inside stream capture (torch warns, HIP aborts).
+    Kept as the documented starting point for a replica-leaf variant.
+
+    Design: dual-stream half-batch step captured as ONE hipGraph — the
+    eager dual-stream variant wins GRU-latency overlap but loses to doubled
+    host launch cost; capture would remove the host from the loop."""
 
     def __init__(self, model, opt: FusedAdam, batch: int, world: int = 1):
         import torch.distributed as dist
