@@ -79,3 +79,58 @@ def test_ddp_grads_match_single_process(tmp_path):
         # mean of per-rank CE losses == CE of the global batch here because
         # both ranks hold equal-sized batches and CE averages over elements
         assert torch.allclose(ddp_grads[n], p.grad, atol=1e-6), n
+
+
+def _infer_worker(rank, world, port, data_path, ckpt_path, out_dir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    from roko_amd.inference import infer
+
+    out = infer(data_path, ckpt_path, None, batch_size=8,
+                device=torch.device("cpu"), log=lambda *a, **k: None)
+    if rank == 0:
+        with open(os.path.join(out_dir, "sharded.pkl"), "wb") as f:
+            pickle.dump(out, f)
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_sharded_inference_matches_single(tmp_path, tiny_assembly):
+    """Window groups sharded over 2 ranks + merged vote tables must produce
+    EXACTLY the single-process polished sequences (votes are associative —
+    SURVEY.md test strategy (e))."""
+    from roko_amd.config import FeatureConfig
+    from roko_amd.features import run as features_run
+    from roko_amd.inference import infer
+
+    data = str(tmp_path / "d.rkw")
+    features_run(tiny_assembly["draft_fasta"], tiny_assembly["reads_bam"],
+                 data, workers=1, cfg=FeatureConfig(seed=0),
+                 log=lambda *a, **k: None)
+    ckpt = str(tmp_path / "m.pth")
+    torch.manual_seed(0)
+    torch.save(RokoModel().state_dict(), ckpt)
+
+    single = infer(data, ckpt, None, batch_size=8,
+                   device=torch.device("cpu"), log=lambda *a, **k: None)
+
+    ctx = mp.get_context("spawn")
+    procs = [
+        ctx.Process(target=_infer_worker,
+                    args=(r, 2, 29533, data, ckpt, str(tmp_path)))
+        for r in range(2)
+    ]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(240)
+        assert p.exitcode == 0
+    with open(str(tmp_path / "sharded.pkl"), "rb") as f:
+        sharded = pickle.load(f)
+    assert sharded.keys() == single.keys()
+    for name in single:
+        assert sharded[name] == single[name], f"contig {name} differs"
