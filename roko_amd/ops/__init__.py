@@ -12,7 +12,9 @@ from __future__ import annotations
 import importlib
 import os
 
-import torch
+# NOTE: torch is imported lazily inside the functions — feature-generation
+# worker processes import roko_amd.ops._pileup through this package and must
+# not pay the ~2 s torch import (measured as half the per-worker wall).
 
 _ext = None
 _tried = False
@@ -31,10 +33,14 @@ def _load():
 
 
 def available() -> bool:
+    import torch
+
     return torch.cuda.is_available() and _load() is not None
 
 
 def require():
+    import torch
+
     if not torch.cuda.is_available():
         raise RuntimeError("roko_amd.ops requires a ROCm GPU")
     if _load() is None:
@@ -50,10 +56,12 @@ def ext():
     return _ext
 
 
-def model_forward(model, x: torch.Tensor) -> torch.Tensor:
+def model_forward(model, x):
     """Full-model forward through the HIP kernels. Eval mode runs the fully
     fused inference path; train mode runs the differentiable path (custom
-    GRU autograd + torch MLP front)."""
+    GRU autograd + fused front)."""
+    import torch
+
     if model.training or torch.is_grad_enabled():
         from .train import train_forward
 

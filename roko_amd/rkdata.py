@@ -79,7 +79,7 @@ class RkwWriter:
             arr = arr.astype(arr.dtype.newbyteorder("<"))
         data = np.ascontiguousarray(arr)
         ref = BlobRef(self._off, _dtype_str(data), data.shape)
-        self._fh.write(data.tobytes())
+        self._fh.write(memoryview(data).cast("B"))  # zero-copy
         self._off += data.nbytes
         return ref
 
