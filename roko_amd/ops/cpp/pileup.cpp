@@ -49,22 +49,19 @@ inline uint8_t base_from_code4(uint8_t c) {
     }
 }
 
+// one flat, append-only event stream instead of 300k per-column vectors
+// (the vector-per-column version spent its time in allocator churn):
+// events are counting-sorted by (column, slot) key in one O(N) pass
 struct Event {
+    uint32_t key;   // (pos - start) * (max_ins + 1) + slot
     uint32_t read;
-    uint8_t slot;  // 0 = aligned/deleted base, 1..max_ins = insertion slots
-    uint8_t base;  // 0..5
+    uint8_t base;   // 0..5
 };
 
 struct ReadMeta {
     int64_t ref_start;
     int64_t ref_end;  // exclusive
     uint8_t offset;   // 0 forward, 6 reverse — added to every emitted base
-};
-
-struct Column {
-    std::vector<Event> events;
-    uint8_t max_slot = 0;
-    bool covered = false;  // has at least one slot-0 event
 };
 
 // splitmix64: deterministic per-region stream derivation
@@ -88,7 +85,9 @@ FeatureResult extract_features(const std::string& bam_path, const std::string& c
     FeatureResult out;
     if (start >= end) return out;
 
-    std::vector<Column> columns(static_cast<size_t>(end - start));
+    const uint32_t SLOTS = uint32_t(P.max_ins) + 1;
+    std::vector<Event> events;
+    events.reserve(1 << 20);
     std::vector<ReadMeta> reads;
 
     // ---- phase 1+2: fetch + CIGAR walk into the column store ----------------
@@ -111,25 +110,23 @@ FeatureResult extract_features(const std::string& bam_path, const std::string& c
                 case CIG_M:
                 case CIG_EQ:
                 case CIG_X: {
-                    for (int64_t k = 0; k < len; ++k) {
-                        int64_t p = rpos + k;
-                        if (p < start || p >= end) continue;
-                        Column& col = columns[size_t(p - start)];
-                        col.events.push_back({rid, 0, base_from_code4(rec.seqi(qpos + k))});
-                        col.covered = true;
+                    int64_t k0 = std::max<int64_t>(0, start - rpos);
+                    int64_t k1 = std::min<int64_t>(len, end - rpos);
+                    for (int64_t k = k0; k < k1; ++k) {
+                        events.push_back(
+                            {uint32_t(rpos + k - start) * SLOTS, rid,
+                             base_from_code4(rec.seqi(qpos + k))});
                     }
                     // insertion following the last aligned base of this chunk
                     if (ci + 1 < nops && (rec.cigar[ci + 1] & 0xf) == CIG_I) {
                         int64_t p = rpos + len - 1;
                         int64_t ins_len = rec.cigar[ci + 1] >> 4;
                         if (p >= start && p < end) {
-                            Column& col = columns[size_t(p - start)];
                             int n = int(std::min<int64_t>(ins_len, P.max_ins));
                             for (int i = 1; i <= n; ++i) {
-                                col.events.push_back(
-                                    {rid, uint8_t(i),
+                                events.push_back(
+                                    {uint32_t(p - start) * SLOTS + uint32_t(i), rid,
                                      base_from_code4(rec.seqi(qpos + len - 1 + i))});
-                                if (uint8_t(i) > col.max_slot) col.max_slot = uint8_t(i);
                             }
                         }
                     }
@@ -145,13 +142,11 @@ FeatureResult extract_features(const std::string& bam_path, const std::string& c
                     qpos += len;
                     break;
                 case CIG_D: {
-                    for (int64_t k = 0; k < len; ++k) {
-                        int64_t p = rpos + k;
-                        if (p < start || p >= end) continue;
-                        Column& col = columns[size_t(p - start)];
-                        col.events.push_back({rid, 0, B_GAP});
-                        col.covered = true;
-                    }
+                    int64_t k0 = std::max<int64_t>(0, start - rpos);
+                    int64_t k1 = std::min<int64_t>(len, end - rpos);
+                    for (int64_t k = k0; k < k1; ++k)
+                        events.push_back(
+                            {uint32_t(rpos + k - start) * SLOTS, rid, B_GAP});
                     rpos += len;
                     break;
                 }
@@ -167,13 +162,27 @@ FeatureResult extract_features(const std::string& bam_path, const std::string& c
         }
     });
 
-    if (reads.empty()) return out;
+    if (reads.empty() || events.empty()) return out;
 
-    // ---- phase 3: sweep columns into the sliding window queue --------------
+    // ---- counting sort by (column, slot) key: one stable O(N) scatter -----
+    const size_t nkeys = size_t(end - start) * SLOTS;
+    std::vector<uint32_t> cnt(nkeys + 1, 0);
+    for (const Event& e : events) cnt[size_t(e.key) + 1]++;
+    for (size_t i = 0; i < nkeys; ++i) cnt[i + 1] += cnt[i];
+    std::vector<Event> sorted(events.size());
+    {
+        std::vector<uint32_t> cur(cnt.begin(), cnt.end() - 1);
+        for (const Event& e : events) sorted[cur[size_t(e.key)]++] = e;
+    }
+    events.clear();
+    events.shrink_to_fit();
+
+    // ---- sweep columns into the sliding window queue -----------------------
     struct Key {
         int32_t pos;
         int32_t ins;
-        std::vector<std::pair<uint32_t, uint8_t>> entries;  // (read, base 0..5)
+        uint32_t beg;   // span into `sorted`
+        uint32_t end_;
     };
     std::deque<Key> queue;
 
@@ -191,12 +200,13 @@ FeatureResult extract_features(const std::string& bam_path, const std::string& c
         while (int64_t(queue.size()) >= W) {
             // -- collect reads visible in these W columns
             valid.clear();
-            for (int s = 0; s < W; ++s) {
-                const Key& k = queue[size_t(s)];
-                for (const auto& e : k.entries) {
-                    if (e.second != B_UNKNOWN && valid_mark[e.first] != window_id) {
-                        valid_mark[e.first] = window_id;
-                        valid.push_back(e.first);
+            for (int sc = 0; sc < W; ++sc) {
+                const Key& k = queue[size_t(sc)];
+                for (uint32_t i = k.beg; i < k.end_; ++i) {
+                    const Event& e = sorted[i];
+                    if (e.base != B_UNKNOWN && valid_mark[e.read] != window_id) {
+                        valid_mark[e.read] = window_id;
+                        valid.push_back(e.read);
                     }
                 }
             }
@@ -213,28 +223,29 @@ FeatureResult extract_features(const std::string& bam_path, const std::string& c
                 for (int v = 0; v < V; ++v) {
                     const ReadMeta& m = reads[valid[size_t(v)]];
                     if (m.ref_start > wpose || m.ref_end < wposs) continue;
-                    // find column range with ref_start <= pos <= ref_end
-                    // (ref_end exclusive-end counted inside — see header note)
+                    // column range with ref_start <= pos <= ref_end
+                    // (ref_end exclusive-end counted inside — header note)
                     int lo = 0, hi = W - 1;
                     while (lo < W && queue[size_t(lo)].pos < m.ref_start) ++lo;
                     while (hi >= 0 && queue[size_t(hi)].pos > m.ref_end) --hi;
                     if (lo <= hi)
                         std::memset(&dense[size_t(v) * W + lo], B_GAP, size_t(hi - lo + 1));
                 }
-                for (int s = 0; s < W; ++s) {
-                    const Key& k = queue[size_t(s)];
-                    for (const auto& e : k.entries) {
-                        int v = row_of[e.first];
-                        if (v >= 0) dense[size_t(v) * W + s] = e.second;
+                for (int sc = 0; sc < W; ++sc) {
+                    const Key& k = queue[size_t(sc)];
+                    for (uint32_t i = k.beg; i < k.end_; ++i) {
+                        const Event& e = sorted[i];
+                        int v = row_of[e.read];
+                        if (v >= 0) dense[size_t(v) * W + sc] = e.base;
                     }
                 }
 
                 // -- emit: positions + R sampled rows
                 size_t pbase = out.positions.size();
                 out.positions.resize(pbase + size_t(W) * 2);
-                for (int s = 0; s < W; ++s) {
-                    out.positions[pbase + 2 * size_t(s)] = queue[size_t(s)].pos;
-                    out.positions[pbase + 2 * size_t(s) + 1] = queue[size_t(s)].ins;
+                for (int sc = 0; sc < W; ++sc) {
+                    out.positions[pbase + 2 * size_t(sc)] = queue[size_t(sc)].pos;
+                    out.positions[pbase + 2 * size_t(sc) + 1] = queue[size_t(sc)].ins;
                 }
                 size_t xbase = out.matrices.size();
                 out.matrices.resize(xbase + size_t(R) * W);
@@ -244,7 +255,7 @@ FeatureResult extract_features(const std::string& bam_path, const std::string& c
                     const uint8_t off = reads[rid].offset;
                     const uint8_t* src = &dense[size_t(row_of[rid]) * W];
                     uint8_t* dst = &out.matrices[xbase + size_t(r) * W];
-                    for (int s = 0; s < W; ++s) dst[s] = uint8_t(src[s] + off);
+                    for (int sc = 0; sc < W; ++sc) dst[sc] = uint8_t(src[sc] + off);
                 }
                 out.n_windows++;
                 for (int v = 0; v < V; ++v) row_of[valid[size_t(v)]] = -1;
@@ -255,16 +266,17 @@ FeatureResult extract_features(const std::string& bam_path, const std::string& c
     };
 
     for (int64_t p = start; p < end; ++p) {
-        Column& col = columns[size_t(p - start)];
-        if (!col.covered) continue;
-        // bucket events by slot into queue keys (slot order 0,1,..,max_slot)
-        size_t kbase = queue.size();
-        for (int slotn = 0; slotn <= col.max_slot; ++slotn)
-            queue.push_back({int32_t(p), int32_t(slotn), {}});
-        for (const Event& e : col.events)
-            queue[kbase + e.slot].entries.emplace_back(e.read, e.base);
-        col.events.clear();
-        col.events.shrink_to_fit();
+        const size_t kb = size_t(p - start) * SLOTS;
+        if (cnt[kb + 1] == cnt[kb]) continue;  // no slot-0 event: not covered
+        int max_slot = 0;
+        for (int sl = int(SLOTS) - 1; sl >= 1; --sl)
+            if (cnt[kb + size_t(sl) + 1] > cnt[kb + size_t(sl)]) {
+                max_slot = sl;
+                break;
+            }
+        for (int sl = 0; sl <= max_slot; ++sl)
+            queue.push_back({int32_t(p), int32_t(sl), cnt[kb + size_t(sl)],
+                             cnt[kb + size_t(sl) + 1]});
         emit_windows();
     }
     // Columns left in the queue (< W of them after the final emit) are
