@@ -21,6 +21,10 @@ import os
 import sys
 import time
 
+# must precede HIP runtime init: 4 default HW queues serialize the pipelined
+# serving streams (profiles/PERF_HISTORY.md — measured +50% at 16 queues)
+os.environ.setdefault("GPU_MAX_HW_QUEUES", "20")
+
 import torch
 
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
@@ -163,8 +167,10 @@ def main():
     p.add_argument("--warmup", type=int, default=50)
     p.add_argument("--batch", type=int, default=C.BATCH_SIZE)
     p.add_argument("--mode", choices=["inference", "train"], default="inference")
-    p.add_argument("--depth", type=int, default=32,
-                   help="in-flight batches / HIP streams (inference mode)")
+    p.add_argument("--depth", type=int, default=48,
+                   help="in-flight batches / HIP streams (inference mode); kept "
+                        "below the default warmup so every hipGraph capture "
+                        "happens untimed")
     args = p.parse_args()
 
     rank, local_rank, world = init_distributed()

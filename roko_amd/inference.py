@@ -150,7 +150,7 @@ def infer(
     if device.type == "cuda":
         from .ops.forward import InferencePipeline
 
-        pipe = InferencePipeline(model, batch_size, depth=32)
+        pipe = InferencePipeline(model, batch_size, depth=48)
     with torch.no_grad():
         if pipe is not None:
             # pipelined: keep `depth` batches in flight; votes are harvested
