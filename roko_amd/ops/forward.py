@@ -181,7 +181,7 @@ class _Ticket:
 
     def materialize(self):
         if self.data is None:
-            self.slot.event.synchronize()
+            self.slot.wait_done()
             self.data = self.slot.host_out[: self.n].clone()
 
     def __call__(self) -> torch.Tensor:
@@ -190,19 +190,44 @@ class _Ticket:
 
 
 class _Slot:
+    """One pipeline slot. Two serving backends:
+
+    * C++ fast path (default): an `ext.ServeSlot` owns the slot stream and
+      enqueues the whole forward (input copy, front kernel, 3x xg GEMM + GRU,
+      fused head+argmax, pinned D2H) in ONE pybind call. The server is
+      host-bound — a torch hipGraph replay costs ~51 us + ~15 us/node of
+      host time on ROCm and threads don't scale (runtime-global enqueue
+      lock) — so per-batch Python work IS the throughput limit
+      (profiles/PERF_HISTORY.md).
+    * hipGraph path (ROKO_SERVE=graph): each slot's forward captured once
+      and replayed per batch; kept for A/B and as the reference backend.
+    """
+
     def __init__(self, pipe: "InferencePipeline", batch: int, dev):
         self.pipe = pipe
         self.pending = None
+        self.host_out = torch.empty(
+            (batch, C.WINDOW_COLS), dtype=torch.uint8, pin_memory=True
+        )
+        self.cpp = None
+        self.graph = None
+        ext = _ext()
+        if pipe.use_graphs and os.environ.get("ROKO_SERVE") != "graph" \
+                and hasattr(ext, "ServeSlot"):
+            self.cpp = ext.ServeSlot(self.pipe.w, batch, self.host_out)
+            # one warm-up pass primes the hipBLASLt workspace/algo cache for
+            # the slot stream so submit() never allocates
+            warm = torch.zeros((batch, C.WINDOW_ROWS, C.WINDOW_COLS),
+                               dtype=torch.uint8, device=dev)
+            self.cpp.run(warm, batch)
+            self.cpp.sync()
+            return
         self.stream = torch.cuda.Stream(device=dev)
         self.x = torch.zeros(
             (batch, C.WINDOW_ROWS, C.WINDOW_COLS), dtype=torch.uint8,
             device=dev
         )
-        self.host_out = torch.empty(
-            (batch, C.WINDOW_COLS), dtype=torch.uint8, pin_memory=True
-        )
         self.event = torch.cuda.Event()
-        self.graph = None
         with torch.cuda.stream(self.stream):
             self.amax = pipe._forward_amax(self.x)  # warm-up + output buffer
             self.host_out.copy_(self.amax, non_blocking=True)
@@ -216,6 +241,12 @@ class _Slot:
                 self.host_out.copy_(self.amax, non_blocking=True)
             self.graph = g
 
+    def wait_done(self):
+        if self.cpp is not None:
+            self.cpp.sync()
+        else:
+            self.event.synchronize()
+
     def submit(self, x: torch.Tensor, copy_out: bool):
         n = x.shape[0]
         # preserve the previous tenant's predictions before the slot buffers
@@ -223,13 +254,23 @@ class _Slot:
         if self.pending is not None:
             self.pending.materialize()
             self.pending = None
+        if x.dtype != torch.uint8:
+            x = x.to(torch.uint8)
+        if self.cpp is not None:
+            # ServeSlot.run orders the slot stream behind the current
+            # (producer) stream itself; backpressure comes from the slot
+            # reuse sync in materialize()/wait_done of the previous ticket
+            self.wait_done()
+            self.cpp.run(x.contiguous(), n)
+            t = _Ticket(self, n)
+            if copy_out:
+                self.pending = t
+            return t
         self.event.synchronize()  # previous tenant fully drained
         # the input may have been produced on another stream (usually the
         # default one): order the slot stream behind it before copying
         self.stream.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(self.stream):
-            if x.dtype != torch.uint8:
-                x = x.to(torch.uint8)
             self.x[:n].copy_(x, non_blocking=True)
             if self.graph is not None:
                 self.graph.replay()  # forward + pinned D2H

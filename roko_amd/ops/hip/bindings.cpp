@@ -1,6 +1,8 @@
 // Torch bindings for the gfx950 kernels (roko_amd.ops._hip_ops).
 
 #include <ATen/cuda/CUDAContext.h>
+#include <c10/cuda/CUDAGuard.h>
+#include <c10/cuda/CUDAStream.h>
 #include <hip/hip_runtime.h>
 #include <torch/extension.h>
 
@@ -372,6 +374,111 @@ torch::Tensor atb_splitk(torch::Tensor A, torch::Tensor B) {
     return Cout;
 }
 
+// ---------------------------------------------------------------------------
+// Serving fast path: ONE C++ call enqueues the whole b-batch inference
+// forward on a slot-owned HIP stream.
+//
+// Why this exists (profiles/PERF_HISTORY.md): the pipelined server is
+// HOST-bound, not GPU-bound. A torch hipGraph replay costs ~51 us host time
+// plus ~15 us per node (ROCm re-enqueues every node at launch), so the
+// 11-node forward graph costs ~200 us of host per batch, and extra Python
+// threads do not help (a runtime-global lock serializes enqueue). Replacing
+// the per-batch Python/graph work with one pybind call of direct kernel
+// enqueues cuts the host cost to the raw HIP launches.
+struct ServeSlot {
+    // weights (kernel-ready layouts from roko_amd.ops.forward._bf16_weights)
+    torch::Tensor w1, b1, w2, b2, emb, w4, b4;
+    std::vector<torch::Tensor> w_ih_t, b_ih, u, bhh;
+    // slot state + workspaces
+    torch::Tensor x_buf;     // (B, 200, 90) u8 static input
+    torch::Tensor seq;       // (90, B, 500) bf16 front output
+    torch::Tensor xg;        // (90*B, 768) bf16 per-layer gate inputs
+    torch::Tensor hseq;      // (90, B, 2, 128) bf16 per-layer output
+    torch::Tensor amax;      // (B, 90) u8 fused-argmax predictions
+    torch::Tensor host_out;  // (B, 90) u8 pinned
+    torch::Tensor seq2d, hseq2d, xg2d;  // cached views for addmm_out
+    int B;
+    at::cuda::CUDAStream stream;
+    hipEvent_t ev_in = nullptr, ev_done = nullptr;
+
+    ServeSlot(py::dict w, int B_, torch::Tensor host_out_)
+        : B(B_), stream(at::cuda::getStreamFromPool(/*high_priority=*/false)) {
+        namespace t = torch;
+        auto need = [&](const char* k) {
+            torch::Tensor v = w[k].cast<torch::Tensor>();
+            TORCH_CHECK(v.is_cuda() && v.is_contiguous(), "weight ", k);
+            return v;
+        };
+        w1 = need("w1"); b1 = need("b1"); w2 = need("w2"); b2 = need("b2");
+        emb = need("emb"); w4 = need("w4"); b4 = need("b4");
+        for (int l = 0; l < 3; ++l) {
+            auto sfx = std::to_string(l);
+            w_ih_t.push_back(need(("w_ih_t" + sfx).c_str()));
+            b_ih.push_back(need(("b_ih" + sfx).c_str()));
+            u.push_back(need(("u" + sfx).c_str()));
+            bhh.push_back(need(("bhh" + sfx).c_str()));
+        }
+        TORCH_CHECK(B % 32 == 0, "serving batch must be a multiple of 32");
+        TORCH_CHECK(host_out_.is_pinned() && host_out_.scalar_type() == t::kUInt8
+                        && host_out_.size(0) == B && host_out_.size(1) == 90,
+                    "host_out must be pinned (B,90) u8");
+        host_out = host_out_;
+        auto dev = w1.options();
+        x_buf = t::zeros({B, 200, 90}, dev.dtype(t::kUInt8));
+        seq = t::empty({90, B, 500}, dev.dtype(t::kBFloat16));
+        xg = t::empty({(int64_t)90 * B, 768}, dev.dtype(t::kBFloat16));
+        hseq = t::empty({90, B, 2, 128}, dev.dtype(t::kBFloat16));
+        amax = t::empty({B, 90}, dev.dtype(t::kUInt8));
+        seq2d = seq.view({(int64_t)90 * B, 500});
+        hseq2d = hseq.view({(int64_t)90 * B, 256});
+        xg2d = xg;
+        (void)hipEventCreateWithFlags(&ev_in, hipEventDisableTiming);
+        (void)hipEventCreateWithFlags(&ev_done, hipEventDisableTiming);
+    }
+    ~ServeSlot() {
+        if (ev_in) (void)hipEventDestroy(ev_in);
+        if (ev_done) (void)hipEventDestroy(ev_done);
+    }
+
+    void run(torch::Tensor x, int64_t n) {
+        TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kUInt8 &&
+                        x.is_contiguous() && n <= B,
+                    "x must be contiguous cuda u8, n <= slot batch");
+        // order the slot stream behind the producer of x, then enqueue the
+        // whole forward on the slot stream
+        auto prod = at::cuda::getCurrentCUDAStream();
+        (void)hipEventRecord(ev_in, prod.stream());
+        (void)hipStreamWaitEvent(stream.stream(), ev_in, 0);
+        {
+            at::cuda::CUDAStreamGuard guard(stream);
+            hipStream_t s = stream.stream();
+            if (n > 0) x_buf.narrow(0, 0, n).copy_(x.narrow(0, 0, n), true);
+            rk::embed_mlp_fwd(x_buf.data_ptr<uint8_t>(), w1.data_ptr(),
+                              b1.data_ptr<float>(), w2.data_ptr(),
+                              b2.data_ptr<float>(), emb.data_ptr(),
+                              seq.data_ptr(), B, s);
+            for (int l = 0; l < 3; ++l) {
+                at::addmm_out(xg2d, b_ih[l], l == 0 ? seq2d : hseq2d,
+                              w_ih_t[l]);
+                rk::gru_layer_fwd(xg.data_ptr(), u[l].data_ptr(),
+                                  bhh[l].data_ptr<float>(), hseq.data_ptr(),
+                                  nullptr, 90, B, s, 0);
+            }
+            rk::head_fwd(hseq.data_ptr(), w4.data_ptr(), b4.data_ptr<float>(),
+                         nullptr, amax.data_ptr<uint8_t>(), 90, B, s);
+            host_out.copy_(amax, true);
+        }
+        (void)hipEventRecord(ev_done, stream.stream());
+    }
+
+    void sync() {
+        hipError_t e = hipEventSynchronize(ev_done);
+        TORCH_CHECK(e == hipSuccess, "serve slot sync failed: ",
+                    hipGetErrorString(e));
+    }
+    bool done() { return hipEventQuery(ev_done) == hipSuccess; }
+};
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -407,4 +514,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("phase_mask") = 0x1F, py::arg("seed_buf") = c10::nullopt);
     m.def("head_fwd", &head_fwd, py::arg("hseq"), py::arg("w4"), py::arg("b4"),
           py::arg("want_logits") = true, py::arg("want_argmax") = false);
+    py::class_<ServeSlot>(m, "ServeSlot")
+        .def(py::init<py::dict, int, torch::Tensor>(), py::arg("weights"),
+             py::arg("batch"), py::arg("host_out"))
+        .def("run", &ServeSlot::run, py::arg("x"), py::arg("n"))
+        .def("sync", &ServeSlot::sync,
+             py::call_guard<py::gil_scoped_release>())
+        .def("done", &ServeSlot::done)
+        .def_readonly("host_out", &ServeSlot::host_out);
 }
