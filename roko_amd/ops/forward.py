@@ -22,15 +22,18 @@ from .. import config as C
 #: less chip-work than the shared train-front kernel; under a deep serving
 #: pipeline the front is throughput-bound, so the one-hot kernel is the
 #: default. ROKO_FRONT=shared switches back for A/B runs.
-_SHARED_FRONT = os.environ.get("ROKO_FRONT") == "shared"
+_FRONT = os.environ.get("ROKO_FRONT", "v2")
 
 
 def _front_eval(ext, ids, w):
-    if _SHARED_FRONT:
+    if _FRONT == "shared":
         return ext.front_fwd(ids, w["w1"], w["b1"], w["w2"], w["b2"],
                              w["emb"], 0, 1.0)
-    return ext.embed_mlp_fwd(ids, w["w1"], w["b1"], w["w2"], w["b2"],
-                             w["emb"])
+    if _FRONT == "v1" or not hasattr(ext, "embed_mlp_fwd2"):
+        return ext.embed_mlp_fwd(ids, w["w1"], w["b1"], w["w2"], w["b2"],
+                                 w["emb"])
+    return ext.embed_mlp_fwd2(ids, w["w1g"], w["b1"], w["w2"], w["b2"],
+                              w["emb"])
 
 
 def _ext():
@@ -57,6 +60,11 @@ def _bf16_weights(model) -> dict:
     g = model.gru
     c = {"ver": ver}
     c["w1"] = model.fc1.weight.detach().to(torch.bfloat16).contiguous()
+    # (112, 232) zero-padded W1 for the chunked v2 front (L2-resident A tile)
+    w1g = torch.zeros(112, 232, dtype=torch.bfloat16,
+                      device=model.fc1.weight.device)
+    w1g[:100, :200] = c["w1"]
+    c["w1g"] = w1g
     c["b1"] = model.fc1.bias.detach().float().contiguous()
     c["w2"] = model.fc2.weight.detach().to(torch.bfloat16).contiguous()
     c["b2"] = model.fc2.bias.detach().float().contiguous()

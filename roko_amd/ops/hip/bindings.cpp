@@ -12,7 +12,11 @@ namespace rk {
 void mfma_probe(const float* a, const float* b, float* d, hipStream_t stream);
 void embed_mlp_fwd(const uint8_t* ids, const void* w1, const float* b1,
                    const void* w2, const float* b2, const void* emb, void* out,
-                   int B, hipStream_t stream);
+                   int B, hipStream_t stream, uint32_t dbg,
+                   unsigned long long* timing);
+void embed_mlp_fwd2(const uint8_t* ids, const void* w1g, const float* b1,
+                    const void* w2, const float* b2, const void* emb,
+                    void* out, int B, hipStream_t stream);
 void gru_layer_fwd(const void* xg, const void* u, const float* bhh, void* hseq,
                    void* cache, int T, int B, hipStream_t stream, uint32_t dbg);
 void gru_layer_bwd(const void* cache, const void* hseq, const void* dhin,
@@ -78,7 +82,9 @@ torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor b) {
 
 // ids (B, 200, 90) u8 -> (90, B, 500) bf16
 torch::Tensor embed_mlp_fwd(torch::Tensor ids, torch::Tensor w1, torch::Tensor b1,
-                            torch::Tensor w2, torch::Tensor b2, torch::Tensor emb) {
+                            torch::Tensor w2, torch::Tensor b2, torch::Tensor emb,
+                            int64_t dbg = 0,
+                            c10::optional<torch::Tensor> timing = c10::nullopt) {
     check(ids, torch::kUInt8, "ids");
     check(w1, torch::kBFloat16, "w1");
     check(b1, torch::kFloat32, "b1");
@@ -93,7 +99,32 @@ torch::Tensor embed_mlp_fwd(torch::Tensor ids, torch::Tensor w1, torch::Tensor b
     auto out = torch::empty({90, B, 500}, ids.options().dtype(torch::kBFloat16));
     rk::embed_mlp_fwd(ids.data_ptr<uint8_t>(), w1.data_ptr(), b1.data_ptr<float>(),
                       w2.data_ptr(), b2.data_ptr<float>(), emb.data_ptr(),
-                      out.data_ptr(), B, cur_stream());
+                      out.data_ptr(), B, cur_stream(), dbg,
+                      timing ? reinterpret_cast<unsigned long long*>(
+                                   timing->data_ptr<int64_t>())
+                             : nullptr);
+    return out;
+}
+
+// ids (B, 200, 90) u8, w1g (112, 232) zero-padded -> (90, B, 500) bf16
+torch::Tensor embed_mlp_fwd2(torch::Tensor ids, torch::Tensor w1g,
+                             torch::Tensor b1, torch::Tensor w2,
+                             torch::Tensor b2, torch::Tensor emb) {
+    check(ids, torch::kUInt8, "ids");
+    check(w1g, torch::kBFloat16, "w1g");
+    check(b1, torch::kFloat32, "b1");
+    check(w2, torch::kBFloat16, "w2");
+    check(b2, torch::kFloat32, "b2");
+    check(emb, torch::kBFloat16, "emb");
+    const int B = ids.size(0);
+    TORCH_CHECK(ids.size(1) == 200 && ids.size(2) == 90, "ids must be (B,200,90)");
+    TORCH_CHECK(w1g.size(0) == 112 && w1g.size(1) == 232,
+                "w1g must be (112,232) zero-padded");
+    auto out = torch::empty({90, B, 500}, ids.options().dtype(torch::kBFloat16));
+    rk::embed_mlp_fwd2(ids.data_ptr<uint8_t>(), w1g.data_ptr(),
+                       b1.data_ptr<float>(), w2.data_ptr(),
+                       b2.data_ptr<float>(), emb.data_ptr(), out.data_ptr(),
+                       B, cur_stream());
     return out;
 }
 
@@ -388,6 +419,7 @@ torch::Tensor atb_splitk(torch::Tensor A, torch::Tensor B) {
 struct ServeSlot {
     // weights (kernel-ready layouts from roko_amd.ops.forward._bf16_weights)
     torch::Tensor w1, b1, w2, b2, emb, w4, b4;
+    torch::Tensor w1gt;  // (112,232) zero-padded W1 for the chunked front
     std::vector<torch::Tensor> w_ih_t, b_ih, u, bhh;
     // slot state + workspaces
     torch::Tensor x_buf;     // (B, 200, 90) u8 static input
@@ -411,6 +443,7 @@ struct ServeSlot {
         };
         w1 = need("w1"); b1 = need("b1"); w2 = need("w2"); b2 = need("b2");
         emb = need("emb"); w4 = need("w4"); b4 = need("b4");
+        if (w.contains("w1g")) w1gt = need("w1g");
         for (int l = 0; l < 3; ++l) {
             auto sfx = std::to_string(l);
             w_ih_t.push_back(need(("w_ih_t" + sfx).c_str()));
@@ -441,9 +474,11 @@ struct ServeSlot {
     }
 
     void run(torch::Tensor x, int64_t n) {
-        TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kUInt8 &&
-                        x.is_contiguous() && n <= B,
-                    "x must be contiguous cuda u8, n <= slot batch");
+        // x may live on the GPU or on (ideally pinned) host memory — the
+        // copy into the static slot buffer below is D2D or H2D accordingly
+        TORCH_CHECK(x.scalar_type() == torch::kUInt8 && x.is_contiguous() &&
+                        n <= B,
+                    "x must be contiguous u8, n <= slot batch");
         // order the slot stream behind the producer of x, then enqueue the
         // whole forward on the slot stream
         auto prod = at::cuda::getCurrentCUDAStream();
@@ -453,10 +488,16 @@ struct ServeSlot {
             at::cuda::CUDAStreamGuard guard(stream);
             hipStream_t s = stream.stream();
             if (n > 0) x_buf.narrow(0, 0, n).copy_(x.narrow(0, 0, n), true);
-            rk::embed_mlp_fwd(x_buf.data_ptr<uint8_t>(), w1.data_ptr(),
-                              b1.data_ptr<float>(), w2.data_ptr(),
-                              b2.data_ptr<float>(), emb.data_ptr(),
-                              seq.data_ptr(), B, s);
+            if (w1gt.defined())
+                rk::embed_mlp_fwd2(x_buf.data_ptr<uint8_t>(), w1gt.data_ptr(),
+                                   b1.data_ptr<float>(), w2.data_ptr(),
+                                   b2.data_ptr<float>(), emb.data_ptr(),
+                                   seq.data_ptr(), B, s);
+            else
+                rk::embed_mlp_fwd(x_buf.data_ptr<uint8_t>(), w1.data_ptr(),
+                                  b1.data_ptr<float>(), w2.data_ptr(),
+                                  b2.data_ptr<float>(), emb.data_ptr(),
+                                  seq.data_ptr(), B, s, 0, nullptr);
             for (int l = 0; l < 3; ++l) {
                 at::addmm_out(xg2d, b_ih[l], l == 0 ? seq2d : hseq2d,
                               w_ih_t[l]);
@@ -484,7 +525,10 @@ struct ServeSlot {
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.doc() = "roko-mi355x CDNA4 kernels (gfx950)";
     m.def("mfma_probe", &mfma_probe);
-    m.def("embed_mlp_fwd", &embed_mlp_fwd);
+    m.def("embed_mlp_fwd2", &embed_mlp_fwd2);
+    m.def("embed_mlp_fwd", &embed_mlp_fwd, py::arg("ids"), py::arg("w1"),
+          py::arg("b1"), py::arg("w2"), py::arg("b2"), py::arg("emb"),
+          py::arg("dbg") = 0, py::arg("timing") = c10::nullopt);
     m.def("gru_layer_fwd", &gru_layer_fwd, py::arg("xg"), py::arg("u"),
           py::arg("bhh"), py::arg("train") = false, py::arg("dbg") = 0);
     m.def("gru_layer_bwd", &gru_layer_bwd);

@@ -48,7 +48,10 @@ __global__ __launch_bounds__(512, 2) void embed_mlp_fwd_kernel(
     const float* __restrict__ b2,     // (F2)
     const bf16* __restrict__ emb,     // (12, E)
     bf16* __restrict__ out,           // (W, B, OUT)
-    int B) {
+    int B,
+    uint32_t dbg,                     // phase-skip bisection: 1 hot, 2 G1,
+                                      // 4 G2, 8 G3, 16 store
+    unsigned long long* timing) {     // optional (5): per-phase cycle sums
     __shared__ struct {
         // shared across both column groups (read-only after staging)
         bf16 w1t[MP][KP_LD];        // A-operand of G1 (zero-padded W1)
@@ -94,32 +97,61 @@ __global__ __launch_bounds__(512, 2) void embed_mlp_fwd_kernel(
     __syncthreads();
 
     // ---- per-column-pair loop (group g owns column wp + g) ----------------
+    unsigned long long tacc[5] = {0, 0, 0, 0, 0};
+#define PH_T0 unsigned long long tp0 = (timing && tid == 0) \
+        ? __builtin_amdgcn_s_memtime() : 0
+#define PH_T1(i) if (timing && tid == 0) \
+        tacc[i] += __builtin_amdgcn_s_memtime() - tp0
     for (int wp = 0; wp < W; wp += 2) {
         const int w = wp + g;
         // build Hot^T: zero, then scatter 200 ones
+        { PH_T0;
+        if (!(dbg & 1u)) {
         for (int e = tidg; e < 16 * KP_LD; e += 256)
             (&lds.hot_t[g][0][0])[e] = f2bf(0.f);
         __syncthreads();
         for (int r = tidg; r < R; r += 256)
             lds.hot_t[g][lds.win[r * W + w]][r] = f2bf(1.0f);
+        }
         __syncthreads();
+        PH_T1(0); }
 
-        // G1: A = W1 · Hot  — the group's waves own m-tiles {wid4, wid4+4}
+        // G1: A = W1 · Hot — each wave owns m-tiles {wid4, wid4+4} and runs
+        // their two k-chains INTERLEAVED on split accumulators with the
+        // shared b-fragment loaded once per k-step: the original
+        // one-accumulator loop serialized load->load->mfma at ~170 cycles
+        // per tile (phase bisection, profiles/) because every MFMA waited
+        // on the previous one's accumulator and on freshly-issued LDS reads
+        { PH_T0;
+        if (!(dbg & 2u)) {
+            const int mt0 = wid4, mt1 = wid4 + 4;
+            const bool has1 = mt1 < 7;
+            f32x4 acc0 = {0.f, 0.f, 0.f, 0.f}, acc1 = {0.f, 0.f, 0.f, 0.f};
+            bf16x8 b0 = lds_load_b_frag_t(&lds.hot_t[g][0][0], 0, 0, KP_LD);
+            bf16x8 a0 = lds_load_a_frag(&lds.w1t[0][0], mt0 * 16, 0, KP_LD);
+            bf16x8 a1 = lds_load_a_frag(&lds.w1t[0][0], mt1 * 16, 0, KP_LD);
 #pragma unroll
-        for (int s = 0; s < 2; ++s) {
-            const int mt = wid4 + s * 4;
-            if (mt < 7) {
-                f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-#pragma unroll
-                for (int kb = 0; kb < 7; ++kb) {
-                    bf16x8 a = lds_load_a_frag(&lds.w1t[0][0], mt * 16, kb * 32, KP_LD);
-                    bf16x8 bb = lds_load_b_frag_t(&lds.hot_t[g][0][0], 0, kb * 32, KP_LD);
-                    acc = mfma16x16x32(a, bb, acc);
+            for (int kb = 0; kb < 7; ++kb) {
+                bf16x8 bn, an0, an1;
+                if (kb + 1 < 7) {  // prefetch next k-step while MFMAs run
+                    bn = lds_load_b_frag_t(&lds.hot_t[g][0][0], 0,
+                                           (kb + 1) * 32, KP_LD);
+                    an0 = lds_load_a_frag(&lds.w1t[0][0], mt0 * 16,
+                                          (kb + 1) * 32, KP_LD);
+                    an1 = lds_load_a_frag(&lds.w1t[0][0], mt1 * 16,
+                                          (kb + 1) * 32, KP_LD);
                 }
+                acc0 = mfma16x16x32(a0, b0, acc0);
+                if (has1) acc1 = mfma16x16x32(a1, b0, acc1);
+                b0 = bn; a0 = an0; a1 = an1;
+            }
+#pragma unroll
+            for (int i = 0; i < 4; ++i)
+                lds.a_lds[g][mt0 * 16 + lrow * 4 + i][lcol] = f2bf(acc0[i]);
+            if (has1)
 #pragma unroll
                 for (int i = 0; i < 4; ++i)
-                    lds.a_lds[g][mt * 16 + lrow * 4 + i][lcol] = f2bf(acc[i]);
-            }
+                    lds.a_lds[g][mt1 * 16 + lrow * 4 + i][lcol] = f2bf(acc1[i]);
         }
         // zero the k-pad of a_lds once: cols 12..15 are written with
         // zero-valued products (Hot cols 12..15 are zero); cols 16..31 are
@@ -129,51 +161,75 @@ __global__ __launch_bounds__(512, 2) void embed_mlp_fwd_kernel(
 #pragma unroll
                 for (int k = 16; k < 32; ++k) lds.a_lds[g][e][k] = f2bf(0.f);
         __syncthreads();
+        PH_T1(1); }
 
         // G2: t1 = relu(A · E + b1) — 7 m-tiles x 4 n-tiles over 4 waves
+        { PH_T0;
+        if (!(dbg & 4u)) {
+            // 7 tiles per wave (tile = wid4 + s*4 is always < 28): load all
+            // operands first, then issue the 7 independent MFMAs, then the
+            // epilogues — straight-line so nothing waits on LDS latency
+            bf16x8 av[7], bv[7];
+            f32x4 accv[7];
 #pragma unroll
-        for (int s = 0; s < 7; ++s) {
-            const int tile = wid4 + s * 4;
-            if (tile < 28) {
+            for (int s = 0; s < 7; ++s) {
+                const int tile = wid4 + s * 4;
+                av[s] = lds_load_a_frag(&lds.a_lds[g][0][0], (tile >> 2) * 16,
+                                        0, 40);
+                bv[s] = lds_load_b_frag_t(&lds.e_t[0][0], (tile & 3) * 16, 0,
+                                          32);
+            }
+#pragma unroll
+            for (int s = 0; s < 7; ++s)
+                accv[s] = mfma16x16x32(av[s], bv[s], f32x4{0.f, 0.f, 0.f, 0.f});
+#pragma unroll
+            for (int s = 0; s < 7; ++s) {
+                const int tile = wid4 + s * 4;
                 const int mt = tile >> 2, nt = tile & 3;
-                bf16x8 a = lds_load_a_frag(&lds.a_lds[g][0][0], mt * 16, 0, 40);
-                bf16x8 bb = lds_load_b_frag_t(&lds.e_t[0][0], nt * 16, 0, 32);
-                f32x4 acc = mfma16x16x32(a, bb, f32x4{0.f, 0.f, 0.f, 0.f});
 #pragma unroll
                 for (int i = 0; i < 4; ++i) {
                     const int f = mt * 16 + lrow * 4 + i;
                     const int e = nt * 16 + lcol;
-                    float v = acc[i] + (f < F1 ? lds.b1s[f] : 0.f);
+                    float v = accv[s][i] + (f < F1 ? lds.b1s[f] : 0.f);
                     lds.t1_t[g][e][f] = f2bf(fmaxf(v, 0.f));
                 }
             }
         }
         __syncthreads();
+        PH_T1(2); }
 
         // G3: t2 = relu(W2 · t1 + b2) — 4 n-tiles, one per wave
-        {
+        { PH_T0;
+        if (!(dbg & 8u)) {
             const int nt = wid4;
-            f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+            // all 8 operand fragments up front, two independent k-chains
+            bf16x8 a[4], bb[4];
 #pragma unroll
             for (int kb = 0; kb < 4; ++kb) {
-                bf16x8 a = lds_load_a_frag(&lds.w2_lds[0][0], 0, kb * 32, 136);
-                bf16x8 bb = lds_load_b_frag_t(&lds.t1_t[g][0][0], nt * 16, kb * 32, 136);
-                acc = mfma16x16x32(a, bb, acc);
+                a[kb] = lds_load_a_frag(&lds.w2_lds[0][0], 0, kb * 32, 136);
+                bb[kb] = lds_load_b_frag_t(&lds.t1_t[g][0][0], nt * 16,
+                                           kb * 32, 136);
             }
+            f32x4 acc0 = mfma16x16x32(a[0], bb[0], f32x4{0.f, 0.f, 0.f, 0.f});
+            f32x4 acc1 = mfma16x16x32(a[1], bb[1], f32x4{0.f, 0.f, 0.f, 0.f});
+            acc0 = mfma16x16x32(a[2], bb[2], acc0);
+            acc1 = mfma16x16x32(a[3], bb[3], acc1);
 #pragma unroll
             for (int i = 0; i < 4; ++i) {
                 const int j = lrow * 4 + i;
                 const int e = nt * 16 + lcol;
                 if (j < F2 && e < E) {
-                    float v = acc[i] + lds.b2s[j];
+                    float v = acc0[i] + acc1[i] + lds.b2s[j];
                     lds.t2[g][e * F2 + j] = f2bf(fmaxf(v, 0.f));
                 }
             }
         }
         __syncthreads();
+        PH_T1(3); }
 
         // coalesced store: out[(w, b), :] = t2 flattened (e*10+j)
-        {
+        { PH_T0;
+        if (!(dbg & 16u)) {
             bf16* dst = out + ((size_t)w * B + b) * OUT;
             if (tidg < 62)
                 *reinterpret_cast<bf16x8*>(dst + tidg * 8) =
@@ -183,16 +239,214 @@ __global__ __launch_bounds__(512, 2) void embed_mlp_fwd_kernel(
                 for (int q = 0; q < 4; ++q) dst[496 + q] = lds.t2[g][496 + q];
         }
         __syncthreads();
+        PH_T1(4); }
     }
+    if (timing && tid == 0)
+#pragma unroll
+        for (int i = 0; i < 5; ++i)
+            atomicAdd(&timing[i], tacc[i]);
+#undef PH_T0
+#undef PH_T1
 }
 
 void embed_mlp_fwd(const uint8_t* ids, const void* w1, const float* b1,
                    const void* w2, const float* b2, const void* emb, void* out,
-                   int B, hipStream_t stream) {
+                   int B, hipStream_t stream, uint32_t dbg,
+                   unsigned long long* timing) {
     hipLaunchKernelGGL(embed_mlp_fwd_kernel, dim3(B), dim3(512), 0, stream,
                        ids, static_cast<const bf16*>(w1), b1,
                        static_cast<const bf16*>(w2), b2,
-                       static_cast<const bf16*>(emb), static_cast<bf16*>(out), B);
+                       static_cast<const bf16*>(emb), static_cast<bf16*>(out), B,
+                       dbg, timing);
+}
+
+// ---------------------------------------------------------------------------
+// v2: chunked-phase variant. The per-column version above runs 93 tiny MFMA
+// tiles per column between 6 barriers and measures 49% issue-stall / 38%
+// parked / 13% active (profiles/pmc_em, phase bisection in PERF_HISTORY.md).
+// This variant processes FOUR columns per iteration as three WIDE GEMM
+// phases over all 8 waves, so each phase is a long stream of independent
+// tiles (fewer barriers per column, deeper load/MFMA pipelining, W1 read
+// straight from L2):
+//   P1: A4   (112 x 48)  = W1g (112x232_L2) · Hot4 (232 x 4*12)   147 MFMA
+//   P2: t1_4 (4*112 x 64)= relu(A4 · E + b1)  per-column stacked  112 MFMA
+//   P3: out  (16 x 208)  = relu(W2 · t1_4 + b2) -> global          91 MFMA
+// Same math, same output, eval-exact.
+namespace v2 {
+
+constexpr int NC = 4;             // columns per chunk
+constexpr int NCH = (W + NC - 1) / NC;  // 23 chunks (last partial, masked)
+constexpr int N1 = NC * 12;       // G1 N = 48 (exactly 3 n-tiles)
+constexpr int LD1 = 136;          // k-pad for the G3 operands (112 -> 136)
+
+__global__ __launch_bounds__(512, 2) void embed_mlp_fwd2_kernel(
+    const uint8_t* __restrict__ ids,  // (B, R, W)
+    const bf16* __restrict__ w1g,     // (MP=112, KP_LD=232) zero-padded W1
+    const float* __restrict__ b1,     // (F1)
+    const bf16* __restrict__ w2,      // (F2, F1)
+    const float* __restrict__ b2,     // (F2)
+    const bf16* __restrict__ emb,     // (12, E)
+    bf16* __restrict__ out,           // (W, B, OUT)
+    int B) {
+    __shared__ struct {
+        uint8_t win[R * W];          // the window (staged once)
+        bf16 hot_t[N1][KP_LD];       // G1 B-operand, [n = col*12+cls][k = r]
+        bf16 a2[NC][MP][40];         // per-col G2 A-operand [f][c] (k-pad 40)
+        bf16 e_t[64][32];            // G2 B-operand [e][c] (zero-padded)
+        bf16 t1_t[208][LD1];         // G3 B-operand [n = col*50+e][k = f]
+        bf16 w2_lds[16][LD1];        // G3 A-operand (zero-padded W2)
+        float b1s[F1];
+        float b2s[16];
+    } lds;
+
+    const int b = blockIdx.x;
+    const int tid = threadIdx.x;
+    const int wid = tid >> 6;
+    const int lane = tid & 63;
+    const int lrow = lane >> 4;
+    const int lcol = lane & 15;
+
+    // ---- one-time staging -------------------------------------------------
+    // NOTE the full zero of a2 and t1_t: their k-pads (a2 cols 12-39, t1_t
+    // cols 112-135) are multiplied by ZERO operands, which is only safe if
+    // they are not NaN — uninitialized LDS can hold NaN bit patterns, and
+    // 0 x NaN = NaN silently turns into 0 at the relu (fmaxf(NaN,0)=0),
+    // zeroing whole output columns data-dependently. The valid regions are
+    // rewritten every chunk; the pads stay zero for the whole kernel.
+    for (int e = tid; e < 64 * 32; e += 512) (&lds.e_t[0][0])[e] = f2bf(0.f);
+    for (int e = tid; e < 16 * LD1; e += 512) (&lds.w2_lds[0][0])[e] = f2bf(0.f);
+    for (int e = tid; e < NC * MP * 40; e += 512)
+        (&lds.a2[0][0][0])[e] = f2bf(0.f);
+    for (int e = tid; e < 208 * LD1; e += 512)
+        (&lds.t1_t[0][0])[e] = f2bf(0.f);
+    for (int e = tid; e < 16; e += 512) lds.b2s[e] = 0.f;
+    __syncthreads();
+    for (int e = tid; e < 12 * E; e += 512) lds.e_t[e % E][e / E] = emb[e];
+    for (int e = tid; e < F2 * F1; e += 512) lds.w2_lds[e / F1][e % F1] = w2[e];
+    for (int e = tid; e < F1; e += 512) lds.b1s[e] = b1[e];
+    for (int e = tid; e < F2; e += 512) lds.b2s[e] = b2[e];
+    {
+        const uint8_t* src = ids + (size_t)b * R * W;
+        for (int e = tid * 8; e < R * W; e += 512 * 8)
+#pragma unroll
+            for (int q = 0; q < 8 && e + q < R * W; ++q) lds.win[e + q] = src[e + q];
+    }
+    __syncthreads();
+
+    for (int ch = 0; ch < NCH; ++ch) {
+        const int c0 = ch * NC;
+        // ---- P0: zero Hot4, scatter 4 x 200 ones --------------------------
+        for (int e = tid * 8; e < N1 * KP_LD; e += 512 * 8)
+            *reinterpret_cast<bf16x8*>(&lds.hot_t[0][0] + e) =
+                bf16x8{f2bf(0.f), f2bf(0.f), f2bf(0.f), f2bf(0.f),
+                       f2bf(0.f), f2bf(0.f), f2bf(0.f), f2bf(0.f)};
+        __syncthreads();
+        for (int idx = tid; idx < NC * R; idx += 512) {
+            const int col = idx / R, r = idx - col * R;
+            const int w = c0 + col;
+            if (w < W)
+                lds.hot_t[col * 12 + lds.win[r * W + w]][r] = f2bf(1.0f);
+        }
+        __syncthreads();
+
+        // ---- P1: A4 = W1 · Hot4 — 21 tiles (7m x 3n) over 8 waves ---------
+        for (int t = wid; t < 21; t += 8) {
+            const int mt = t / 3, nt = t - mt * 3;
+            bf16x8 av[7], bv[7];
+#pragma unroll
+            for (int kb = 0; kb < 7; ++kb) {
+                av[kb] = global_load_a_frag(w1g, mt * 16, kb * 32, KP_LD);
+                bv[kb] = lds_load_b_frag_t(&lds.hot_t[0][0], nt * 16, kb * 32,
+                                           KP_LD);
+            }
+            f32x4 acc0 = {0.f, 0.f, 0.f, 0.f}, acc1 = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+            for (int kb = 0; kb < 7; kb += 2) acc0 = mfma16x16x32(av[kb], bv[kb], acc0);
+#pragma unroll
+            for (int kb = 1; kb < 7; kb += 2) acc1 = mfma16x16x32(av[kb], bv[kb], acc1);
+            const int n = nt * 16 + lcol;
+            const int col = n / 12, c = n - col * 12;
+#pragma unroll
+            for (int i = 0; i < 4; ++i)
+                lds.a2[col][mt * 16 + lrow * 4 + i][c] = f2bf(acc0[i] + acc1[i]);
+        }
+        __syncthreads();
+
+        // ---- P2: t1 = relu(A4 · E + b1) — 112 single-MFMA tiles -----------
+        {
+            bf16x8 av[14], bv[14];
+            f32x4 accv[14];
+#pragma unroll
+            for (int s = 0; s < 14; ++s) {
+                const int t = wid + s * 8;
+                const int col = t / 28, rem = t - col * 28;
+                const int mt = rem >> 2, nt = rem & 3;
+                av[s] = lds_load_a_frag(&lds.a2[col][0][0], mt * 16, 0, 40);
+                bv[s] = lds_load_b_frag_t(&lds.e_t[0][0], nt * 16, 0, 32);
+            }
+#pragma unroll
+            for (int s = 0; s < 14; ++s)
+                accv[s] = mfma16x16x32(av[s], bv[s], f32x4{0.f, 0.f, 0.f, 0.f});
+#pragma unroll
+            for (int s = 0; s < 14; ++s) {
+                const int t = wid + s * 8;
+                const int col = t / 28, rem = t - col * 28;
+                const int mt = rem >> 2, nt = rem & 3;
+                const int e = nt * 16 + lcol;
+                if (e < E) {
+#pragma unroll
+                    for (int i = 0; i < 4; ++i) {
+                        const int f = mt * 16 + lrow * 4 + i;
+                        float v = accv[s][i] + (f < F1 ? lds.b1s[f] : 0.f);
+                        lds.t1_t[col * E + e][f] = f2bf(fmaxf(v, 0.f));
+                    }
+                }
+            }
+        }
+        __syncthreads();
+
+        // ---- P3: out = relu(W2 · t1 + b2), stored straight to global ------
+        for (int t = wid; t < 13; t += 8) {
+            bf16x8 av[4], bv[4];
+#pragma unroll
+            for (int kb = 0; kb < 4; ++kb) {
+                av[kb] = lds_load_a_frag(&lds.w2_lds[0][0], 0, kb * 32, LD1);
+                bv[kb] = lds_load_b_frag_t(&lds.t1_t[0][0], t * 16, kb * 32,
+                                           LD1);
+            }
+            f32x4 acc0 = mfma16x16x32(av[0], bv[0], f32x4{0.f, 0.f, 0.f, 0.f});
+            f32x4 acc1 = mfma16x16x32(av[1], bv[1], f32x4{0.f, 0.f, 0.f, 0.f});
+            acc0 = mfma16x16x32(av[2], bv[2], acc0);
+            acc1 = mfma16x16x32(av[3], bv[3], acc1);
+            const int n = t * 16 + lcol;
+            const int col = n / E, e = n - col * E;
+            const int w = c0 + col;
+            if (n < NC * E && w < W) {
+                bf16* dst = out + ((size_t)w * B + b) * OUT + e * F2;
+#pragma unroll
+                for (int i = 0; i < 4; ++i) {
+                    const int j = lrow * 4 + i;
+                    if (j < F2) {
+                        float v = acc0[i] + acc1[i] + lds.b2s[j];
+                        dst[j] = f2bf(fmaxf(v, 0.f));
+                    }
+                }
+            }
+        }
+        __syncthreads();
+    }
+}
+
+}  // namespace v2
+
+void embed_mlp_fwd2(const uint8_t* ids, const void* w1g, const float* b1,
+                    const void* w2, const float* b2, const void* emb,
+                    void* out, int B, hipStream_t stream) {
+    hipLaunchKernelGGL(v2::embed_mlp_fwd2_kernel, dim3(B), dim3(512), 0,
+                       stream, ids, static_cast<const bf16*>(w1g), b1,
+                       static_cast<const bf16*>(w2), b2,
+                       static_cast<const bf16*>(emb), static_cast<bf16*>(out),
+                       B);
 }
 
 }  // namespace rk
