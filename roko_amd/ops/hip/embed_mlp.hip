@@ -333,14 +333,22 @@ __global__ __launch_bounds__(512, 2) void embed_mlp_fwd2_kernel(
     }
     __syncthreads();
 
-    for (int ch = 0; ch < NCH; ++ch) {
-        const int c0 = ch * NC;
-        // ---- P0: zero Hot4, scatter 4 x 200 ones --------------------------
+    // hot is zeroed for chunk 0 here and for chunk ch+1 inside chunk ch's
+    // P3 (different LDS arrays, so the stores overlap P3's MFMA streams and
+    // the zero->scatter ordering rides P3's end barrier — one barrier saved
+    // per chunk)
+    auto zero_hot = [&]() {
         for (int e = tid * 8; e < N1 * KP_LD; e += 512 * 8)
             *reinterpret_cast<bf16x8*>(&lds.hot_t[0][0] + e) =
                 bf16x8{f2bf(0.f), f2bf(0.f), f2bf(0.f), f2bf(0.f),
                        f2bf(0.f), f2bf(0.f), f2bf(0.f), f2bf(0.f)};
-        __syncthreads();
+    };
+    zero_hot();
+    __syncthreads();
+
+    for (int ch = 0; ch < NCH; ++ch) {
+        const int c0 = ch * NC;
+        // ---- P0: scatter 4 x 200 ones (hot zeroed by the previous chunk) --
         for (int idx = tid; idx < NC * R; idx += 512) {
             const int col = idx / R, r = idx - col * R;
             const int w = c0 + col;
@@ -406,6 +414,7 @@ __global__ __launch_bounds__(512, 2) void embed_mlp_fwd2_kernel(
         __syncthreads();
 
         // ---- P3: out = relu(W2 · t1 + b2), stored straight to global ------
+        if (ch + 1 < NCH) zero_hot();
         for (int t = wid; t < 13; t += 8) {
             bf16x8 av[4], bv[4];
 #pragma unroll
