@@ -30,6 +30,63 @@ def _ext():
     return ext()
 
 
+_DEFER = False
+_PENDING: list = []
+_SIDE = None
+
+
+def _side_stream(dev) -> "torch.cuda.Stream":
+    global _SIDE
+    if _SIDE is None:
+        _SIDE = torch.cuda.Stream(device=dev)
+    return _SIDE
+
+
+class deferred_weight_grads:
+    """Scope in which the weight-gradient reductions (split-K AtB kernels,
+    bias GEMVs, the whole recompute front backward) are enqueued on a SIDE
+    HIP stream instead of the autograd (main) stream, and parameters receive
+    `.grad` at `drain_deferred_grads()`.
+
+    Why: the BPTT/GRU kernels on the main stream are latency-bound at 8
+    workgroups (3% of the chip), while the weight-grad GEMMs between them
+    are wide — serial they add ~0.9 ms to the 3.85 ms step; on the side
+    stream they fill the idle CUs under the recurrence (profiles/
+    PERF_HISTORY.md). Opt-in (used by fused_train_step) so the default
+    autograd semantics — p.grad populated by backward() — stay intact for
+    every other caller."""
+
+    def __enter__(self):
+        global _DEFER
+        self._prev = _DEFER
+        _DEFER = True
+        return self
+
+    def __exit__(self, *a):
+        global _DEFER
+        _DEFER = self._prev
+
+
+def _defer_active() -> bool:
+    return _DEFER and not torch.cuda.is_current_stream_capturing()
+
+
+def drain_deferred_grads():
+    """Join the side stream and attach the stashed gradients to params."""
+    global _PENDING
+    if not _PENDING:
+        return
+    cur = torch.cuda.current_stream()
+    cur.wait_stream(_SIDE)
+    for p, g in _PENDING:
+        g.record_stream(cur)
+        if p.grad is None:
+            p.grad = g
+        else:
+            p.grad = p.grad + g
+    _PENDING = []
+
+
 def train_step_available() -> bool:
     from . import available
 
@@ -40,9 +97,11 @@ class GruLayerFn(torch.autograd.Function):
     """One bidirectional GRU layer over (T, B, in) through the HIP kernels."""
 
     @staticmethod
-    def forward(ctx, x_seq, w_ih, b_ih_all, u, bhh):
+    def forward(ctx, x_seq, w_ih, b_ih_all, u, bhh, prefs=None):
         # x_seq (T, B, in) fp32/bf16; w_ih (768, in); b_ih_all (768,)
-        # u (2, 384, 128); bhh (2, 384)
+        # u (2, 384, 128); bhh (2, 384); prefs: the 8 underlying per-direction
+        # parameters (w_ih f/r, b_ih f/r, w_hh f/r, b_hh f/r) for the
+        # deferred-weight-grad path (w_ih/u here are cat/stack composites)
         ext = _ext()
         T, B, _ = x_seq.shape
         x_bf = x_seq.to(torch.bfloat16)
@@ -57,6 +116,7 @@ class GruLayerFn(torch.autograd.Function):
         hseq, cache = ext.gru_layer_fwd(xg, u_bf, bhh_f, True)
         ctx.save_for_backward(x_bf, w_ih_bf, u_bf, hseq, cache)
         ctx.in_dtype = x_seq.dtype
+        ctx.prefs = prefs
         return hseq.view(T, B, 2 * C.HIDDEN_SIZE)
 
     @staticmethod
@@ -71,33 +131,53 @@ class GruLayerFn(torch.autograd.Function):
         # dhg (2,T,B,384) -> contiguous per-direction (TB,384) slices
         dxg, dhg = ext.gru_layer_bwd(cache, hseq, dhin, ut)
 
-        # h_prev sequences per direction
-        zeros = hseq.new_zeros(1, B, H)
-        hp_f = torch.cat([zeros, hseq[:-1, :, 0, :]], dim=0)  # (T,B,H)
-        hp_r = torch.cat([hseq[1:, :, 1, :], zeros], dim=0)
-
         TB = T * B
-        dhg_f = dhg[0].reshape(TB, 3 * H)
-        dhg_r = dhg[1].reshape(TB, 3 * H)
-        # custom split-K A^T·B kernels: hipBLASLt schedules these K=11520
-        # transpose-A reductions on 6 workgroups (81 us each)
-        du = torch.stack(
-            [ext.atb_splitk(dhg_f, hp_f.reshape(TB, H).contiguous()),
-             ext.atb_splitk(dhg_r, hp_r.reshape(TB, H).contiguous())]
-        )
-
         dxg_cat = dxg.reshape(TB, 2 * 3 * H)  # zero-copy (TB, 768)
-        x_flat = x_bf.reshape(TB, -1)
-        dw_ih = ext.atb_splitk(dxg_cat, x_flat.contiguous())  # (768, in) f32
-        # column sums as one hipBLASLt GEMV each
-        ones = x_bf.new_ones(1, TB)
-        dbhh = torch.stack(
-            [ones.mm(dhg_f).squeeze(0), ones.mm(dhg_r).squeeze(0)]
-        ).float()
-        db_ih = ones.mm(dxg_cat).squeeze(0).float()      # (768,)
         dx = dxg_cat.mm(w_ih_bf).to(ctx.in_dtype).view(T, B, -1)
 
-        return dx, dw_ih, db_ih, du, dbhh
+        def weight_grads():
+            # h_prev sequences per direction
+            zeros = hseq.new_zeros(1, B, H)
+            hp_f = torch.cat([zeros, hseq[:-1, :, 0, :]], dim=0)  # (T,B,H)
+            hp_r = torch.cat([hseq[1:, :, 1, :], zeros], dim=0)
+            dhg_f = dhg[0].reshape(TB, 3 * H)
+            dhg_r = dhg[1].reshape(TB, 3 * H)
+            # custom split-K A^T·B kernels: hipBLASLt schedules these
+            # K=11520 transpose-A reductions on 6 workgroups (81 us each)
+            du = torch.stack(
+                [ext.atb_splitk(dhg_f, hp_f.reshape(TB, H).contiguous()),
+                 ext.atb_splitk(dhg_r, hp_r.reshape(TB, H).contiguous())]
+            )
+            x_flat = x_bf.reshape(TB, -1)
+            dw_ih = ext.atb_splitk(dxg_cat, x_flat.contiguous())  # (768,in)
+            # column sums as one hipBLASLt GEMV each
+            ones = x_bf.new_ones(1, TB)
+            dbhh = torch.stack(
+                [ones.mm(dhg_f).squeeze(0), ones.mm(dhg_r).squeeze(0)]
+            ).float()
+            db_ih = ones.mm(dxg_cat).squeeze(0).float()  # (768,)
+            return du, dw_ih, dbhh, db_ih
+
+        if _defer_active() and ctx.prefs is not None:
+            cur = torch.cuda.current_stream()
+            side = _side_stream(dx.device)
+            side.wait_stream(cur)
+            with torch.cuda.stream(side):
+                du, dw_ih, dbhh, db_ih = weight_grads()
+            for t in (hseq, dhg, dxg, x_bf):
+                t.record_stream(side)
+            wf, wr, bf_, br, uf, ur, bhf, bhr = ctx.prefs
+            G3 = 3 * H
+            _PENDING.extend([
+                (wf, dw_ih[:G3]), (wr, dw_ih[G3:]),
+                (bf_, db_ih[:G3]), (br, db_ih[G3:]),
+                (uf, du[0]), (ur, du[1]),
+                (bhf, dbhh[0]), (bhr, dbhh[1]),
+            ])
+            return dx, None, None, None, None, None
+
+        du, dw_ih, dbhh, db_ih = weight_grads()
+        return dx, dw_ih, db_ih, du, dbhh, None
 
 
 class FusedCrossEntropy(torch.autograd.Function):
@@ -232,12 +312,29 @@ class FrontFn(torch.autograd.Function):
                             seed_buf)
         ctx.save_for_backward(ids_u8, w1b, b1f, w2b, b2f, embb)
         ctx.seed, ctx.keep, ctx.seed_buf = seed, keep, seed_buf
+        ctx.prefs = (emb_w, w1, b1, w2, b2)
         return out  # (90, B, 500) bf16
 
     @staticmethod
     def backward(ctx, dseq):
         ext = _ext()
         ids_u8, w1b, b1f, w2b, b2f, embb = ctx.saved_tensors
+        if _defer_active():
+            # the front produces ONLY weight gradients (it is the first
+            # layer): the whole recompute backward runs on the side stream,
+            # overlapping the GRU BPTT kernels still on the main stream
+            cur = torch.cuda.current_stream()
+            side = _side_stream(dseq.device)
+            side.wait_stream(cur)
+            with torch.cuda.stream(side):
+                de, dw1, db1, dw2, db2 = ext.front_bwd(
+                    ids_u8, dseq.to(torch.bfloat16).contiguous(), w1b, b1f,
+                    w2b, b2f, embb, ctx.seed, ctx.keep, 0x1F, ctx.seed_buf,
+                )
+            for t in (ids_u8, w1b, b1f, w2b, b2f, embb, dseq):
+                t.record_stream(side)
+            _PENDING.extend(zip(ctx.prefs, (de, dw1, db1, dw2, db2)))
+            return None, None, None, None, None, None, None, None, None
         de, dw1, db1, dw2, db2 = ext.front_bwd(
             ids_u8, dseq.to(torch.bfloat16).contiguous(), w1b, b1f, w2b, b2f,
             embb, ctx.seed, ctx.keep, 0x1F, ctx.seed_buf,
@@ -299,7 +396,13 @@ def train_forward(model, x: torch.Tensor, seed_buf=None) -> torch.Tensor:
         )
         if l > 0 and drop_p > 0:
             seq = torch.nn.functional.dropout(seq, drop_p, model.training)
-        seq = GruLayerFn.apply(seq, w_ih, b_ih, u, bhh)
+        prefs = (
+            getattr(g, f"weight_ih_l{l}"), getattr(g, f"weight_ih_l{l}_reverse"),
+            getattr(g, f"bias_ih_l{l}"), getattr(g, f"bias_ih_l{l}_reverse"),
+            getattr(g, f"weight_hh_l{l}"), getattr(g, f"weight_hh_l{l}_reverse"),
+            getattr(g, f"bias_hh_l{l}"), getattr(g, f"bias_hh_l{l}_reverse"),
+        )
+        seq = GruLayerFn.apply(seq, w_ih, b_ih, u, bhh, prefs)
     logits = model.fc4(seq.float()).transpose(0, 1)  # (B, T, 5)
     return logits
 
@@ -313,7 +416,9 @@ def fused_train_step(model, x, y, opt: Optional[FusedAdam] = None,
         opt.zero_grad()
     else:
         model.zero_grad(set_to_none=False)
-    loss.backward()
+    with deferred_weight_grads():
+        loss.backward()
+    drain_deferred_grads()
     if opt is not None:
         opt.allreduce_grads()
         opt.step()

@@ -395,3 +395,45 @@ def test_dual_stream_step_matches_single():
         assert rel < 5e-3, (n1, rel)
 
 
+
+
+@requires_gpu
+def test_deferred_weight_grads_match_inline():
+    """The side-stream deferred weight-grad path (used by fused_train_step)
+    produces the same gradients as the inline autograd path."""
+    import numpy as np
+    from roko_amd.model import RokoModel
+    from roko_amd.ops.train import (deferred_weight_grads,
+                                    drain_deferred_grads, fused_cross_entropy,
+                                    train_forward)
+
+    torch.manual_seed(0)
+    model = RokoModel().cuda().train()
+    g = torch.Generator().manual_seed(5)
+    x = torch.randint(0, 12, (32, 200, 90), generator=g,
+                      dtype=torch.uint8).cuda()
+    y = torch.randint(0, 5, (32, 90), generator=g).cuda()
+
+    def run(defer):
+        model.zero_grad(set_to_none=True)
+        torch.manual_seed(123)  # same dropout seeds both ways
+        logits = train_forward(model, x)
+        loss = fused_cross_entropy(logits, y)
+        if defer:
+            with deferred_weight_grads():
+                loss.backward()
+            drain_deferred_grads()
+        else:
+            loss.backward()
+        torch.cuda.synchronize()
+        return {n: p.grad.detach().clone()
+                for n, p in model.named_parameters()}, float(loss)
+
+    g0, l0 = run(False)
+    g1, l1 = run(True)
+    assert abs(l0 - l1) < 1e-5, (l0, l1)
+    for n in g0:
+        ref, got = g0[n].float(), g1[n].float()
+        tol = 1e-4 + 1e-3 * ref.abs().max()
+        assert (ref - got).abs().max() <= tol, (
+            n, float((ref - got).abs().max()))
