@@ -12,7 +12,7 @@ reference's hard-coded module constants):
 Nothing else in the package may re-define these numbers.
 """
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 
 # ---------------------------------------------------------------------------
 # Window geometry (reference: include/generate.h:19-23, features.py:16)

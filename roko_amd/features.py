@@ -13,10 +13,9 @@ Usage:
 from __future__ import annotations
 
 import argparse
-import sys
 import time
 from multiprocessing import Pool
-from typing import Iterator, List, Optional, Tuple
+from typing import Iterator, Optional, Tuple
 
 import numpy as np
 

@@ -19,7 +19,6 @@ then append the draft suffix after the last voted position.
 from __future__ import annotations
 
 import argparse
-import sys
 import time
 from collections import defaultdict
 from typing import Dict, List, Optional, Tuple

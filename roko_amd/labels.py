@@ -15,7 +15,7 @@ leak into labels at alignment ends; reference labels.py:121-189).
 from __future__ import annotations
 
 import itertools
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Iterator, List, Optional, Tuple
 
 import numpy as np

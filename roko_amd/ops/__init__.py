@@ -10,7 +10,6 @@ it — a silent eager/PyTorch fallback would invalidate every benchmark number
 from __future__ import annotations
 
 import importlib
-import os
 
 # NOTE: torch is imported lazily inside the functions — feature-generation
 # worker processes import roko_amd.ops._pileup through this package and must

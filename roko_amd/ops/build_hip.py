@@ -10,7 +10,6 @@ Cross-compiles fine on GPU-less hosts (PYTORCH_ROCM_ARCH=gfx950).
 import glob
 import os
 import shutil
-import sys
 
 HERE = os.path.dirname(os.path.abspath(__file__))
 

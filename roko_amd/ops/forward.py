@@ -9,8 +9,6 @@ GEMV(+argmax) kernel. No per-step launches, no MIOpen RNN, no eager fallback.
 
 from __future__ import annotations
 
-from typing import Optional
-
 import os
 
 import torch
