@@ -263,11 +263,12 @@ class _Slot:
         if x.dtype != torch.uint8:
             x = x.to(torch.uint8)
         if self.cpp is not None:
-            # ServeSlot.run orders the slot stream behind the current
-            # (producer) stream itself; backpressure comes from the slot
-            # reuse sync in materialize()/wait_done of the previous ticket
-            self.wait_done()
-            self.cpp.run(x.contiguous(), n)
+            # ONE C++ call: drain previous tenant + enqueue the forward
+            # (ServeSlot.submit, GIL released). The slot stream is ordered
+            # behind the producer stream inside; the Python wrapper around
+            # separate sync/run calls cost 81 us/batch — the serving bound
+            # is host submit cost (profiles/PERF_HISTORY.md)
+            self.cpp.submit(x if x.is_contiguous() else x.contiguous(), n)
             t = _Ticket(self, n)
             if copy_out:
                 self.pending = t

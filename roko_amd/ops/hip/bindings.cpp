@@ -517,6 +517,17 @@ struct ServeSlot {
         TORCH_CHECK(e == hipSuccess, "serve slot sync failed: ",
                     hipGetErrorString(e));
     }
+
+    // one-call hot path. There is deliberately NO event sync here:
+    // hipEventSynchronize costs ~80 us with ~32 live streams (measured —
+    // it was the serving throughput bound), and stream ordering already
+    // makes slot reuse safe (x_buf/host_out overwrites are enqueued on the
+    // same slot stream, and tickets materialize() — which does sync —
+    // before the pipeline reuses a slot's host_out). If a caller outruns
+    // the GPU the AQL ring eventually blocks the enqueue: bounded, safe.
+    void submit(torch::Tensor x, int64_t n) {
+        run(std::move(x), n);
+    }
     bool done() { return hipEventQuery(ev_done) == hipSuccess; }
 };
 
@@ -562,6 +573,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         .def(py::init<py::dict, int, torch::Tensor>(), py::arg("weights"),
              py::arg("batch"), py::arg("host_out"))
         .def("run", &ServeSlot::run, py::arg("x"), py::arg("n"))
+        .def("submit", &ServeSlot::submit, py::arg("x"), py::arg("n"),
+             py::call_guard<py::gil_scoped_release>())
         .def("sync", &ServeSlot::sync,
              py::call_guard<py::gil_scoped_release>())
         .def("done", &ServeSlot::done)
