@@ -118,14 +118,16 @@ class InferencePipeline:
     """Serving-style pipelined inference on one GPU.
 
     The GRU recurrence is latency-bound: at b=128 one forward occupies only
-    8 workgroups of the 256 CUs (profiles/infer_r01_kernel_stats.txt — 67%
-    of step time in gru_layer_fwd at ~3% chip occupancy). Independent
-    batches therefore overlap almost perfectly, so the engine keeps `depth`
-    in-flight batches on `depth` HIP streams, and each slot's whole forward
-    (embed+MLP kernel -> 3x(xg GEMM + persistent GRU kernel) -> fused
-    head+argmax) is captured once into a hipGraph and replayed per step —
-    one launch per batch instead of ~11 (BASELINE.json config 4's
-    "hipGraph-captured GRU steps").
+    8 workgroups of the 256 CUs (profiles/ kernel stats — gru_layer_fwd
+    dominates chain time at ~3% chip occupancy). Independent batches
+    therefore overlap, so the engine keeps `depth` in-flight batches on
+    per-slot HIP streams. Each slot's whole forward (front kernel -> 3x
+    (xg GEMM + persistent GRU kernel) -> fused head+argmax -> pinned D2H)
+    is enqueued by ONE C++ call (`ext.ServeSlot.submit`) — host submit cost
+    is a first-order serving bound on ROCm (profiles/PERF_HISTORY.md). A
+    hipGraph backend (each slot captured once, replayed per step —
+    BASELINE.json config 4's "hipGraph-captured GRU steps") is kept as
+    ROKO_SERVE=graph; measured equal throughput at higher host cost.
 
     Every submitted batch runs the full model; nothing is cached or skipped
     — pipelining only overlaps independent batches, as a serving deployment
