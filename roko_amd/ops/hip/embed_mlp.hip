@@ -10,14 +10,15 @@
 // the per-column MAC count ~8x (200->12 contraction) and turns the gather
 // into a 200-entry scatter of ones. Exact in eval mode (no dropout).
 //
-// One workgroup (EIGHT waves, two per SIMD) owns one window and processes
-// TWO columns per iteration: waves 0-3 run column w, waves 4-7 column w+1,
-// with private hot/a/t1/t2 tiles per column group and the weights staged
-// once in shared LDS. The kernel is latency-bound, not MFMA-bound (~6k
-// cycles per column against ~400 cycles of MFMA issue), so the win is the
-// CDNA4 arbitration rule (docs/KERNELS.md "Measured lessons" #1): a
-// co-resident partner wave's MFMAs fill the stall slots of the other
-// column's dependent LDS/VALU chains. Each column is three small MFMA GEMMs:
+// TWO kernels live in this file. The first (v1) is the per-column version:
+// one workgroup (eight waves, two per SIMD) owns one window, waves 0-3 run
+// column w and waves 4-7 column w+1 with private hot/a/t1/t2 tiles. It is
+// KEPT as the reference implementation and A/B fallback (ROKO_FRONT=v1):
+// phase bisection + PMC showed it 49% issue-stall / 13% active, and neither
+// the 2-column split, accumulator splitting nor load hoisting moved it —
+// the cost is the barrier-fenced tiny-GEMM phase structure itself, which
+// the chunked v2 kernel below replaces (1.85x). Each column is three small
+// MFMA GEMMs:
 //   G1: A   (100x12) = W1 (100x200) · Hot (200x12)        49 MFMA
 //   G2: t1  (100x50) = relu(A · E (12x50) + b1)           28 MFMA
 //   G3: t2  (10x50)  = relu(W2 (10x100) · t1 + b2)        16 MFMA
