@@ -7,6 +7,7 @@
 #include <torch/extension.h>
 
 #include <cstdint>
+#include <vector>
 
 namespace rk {
 void mfma_probe(const float* a, const float* b, float* d, hipStream_t stream);
@@ -531,6 +532,30 @@ struct ServeSlot {
     bool done() { return hipEventQuery(ev_done) == hipSuccess; }
 };
 
+// Create a HIP stream restricted to all but the first `reserve_cus` CUs.
+// Chip-wide kernels launched on it leave headroom for the latency-bound
+// 8-workgroup GRU chain on the default/main stream (the deferred weight-grad
+// GEMMs measured as BLOCKING the BPTT kernels when launched unmasked —
+// profiles/PERF_HISTORY.md). Returned as an integer handle for
+// torch.cuda.ExternalStream; lives for the process lifetime.
+uint64_t cu_masked_stream(int64_t reserve_cus) {
+    hipDeviceProp_t prop;
+    int dev = 0;
+    (void)hipGetDevice(&dev);
+    (void)hipGetDeviceProperties(&prop, dev);
+    const int ncu = prop.multiProcessorCount;
+    TORCH_CHECK(reserve_cus >= 0 && reserve_cus < ncu, "bad reserve_cus");
+    const int words = (ncu + 31) / 32;
+    std::vector<uint32_t> mask(words, 0u);
+    for (int cu = (int)reserve_cus; cu < ncu; ++cu)
+        mask[cu / 32] |= (1u << (cu % 32));
+    hipStream_t s = nullptr;
+    hipError_t e = hipExtStreamCreateWithCUMask(&s, words, mask.data());
+    TORCH_CHECK(e == hipSuccess, "hipExtStreamCreateWithCUMask: ",
+                hipGetErrorString(e));
+    return reinterpret_cast<uint64_t>(s);
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -569,6 +594,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("phase_mask") = 0x1F, py::arg("seed_buf") = c10::nullopt);
     m.def("head_fwd", &head_fwd, py::arg("hseq"), py::arg("w4"), py::arg("b4"),
           py::arg("want_logits") = true, py::arg("want_argmax") = false);
+    m.def("cu_masked_stream", &cu_masked_stream, py::arg("reserve_cus"));
     py::class_<ServeSlot>(m, "ServeSlot")
         .def(py::init<py::dict, int, torch::Tensor>(), py::arg("weights"),
              py::arg("batch"), py::arg("host_out"))
