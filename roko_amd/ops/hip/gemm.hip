@@ -176,7 +176,7 @@ constexpr int BM = 128;  // C rows (= A cols) per workgroup
 constexpr int BN = 128;  // C cols per workgroup
 constexpr int BK = 64;   // K slice per stage
 constexpr int KSLICE = 256;  // K rows per workgroup (KSLICE/BK stages)
-constexpr int LD = BK + 8;
+constexpr int LDM = BM + 8;  // [k][m] row stride (staged UNtransposed)
 constexpr int WAVES = 8;
 
 __global__ __launch_bounds__(WAVES * 64, 2) void atb_splitk_kernel(
@@ -186,9 +186,16 @@ __global__ __launch_bounds__(WAVES * 64, 2) void atb_splitk_kernel(
                                  // the atomic-commit version spent ~50 us in
                                  // 45-way same-address contention on dW_ih)
     int M, int N, int K) {
+    // staged in the GLOBAL orientation [k][m] / [k][n]: the transposed
+    // staging this replaced wrote 8 scalar u16 per thread at an 8-row lane
+    // stride whose dword period gcd'd with the 64 banks to 32 — 8-way bank
+    // conflicts on every write, 88% LDS-conflict rate and 7.6% ACTIVE in
+    // PMC (profiles/pmc_train_r01.txt). Un-transposed staging is straight
+    // b128 stores; the fragment reads below pay scalar u16 loads instead
+    // (2-cyc issues that hide under the MFMAs — docs/KERNELS.md lesson 3).
     __shared__ struct {
-        bf16 at[2][BM][LD];  // [m][k]
-        bf16 bt[2][BN][LD];  // [n][k]
+        bf16 at[2][BK][LDM];  // [k][m]
+        bf16 bt[2][BK][LDM];  // [k][n]
     } lds;
 
     const int k_begin = blockIdx.x * KSLICE;
@@ -224,8 +231,8 @@ __global__ __launch_bounds__(WAVES * 64, 2) void atb_splitk_kernel(
                     v[q] = (krow < K && m0 + m + q < M)
                                ? A[(size_t)krow * M + m0 + m + q] : f2bf(0.f);
             }
-#pragma unroll
-            for (int q = 0; q < 8; ++q) lds.at[buf][m + q][kk] = v[q];
+            *reinterpret_cast<bf16x8*>(&lds.at[buf][kk][m]) =
+                *reinterpret_cast<const bf16x8*>(v);
         }
 #pragma unroll
         for (int p = 0; p < 2; ++p) {
@@ -242,8 +249,8 @@ __global__ __launch_bounds__(WAVES * 64, 2) void atb_splitk_kernel(
                     v[q] = (krow < K && n0 + n + q < N)
                                ? B[(size_t)krow * N + n0 + n + q] : f2bf(0.f);
             }
-#pragma unroll
-            for (int q = 0; q < 8; ++q) lds.bt[buf][n + q][kk] = v[q];
+            *reinterpret_cast<bf16x8*>(&lds.bt[buf][kk][n]) =
+                *reinterpret_cast<const bf16x8*>(v);
         }
     };
 
@@ -254,10 +261,12 @@ __global__ __launch_bounds__(WAVES * 64, 2) void atb_splitk_kernel(
         if (s + 1 < stages) stage((s + 1) & 1, k_begin + (s + 1) * BK);
 #pragma unroll
         for (int kb = 0; kb < BK / 32; ++kb) {
-            bf16x8 af = lds_load_a_frag(&lds.at[s & 1][0][0], wid * 16, kb * 32, LD);
+            bf16x8 af = lds_load_a_frag_t(&lds.at[s & 1][0][0], wid * 16,
+                                          kb * 32, LDM);
 #pragma unroll
             for (int b_ = 0; b_ < 8; ++b_) {
-                bf16x8 bf_ = lds_load_b_frag_t(&lds.bt[s & 1][0][0], b_ * 16, kb * 32, LD);
+                bf16x8 bf_ = lds_load_b_frag_km(&lds.bt[s & 1][0][0], kb * 32,
+                                                b_ * 16, LDM);
                 acc[b_] = mfma16x16x32(af, bf_, acc[b_]);
             }
         }
