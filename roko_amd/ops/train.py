@@ -71,6 +71,9 @@ class deferred_weight_grads:
 
     def __enter__(self):
         global _DEFER
+        # discard stashed grads from a step that died mid-backward — they
+        # would otherwise be attached to the params on the next drain
+        _PENDING.clear()
         self._prev = _DEFER
         _DEFER = True
         return self
