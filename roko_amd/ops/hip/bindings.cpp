@@ -44,7 +44,7 @@ void front_fwd(const uint8_t* ids, const void* w1, const float* b1,
 void front_bwd(const uint8_t* ids, const void* dseq, const void* w1,
                const float* b1, const void* w2, const float* b2,
                const void* emb, float* dw1, float* db1, float* dw2, float* db2,
-               void* dt1g, int B, uint32_t seed, float keep,
+               const void* w1t_g, float* de, int B, uint32_t seed, float keep,
                hipStream_t stream, uint32_t phase_mask,
                const uint32_t* seed_ptr);
 void gemm_bias(const void* A, const void* B, const float* bias, void* C,
@@ -334,22 +334,16 @@ std::vector<torch::Tensor> front_bwd(torch::Tensor ids, torch::Tensor dseq,
     auto dw2 = torch::zeros({10, 100}, opt);
     auto db2 = torch::zeros({10}, opt);
     auto de = torch::zeros({12, 50}, opt);
-    // dt1 workspace feeds the standalone embedding-grad kernel
-    auto dt1g = torch::empty({B, 90, 112, 64}, dseq.options());
+    // zero-padded W1^T image: the merged dm phase's L2-read B-fragments
+    auto w1t_g = torch::zeros({208, 128}, w1.options());
+    w1t_g.slice(0, 0, 200).slice(1, 0, 100).copy_(w1.t());
     rk::front_bwd(ids.data_ptr<uint8_t>(), dseq.data_ptr(), w1.data_ptr(),
                   b1.data_ptr<float>(), w2.data_ptr(), b2.data_ptr<float>(),
                   emb.data_ptr(), dw1.data_ptr<float>(), db1.data_ptr<float>(),
                   dw2.data_ptr<float>(), db2.data_ptr<float>(),
-                  dt1g.data_ptr(), B, (uint32_t)seed, (float)keep,
-                  cur_stream(), (uint32_t)phase_mask, seed_ptr_of(seed_buf));
-    if (phase_mask == 0x1F) {
-        // zero-padded W1^T image for the de kernel's L2-read A-fragments
-        auto w1t_g = torch::zeros({208, 128}, w1.options());
-        w1t_g.slice(0, 0, 200).slice(1, 0, 100).copy_(w1.t());
-        rk::front_de(ids.data_ptr<uint8_t>(), dt1g.data_ptr(), w1t_g.data_ptr(),
-                     de.data_ptr<float>(), B, (uint32_t)seed, (float)keep,
-                     cur_stream(), nullptr, 0, seed_ptr_of(seed_buf));
-    }
+                  w1t_g.data_ptr(), de.data_ptr<float>(), B, (uint32_t)seed,
+                  (float)keep, cur_stream(), (uint32_t)phase_mask,
+                  seed_ptr_of(seed_buf));
     return {de, dw1, db1, dw2, db2};
 }
 

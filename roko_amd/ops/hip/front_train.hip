@@ -294,12 +294,12 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
     float* __restrict__ db1,  // (F1)
     float* __restrict__ dw2,  // (F2, F1)
     float* __restrict__ db2,  // (F2)
-    bf16* __restrict__ dt1g,  // (B, W, MP, EP) dt1 pre-activation grads —
-                              // consumed by front_de_kernel (dm/de split off
-                              // into its own kernel where a transposed W1
-                              // tile fits LDS; scalar W1 reads made the dm
-                              // phase HALF this kernel's time,
-                              // profiles/front_bwd_phases_r01)
+    const bf16* __restrict__ w1t_g,  // (208, 128) zero-padded W1^T in L2:
+                              // the dm GEMM's B-operand (a second transposed
+                              // W1 LDS copy would not fit; scalar W1 reads
+                              // made the dm phase HALF this kernel's time
+                              // in v1 — profiles/front_bwd_phases_r01)
+    float* __restrict__ de,   // (12, E) pre-zeroed, atomic-accumulated
     int B, uint32_t seed, float keep, uint32_t phase_mask,
     const uint32_t* __restrict__ seed_ptr) {
     // phase_mask: timing-experiment switch (default 0x1F = all phases).
@@ -315,13 +315,26 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
                 bf16 dt2_je[16][72];   // [j][e]: dW2 A-operand
                 bf16 dt2_ej[EP][40];   // [e][j]: dt1 A-operand
             } g;
+            bf16 dmm[EP][KP_LD];  // [e][r]: masked dm — the de GEMM's
+                                  // B-operand (dead dt2 region, same size)
         } u1;
         union {
             bf16 t1_t[EP][136];   // [e][f]: G3 B-operand (dead after G3)
-            bf16 dt1_fe[MP][72];  // [f][e]: dW1 A-operand
+            // 128 rows: the dm A-fragment reads k = f up to 127; rows
+            // 112..119 alias t1_t's tail (finite values x zero W1^T pad),
+            // rows 120..127 are zeroed once at staging (NaN hazard:
+            // 0 x uninitialized-LDS-NaN = NaN — docs/KERNELS.md lesson 9)
+            bf16 dt1_fe[128][72]; // [f][e]: dW1 + dm A-operand
         } u2;
         bf16 m_rt[R + 8][72];    // [r][e]: dW1 B-operand
-        bf16 t1_fe[MP][72];      // [f][e]: dW2 B-operand + relu/drop mask
+        union {
+            bf16 t1_fe[MP][72];     // [f][e]: dW2 B-operand + relu/drop mask
+            // hot aliases t1_fe's first 3712 elements: t1_fe is dead after
+            // the dt1/dW2 reads of the SAME column and fully rewritten by
+            // the next column's G1 epilogue; its e-pad columns 64..71 are
+            // never fragment-read (k stops at 63)
+            bf16 hot_t[16][KP_LD];  // [c][r]: de GEMM A-operand
+        } u4;
         bf16 w2_lds[12][136];    // [j][f]: G3 A-operand. Only 12 rows: the
                                  // A-fragment's row 12..15 reads run past
                                  // the array (garbage), feeding accumulator
@@ -351,7 +364,9 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
     for (int e = tid; e < EP * KP_LD; e += 512) (&lds.u1.m_t[0][0])[e] = f2bf(0.f);
     for (int e = tid; e < EP * 136; e += 512) (&lds.u2.t1_t[0][0])[e] = f2bf(0.f);
     for (int e = tid; e < (R + 8) * 72; e += 512) (&lds.m_rt[0][0])[e] = f2bf(0.f);
-    for (int e = tid; e < MP * 72; e += 512) (&lds.t1_fe[0][0])[e] = f2bf(0.f);
+    for (int e = tid; e < 16 * 72; e += 512)
+        lds.u2.dt1_fe[112 + e / 72][e % 72] = f2bf(0.f);
+    for (int e = tid; e < MP * 72; e += 512) (&lds.u4.t1_fe[0][0])[e] = f2bf(0.f);
     for (int e = tid; e < 12 * 136; e += 512) (&lds.w2_lds[0][0])[e] = f2bf(0.f);
     for (int e = tid; e < MP * 40; e += 512) (&lds.w2t_t[0][0])[e] = f2bf(0.f);
     for (int e = tid; e < MP; e += 512) lds.b1s[e] = f2bf(0.f);
@@ -375,6 +390,7 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
     float db2acc[4] = {0.f, 0.f, 0.f, 0.f};
     __syncthreads();
 
+    f32x4 de_acc = {0.f, 0.f, 0.f, 0.f};  // embedding-grad tile fragments
     for (int w = w_begin; w < w_end; ++w) {
         // ---- stage ids + dseq; rebuild m in BOTH layouts; re-zero the pad
         // regions of the aliased tiles (trashed by last column's grads) ----
@@ -465,8 +481,8 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
                     }
                     *reinterpret_cast<uint32_t*>(&lds.u2.t1_t[e][f]) =
                         *reinterpret_cast<const uint32_t*>(pk);
-                    lds.t1_fe[f][e] = pk[0];
-                    lds.t1_fe[f + 1][e] = pk[1];
+                    lds.u4.t1_fe[f][e] = pk[0];
+                    lds.u4.t1_fe[f + 1][e] = pk[1];
                 }
             }
         }
@@ -523,7 +539,7 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
                 const int e0 = emt * 16 + lrow * 4;
                 // one 8-byte read for the lane's 4 consecutive t1post values
                 const bf16x4 t1p4 = *reinterpret_cast<const bf16x4*>(
-                    &lds.t1_fe[f][e0]);
+                    &lds.u4.t1_fe[f][e0]);
                 bf16x4 pk;
                 float fsum = 0.f;
 #pragma unroll
@@ -535,11 +551,6 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
                     fsum += g;
                 }
                 *reinterpret_cast<bf16x4*>(&lds.u2.dt1_fe[f][e0]) = pk;
-                {
-                    bf16* dst = dt1g +
-                        (((size_t)b * W + w) * MP + f) * EP + emt * 16 + lrow * 4;
-                    *reinterpret_cast<bf16x4*>(dst) = pk;
-                }
                 db1acc[s] += fsum;  // per-lane partial; reduced at the end
             }
         }
@@ -551,7 +562,7 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
 #pragma unroll
             for (int kb = 0; kb < 2; ++kb) {
                 bf16x8 a = lds_load_a_frag(&lds.u1.g.dt2_je[0][0], 0, kb * 32, 72);
-                bf16x8 bb = lds_load_b_frag_t(&lds.t1_fe[0][0], nt * 16, kb * 32, 72);
+                bf16x8 bb = lds_load_b_frag_t(&lds.u4.t1_fe[0][0], nt * 16, kb * 32, 72);
                 dw2acc = mfma16x16x32(a, bb, dw2acc);
             }
         }
@@ -568,10 +579,80 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
                 }
             }
         }
+        __syncthreads();  // dW phases done: u1 (dt2) and u4 (t1_fe) are dead
+
+        // ---- dm/de (merged from the former front_de kernel: the dt1 tile
+        // is already in LDS here, so the (B,W,112,64) dt1g workspace round-
+        // trip through HBM and the second kernel's launch/staging/barriers
+        // all disappear) --------------------------------------------------
+        if (phase_mask & 16u) {
+            for (int z = tid; z < 16 * KP_LD / 8; z += 512)
+                *reinterpret_cast<bf16x8*>(&lds.u4.hot_t[0][0] + z * 8) =
+                    bf16x8{};
+            __syncthreads();
+            if (tid < R) lds.u4.hot_t[lds.col_ids[tid]][tid] = f2bf(1.0f);
+
+            // dm^T[e][r] = dt1^T · W1: A from dt1_fe ([f][e], transposed
+            // fragment reads), B = W1^T straight from L2
+#pragma clang loop unroll(disable)
+            for (int s = 0; s < 7; ++s) {
+                const int tile = wid + s * 8;
+                if (tile < 52) {
+                    const int emt = tile & 3, rnt = tile >> 2;
+                    const int r = rnt * 16 + lcol;
+                    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+                    for (int kb = 0; kb < 4; ++kb) {
+                        bf16x8 a = lds_load_a_frag_t(&lds.u2.dt1_fe[0][0],
+                                                     emt * 16, kb * 32, 72);
+                        bf16x8 bb = global_load_a_frag(w1t_g, rnt * 16,
+                                                       kb * 32, 128);
+                        acc = mfma16x16x32(a, bb, acc);
+                    }
+#pragma unroll
+                    for (int i = 0; i < 4; ++i) {
+                        const int e = emt * 16 + lrow * 4 + i;
+                        float v = 0.f;
+                        if (r < R && e < E) {
+                            const uint32_t h =
+                                mask_hash(seed, ekey(b, w, r >> 1, e));
+                            if (keep_half(h, r & 1, thresh16))
+                                v = acc[i] * inv_keep;
+                        }
+                        lds.u1.dmm[e][r] = f2bf(v);
+                    }
+                }
+            }
+            __syncthreads();
+            // de += Hot^T · dmm — k-range split over both wave halves so
+            // no half idles at the phase barrier
+            {
+                const int nt = wid & 3;
+                const int kb0 = (wid < 4) ? 0 : 4;
+                const int kb1 = (wid < 4) ? 4 : 7;
+                for (int kb = kb0; kb < kb1; ++kb) {
+                    bf16x8 a = lds_load_a_frag(&lds.u4.hot_t[0][0], 0,
+                                               kb * 32, KP_LD);
+                    bf16x8 bb = lds_load_b_frag_t(&lds.u1.dmm[0][0], nt * 16,
+                                                  kb * 32, KP_LD);
+                    de_acc = mfma16x16x32(a, bb, de_acc);
+                }
+            }
+        }
         __syncthreads();
     }
 
     // ---- commit the per-workgroup accumulators ----------------------------
+    if (phase_mask & 16u) {
+        const int nt = wid & 3;
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+            const int c = lrow * 4 + i;
+            const int e = nt * 16 + lcol;
+            if (c < 12 && e < E && de_acc[i] != 0.f)
+                agent_atomic_add(&de[c * E + e], de_acc[i]);
+        }
+    }
 #pragma unroll
     for (int s = 0; s < 12; ++s) {
         const int tile = wid + s * 8;
@@ -798,7 +879,7 @@ void front_de(const uint8_t* ids, const void* dt1g, const void* w1t_g,
 void front_bwd(const uint8_t* ids, const void* dseq, const void* w1,
                const float* b1, const void* w2, const float* b2,
                const void* emb, float* dw1, float* db1, float* dw2, float* db2,
-               void* dt1g, int B, uint32_t seed, float keep,
+               const void* w1t_g, float* de, int B, uint32_t seed, float keep,
                hipStream_t stream, uint32_t phase_mask,
                const uint32_t* seed_ptr) {
     hipLaunchKernelGGL(front::front_bwd_kernel,
@@ -807,8 +888,8 @@ void front_bwd(const uint8_t* ids, const void* dseq, const void* w1,
                        static_cast<const bf16*>(w1), b1,
                        static_cast<const bf16*>(w2), b2,
                        static_cast<const bf16*>(emb), dw1, db1, dw2, db2,
-                       static_cast<bf16*>(dt1g), B, seed, keep, phase_mask,
-                       seed_ptr);
+                       static_cast<const bf16*>(w1t_g), de, B, seed, keep,
+                       phase_mask, seed_ptr);
 }
 
 }  // namespace rk
