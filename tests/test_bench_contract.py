@@ -1,0 +1,50 @@
+"""The driver depends on bench.py's exact output contract: one JSON line on
+stdout with the documented schema. Run it end-to-end on CPU (tiny step
+counts; eager torch path) and validate every field the driver reads."""
+
+import json
+import os
+import subprocess
+import sys
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def test_bench_json_contract_cpu():
+    out = subprocess.run(
+        [sys.executable, os.path.join(ROOT, "bench.py"),
+         "--steps", "2", "--warmup", "1", "--batch", "8", "--depth", "2"],
+        cwd=ROOT, capture_output=True, text=True, timeout=600,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [l for l in out.stdout.strip().splitlines() if l.startswith("{")]
+    assert len(lines) == 1, out.stdout
+    r = json.loads(lines[0])
+    for key in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+                "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+                "dtype", "data", "config"):
+        assert key in r, key
+    assert r["metric"] == "inference_bases_per_sec"
+    assert r["n_gpus"] == 1 and r["steps"] == 2 and r["warmup"] == 1
+    assert r["higher_is_better"] is True
+    assert r["scaling"] == "weak"
+    assert r["data"] == "synthetic"
+    assert r["value"] > 0 and r["ms_per_step"] > 0
+    cfg = r["config"]
+    for key in ("model", "global_batch", "seq_len", "parallelism"):
+        assert key in cfg, key
+    assert cfg["global_batch"] == 8 and cfg["seq_len"] == 90
+
+
+def test_hw_queue_default_set_on_import():
+    """The 4-HW-queue default serializes the serving streams; the package
+    must raise it before HIP init (PERF_HISTORY.md)."""
+    out = subprocess.run(
+        [sys.executable, "-c",
+         "import roko_amd, os; print(os.environ['GPU_MAX_HW_QUEUES'])"],
+        cwd=ROOT, capture_output=True, text=True, timeout=120,
+        env={k: v for k, v in os.environ.items()
+             if k != "GPU_MAX_HW_QUEUES"},
+    )
+    assert out.returncode == 0, out.stderr[-500:]
+    assert out.stdout.strip() == "20"
