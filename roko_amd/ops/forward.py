@@ -265,11 +265,11 @@ class _Slot:
         if x.dtype != torch.uint8:
             x = x.to(torch.uint8)
         if self.cpp is not None:
-            # ONE C++ call: drain previous tenant + enqueue the forward
-            # (ServeSlot.submit, GIL released). The slot stream is ordered
-            # behind the producer stream inside; the Python wrapper around
-            # separate sync/run calls cost 81 us/batch — the serving bound
-            # is host submit cost (profiles/PERF_HISTORY.md)
+            # ONE C++ call enqueues the whole forward (GIL released). No
+            # event sync in the hot path: stream ordering makes buffer
+            # reuse safe, tickets sync in materialize(), and
+            # hipEventSynchronize itself measured ~80 us with ~32 live
+            # streams (profiles/PERF_HISTORY.md)
             self.cpp.submit(x if x.is_contiguous() else x.contiguous(), n)
             t = _Ticket(self, n)
             if copy_out:
