@@ -1,16 +1,23 @@
 """Benchmark harness (driver contract — see project brief).
 
-Measures the flagship inference step on N GPUs of one node: batches of 128
-sampled-read windows (200x90, synthetic, random-init weights) through the
-hand-written gfx950 kernel path, fused argmax, predictions copied to host —
-i.e. the per-window work of the polishing pipeline (BASELINE.json metric:
-inference bases/sec at b=128; bases = windows * 30-column stride).
+Measures BOTH halves of the BASELINE.json metric ("inference bases/sec +
+train windows/sec at b=128") on N GPUs of one node, by default as two timed
+sections of one invocation:
 
-  python bench.py --gpus N --steps K --warmup W [--mode inference|train]
+  * train: the fused HIP train step (the same path the roko_amd.train CLI
+    runs on GPU) — full forward, backward, FusedAdam update per step;
+  * inference: pipelined serving of b=128 windows through the gfx950 kernel
+    path, fused argmax, predictions copied to host (bases = windows *
+    30-column stride).
 
-Under torchrun (one rank per GPU, RCCL) ranks run independent streams of
-batches (weak scaling; inference needs no collectives — SURVEY.md §2.5) and
-the max elapsed over ranks is used.
+  python bench.py --gpus N --steps K --warmup W [--mode both|inference|train]
+
+One JSON line is printed per section; the LAST line is the inference record
+(comparable round to round) and it carries the train numbers in its config.
+Under torchrun (one rank per GPU, RCCL) inference ranks run independent
+batch streams (weak scaling; no collectives — SURVEY.md §2.5); train ranks
+all-reduce gradients each step (weak scaling: global batch = 128 * N). Max
+elapsed over ranks is used everywhere.
 """
 
 from __future__ import annotations
@@ -160,13 +167,47 @@ def bench_train(args, rank, world, device):
     }
 
 
+def _record(res, args, world, device, mode, extra=None):
+    out = {
+        "metric": res["metric"],
+        "value": res["value"],
+        "unit": res["unit"],
+        "n_gpus": world,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": res["ms_per_step"],
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,
+        "dtype": "bf16" if device.type == "cuda" else "float32",
+        "data": "synthetic",
+        "config": {
+            "model": "roko bi-GRU polisher (r10-shape: 200x90 windows, "
+                     "emb50, fc 200->100->10, GRU 500/128x3 bidir, 5-class)",
+            "global_batch": args.batch * world,
+            "seq_len": C.WINDOW_COLS,
+            "parallelism": f"dp{world}",
+            "mode": mode,
+            "bases_per_window": C.WINDOW_STRIDE,
+        },
+    }
+    if extra:
+        out["config"].update(extra)
+    return out
+
+
 def main():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=400)
     p.add_argument("--warmup", type=int, default=50)
     p.add_argument("--batch", type=int, default=C.BATCH_SIZE)
-    p.add_argument("--mode", choices=["inference", "train"], default="inference")
+    p.add_argument("--mode", choices=["both", "inference", "train"],
+                   default="both",
+                   help="'both' (default) measures train windows/s first, "
+                        "then inference bases/s — the two halves of the "
+                        "BASELINE metric — and prints one JSON line each "
+                        "(inference last)")
     p.add_argument("--depth", type=int, default=48,
                    help="in-flight batches / HIP streams (inference mode); kept "
                         "below the default warmup so every hipGraph capture "
@@ -174,40 +215,45 @@ def main():
     args = p.parse_args()
 
     rank, local_rank, world = init_distributed()
+    if args.gpus != world:
+        raise SystemExit(
+            f"--gpus {args.gpus} but WORLD_SIZE is {world}: for N>1 launch "
+            "via torchrun --nproc-per-node N (the flag is validated, not a "
+            "process launcher)")
     device = (
         torch.device("cuda", local_rank)
         if torch.cuda.is_available()
         else torch.device("cpu")
     )
 
-    fn = bench_inference if args.mode == "inference" else bench_train
-    res = fn(args, rank, world, device)
+    from roko_amd.ops.train import train_step_available
+
+    can_train = device.type == "cuda" and train_step_available()
+    records = []
+    if args.mode == "train":
+        res = bench_train(args, rank, world, device)
+        records.append(_record(res, args, world, device, "train"))
+    elif args.mode == "inference":
+        res = bench_inference(args, rank, world, device)
+        records.append(_record(res, args, world, device, "inference"))
+    else:  # both
+        tres = None
+        if can_train:
+            tres = bench_train(args, rank, world, device)
+            records.append(_record(tres, args, world, device, "train"))
+        elif rank == 0:
+            print("train section skipped: fused HIP step needs a GPU",
+                  file=sys.stderr)
+        res = bench_inference(args, rank, world, device)
+        extra = {}
+        if tres is not None:
+            extra = {"train_windows_per_sec": tres["value"],
+                     "train_ms_per_step": tres["ms_per_step"]}
+        records.append(_record(res, args, world, device, "inference", extra))
 
     if rank == 0:
-        out = {
-            "metric": res["metric"],
-            "value": res["value"],
-            "unit": res["unit"],
-            "n_gpus": world,
-            "steps": args.steps,
-            "warmup": args.warmup,
-            "ms_per_step": res["ms_per_step"],
-            "higher_is_better": True,
-            "scaling": "weak",
-            "vs_baseline": None,
-            "dtype": "bf16" if device.type == "cuda" else "float32",
-            "data": "synthetic",
-            "config": {
-                "model": "roko bi-GRU polisher (r10-shape: 200x90 windows, "
-                         "emb50, fc 200->100->10, GRU 500/128x3 bidir, 5-class)",
-                "global_batch": args.batch * world,
-                "seq_len": C.WINDOW_COLS,
-                "parallelism": f"dp{world}",
-                "mode": args.mode,
-                "bases_per_window": C.WINDOW_STRIDE,
-            },
-        }
-        print(json.dumps(out))
+        for out in records:
+            print(json.dumps(out))
 
 
 if __name__ == "__main__":

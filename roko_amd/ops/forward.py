@@ -279,6 +279,10 @@ class _Slot:
         # the input may have been produced on another stream (usually the
         # default one): order the slot stream behind it before copying
         self.stream.wait_stream(torch.cuda.current_stream())
+        if x.is_cuda:
+            # x may be the dtype/contiguity temp created above, freed as soon
+            # as submit() returns — keep its memory until the copy runs
+            x.record_stream(self.stream)
         with torch.cuda.stream(self.stream):
             self.x[:n].copy_(x, non_blocking=True)
             if self.graph is not None:

@@ -479,6 +479,13 @@ struct ServeSlot {
         auto prod = at::cuda::getCurrentCUDAStream();
         (void)hipEventRecord(ev_in, prod.stream());
         (void)hipStreamWaitEvent(stream.stream(), ev_in, 0);
+        // x may be a caller-side temporary (dtype/contiguity conversion in
+        // Python) that dies when submit() returns while the copy below is
+        // still queued on the slot stream; pin its storage to this stream so
+        // the caching allocator cannot hand the memory out early.
+        if (x.is_cuda())
+            c10::cuda::CUDACachingAllocator::recordStream(
+                x.storage().data_ptr(), stream);
         {
             at::cuda::CUDAStreamGuard guard(stream);
             hipStream_t s = stream.stream();

@@ -59,6 +59,10 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gemm_bias_kernel(
 
     // cooperative staging: A tile BM x BK (8192 elems = 16/thread = 2 x b128),
     // B tile BK x BN (2048 elems = 4/thread, transposed scalar writes)
+    // b128 loads/stores need 16-byte alignment: row starts are only aligned
+    // when the row stride is a multiple of 8 bf16 elements (live shapes like
+    // K=500 are not) — take the scalar tail path otherwise
+    const bool a_aligned = (K % 8) == 0;
     auto stage = [&](int buf, int k0) {
         const bool full_k = (k0 + BK <= K);
 #pragma unroll
@@ -67,7 +71,7 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gemm_bias_kernel(
             const int r = e / BK, kk = e % BK;
             const int row = m0 + r;
             bf16 v[8];
-            if (full_k && row < M) {
+            if (full_k && a_aligned && row < M) {
                 *reinterpret_cast<bf16x8*>(v) =
                     *reinterpret_cast<const bf16x8*>(A + (size_t)row * K + k0 + kk);
             } else {
@@ -127,7 +131,7 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gemm_bias_kernel(
     }
     __syncthreads();
     {
-        const int n_full = (n0 + BN <= N);
+        const int n_full = (n0 + BN <= N) && ((N % 8) == 0);
 #pragma unroll
         for (int p = 0; p < 4; ++p) {
             const int e = (p * 512 + tid) * 8;
@@ -215,6 +219,10 @@ __global__ __launch_bounds__(WAVES * 64, 2) void atb_splitk_kernel(
 
     // stage one BK-slice: A rows [k0, k0+BK) cols [m0, m0+BM) transposed to
     // [m][k]; same for B. 128*64 elems each = 16/thread.
+    // same alignment rule as gemm_bias_kernel: vector loads only when the
+    // row stride keeps every row 16-byte aligned
+    const bool m_aligned = (M % 8) == 0;
+    const bool n_aligned = (N % 8) == 0;
     auto stage = [&](int buf, int k0) {
 #pragma unroll
         for (int p = 0; p < 2; ++p) {
@@ -222,7 +230,7 @@ __global__ __launch_bounds__(WAVES * 64, 2) void atb_splitk_kernel(
             const int kk = e / BM, m = e % BM;   // 8 consecutive m per thread
             bf16 v[8];
             const int krow = k0 + kk;
-            if (krow < K && m0 + m + 7 < M) {
+            if (krow < K && m_aligned && m0 + m + 7 < M) {
                 *reinterpret_cast<bf16x8*>(v) = *reinterpret_cast<const bf16x8*>(
                     A + (size_t)krow * M + m0 + m);
             } else {
@@ -240,7 +248,7 @@ __global__ __launch_bounds__(WAVES * 64, 2) void atb_splitk_kernel(
             const int kk = e / BN, n = e % BN;
             bf16 v[8];
             const int krow = k0 + kk;
-            if (krow < K && n0 + n + 7 < N) {
+            if (krow < K && n_aligned && n0 + n + 7 < N) {
                 *reinterpret_cast<bf16x8*>(v) = *reinterpret_cast<const bf16x8*>(
                     B + (size_t)krow * N + n0 + n);
             } else {
@@ -325,9 +333,11 @@ void atb_splitk(const void* A, const void* B, float* ws, float* C, int M,
                        static_cast<const bf16*>(A), static_cast<const bf16*>(B),
                        ws, M, N, K);
     const int64_t MN = (int64_t)M * N;
+    // ceil over the 4-element stride: floor(MN/4) left the last MN%4
+    // outputs unwritten for odd shapes
     hipLaunchKernelGGL(gemmatb::slice_sum_kernel,
-                       dim3((MN / 4 + 255) / 256), dim3(256), 0, stream, ws, C,
-                       S, MN);
+                       dim3(((MN + 3) / 4 + 255) / 256), dim3(256), 0, stream,
+                       ws, C, S, MN);
 }
 
 }  // namespace rk

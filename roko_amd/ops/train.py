@@ -231,7 +231,11 @@ class FusedAdam:
     parameter, which cost ~123 us/step as 26 tiny launches) and the update is
     one multi-tensor kernel driven by a [grad_ptr, offset, numel] table. In
     DP a single gather kernel packs the grads into one flat tensor for ONE
-    all-reduce."""
+    all-reduce.
+
+    On CPU (no HIP extension) the same flat-buffer semantics run through
+    plain torch ops, so the DP sync path (`allreduce_grads`) is exercised by
+    the multi-process gloo tests byte-for-byte as the GPU path would be."""
 
     def __init__(self, params: List[torch.nn.Parameter], lr: float = C.LR,
                  betas=(0.9, 0.999), eps: float = 1e-8):
@@ -240,6 +244,7 @@ class FusedAdam:
         self.step_count = 0
         n = sum(p.numel() for p in self.params)
         dev = self.params[0].device
+        self.on_gpu = dev.type == "cuda"
         self.flat_p = torch.empty(n, dtype=torch.float32, device=dev)
         self.flat_g = torch.zeros(n, dtype=torch.float32, device=dev)
         self.m = torch.zeros(n, dtype=torch.float32, device=dev)
@@ -256,6 +261,47 @@ class FusedAdam:
     def zero_grad(self):
         for p in self.params:
             p.grad = None
+
+    def state_dict(self) -> dict:
+        """Resume sidecar payload (roko_amd.train CheckpointManager)."""
+        return {
+            "kind": "fused_adam",
+            "step_count": self.step_count,
+            "m": self.m.detach().cpu(),
+            "v": self.v.detach().cpu(),
+            "lr": self.lr, "betas": self.betas, "eps": self.eps,
+        }
+
+    def load_state_dict(self, st: dict) -> None:
+        if st.get("kind") != "fused_adam":
+            raise ValueError("not a FusedAdam state dict")
+        self.step_count = int(st["step_count"])
+        self.m.copy_(st["m"].to(self.m.device))
+        self.v.copy_(st["v"].to(self.v.device))
+        self.lr = st.get("lr", self.lr)
+        self.betas = tuple(st.get("betas", self.betas))
+        self.eps = st.get("eps", self.eps)
+
+    def _gather_flat(self) -> None:
+        """Pack per-param grads into flat_g (one kernel on GPU)."""
+        if self.on_gpu:
+            tab = self._table()
+            _ext().grad_gather(tab, len(self.params), self.flat_g)
+            return
+        for p, off in zip(self.params, self.offs):
+            assert p.grad is not None, "param missing grad"
+            self.flat_g[off : off + p.numel()].copy_(
+                p.grad.detach().reshape(-1).float())
+
+    def _torch_adam(self, g: torch.Tensor) -> None:
+        """CPU fallback of the fused update, same math as adam.hip."""
+        b1, b2 = self.betas
+        self.m.mul_(b1).add_(g, alpha=1 - b1)
+        self.v.mul_(b2).addcmul_(g, g, value=1 - b2)
+        bc1 = 1 - b1 ** self.step_count
+        bc2 = 1 - b2 ** self.step_count
+        denom = (self.v / bc2).sqrt_().add_(self.eps)
+        self.flat_p.addcdiv_(self.m / bc1, denom, value=-self.lr)
 
     def _flat_g2(self):
         if not hasattr(self, "_g2"):
@@ -285,8 +331,7 @@ class FusedAdam:
         import torch.distributed as dist
 
         if dist.is_initialized() and dist.get_world_size() > 1:
-            tab = self._table()
-            _ext().grad_gather(tab, len(self.params), self.flat_g)
+            self._gather_flat()
             dist.all_reduce(self.flat_g)
             self.flat_g /= dist.get_world_size()
             self._synced = True
@@ -295,7 +340,11 @@ class FusedAdam:
 
     def step(self):
         self.step_count += 1
-        if getattr(self, "_synced", False):
+        if not self.on_gpu:
+            if not getattr(self, "_synced", False):
+                self._gather_flat()
+            self._torch_adam(self.flat_g)
+        elif getattr(self, "_synced", False):
             _ext().adam_step(self.flat_p, self.flat_g, self.m, self.v, self.lr,
                              self.betas[0], self.betas[1], self.eps,
                              self.step_count)
@@ -428,7 +477,15 @@ def train_forward(model, x: torch.Tensor, seed_buf=None) -> torch.Tensor:
 
 def fused_train_step(model, x, y, opt: Optional[FusedAdam] = None,
                      reducer=None) -> torch.Tensor:
-    """One full training step through the fused path; returns the loss."""
+    """One full training step through the fused path; returns the loss.
+
+    DP gradient sync happens inside ``opt.allreduce_grads()`` (FusedAdam) —
+    a GradReducer's autograd hooks never fire here because the fused
+    backward ASSIGNS grads, so passing one would silently skip the sync."""
+    if reducer is not None and getattr(reducer, "enabled", False):
+        raise ValueError(
+            "fused_train_step does not support GradReducer; DP sync runs "
+            "through FusedAdam.allreduce_grads (pass opt=FusedAdam(...))")
     logits = train_forward(model, x)
     loss = fused_cross_entropy(logits, y)
     if opt is not None:
@@ -485,6 +542,12 @@ class GraphedTrainStep:
                              self.step_buf)
             return loss
 
+        # warmup runs one_step() for real — snapshot optimizer/model state so
+        # training starts from the caller's weights and step 0, not from 3
+        # all-zero-batch Adam updates (restored after capture; capture itself
+        # only records, it does not execute)
+        snap = (opt.flat_p.clone(), opt.m.clone(), opt.v.clone(),
+                self.seed_buf.clone())
         # warmup on a side stream (torch full-network capture recipe)
         s = torch.cuda.Stream(device=dev)
         s.wait_stream(torch.cuda.current_stream())
@@ -508,6 +571,13 @@ class GraphedTrainStep:
         finally:
             if gc_was_enabled:
                 gc.enable()
+        # undo the warmup's 3 real updates (see snapshot above)
+        opt.flat_p.copy_(snap[0])
+        opt.m.copy_(snap[1])
+        opt.v.copy_(snap[2])
+        self.seed_buf.copy_(snap[3])
+        self.step_buf.zero_()
+        torch.cuda.synchronize()
 
     def __call__(self, x, y):
         self.x.copy_(x, non_blocking=True)
@@ -665,6 +735,8 @@ class GraphedDualTrainStep:
                           self.step_buf)
             return (loss0 + loss1) * 0.5, grads_a, grads_b
 
+        snap = (opt.flat_p.clone(), opt.m.clone(), opt.v.clone(),
+                self.seed0.clone(), self.seed1.clone())
         side = torch.cuda.Stream(device=dev)
         side.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(side):
@@ -688,6 +760,14 @@ class GraphedDualTrainStep:
         # now the grad addresses are final: fill the recorded tables once
         self.tab_a.copy_(host_table(self._ga))
         self.tab_b.copy_(host_table(self._gb))
+        # undo the warmup's 3 real updates (see snapshot above)
+        opt.flat_p.copy_(snap[0])
+        opt.m.copy_(snap[1])
+        opt.v.copy_(snap[2])
+        self.seed0.copy_(snap[3])
+        self.seed1.copy_(snap[4])
+        self.step_buf.zero_()
+        torch.cuda.synchronize()
 
     def __call__(self, x, y):
         self.x.copy_(x, non_blocking=True)

@@ -5,13 +5,16 @@ cross-entropy -> Adam, val accuracy, EarlyStopping(patience=7), per-epoch
 best-by-val-acc ``.pth`` checkpoints), re-built as an explicit engine with
 distributed data parallelism:
 
-  * one process per GPU (torchrun), RCCL all-reduce over xGMI with
-    backward-overlapped gradient buckets (roko_amd.parallel.ddp);
+  * one process per GPU (torchrun), RCCL all-reduce over xGMI
+    (roko_amd.parallel.ddp / FusedAdam.allreduce_grads);
   * checkpoints are plain reference-format state_dicts
     (``rnn_model_<epoch>_acc=<acc>.pth`` — SURVEY.md §5.4), so they load in
     either framework;
-  * on ROCm GPUs the fused HIP train step (ops.train_step) is used when the
-    extension is built; CPU runs the autograd reference path.
+  * on GPU the DEFAULT step is the fused HIP path benchmarked in bench.py
+    (fused front/CE kernels, side-stream weight grads, FusedAdam with one
+    flat all-reduce — ops/train.py); ``ROKO_TRAIN_PATH=autograd`` or a batch
+    size not divisible by 32 falls back to autograd + torch Adam +
+    GradReducer. CPU always runs the autograd reference path.
 
 Usage:
   python -m roko_amd.train <train.rkw|dir> <out_dir> [--val v.rkw] [--memory]
@@ -119,12 +122,21 @@ class CheckpointManager:
         return int(st["epoch"]) + 1
 
 
-def evaluate(model, loader, device) -> tuple[float, float]:
-    """(accuracy, loss) over a loader (reference: train.py:57-63,69-71)."""
+def evaluate(model, loader, device, rank: int = 0, world: int = 1) -> tuple[float, float]:
+    """(accuracy, loss) over a loader (reference: train.py:57-63,69-71).
+
+    With world > 1 the batches are sharded round-robin across ranks and the
+    counts all-reduced, so every rank returns the same full-set numbers at
+    1/world of the per-rank work (the reference runs the whole set on every
+    rank)."""
+    import torch.distributed as dist
+
     model.eval()
     correct, total, loss_sum, batches = 0, 0, 0.0, 0
     with torch.no_grad():
-        for x, y in loader:
+        for i, (x, y) in enumerate(loader):
+            if world > 1 and i % world != rank:
+                continue
             x, y = x.to(device), y.to(device)
             logits = model(x)
             loss = F.cross_entropy(logits.transpose(1, 2), y)
@@ -134,9 +146,39 @@ def evaluate(model, loader, device) -> tuple[float, float]:
             loss_sum += loss.item()
             batches += 1
     model.train()
+    if world > 1 and dist.is_initialized():
+        t = torch.tensor([correct, total, loss_sum, batches],
+                         dtype=torch.float64, device=device)
+        dist.all_reduce(t)
+        correct, total, loss_sum, batches = t.tolist()
     if total == 0:
         return 0.0, 0.0
     return correct / total, loss_sum / max(batches, 1)
+
+
+def _select_fused_path(device: torch.device, cfg: TrainConfig) -> bool:
+    """GPU default = the fused HIP step bench.py measures; autograd only on
+    CPU, on request (ROKO_TRAIN_PATH=autograd), or for batch sizes the
+    kernels reject (not a multiple of 32)."""
+    if device.type != "cuda":
+        return False
+    if os.environ.get("ROKO_TRAIN_PATH", "fused") == "autograd":
+        return False
+    if cfg.batch_size % 32 != 0:
+        return False
+    from .ops.train import train_step_available
+
+    return train_step_available()
+
+
+def _broadcast_initial_state(model: torch.nn.Module) -> None:
+    """Rank-0 weights to all ranks (fused path has no GradReducer)."""
+    import torch.distributed as dist
+
+    if dist.is_initialized() and dist.get_world_size() > 1:
+        for t in model.state_dict().values():
+            if t.is_floating_point() or t.dtype in (torch.int64, torch.int32):
+                dist.broadcast(t, src=0)
 
 
 def train(
@@ -176,9 +218,22 @@ def train(
         val_dl = DataLoader(val_ds, batch_size=cfg.batch_size, num_workers=cfg.workers)
 
     model = RokoModel().to(device)
-    reducer = GradReducer(list(model.parameters()), cfg.bucket_bytes)
-    reducer.sync_module_buffers_and_params(model)
-    opt = torch.optim.Adam(model.parameters(), lr=cfg.lr)
+    use_fused = _select_fused_path(device, cfg)
+    if use_fused:
+        from .ops.train import FusedAdam
+
+        # DP sync for the fused path is FusedAdam.allreduce_grads (one flat
+        # all-reduce per step) — no GradReducer hooks (the fused backward
+        # assigns grads outside autograd's accumulate hooks)
+        _broadcast_initial_state(model)
+        opt = FusedAdam(list(model.parameters()), lr=cfg.lr)
+        reducer = None
+    else:
+        reducer = GradReducer(list(model.parameters()), cfg.bucket_bytes)
+        reducer.sync_module_buffers_and_params(model)
+        opt = torch.optim.Adam(model.parameters(), lr=cfg.lr)
+    if rank == 0:
+        log(f"train path: {'fused HIP step' if use_fused else 'autograd'}")
 
     stopper = EarlyStopper(cfg.patience)
     ckpt = CheckpointManager(out_dir)
@@ -197,27 +252,38 @@ def train(
         if sampler is not None:
             sampler.set_epoch(epoch)
         t0 = time.time()
-        run_loss, n_batches = 0.0, 0
+        n_batches = 0
+        # device-side loss accumulator: a per-step .item() would sync the
+        # stream and stall the fused step's side-stream overlap
+        loss_acc = torch.zeros((), device=device)
         for x, y in train_dl:
             x = x.to(device, non_blocking=True)
             y = y.to(device, non_blocking=True)
-            logits = model(x)
-            loss = F.cross_entropy(logits.transpose(1, 2), y)
-            opt.zero_grad(set_to_none=False)
-            loss.backward()
-            reducer.finish()
-            opt.step()
-            run_loss += loss.item()
+            if use_fused:
+                from .ops.train import fused_train_step
+
+                loss = fused_train_step(model, x, y, opt)
+            else:
+                logits = model(x)
+                loss = F.cross_entropy(logits.transpose(1, 2), y)
+                opt.zero_grad(set_to_none=False)
+                loss.backward()
+                reducer.finish()
+                opt.step()
+            loss_acc += loss.detach()
             n_batches += 1
             step += 1
             meter.add(windows=len(x))
             if max_steps is not None and step >= max_steps:
                 break
+        if device.type == "cuda":
+            torch.cuda.synchronize()
         dt = time.time() - t0
         wps = n_batches * cfg.batch_size * world / max(dt, 1e-9)
+        run_loss = float(loss_acc.item())
 
         if val_dl is not None:
-            acc, vloss = evaluate(model, val_dl, device)
+            acc, vloss = evaluate(model, val_dl, device, rank=rank, world=world)
         else:
             acc, vloss = float("nan"), float("nan")
         score = acc if val_dl is not None else -run_loss / max(n_batches, 1)
@@ -240,7 +306,8 @@ def train(
             break
 
     meter.close()
-    reducer.remove()
+    if reducer is not None:
+        reducer.remove()
     return model, history
 
 

@@ -81,6 +81,101 @@ def test_ddp_grads_match_single_process(tmp_path):
         assert torch.allclose(ddp_grads[n], p.grad, atol=1e-6), n
 
 
+def _fused_adam_worker(rank, world, port, tmpdir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    torch.manual_seed(0)
+    from roko_amd.ops.train import FusedAdam
+
+    model = RokoModel()
+    model.eval()
+    opt = FusedAdam(list(model.parameters()), lr=1e-3)
+
+    g = torch.Generator().manual_seed(7)
+    x_all = torch.randint(0, 12, (4, C.WINDOW_ROWS, C.WINDOW_COLS), generator=g)
+    y_all = torch.randint(0, 5, (4, C.WINDOW_COLS), generator=g)
+    x = x_all[rank * 2 : rank * 2 + 2]
+    y = y_all[rank * 2 : rank * 2 + 2]
+
+    for _ in range(3):  # multiple steps: moments must stay rank-identical
+        logits = model(x)
+        loss = F.cross_entropy(logits.transpose(1, 2), y)
+        opt.zero_grad()
+        loss.backward()
+        opt.allreduce_grads()
+        opt.step()
+
+    if rank == 0:
+        with open(os.path.join(tmpdir, "fused_params.pt"), "wb") as f:
+            torch.save({"flat_p": opt.flat_p.detach(), "m": opt.m, "v": opt.v}, f)
+    else:
+        with open(os.path.join(tmpdir, "fused_params_r1.pt"), "wb") as f:
+            torch.save({"flat_p": opt.flat_p.detach()}, f)
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_fused_adam_allreduce_matches_single_process(tmp_path):
+    """The fused train path's DP mechanism (FusedAdam.allreduce_grads: one
+    flat gather + ONE all-reduce, then the flat Adam update) must produce
+    the same parameters as a single process on the full global batch — the
+    mirror of test_ddp_grads_match_single_process for the second DP code
+    path."""
+    world = 2
+    ctx = mp.get_context("spawn")
+    procs = [
+        ctx.Process(target=_fused_adam_worker, args=(r, world, 29537, str(tmp_path)))
+        for r in range(world)
+    ]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(240)
+        assert p.exitcode == 0
+
+    from roko_amd.ops.train import FusedAdam
+
+    torch.manual_seed(0)
+    model = RokoModel()
+    model.eval()
+    opt = FusedAdam(list(model.parameters()), lr=1e-3)
+    g = torch.Generator().manual_seed(7)
+    x = torch.randint(0, 12, (4, C.WINDOW_ROWS, C.WINDOW_COLS), generator=g)
+    y = torch.randint(0, 5, (4, C.WINDOW_COLS), generator=g)
+    for _ in range(3):
+        logits = model(x)
+        loss = F.cross_entropy(logits.transpose(1, 2), y)
+        opt.zero_grad()
+        loss.backward()
+        opt.allreduce_grads()  # world 1: no-op marker
+        opt.step()
+
+    got = torch.load(tmp_path / "fused_params.pt", weights_only=True)
+    got_r1 = torch.load(tmp_path / "fused_params_r1.pt", weights_only=True)
+    # ranks ended identical (the sync really ran)
+    assert torch.equal(got["flat_p"], got_r1["flat_p"])
+    # and match the single-process full-batch result (not bit-equal: the
+    # all-reduce averages two half-batch means, the reference one full mean —
+    # different fp summation order, amplified through 3 Adam sqrt/divides)
+    assert torch.allclose(got["flat_p"], opt.flat_p, atol=3e-5, rtol=1e-4)
+    assert torch.allclose(got["m"], opt.m, atol=1e-5, rtol=1e-4)
+
+
+def test_fused_train_step_rejects_grad_reducer():
+    """fused_train_step must not silently skip DP sync when handed a
+    GradReducer (its hooks never fire on the fused backward)."""
+    from roko_amd.ops.train import fused_train_step
+
+    class FakeReducer:
+        enabled = True
+
+    model = RokoModel()
+    with pytest.raises(ValueError, match="GradReducer"):
+        fused_train_step(model, None, None, None, reducer=FakeReducer())
+
+
 def _infer_worker(rank, world, port, data_path, ckpt_path, out_dir):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
