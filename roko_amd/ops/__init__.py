@@ -55,6 +55,12 @@ def ext():
     return _ext
 
 
+def pileup_ext():
+    """The C++ data-path module (_pileup): BAM reader, window builder,
+    alignment stats. CPU-only — no torch, no GPU required."""
+    return importlib.import_module("roko_amd.ops._pileup")
+
+
 def model_forward(model, x):
     """Full-model forward through the HIP kernels. Eval mode runs the fully
     fused inference path; train mode runs the differentiable path (custom

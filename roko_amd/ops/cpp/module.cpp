@@ -9,6 +9,7 @@
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
 
+#include "align.h"
 #include "bam.h"
 #include "pileup.h"
 
@@ -78,4 +79,22 @@ PYBIND11_MODULE(_pileup, m) {
     m.def("bam_references", &bam_references, py::arg("path"));
     m.def("fetch_records", &fetch_records, py::arg("path"), py::arg("contig"),
           py::arg("start"), py::arg("end"));
+    m.def(
+        "align_stats",
+        [](const std::string& query, const std::string& target, int band) {
+            rk::AlignStats s;
+            {
+                py::gil_scoped_release release;
+                s = rk::align_stats(query, target, band);
+            }
+            py::dict d;
+            d["edit_distance"] = s.edit_distance;
+            d["matches"] = s.matches;
+            d["mismatches"] = s.mismatches;
+            d["insertions"] = s.insertions;
+            d["deletions"] = s.deletions;
+            return d;
+        },
+        py::arg("query"), py::arg("target"), py::arg("band") = 128,
+        "Banded global alignment stats of query vs target (assembly QC)");
 }

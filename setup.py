@@ -24,7 +24,7 @@ ext_modules = [
     Pybind11Extension(
         "roko_amd.ops._pileup",
         ["roko_amd/ops/cpp/bam.cpp", "roko_amd/ops/cpp/pileup.cpp",
-         "roko_amd/ops/cpp/module.cpp"],
+         "roko_amd/ops/cpp/align.cpp", "roko_amd/ops/cpp/module.cpp"],
         cxx_std=17,
         libraries=["z"],
         extra_compile_args=["-O3", "-Wall"],

@@ -352,22 +352,40 @@ def test_atb_splitk_vs_mm():
 
 
 @requires_gpu
-def test_graphed_train_step_learns():
-    """Whole-iteration hipGraph capture: replays must advance the seed/step
-    buffers and reduce the loss on a fixed batch like the eager path."""
-    from roko_amd.ops.train import FusedAdam, GraphedTrainStep
+def test_graphed_train_step_matches_eager_and_learns():
+    """Whole-iteration hipGraph capture: construction must leave the weights
+    untouched (warmup state is snapshotted/restored — ADVICE r1), replays
+    must track the eager fused step trajectory from the same init, advance
+    the device counters once per replay, and learn a learnable task.
+
+    Labels are the per-column consensus of the window (the real polishing
+    signal) — purely random labels are NOT memorisable by this 1.27M-param
+    model in 80 steps; the round-1 version of this test only passed because
+    the warmup perturbation inflated losses[0]."""
+    from roko_amd.ops.train import FusedAdam, GraphedTrainStep, fused_train_step
 
     torch.manual_seed(13)
-    m = RokoModel().cuda().train()
-    opt = FusedAdam(list(m.parameters()), lr=3e-3)
-    step = GraphedTrainStep(m, opt, batch=32)
+    m1 = RokoModel().cuda().eval()  # dropout off: deterministic trajectories
+    m2 = RokoModel().cuda().eval()
+    m2.load_state_dict(m1.state_dict())
+    o1 = FusedAdam(list(m1.parameters()), lr=2e-3)
+    o2 = FusedAdam(list(m2.parameters()), lr=2e-3)
+    p0 = o1.flat_p.clone()
+    step = GraphedTrainStep(m1, o1, batch=32)
+    assert torch.equal(o1.flat_p, p0), "capture warmup must not move weights"
+    assert int(step.step_buf.item()) == 0
+    assert torch.equal(o1.m, torch.zeros_like(o1.m))
+
     x = torch.randint(0, 12, (32, 200, 90), dtype=torch.uint8, device="cuda")
-    y = torch.randint(0, 5, (32, 90), device="cuda")
-    losses = [float(step(x, y)) for _ in range(80)]
-    assert losses[-1] < losses[0] * 0.7, (losses[0], losses[-1])
-    # device counters advance once per replay + 3 warmup runs (the capture
-    # itself records without executing)
-    assert int(step.step_buf.item()) == 80 + 3
+    y = torch.clamp(torch.mode(x.long() % 6, dim=1).values, max=4)  # consensus
+    gl = [float(step(x, y)) for _ in range(60)]
+    el = [float(fused_train_step(m2, x, y, o2)) for _ in range(60)]
+    assert int(step.step_buf.item()) == 60
+    # same math, two kernel schedules: trajectories must track closely
+    for i in (0, 1, 5, 20, 59):
+        assert abs(gl[i] - el[i]) < 5e-3 + 0.01 * el[i], (i, gl[i], el[i])
+    # and the consensus task is learnable
+    assert gl[-1] < gl[0] * 0.7, (gl[0], gl[-1])
 
 
 @requires_gpu
