@@ -60,6 +60,47 @@ def accumulate_votes(
     return out_keys, counts
 
 
+class StreamingVotes:
+    """Bounded-memory vote accumulation (VERDICT r1 item 7 / BASELINE
+    config 5): windows are buffered per contig and folded into a per-contig
+    (keys, counts) table every `chunk_windows`, so peak host memory is the
+    vote table (O(contig positions)) plus one chunk — NOT all windows'
+    positions+predictions (~7 GB transient for a chr1-sized contig in the
+    round-1 design; the reference's per-base Python Counter loop is
+    inference.py:119-124)."""
+
+    def __init__(self, chunk_windows: int = 4096):
+        self.chunk = chunk_windows
+        self.tables: VoteTable = {}
+        self._pos: Dict[str, List[np.ndarray]] = defaultdict(list)
+        self._pred: Dict[str, List[np.ndarray]] = defaultdict(list)
+
+    def add(self, contig: str, positions: np.ndarray, preds: np.ndarray) -> None:
+        """One window: positions (W, 2), preds (W,)."""
+        self._pos[contig].append(positions)
+        self._pred[contig].append(preds)
+        if len(self._pos[contig]) >= self.chunk:
+            self._flush(contig)
+
+    def _flush(self, contig: str) -> None:
+        if not self._pos[contig]:
+            return
+        t = accumulate_votes(np.stack(self._pos[contig]),
+                             np.stack(self._pred[contig]))
+        self._pos[contig].clear()
+        self._pred[contig].clear()
+        prev = self.tables.get(contig)
+        self.tables[contig] = t if prev is None else merge_votes([prev, t])
+
+    def buffered(self, contig: str) -> int:
+        return len(self._pos[contig])
+
+    def finalize(self) -> VoteTable:
+        for contig in list(self._pos):
+            self._flush(contig)
+        return self.tables
+
+
 def merge_votes(tables: List[Tuple[np.ndarray, np.ndarray]]) -> Tuple[np.ndarray, np.ndarray]:
     tables = [t for t in tables if len(t[0])]
     if not tables:
@@ -132,15 +173,13 @@ def infer(
     dl = DataLoader(ds, batch_size=batch_size, num_workers=workers,
                     pin_memory=torch.cuda.is_available())
 
-    # per-contig accumulation of (positions, preds) then one vectorised vote
-    per_contig_pos: Dict[str, List[np.ndarray]] = defaultdict(list)
-    per_contig_pred: Dict[str, List[np.ndarray]] = defaultdict(list)
+    # streaming per-contig vote accumulation (bounded memory)
+    sv = StreamingVotes()
 
     def account(gis, js, preds):
         for k in range(len(gis)):
             g, pos_arr, _, _ = rkw.group_arrays(int(gis[k]))
-            per_contig_pos[g["contig"]].append(np.asarray(pos_arr[js[k]]))
-            per_contig_pred[g["contig"]].append(preds[k])
+            sv.add(g["contig"], np.asarray(pos_arr[js[k]]), preds[k])
 
     t0 = time.time()
     n_windows = 0
@@ -182,11 +221,7 @@ def infer(
         f"({bases / max(dt, 1e-9):.0f} bases/s)"
     )
 
-    votes: VoteTable = {}
-    for contig in per_contig_pos:
-        P = np.stack(per_contig_pos[contig])
-        V = np.stack(per_contig_pred[contig])
-        votes[contig] = accumulate_votes(P, V)
+    votes: VoteTable = sv.finalize()
 
     if world > 1:
         gathered: List[Optional[VoteTable]] = [None] * world if rank == 0 else None
