@@ -20,6 +20,14 @@ from pybind11.setup_helpers import Pybind11Extension, build_ext
 
 ROOT = os.path.dirname(os.path.abspath(__file__))
 
+# Sanitizer build mode (SURVEY.md §5.2 — the reference ships none):
+#   ROKO_SANITIZE=address|undefined|thread python setup.py build_ext --inplace
+# builds the C++ data path instrumented; run the data-path tests with
+# LD_PRELOAD=$(gcc -print-file-name=libasan.so) (see scripts/sanitize.sh).
+_SAN = os.environ.get("ROKO_SANITIZE", "")
+_san_args = ([f"-fsanitize={_SAN}", "-fno-omit-frame-pointer", "-g", "-O1"]
+             if _SAN else [])
+
 ext_modules = [
     Pybind11Extension(
         "roko_amd.ops._pileup",
@@ -27,7 +35,8 @@ ext_modules = [
          "roko_amd/ops/cpp/align.cpp", "roko_amd/ops/cpp/module.cpp"],
         cxx_std=17,
         libraries=["z"],
-        extra_compile_args=["-O3", "-Wall"],
+        extra_compile_args=(["-O3", "-Wall"] + _san_args),
+        extra_link_args=_san_args,
     ),
 ]
 
@@ -35,7 +44,7 @@ ext_modules = [
 class BuildExt(build_ext):
     def run(self):
         super().run()
-        if shutil.which("hipcc"):
+        if shutil.which("hipcc") and not os.environ.get("SKIP_HIP"):
             build_hip()
 
 
