@@ -30,6 +30,9 @@ def _front_eval(ext, ids, w):
     if _FRONT == "v1" or not hasattr(ext, "embed_mlp_fwd2"):
         return ext.embed_mlp_fwd(ids, w["w1"], w["b1"], w["w2"], w["b2"],
                                  w["emb"])
+    if _FRONT == "v3" and hasattr(ext, "embed_mlp_fwd3"):
+        return ext.embed_mlp_fwd3(ids, w["w1g"], w["b1"], w["w2"], w["b2"],
+                                  w["emb"])
     return ext.embed_mlp_fwd2(ids, w["w1g"], w["b1"], w["w2"], w["b2"],
                               w["emb"])
 

@@ -7,6 +7,8 @@
 #include <torch/extension.h>
 
 #include <cstdint>
+#include <cstdlib>
+#include <string>
 #include <vector>
 
 namespace rk {
@@ -16,6 +18,9 @@ void embed_mlp_fwd(const uint8_t* ids, const void* w1, const float* b1,
                    int B, hipStream_t stream, uint32_t dbg,
                    unsigned long long* timing);
 void embed_mlp_fwd2(const uint8_t* ids, const void* w1g, const float* b1,
+                    const void* w2, const float* b2, const void* emb,
+                    void* out, int B, hipStream_t stream);
+void embed_mlp_fwd3(const uint8_t* ids, const void* w1g, const float* b1,
                     const void* w2, const float* b2, const void* emb,
                     void* out, int B, hipStream_t stream);
 void gru_layer_fwd(const void* xg, const void* u, const float* bhh, void* hseq,
@@ -123,6 +128,28 @@ torch::Tensor embed_mlp_fwd2(torch::Tensor ids, torch::Tensor w1g,
                 "w1g must be (112,232) zero-padded");
     auto out = torch::empty({90, B, 500}, ids.options().dtype(torch::kBFloat16));
     rk::embed_mlp_fwd2(ids.data_ptr<uint8_t>(), w1g.data_ptr(),
+                       b1.data_ptr<float>(), w2.data_ptr(),
+                       b2.data_ptr<float>(), emb.data_ptr(), out.data_ptr(),
+                       B, cur_stream());
+    return out;
+}
+
+// same contract as embed_mlp_fwd2, wave-private-column kernel (v3)
+torch::Tensor embed_mlp_fwd3(torch::Tensor ids, torch::Tensor w1g,
+                             torch::Tensor b1, torch::Tensor w2,
+                             torch::Tensor b2, torch::Tensor emb) {
+    check(ids, torch::kUInt8, "ids");
+    check(w1g, torch::kBFloat16, "w1g");
+    check(b1, torch::kFloat32, "b1");
+    check(w2, torch::kBFloat16, "w2");
+    check(b2, torch::kFloat32, "b2");
+    check(emb, torch::kBFloat16, "emb");
+    const int B = ids.size(0);
+    TORCH_CHECK(ids.size(1) == 200 && ids.size(2) == 90, "ids must be (B,200,90)");
+    TORCH_CHECK(w1g.size(0) == 112 && w1g.size(1) == 232,
+                "w1g must be (112,232) zero-padded");
+    auto out = torch::empty({90, B, 500}, ids.options().dtype(torch::kBFloat16));
+    rk::embed_mlp_fwd3(ids.data_ptr<uint8_t>(), w1g.data_ptr(),
                        b1.data_ptr<float>(), w2.data_ptr(),
                        b2.data_ptr<float>(), emb.data_ptr(), out.data_ptr(),
                        B, cur_stream());
@@ -427,6 +454,12 @@ struct ServeSlot {
     int B;
     at::cuda::CUDAStream stream;
     hipEvent_t ev_in = nullptr, ev_done = nullptr;
+    // front kernel choice (ROKO_FRONT env; v2 = round-1 default until v3
+    // is A/B-measured faster on hardware)
+    bool use_v3 = [] {
+        const char* f = getenv("ROKO_FRONT");
+        return f && std::string(f) == "v3";
+    }();
 
     ServeSlot(py::dict w, int B_, torch::Tensor host_out_)
         : B(B_), stream(at::cuda::getStreamFromPool(/*high_priority=*/false)) {
@@ -491,10 +524,11 @@ struct ServeSlot {
             hipStream_t s = stream.stream();
             if (n > 0) x_buf.narrow(0, 0, n).copy_(x.narrow(0, 0, n), true);
             if (w1gt.defined())
-                rk::embed_mlp_fwd2(x_buf.data_ptr<uint8_t>(), w1gt.data_ptr(),
-                                   b1.data_ptr<float>(), w2.data_ptr(),
-                                   b2.data_ptr<float>(), emb.data_ptr(),
-                                   seq.data_ptr(), B, s);
+                (use_v3 ? rk::embed_mlp_fwd3 : rk::embed_mlp_fwd2)(
+                    x_buf.data_ptr<uint8_t>(), w1gt.data_ptr(),
+                    b1.data_ptr<float>(), w2.data_ptr(),
+                    b2.data_ptr<float>(), emb.data_ptr(),
+                    seq.data_ptr(), B, s);
             else
                 rk::embed_mlp_fwd(x_buf.data_ptr<uint8_t>(), w1.data_ptr(),
                                   b1.data_ptr<float>(), w2.data_ptr(),
@@ -563,6 +597,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.doc() = "roko-mi355x CDNA4 kernels (gfx950)";
     m.def("mfma_probe", &mfma_probe);
     m.def("embed_mlp_fwd2", &embed_mlp_fwd2);
+    m.def("embed_mlp_fwd3", &embed_mlp_fwd3);
     m.def("embed_mlp_fwd", &embed_mlp_fwd, py::arg("ids"), py::arg("w1"),
           py::arg("b1"), py::arg("w2"), py::arg("b2"), py::arg("emb"),
           py::arg("dbg") = 0, py::arg("timing") = c10::nullopt);

@@ -123,12 +123,19 @@ RK_DEV void glds_b128(const void* g, void* l) {
 #pragma clang diagnostic pop
 }
 
-RK_DEV float sigmoidf_dev(float x) { return 1.0f / (1.0f + __expf(-x)); }
+// v_rcp_f32 (~1 ulp) instead of the IEEE division sequence: hipcc lowers
+// `a / b` to div_scale/rcp/6xfma/div_fmas/div_fixup (~12 dependent VALU) —
+// 24 of those per lane per GRU step measured as 55% of the whole recurrence
+// kernel (scripts/gru_timing.py dbg=4 bisection). The 1-ulp rcp error is
+// orders of magnitude below the bf16 state rounding.
+RK_DEV float fast_rcp(float x) { return __builtin_amdgcn_rcpf(x); }
+
+RK_DEV float sigmoidf_dev(float x) { return fast_rcp(1.0f + __expf(-x)); }
 
 // overflow-stable tanh: tanh(x) = sign(x) * (1 - e) / (1 + e), e = exp(-2|x|)
 RK_DEV float tanhf_dev(float x) {
     float e = __expf(-2.0f * fabsf(x));
-    float t = (1.0f - e) / (1.0f + e);
+    float t = (1.0f - e) * fast_rcp(1.0f + e);
     return copysignf(t, x);
 }
 

@@ -449,6 +449,215 @@ __global__ __launch_bounds__(512, 2) void embed_mlp_fwd2_kernel(
 
 }  // namespace v2
 
+// ---------------------------------------------------------------------------
+// v3: wave-private columns. v2's cost is NOT MFMA (the ~7.2k MFMAs per
+// window are ~13 us of pipe time vs 156 us measured): its 4 barriers x 23
+// chunks park every wave on the slowest phase and the staged tiles fight
+// over LDS banks (PMC: ~50% parked / 43% conflicts). Here each WAVE owns a
+// whole column with PRIVATE hot/a/t1/t2 tiles — there is no barrier between
+// G1/G2/G3 at all (within-wave lgkmcnt ordering is free), so a wave streams
+// 93 MFMAs per column back to back with register prefetch across phases.
+// 4 waves per workgroup (LDS: 4 x ~35 KB private + ~9 KB shared = 148 KB);
+// W1 A-fragments come straight from L2 like v2's P1. One wave per SIMD —
+// the stream is MFMA-ILP-rich, so no partner wave is needed for cover.
+namespace v3 {
+
+constexpr int WAVES3 = 4;
+constexpr int LD1 = 136;  // t1_t/w2 row stride (k-pad 112 -> 136)
+
+__global__ __launch_bounds__(WAVES3 * 64, 1) void embed_mlp_fwd3_kernel(
+    const uint8_t* __restrict__ ids,  // (B, R, W)
+    const bf16* __restrict__ w1g,     // (MP=112, KP_LD=232) zero-padded W1
+    const float* __restrict__ b1,     // (F1)
+    const bf16* __restrict__ w2,      // (F2, F1)
+    const float* __restrict__ b2,     // (F2)
+    const bf16* __restrict__ emb,     // (12, E)
+    bf16* __restrict__ out,           // (W, B, OUT)
+    int B) {
+    __shared__ struct {
+        // shared, read-only after staging
+        bf16 e_t[64][32];        // G2 B-operand [e][k=c] (zero-padded)
+        bf16 w2_lds[16][LD1];    // G3 A-operand (zero-padded W2)
+        float b1s[F1];
+        float b2s[16];
+        // per-wave private tiles
+        bf16 hot_t[WAVES3][16][KP_LD];  // G1 B-operand [cls][k=r]
+        bf16 a[WAVES3][MP][40];         // G1 out / G2 A-operand [f][c]
+        bf16 t1_t[WAVES3][64][LD1];     // G3 B-operand [e][k=f]
+        bf16 t2[WAVES3][512];           // output staging
+    } lds;
+
+    const int b = blockIdx.x;
+    const int tid = threadIdx.x;
+    const int wid = tid >> 6;
+    const int lane = tid & 63;
+    const int lrow = lane >> 4;
+    const int lcol = lane & 15;
+
+    // ---- one-time staging (zero pads once; see v2's NaN note) -------------
+    for (int e = tid; e < 64 * 32; e += WAVES3 * 64) (&lds.e_t[0][0])[e] = f2bf(0.f);
+    for (int e = tid; e < 16 * LD1; e += WAVES3 * 64)
+        (&lds.w2_lds[0][0])[e] = f2bf(0.f);
+    for (int e = tid; e < WAVES3 * MP * 40; e += WAVES3 * 64)
+        (&lds.a[0][0][0])[e] = f2bf(0.f);
+    for (int e = tid; e < WAVES3 * 64 * LD1; e += WAVES3 * 64)
+        (&lds.t1_t[0][0][0])[e] = f2bf(0.f);
+    for (int e = tid; e < 16; e += WAVES3 * 64) lds.b2s[e] = 0.f;
+    __syncthreads();
+    for (int e = tid; e < 12 * E; e += WAVES3 * 64) lds.e_t[e % E][e / E] = emb[e];
+    for (int e = tid; e < F2 * F1; e += WAVES3 * 64)
+        lds.w2_lds[e / F1][e % F1] = w2[e];
+    for (int e = tid; e < F1; e += WAVES3 * 64) lds.b1s[e] = b1[e];
+    for (int e = tid; e < F2; e += WAVES3 * 64) lds.b2s[e] = b2[e];
+    // private hot tiles: zero whole (incl. pad rows 12-15) once; the scatter
+    // epilogue below re-zeroes exactly the 200 entries it set
+    for (int e = tid; e < WAVES3 * 16 * KP_LD; e += WAVES3 * 64)
+        (&lds.hot_t[0][0][0])[e] = f2bf(0.f);
+    __syncthreads();  // the ONLY barrier (staging); columns are wave-private
+
+    bf16(&hot)[16][KP_LD] = lds.hot_t[wid];
+    bf16(&at)[MP][40] = lds.a[wid];
+    bf16(&t1)[64][LD1] = lds.t1_t[wid];
+    bf16* t2 = lds.t2[wid];
+    const uint8_t* win = ids + (size_t)b * R * W;
+
+    // wave w handles columns w, w+WAVES3, ... — each fully privately
+    for (int w = wid; w < W; w += WAVES3) {
+        // ---- scatter: 200 one-hots (ids read straight through L1) ---------
+        // lane r and r+64... handle reads r, r+64, r+128 (200 = 3*64 + 8)
+        uint8_t myid[4];
+        int nr = 0;
+#pragma unroll
+        for (int q = 0; q < 4; ++q) {
+            const int r = lane + q * 64;
+            if (r < R) myid[nr++] = win[(size_t)r * W + w];
+        }
+#pragma unroll
+        for (int q = 0; q < 4; ++q)
+            if (q < nr) hot[myid[q]][lane + q * 64] = f2bf(1.0f);
+
+        // ---- G1: A = W1 · Hot  (7 m-tiles x 7 k-steps, 49 MFMA) -----------
+        // 7 independent accumulator chains; B-frag per k shared across m.
+        {
+            f32x4 acc[7];
+#pragma unroll
+            for (int mt = 0; mt < 7; ++mt) acc[mt] = f32x4{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+            for (int kb = 0; kb < 7; ++kb) {
+                const bf16x8 bf_ = lds_load_b_frag_t(&hot[0][0], 0, kb * 32,
+                                                     KP_LD);
+#pragma unroll
+                for (int mt = 0; mt < 7; ++mt) {
+                    const bf16x8 af = global_load_a_frag(w1g, mt * 16, kb * 32,
+                                                         KP_LD);
+                    acc[mt] = mfma16x16x32(af, bf_, acc[mt]);
+                }
+            }
+            // epilogue: a[f][c] (c = lcol < 12 live, 12..15 zero products)
+#pragma unroll
+            for (int mt = 0; mt < 7; ++mt)
+#pragma unroll
+                for (int i = 0; i < 4; ++i)
+                    at[mt * 16 + lrow * 4 + i][lcol] = f2bf(acc[mt][i]);
+            // un-scatter the hot ones (cheaper than re-zeroing 7.4 KB)
+#pragma unroll
+            for (int q = 0; q < 4; ++q)
+                if (q < nr) hot[myid[q]][lane + q * 64] = f2bf(0.0f);
+        }
+
+        // ---- G2: t1 = relu(A · E + b1)  (7m x 4n single-K, 28 MFMA) -------
+        {
+            f32x4 acc[7][4];
+#pragma unroll
+            for (int mt = 0; mt < 7; ++mt) {
+                const bf16x8 af = lds_load_a_frag(&at[0][0], mt * 16, 0, 40);
+#pragma unroll
+                for (int nt = 0; nt < 4; ++nt) {
+                    const bf16x8 bf_ = lds_load_b_frag_t(&lds.e_t[0][0],
+                                                         nt * 16, 0, 32);
+                    acc[mt][nt] = mfma16x16x32(af, bf_,
+                                               f32x4{0.f, 0.f, 0.f, 0.f});
+                }
+            }
+#pragma unroll
+            for (int mt = 0; mt < 7; ++mt)
+#pragma unroll
+                for (int nt = 0; nt < 4; ++nt) {
+                    const int e = nt * 16 + lcol;
+                    if (e < E) {
+                        // f = mt*16 + lrow*4 + i contiguous over i: one b64
+                        bf16x4 pk;
+#pragma unroll
+                        for (int i = 0; i < 4; ++i) {
+                            const int f = mt * 16 + lrow * 4 + i;
+                            const float v =
+                                acc[mt][nt][i] + (f < F1 ? lds.b1s[f] : 0.f);
+                            pk[i] = f2bf(fmaxf(v, 0.f));
+                        }
+                        *reinterpret_cast<bf16x4*>(
+                            &t1[e][mt * 16 + lrow * 4]) = pk;
+                    }
+                }
+        }
+
+        // ---- G3: t2 = relu(W2 · t1 + b2)  (4 n-tiles x K=112, 16 MFMA) ----
+        {
+#pragma unroll
+            for (int nt = 0; nt < 4; ++nt) {
+                f32x4 acc0 = {0.f, 0.f, 0.f, 0.f}, acc1 = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+                for (int kb = 0; kb < 4; kb += 2) {
+                    acc0 = mfma16x16x32(
+                        lds_load_a_frag(&lds.w2_lds[0][0], 0, kb * 32, LD1),
+                        lds_load_b_frag_t(&t1[0][0], nt * 16, kb * 32, LD1),
+                        acc0);
+                    acc1 = mfma16x16x32(
+                        lds_load_a_frag(&lds.w2_lds[0][0], 0, (kb + 1) * 32,
+                                        LD1),
+                        lds_load_b_frag_t(&t1[0][0], nt * 16, (kb + 1) * 32,
+                                          LD1),
+                        acc1);
+                }
+                const int e = nt * 16 + lcol;
+                const int j = lrow * 4;  // rows j..j+3; only j<10 live
+                if (e < E) {
+#pragma unroll
+                    for (int i = 0; i < 4; ++i)
+                        if (j + i < F2) {
+                            const float v =
+                                acc0[i] + acc1[i] + lds.b2s[j + i];
+                            t2[e * F2 + j + i] = f2bf(fmaxf(v, 0.f));
+                        }
+                }
+            }
+        }
+
+        // ---- coalesced store (wave-wide b128) -----------------------------
+        {
+            bf16* dst = out + ((size_t)w * B + b) * OUT;
+            const int e8 = lane * 8;
+            if (e8 + 8 <= OUT)
+                *reinterpret_cast<bf16x8*>(dst + e8) =
+                    *reinterpret_cast<const bf16x8*>(&t2[e8]);
+            else if (e8 < OUT)
+                for (int q = e8; q < OUT; ++q) dst[q] = t2[q];
+        }
+    }
+}
+
+}  // namespace v3
+
+void embed_mlp_fwd3(const uint8_t* ids, const void* w1g, const float* b1,
+                    const void* w2, const float* b2, const void* emb,
+                    void* out, int B, hipStream_t stream) {
+    hipLaunchKernelGGL(v3::embed_mlp_fwd3_kernel, dim3(B),
+                       dim3(v3::WAVES3 * 64), 0, stream, ids,
+                       static_cast<const bf16*>(w1g), b1,
+                       static_cast<const bf16*>(w2), b2,
+                       static_cast<const bf16*>(emb), static_cast<bf16*>(out),
+                       B);
+}
+
 void embed_mlp_fwd2(const uint8_t* ids, const void* w1g, const float* b1,
                     const void* w2, const float* b2, const void* emb,
                     void* out, int B, hipStream_t stream) {

@@ -67,6 +67,34 @@ def test_embed_mlp_fwd_vs_torch():
 
 
 @requires_gpu
+@pytest.mark.parametrize("version", ["v2", "v3"])
+def test_embed_mlp_fwd_v2_v3_vs_torch(version):
+    """The chunked (v2) and wave-private-column (v3) eval fronts against the
+    fp32 torch reference, incl. the padded-batch path."""
+    torch.manual_seed(21)
+    m = RokoModel().eval()
+    B = 32
+    x = torch.randint(0, 12, (B, 200, 90))
+    with torch.no_grad():
+        e = m.embedding(x)
+        t = torch.relu(m.fc1(e.permute(0, 2, 3, 1)))
+        t = torch.relu(m.fc2(t))
+        ref = t.reshape(B, 90, 500).transpose(0, 1)
+
+    m = m.cuda()
+    w = fwd._bf16_weights(m)
+    ext = ops.ext()
+    fn = ext.embed_mlp_fwd2 if version == "v2" else ext.embed_mlp_fwd3
+    out = fn(x.to(torch.uint8).cuda(), w["w1g"], w["b1"], w["w2"], w["b2"],
+             w["emb"])
+    got = out.float().cpu()
+    err = (got - ref).abs()
+    scale = ref.abs().mean().item() + 1e-6
+    assert err.max().item() < 0.08, err.max().item()
+    assert err.mean().item() / scale < 0.02
+
+
+@requires_gpu
 def test_gru_layer_fwd_vs_torch():
     torch.manual_seed(2)
     T, B, H = 90, 32, 128
