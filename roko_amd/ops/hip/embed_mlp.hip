@@ -538,20 +538,35 @@ __global__ __launch_bounds__(WAVES3 * 64, 1) void embed_mlp_fwd3_kernel(
 
         // ---- G1: A = W1 · Hot  (7 m-tiles x 7 k-steps, 49 MFMA) -----------
         // 7 independent accumulator chains; B-frag per k shared across m.
+        // Software-pipelined one k-step ahead: the L2 A-fragment loads for
+        // kb+1 issue before kb's MFMAs, so their ~200-cycle L2 latency hides
+        // under 7 MFMAs + the LDS B read instead of stalling each chain.
         {
             f32x4 acc[7];
 #pragma unroll
             for (int mt = 0; mt < 7; ++mt) acc[mt] = f32x4{0.f, 0.f, 0.f, 0.f};
+            bf16x8 afA[7], afB[7];
+            bf16x8 bfA, bfB;
+#pragma unroll
+            for (int mt = 0; mt < 7; ++mt)
+                afA[mt] = global_load_a_frag(w1g, mt * 16, 0, KP_LD);
+            bfA = lds_load_b_frag_t(&hot[0][0], 0, 0, KP_LD);
 #pragma unroll
             for (int kb = 0; kb < 7; ++kb) {
-                const bf16x8 bf_ = lds_load_b_frag_t(&hot[0][0], 0, kb * 32,
-                                                     KP_LD);
+                bf16x8(&cur)[7] = (kb & 1) ? afB : afA;
+                bf16x8(&nxt)[7] = (kb & 1) ? afA : afB;
+                const bf16x8 bf_ = (kb & 1) ? bfB : bfA;
+                if (kb + 1 < 7) {
 #pragma unroll
-                for (int mt = 0; mt < 7; ++mt) {
-                    const bf16x8 af = global_load_a_frag(w1g, mt * 16, kb * 32,
-                                                         KP_LD);
-                    acc[mt] = mfma16x16x32(af, bf_, acc[mt]);
+                    for (int mt = 0; mt < 7; ++mt)
+                        nxt[mt] = global_load_a_frag(w1g, mt * 16,
+                                                     (kb + 1) * 32, KP_LD);
+                    ((kb & 1) ? bfA : bfB) = lds_load_b_frag_t(
+                        &hot[0][0], 0, (kb + 1) * 32, KP_LD);
                 }
+#pragma unroll
+                for (int mt = 0; mt < 7; ++mt)
+                    acc[mt] = mfma16x16x32(cur[mt], bf_, acc[mt]);
             }
             // epilogue: a[f][c] (c = lcol < 12 live, 12..15 zero products)
 #pragma unroll

@@ -174,35 +174,75 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gru_layer_fwd_kernel(
         }
 
         // fused gate math; updates the fp32 register hidden state and writes
-        // the bf16 mirror into the BACK h buffer (no reader until barrier)
+        // the bf16 mirror into the BACK h buffer (no reader until barrier).
+        // Structured as: batch ALL 24 xg reads first, then run the 8
+        // elements' serial exp/rcp chains as INDEPENDENT interleavable
+        // streams — the original per-element form (reads + chain + a dbg
+        // branch per element) compiled to 16 fenced basic blocks whose
+        // ~200-cycle dependency chains executed back to back: the gate
+        // phase alone was 55% of the kernel (scripts/gru_timing.py).
+        if (dbg & 4u) {  // timing: gate math stripped (separate cold loop)
 #pragma unroll
-        for (int mt = 0; mt < 2; ++mt) {
+            for (int mt = 0; mt < 2; ++mt)
 #pragma unroll
-            for (int i = 0; i < 4; ++i) {
-                const int row = mt * 16 + lrow * 4 + i;
-                const int j = j0 + lcol;
-                float hnew, r, z, n, hgn;
-                if (dbg & 4u) {  // timing: gate math stripped
-                    hnew = acc[mt][0][i] + acc[mt][1][i] + acc[mt][2][i];
-                    r = z = n = hgn = hnew;
-                } else {
-                    const float xr = bf2f(lds.xgb[curp][row][0 * H + j]);
-                    const float xz = bf2f(lds.xgb[curp][row][1 * H + j]);
-                    const float xn = bf2f(lds.xgb[curp][row][2 * H + j]);
-                    hgn = acc[mt][2][i] + bhh_reg[2];
-                    r = sigmoidf_dev(xr + acc[mt][0][i] + bhh_reg[0]);
-                    z = sigmoidf_dev(xz + acc[mt][1][i] + bhh_reg[1]);
-                    n = tanhf_dev(xn + r * hgn);
-                    hnew = (1.0f - z) * n + z * hreg[mt][i];
+                for (int i = 0; i < 4; ++i) {
+                    const int row = mt * 16 + lrow * 4 + i;
+                    const int j = j0 + lcol;
+                    const float hnew =
+                        acc[mt][0][i] + acc[mt][1][i] + acc[mt][2][i];
+                    hreg[mt][i] = hnew;
+                    lds.h[curp ^ 1][row][j] = f2bf(hnew);
+                    if constexpr (TRAIN) {
+                        bf16x4 pk = {f2bf(hnew), f2bf(hnew), f2bf(hnew),
+                                     f2bf(hnew)};
+                        *reinterpret_cast<bf16x4*>(
+                            &lds.cache_st[curp ^ 1][row][4 * j]) = pk;
+                    }
                 }
-                hreg[mt][i] = hnew;
-                lds.h[curp ^ 1][row][j] = f2bf(hnew);
-                if constexpr (TRAIN) {
-                    bf16x4 pk = {f2bf(r), f2bf(z), f2bf(n), f2bf(hgn)};
-                    *reinterpret_cast<bf16x4*>(
-                        &lds.cache_st[curp ^ 1][row][4 * j]) = pk;
+        } else {
+            const int j = j0 + lcol;
+            float xr[2][4], xz[2][4], xn[2][4];
+#pragma unroll
+            for (int mt = 0; mt < 2; ++mt)
+#pragma unroll
+                for (int i = 0; i < 4; ++i) {
+                    const int row = mt * 16 + lrow * 4 + i;
+                    xr[mt][i] = bf2f(lds.xgb[curp][row][0 * H + j]);
+                    xz[mt][i] = bf2f(lds.xgb[curp][row][1 * H + j]);
+                    xn[mt][i] = bf2f(lds.xgb[curp][row][2 * H + j]);
                 }
-            }
+            float r8[2][4], z8[2][4], n8[2][4], hg8[2][4];
+#pragma unroll
+            for (int mt = 0; mt < 2; ++mt)
+#pragma unroll
+                for (int i = 0; i < 4; ++i) {
+                    hg8[mt][i] = acc[mt][2][i] + bhh_reg[2];
+                    r8[mt][i] =
+                        sigmoidf_dev(xr[mt][i] + acc[mt][0][i] + bhh_reg[0]);
+                    z8[mt][i] =
+                        sigmoidf_dev(xz[mt][i] + acc[mt][1][i] + bhh_reg[1]);
+                }
+#pragma unroll
+            for (int mt = 0; mt < 2; ++mt)
+#pragma unroll
+                for (int i = 0; i < 4; ++i)
+                    n8[mt][i] = tanhf_dev(xn[mt][i] + r8[mt][i] * hg8[mt][i]);
+#pragma unroll
+            for (int mt = 0; mt < 2; ++mt)
+#pragma unroll
+                for (int i = 0; i < 4; ++i) {
+                    const int row = mt * 16 + lrow * 4 + i;
+                    const float hnew = (1.0f - z8[mt][i]) * n8[mt][i] +
+                                       z8[mt][i] * hreg[mt][i];
+                    hreg[mt][i] = hnew;
+                    lds.h[curp ^ 1][row][j] = f2bf(hnew);
+                    if constexpr (TRAIN) {
+                        bf16x4 pk = {f2bf(r8[mt][i]), f2bf(z8[mt][i]),
+                                     f2bf(n8[mt][i]), f2bf(hg8[mt][i])};
+                        *reinterpret_cast<bf16x4*>(
+                            &lds.cache_st[curp ^ 1][row][4 * j]) = pk;
+                    }
+                }
         }
         __syncthreads();  // h[curp^1] (+ cache staging) published
 
@@ -367,37 +407,51 @@ __global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
         if (sidx + 1 < T) write_stage(curp ^ 1, rc_wr, rdh_wr, rhp_wr);
 
         // ---- gate gradients from stage[curp] ------------------------------
+        // reads batched first, then 8 independent arithmetic chains, then
+        // the stores — same restructure as the forward's gate phase (the
+        // interleaved per-element form serialized the read latencies)
         float dhp_part[2][4];
+        {
+            const int j = j0 + lcol;
+            float dh8[2][4], hp8[2][4];
+            bf16x4 pk8[2][4];
 #pragma unroll
-        for (int mt = 0; mt < 2; ++mt) {
+            for (int mt = 0; mt < 2; ++mt)
 #pragma unroll
-            for (int i = 0; i < 4; ++i) {
-                const int row = mt * 16 + lrow * 4 + i;
-                const int j = j0 + lcol;
-                const float dh = dhc[mt][i] + bf2f(lds.dhin_st[curp][row][j]);
-                const bf16x4 pk = *reinterpret_cast<const bf16x4*>(
-                    &lds.cache_st[curp][row][4 * j]);
-                const float r = bf2f(pk[0]);
-                const float z = bf2f(pk[1]);
-                const float n = bf2f(pk[2]);
-                const float hgn = bf2f(pk[3]);
-                const float hp = bf2f(lds.hprev_st[curp][row][j]);
-                const float dn = dh * (1.0f - z);
-                const float dz = dh * (hp - n);
-                const float dan = dn * (1.0f - n * n);
-                const float dxn = dan;
-                const float dhgn = dan * r;
-                const float dxr = dan * hgn * r * (1.0f - r);
-                const float dxz = dz * z * (1.0f - z);
-                dhp_part[mt][i] = dh * z;
-                lds.dhg[row][0 * H + j] = f2bf(dxr);
-                lds.dhg[row][1 * H + j] = f2bf(dxz);
-                lds.dhg[row][2 * H + j] = f2bf(dhgn);
-                lds.dg_st[row][0 * H + j] = f2bf(dxr);
-                lds.dg_st[row][1 * H + j] = f2bf(dxz);
-                lds.dg_st[row][2 * H + j] = f2bf(dxn);
-                lds.dg_st[row][3 * H + j] = f2bf(dhgn);
-            }
+                for (int i = 0; i < 4; ++i) {
+                    const int row = mt * 16 + lrow * 4 + i;
+                    dh8[mt][i] =
+                        dhc[mt][i] + bf2f(lds.dhin_st[curp][row][j]);
+                    pk8[mt][i] = *reinterpret_cast<const bf16x4*>(
+                        &lds.cache_st[curp][row][4 * j]);
+                    hp8[mt][i] = bf2f(lds.hprev_st[curp][row][j]);
+                }
+#pragma unroll
+            for (int mt = 0; mt < 2; ++mt)
+#pragma unroll
+                for (int i = 0; i < 4; ++i) {
+                    const int row = mt * 16 + lrow * 4 + i;
+                    const float dh = dh8[mt][i];
+                    const float r = bf2f(pk8[mt][i][0]);
+                    const float z = bf2f(pk8[mt][i][1]);
+                    const float n = bf2f(pk8[mt][i][2]);
+                    const float hgn = bf2f(pk8[mt][i][3]);
+                    const float dn = dh * (1.0f - z);
+                    const float dz = dh * (hp8[mt][i] - n);
+                    const float dan = dn * (1.0f - n * n);
+                    const float dxn = dan;
+                    const float dhgn = dan * r;
+                    const float dxr = dan * hgn * r * (1.0f - r);
+                    const float dxz = dz * z * (1.0f - z);
+                    dhp_part[mt][i] = dh * z;
+                    lds.dhg[row][0 * H + j] = f2bf(dxr);
+                    lds.dhg[row][1 * H + j] = f2bf(dxz);
+                    lds.dhg[row][2 * H + j] = f2bf(dhgn);
+                    lds.dg_st[row][0 * H + j] = f2bf(dxr);
+                    lds.dg_st[row][1 * H + j] = f2bf(dxz);
+                    lds.dg_st[row][2 * H + j] = f2bf(dxn);
+                    lds.dg_st[row][3 * H + j] = f2bf(dhgn);
+                }
         }
         __syncthreads();  // dhg/dg_st and stage[curp^1] published
 
