@@ -78,9 +78,16 @@ def _bf16_weights(model) -> dict:
         bf = getattr(g, f"bias_ih_l{l}").detach()
         br = getattr(g, f"bias_ih_l{l}_reverse").detach()
         # (in, 768) so xg = x @ w_ih_t is one GEMM covering both directions
-        c[f"w_ih_t{l}"] = (
-            torch.cat([wf, wr], dim=0).to(torch.bfloat16).t().contiguous()
-        )
+        w_cat = torch.cat([wf, wr], dim=0).to(torch.bfloat16)
+        c[f"w_ih_t{l}"] = w_cat.t().contiguous()
+        # (768, KP) row-padded copy for the in-kernel xg GEMM (KP multiple of
+        # 32 and 16-byte-aligned rows; zero pad cols are multiplied by the
+        # zero-padded x stage)
+        kin = w_cat.shape[1]
+        kp = 512 if kin > 256 else 256
+        wp = torch.zeros(768, kp, dtype=torch.bfloat16, device=w_cat.device)
+        wp[:, :kin] = w_cat
+        c[f"w_ih_p{l}"] = wp.contiguous()
         c[f"b_ih{l}"] = torch.cat([bf, br]).to(torch.bfloat16).contiguous()
         c[f"u{l}"] = torch.stack(
             [getattr(g, f"weight_hh_l{l}").detach(),

@@ -25,6 +25,10 @@ void embed_mlp_fwd3(const uint8_t* ids, const void* w1g, const float* b1,
                     void* out, int B, hipStream_t stream);
 void gru_layer_fwd(const void* xg, const void* u, const float* bhh, void* hseq,
                    void* cache, int T, int B, hipStream_t stream, uint32_t dbg);
+void gru_layer_fwd_fused(const void* x, const void* w_ih_p, const void* b_ih,
+                         const void* u, const float* bhh, void* xg_ws,
+                         void* hseq, int T, int B, int IN, int KP,
+                         hipStream_t stream);
 void gru_layer_bwd(const void* cache, const void* hseq, const void* dhin,
                    const void* ut, void* dxg, void* dhg, int T, int B,
                    hipStream_t stream);
@@ -154,6 +158,31 @@ torch::Tensor embed_mlp_fwd3(torch::Tensor ids, torch::Tensor w1g,
                        b2.data_ptr<float>(), emb.data_ptr(), out.data_ptr(),
                        B, cur_stream());
     return out;
+}
+
+// Serving fused variant: x (T, B, IN) bf16 + row-padded W_ih (768, KP) +
+// b_ih (768) bf16 -> hseq; the xg GEMM runs inside the kernel.
+torch::Tensor gru_layer_fused(torch::Tensor x, torch::Tensor w_ih_p,
+                              torch::Tensor b_ih, torch::Tensor u,
+                              torch::Tensor bhh) {
+    check(x, torch::kBFloat16, "x");
+    check(w_ih_p, torch::kBFloat16, "w_ih_p");
+    check(b_ih, torch::kBFloat16, "b_ih");
+    check(u, torch::kBFloat16, "u");
+    check(bhh, torch::kFloat32, "bhh");
+    const int T = x.size(0), B = x.size(1), IN = x.size(2);
+    const int KP = w_ih_p.size(1);
+    TORCH_CHECK(w_ih_p.size(0) == 768 && KP % 32 == 0 && KP >= IN && KP <= 512,
+                "w_ih_p must be (768, KP<=512), KP %% 32 == 0");
+    TORCH_CHECK(IN % 4 == 0, "IN must be a multiple of 4");
+    TORCH_CHECK(B % 32 == 0, "batch must be a multiple of 32 (pad on host)");
+    auto xg_ws = torch::empty({T, B, 2, 384}, x.options());
+    auto hseq = torch::empty({T, B, 2, 128}, x.options());
+    rk::gru_layer_fwd_fused(x.data_ptr(), w_ih_p.data_ptr(), b_ih.data_ptr(),
+                            u.data_ptr(), bhh.data_ptr<float>(),
+                            xg_ws.data_ptr(), hseq.data_ptr(), T, B, IN, KP,
+                            cur_stream());
+    return hseq;
 }
 
 // xg (T, B, 2, 384) bf16, u (2, 384, 128) bf16, bhh (2, 384) f32
@@ -443,6 +472,8 @@ struct ServeSlot {
     torch::Tensor w1, b1, w2, b2, emb, w4, b4;
     torch::Tensor w1gt;  // (112,232) zero-padded W1 for the chunked front
     std::vector<torch::Tensor> w_ih_t, b_ih, u, bhh;
+    std::vector<torch::Tensor> w_ih_p;  // (768, KP) row-padded, xg-fold path
+    torch::Tensor hseq2;                // ping-pong buffer for the fold path
     // slot state + workspaces
     torch::Tensor x_buf;     // (B, 200, 90) u8 static input
     torch::Tensor seq;       // (90, B, 500) bf16 front output
@@ -459,6 +490,11 @@ struct ServeSlot {
     bool use_v3 = [] {
         const char* f = getenv("ROKO_FRONT");
         return f && std::string(f) == "v3";
+    }();
+    // xg-GEMM fold into the GRU kernel (ROKO_XGFOLD=1 to enable for A/B)
+    bool use_fold = [] {
+        const char* f = getenv("ROKO_XGFOLD");
+        return f && std::string(f) == "1";
     }();
 
     ServeSlot(py::dict w, int B_, torch::Tensor host_out_)
@@ -478,6 +514,12 @@ struct ServeSlot {
             b_ih.push_back(need(("b_ih" + sfx).c_str()));
             u.push_back(need(("u" + sfx).c_str()));
             bhh.push_back(need(("bhh" + sfx).c_str()));
+            if (w.contains(("w_ih_p" + sfx).c_str()))
+                w_ih_p.push_back(need(("w_ih_p" + sfx).c_str()));
+        }
+        if (w_ih_p.size() != 3) {
+            w_ih_p.clear();
+            use_fold = false;
         }
         TORCH_CHECK(B % 32 == 0, "serving batch must be a multiple of 32");
         TORCH_CHECK(host_out_.is_pinned() && host_out_.scalar_type() == t::kUInt8
@@ -489,6 +531,7 @@ struct ServeSlot {
         seq = t::empty({90, B, 500}, dev.dtype(t::kBFloat16));
         xg = t::empty({(int64_t)90 * B, 768}, dev.dtype(t::kBFloat16));
         hseq = t::empty({90, B, 2, 128}, dev.dtype(t::kBFloat16));
+        if (use_fold) hseq2 = t::empty({90, B, 2, 128}, dev.dtype(t::kBFloat16));
         amax = t::empty({B, 90}, dev.dtype(t::kUInt8));
         seq2d = seq.view({(int64_t)90 * B, 500});
         hseq2d = hseq.view({(int64_t)90 * B, 256});
@@ -534,12 +577,30 @@ struct ServeSlot {
                                   b1.data_ptr<float>(), w2.data_ptr(),
                                   b2.data_ptr<float>(), emb.data_ptr(),
                                   seq.data_ptr(), B, s, 0, nullptr);
-            for (int l = 0; l < 3; ++l) {
-                at::addmm_out(xg2d, b_ih[l], l == 0 ? seq2d : hseq2d,
-                              w_ih_t[l]);
-                rk::gru_layer_fwd(xg.data_ptr(), u[l].data_ptr(),
-                                  bhh[l].data_ptr<float>(), hseq.data_ptr(),
-                                  nullptr, 90, B, s, 0);
+            if (use_fold) {
+                // xg GEMM folded into the GRU kernel; layers ping-pong
+                // between hseq buffers (phase 1 of layer l reads the other
+                // dir's workgroup output, so in-place would race)
+                const void* xin = seq.data_ptr();
+                int in_dim = 500;
+                for (int l = 0; l < 3; ++l) {
+                    torch::Tensor& out_t = (l % 2) ? hseq2 : hseq;
+                    rk::gru_layer_fwd_fused(
+                        xin, w_ih_p[l].data_ptr(), b_ih[l].data_ptr(),
+                        u[l].data_ptr(), bhh[l].data_ptr<float>(),
+                        xg.data_ptr(), out_t.data_ptr(), 90, B, in_dim,
+                        (int)w_ih_p[l].size(1), s);
+                    xin = out_t.data_ptr();
+                    in_dim = 256;
+                }
+            } else {
+                for (int l = 0; l < 3; ++l) {
+                    at::addmm_out(xg2d, b_ih[l], l == 0 ? seq2d : hseq2d,
+                                  w_ih_t[l]);
+                    rk::gru_layer_fwd(xg.data_ptr(), u[l].data_ptr(),
+                                      bhh[l].data_ptr<float>(), hseq.data_ptr(),
+                                      nullptr, 90, B, s, 0);
+                }
             }
             rk::head_fwd(hseq.data_ptr(), w4.data_ptr(), b4.data_ptr<float>(),
                          nullptr, amax.data_ptr<uint8_t>(), 90, B, s);
@@ -603,6 +664,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("dbg") = 0, py::arg("timing") = c10::nullopt);
     m.def("gru_layer_fwd", &gru_layer_fwd, py::arg("xg"), py::arg("u"),
           py::arg("bhh"), py::arg("train") = false, py::arg("dbg") = 0);
+    m.def("gru_layer_fused", &gru_layer_fused, py::arg("x"), py::arg("w_ih_p"),
+          py::arg("b_ih"), py::arg("u"), py::arg("bhh"));
     m.def("gru_layer_bwd", &gru_layer_bwd);
     m.def("ce_fwd_bwd", &ce_fwd_bwd);
     m.def("adam_step", &adam_step, py::arg("p"), py::arg("g"),

@@ -45,20 +45,39 @@ constexpr int WAVES = 8;     // 512 threads, two waves per SIMD: a wave's
 constexpr int HPAD = H + 8;  // LDS row padding (bank-conflict fix)
 constexpr int XCH = MB * G3 / (WAVES * 64 * 8);  // xg chunks per thread (3)
 
-template <bool TRAIN>
+constexpr int XKP = 512;        // max padded input width for the fused xg GEMM
+constexpr int XLD = XKP + 8;    // x-stage LDS row stride (bank pad)
+
+// FUSEXG (serving variant): the input-projection GEMM xg = x·W_ihᵀ + b_ih
+// runs as PHASE 1 of this kernel instead of a separate hipBLASLt launch.
+// Rationale (profiles/infer_r02_v3_kernel_stats.csv): hipBLASLt runs the
+// (11520, 768, K<=500) GEMM at ~543 TF/s on ~225 workgroups — ~105 CU·us
+// per window of aggregate chip time, tied with the front as the top serving
+// cost. Here each of the 8 workgroups computes its OWN (90x32, 384) slice
+// with the W slice L2-resident and reused across all 90 steps, at a small
+// fraction of the aggregate CU time; the per-chain latency rises (~+60 us),
+// which deep pipelining absorbs. Not used by training: there the step's
+// LATENCY is the metric and 225 workgroups beat 8.
+template <bool TRAIN, bool FUSEXG = false>
 __global__ __launch_bounds__(WAVES * 64, 2) void gru_layer_fwd_kernel(
     const bf16* __restrict__ xg,   // (T, B, 2, 3H)  W_ih·x + b_ih
+                                   // (FUSEXG: workspace this kernel fills)
     const bf16* __restrict__ u,    // (2, 3H, H)     weight_hh
     const float* __restrict__ bhh, // (2, 3H)        bias_hh
     bf16* __restrict__ hseq,       // (T, B, 2, H)   output
     bf16* __restrict__ cache,      // (T, B, 2, H, 4) [r z n hgn] or nullptr
     int T, int B,
-    uint32_t dbg) {  // timing bisection: 1 no hseq store, 2 no cache store,
-                     // 4 no gate VALU, 8 no xg staging, 16 no MFMA
+    uint32_t dbg,   // timing bisection: 1 no hseq store, 2 no cache store,
+                    // 4 no gate VALU, 8 no xg staging, 16 no MFMA
+    const bf16* __restrict__ x = nullptr,     // (T, B, IN) layer input
+    const bf16* __restrict__ w_ih_p = nullptr,  // (768, KP) row-padded W_ih
+    const bf16* __restrict__ b_ih = nullptr,   // (768)
+    int IN = 0, int KP = 0) {
     __shared__ struct {
         bf16 h[2][MB][HPAD];         // double-buffered hidden-state mirror
         bf16 xgb[2][MB][G3];         // double-buffered step gate inputs
         bf16 cache_st[TRAIN ? 2 : 1][TRAIN ? MB : 1][TRAIN ? 4 * H : 1];
+        bf16 x_st[FUSEXG ? 2 : 1][FUSEXG ? MB : 1][FUSEXG ? XLD : 1];
     } lds;
 
     const int dir = blockIdx.y;
@@ -69,6 +88,79 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gru_layer_fwd_kernel(
     const int j0 = wid * 16;         // this wave's 16 hidden columns
     const int lrow = lane >> 4;      // fragment row group (0..3)
     const int lcol = lane & 15;      // fragment column
+
+    // ---- PHASE 1 (FUSEXG): xg[t, b0:b0+MB, dir, :] = x·Wᵀ + b_ih ----------
+    if constexpr (FUSEXG) {
+        // x[t] tile (MB, KP) double-buffered through LDS; W B-fragments read
+        // straight from L2 ((768, KP) row-major = [col][k], so the B-frag
+        // load is the A-frag address pattern). Wave owns 3 n-tiles
+        // (nt = wid + 8s) x 2 m-tiles; one barrier per t.
+        const int KB = KP / 32;
+        auto stage_x = [&](int t, int buf) {
+            const bf16* src = x + ((size_t)t * B + b0) * IN;
+            for (int c = tid; c < MB * KP / 4; c += WAVES * 64) {
+                const int row = (c * 4) / KP, col = (c * 4) % KP;
+                uint64_t v = 0;  // 4 bf16 (zero pads cols IN..KP)
+                if (col + 4 <= IN)
+                    v = *reinterpret_cast<const uint64_t*>(
+                        src + (size_t)row * IN + col);
+                *reinterpret_cast<uint64_t*>(&lds.x_st[buf][row][col]) = v;
+            }
+        };
+        float bih_reg[3][4];  // [s][i of 4-row group]? bias is per-COLUMN
+#pragma unroll
+        for (int s = 0; s < 3; ++s) {
+            const int col = (wid + 8 * s) * 16 + lcol;
+            const float bv = b_ih ? bf2f(b_ih[dir * G3 + col]) : 0.f;
+#pragma unroll
+            for (int i = 0; i < 4; ++i) bih_reg[s][i] = bv;
+        }
+        stage_x(0, 0);
+        __syncthreads();
+        for (int t = 0; t < T; ++t) {
+            if (t + 1 < T) stage_x(t + 1, (t + 1) & 1);
+            f32x4 acc[3][2];
+#pragma unroll
+            for (int s = 0; s < 3; ++s)
+#pragma unroll
+                for (int mt = 0; mt < 2; ++mt)
+                    acc[s][mt] = f32x4{0.f, 0.f, 0.f, 0.f};
+            for (int kb = 0; kb < KB; ++kb) {
+                bf16x8 a0 = lds_load_a_frag(&lds.x_st[t & 1][0][0], 0,
+                                            kb * 32, XLD);
+                bf16x8 a1 = lds_load_a_frag(&lds.x_st[t & 1][0][0], 16,
+                                            kb * 32, XLD);
+#pragma unroll
+                for (int s = 0; s < 3; ++s) {
+                    const int col0 = dir * G3 + (wid + 8 * s) * 16;
+                    const bf16x8 w = global_load_a_frag(w_ih_p, col0,
+                                                        kb * 32, KP);
+                    acc[s][0] = mfma16x16x32(a0, w, acc[s][0]);
+                    acc[s][1] = mfma16x16x32(a1, w, acc[s][1]);
+                }
+            }
+            // epilogue: + b_ih, scalar bf16 stores into the xg workspace
+            // (xg is a mutable workspace in the FUSEXG instantiation)
+            bf16* dst = const_cast<bf16*>(xg) +
+                        (((size_t)t * B + b0) * 2 + dir) * G3;
+#pragma unroll
+            for (int s = 0; s < 3; ++s) {
+                const int col = (wid + 8 * s) * 16 + lcol;
+#pragma unroll
+                for (int mt = 0; mt < 2; ++mt)
+#pragma unroll
+                    for (int i = 0; i < 4; ++i) {
+                        const int row = mt * 16 + lrow * 4 + i;
+                        dst[(size_t)row * 2 * G3 + col] =
+                            f2bf(acc[s][mt][i] + bih_reg[s][i]);
+                    }
+            }
+            __syncthreads();
+        }
+        // all xg stores must land before phase 2 reads them back
+        __builtin_amdgcn_s_waitcnt(0);  // vmcnt(0) & lgkmcnt(0)
+        __syncthreads();
+    }
 
     // ---- load U fragments (kept in registers for all T steps) -------------
     // B-fragment for gates = h·U^T: B[k][col] = U[gate*H + j0 + 16ct + col][k]
@@ -283,14 +375,34 @@ void gru_layer_fwd(const void* xg, const void* u, const float* bhh, void* hseq,
     dim3 grid(B / MB, 2);
     dim3 block(WAVES * 64);
     if (cache)
-        hipLaunchKernelGGL(gru_layer_fwd_kernel<true>, grid, block, 0, stream,
+        hipLaunchKernelGGL((gru_layer_fwd_kernel<true, false>), grid, block, 0,
+                           stream,
                            static_cast<const bf16*>(xg), static_cast<const bf16*>(u),
                            bhh, static_cast<bf16*>(hseq), static_cast<bf16*>(cache),
-                           T, B, dbg);
+                           T, B, dbg, nullptr, nullptr, nullptr, 0, 0);
     else
-        hipLaunchKernelGGL(gru_layer_fwd_kernel<false>, grid, block, 0, stream,
+        hipLaunchKernelGGL((gru_layer_fwd_kernel<false, false>), grid, block, 0,
+                           stream,
                            static_cast<const bf16*>(xg), static_cast<const bf16*>(u),
-                           bhh, static_cast<bf16*>(hseq), nullptr, T, B, dbg);
+                           bhh, static_cast<bf16*>(hseq), nullptr, T, B, dbg,
+                           nullptr, nullptr, nullptr, 0, 0);
+}
+
+// serving variant: xg computed in-kernel from (x, W_ih, b_ih); `xg` is a
+// (T, B, 2, 3H) workspace
+void gru_layer_fwd_fused(const void* x, const void* w_ih_p, const void* b_ih,
+                         const void* u, const float* bhh, void* xg_ws,
+                         void* hseq, int T, int B, int IN, int KP,
+                         hipStream_t stream) {
+    dim3 grid(B / MB, 2);
+    dim3 block(WAVES * 64);
+    hipLaunchKernelGGL((gru_layer_fwd_kernel<false, true>), grid, block, 0,
+                       stream, static_cast<bf16*>(xg_ws),
+                       static_cast<const bf16*>(u), static_cast<const float*>(bhh),
+                       static_cast<bf16*>(hseq), nullptr, T, B, 0u,
+                       static_cast<const bf16*>(x),
+                       static_cast<const bf16*>(w_ih_p),
+                       static_cast<const bf16*>(b_ih), IN, KP);
 }
 
 // ---------------------------------------------------------------------------

@@ -365,6 +365,31 @@ def test_gemm_bias_vs_addmm():
 
 
 @requires_gpu
+def test_gru_layer_fused_matches_split():
+    """The serving fold (xg GEMM inside the GRU kernel) must match the
+    split path (hipBLASLt addmm + gru_layer_fwd) for all three layer
+    shapes within bf16 accumulation-order noise."""
+    ext = ops.ext()
+    torch.manual_seed(31)
+    m = RokoModel().cuda().eval()
+    w = fwd._bf16_weights(m)
+    T, B = 90, 64
+    x = (torch.randn(T, B, 500, device="cuda") * 0.4).to(torch.bfloat16)
+    seq = x
+    for l in range(3):
+        xg = torch.addmm(
+            w[f"b_ih{l}"], seq.reshape(T * B, -1), w[f"w_ih_t{l}"]
+        ).view(T, B, 2, 384)
+        (ref,) = ext.gru_layer_fwd(xg.contiguous(), w[f"u{l}"], w[f"bhh{l}"],
+                                   False)
+        got = ext.gru_layer_fused(seq.contiguous(), w[f"w_ih_p{l}"],
+                                  w[f"b_ih{l}"], w[f"u{l}"], w[f"bhh{l}"])
+        d = (got.float() - ref.float()).abs()
+        assert d.max().item() < 0.03, (l, d.max().item())
+        seq = ref.view(T, B, 256)
+
+
+@requires_gpu
 def test_atb_splitk_vs_mm():
     """Split-K A^T·B kernel vs hipBLASLt for the weight-grad shapes."""
     ext = ops.ext()
