@@ -17,10 +17,10 @@ from .. import config as C
 
 #: eval front kernel: the one-hot embed+MLP (factors the read reduction
 #: through the 12 base classes — ~2.3x fewer MFMAs per column) does ~1.45x
-#: less chip-work than the shared train-front kernel; under a deep serving
-#: pipeline the front is throughput-bound, so the one-hot kernel is the
-#: default. ROKO_FRONT=shared switches back for A/B runs.
-_FRONT = os.environ.get("ROKO_FRONT", "v2")
+#: less chip-work than the shared train-front kernel; v3 (wave-private
+#: columns, no per-phase barriers) is the measured-fastest default
+#: (106.7 us vs v2's 152.4 at b=128). ROKO_FRONT=v2/v1/shared for A/B.
+_FRONT = os.environ.get("ROKO_FRONT", "v3")
 
 
 def _front_eval(ext, ids, w):

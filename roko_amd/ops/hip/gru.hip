@@ -58,7 +58,8 @@ constexpr int XLD = XKP + 8;    // x-stage LDS row stride (bank pad)
 // fraction of the aggregate CU time; the per-chain latency rises (~+60 us),
 // which deep pipelining absorbs. Not used by training: there the step's
 // LATENCY is the metric and 225 workgroups beat 8.
-template <bool TRAIN, bool FUSEXG = false>
+template <bool TRAIN, int FUSEKP = 0>  // FUSEKP: 0 = xg precomputed;
+                                       // 256/512 = in-kernel xg GEMM width
 __global__ __launch_bounds__(WAVES * 64, 2) void gru_layer_fwd_kernel(
     const bf16* __restrict__ xg,   // (T, B, 2, 3H)  W_ih·x + b_ih
                                    // (FUSEXG: workspace this kernel fills)
@@ -72,12 +73,14 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gru_layer_fwd_kernel(
     const bf16* __restrict__ x = nullptr,     // (T, B, IN) layer input
     const bf16* __restrict__ w_ih_p = nullptr,  // (768, KP) row-padded W_ih
     const bf16* __restrict__ b_ih = nullptr,   // (768)
-    int IN = 0, int KP = 0) {
+    int IN = 0) {
+    constexpr bool FUSEXG = FUSEKP > 0;
+    constexpr int XLDK = FUSEXG ? FUSEKP + 8 : 1;
     __shared__ struct {
         bf16 h[2][MB][HPAD];         // double-buffered hidden-state mirror
         bf16 xgb[2][MB][G3];         // double-buffered step gate inputs
         bf16 cache_st[TRAIN ? 2 : 1][TRAIN ? MB : 1][TRAIN ? 4 * H : 1];
-        bf16 x_st[FUSEXG ? 2 : 1][FUSEXG ? MB : 1][FUSEXG ? XLD : 1];
+        bf16 x_st[FUSEXG ? 2 : 1][FUSEXG ? MB : 1][XLDK];
     } lds;
 
     const int dir = blockIdx.y;
@@ -94,12 +97,17 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gru_layer_fwd_kernel(
         // x[t] tile (MB, KP) double-buffered through LDS; W B-fragments read
         // straight from L2 ((768, KP) row-major = [col][k], so the B-frag
         // load is the A-frag address pattern). Wave owns 3 n-tiles
-        // (nt = wid + 8s) x 2 m-tiles; one barrier per t.
-        const int KB = KP / 32;
+        // (nt = wid + 8s) x 2 m-tiles; one barrier per t. KP is a template
+        // constant: the k-loop fully unrolls and the W/A fragment loads for
+        // kb+1 issue under kb's MFMAs (a runtime trip count compiled to a
+        // serial load->mfma->load chain — 1.47 ms/layer vs ~210 split).
+        constexpr int KB = FUSEKP / 32;
         auto stage_x = [&](int t, int buf) {
             const bf16* src = x + ((size_t)t * B + b0) * IN;
-            for (int c = tid; c < MB * KP / 4; c += WAVES * 64) {
-                const int row = (c * 4) / KP, col = (c * 4) % KP;
+#pragma unroll
+            for (int p = 0; p < MB * FUSEKP / 4 / (WAVES * 64); ++p) {
+                const int c = p * WAVES * 64 + tid;
+                const int row = (c * 4) / FUSEKP, col = (c * 4) % FUSEKP;
                 uint64_t v = 0;  // 4 bf16 (zero pads cols IN..KP)
                 if (col + 4 <= IN)
                     v = *reinterpret_cast<const uint64_t*>(
@@ -107,14 +115,13 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gru_layer_fwd_kernel(
                 *reinterpret_cast<uint64_t*>(&lds.x_st[buf][row][col]) = v;
             }
         };
-        float bih_reg[3][4];  // [s][i of 4-row group]? bias is per-COLUMN
+        float bih_reg[3];
 #pragma unroll
         for (int s = 0; s < 3; ++s) {
             const int col = (wid + 8 * s) * 16 + lcol;
-            const float bv = b_ih ? bf2f(b_ih[dir * G3 + col]) : 0.f;
-#pragma unroll
-            for (int i = 0; i < 4; ++i) bih_reg[s][i] = bv;
+            bih_reg[s] = b_ih ? bf2f(b_ih[dir * G3 + col]) : 0.f;
         }
+        const int col0w = dir * G3 + wid * 16;
         stage_x(0, 0);
         __syncthreads();
         for (int t = 0; t < T; ++t) {
@@ -125,18 +132,33 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gru_layer_fwd_kernel(
 #pragma unroll
                 for (int mt = 0; mt < 2; ++mt)
                     acc[s][mt] = f32x4{0.f, 0.f, 0.f, 0.f};
+            bf16x8 wfA[3], wfB[3], a0A, a1A, a0B, a1B;
+#pragma unroll
+            for (int s = 0; s < 3; ++s)
+                wfA[s] = global_load_a_frag(w_ih_p, col0w + s * 128, 0,
+                                            FUSEKP);
+            a0A = lds_load_a_frag(&lds.x_st[t & 1][0][0], 0, 0, XLDK);
+            a1A = lds_load_a_frag(&lds.x_st[t & 1][0][0], 16, 0, XLDK);
+#pragma unroll
             for (int kb = 0; kb < KB; ++kb) {
-                bf16x8 a0 = lds_load_a_frag(&lds.x_st[t & 1][0][0], 0,
-                                            kb * 32, XLD);
-                bf16x8 a1 = lds_load_a_frag(&lds.x_st[t & 1][0][0], 16,
-                                            kb * 32, XLD);
+                bf16x8(&wc)[3] = (kb & 1) ? wfB : wfA;
+                bf16x8(&wn)[3] = (kb & 1) ? wfA : wfB;
+                const bf16x8 a0 = (kb & 1) ? a0B : a0A;
+                const bf16x8 a1 = (kb & 1) ? a1B : a1A;
+                if (kb + 1 < KB) {
+#pragma unroll
+                    for (int s = 0; s < 3; ++s)
+                        wn[s] = global_load_a_frag(w_ih_p, col0w + s * 128,
+                                                   (kb + 1) * 32, FUSEKP);
+                    ((kb & 1) ? a0A : a0B) = lds_load_a_frag(
+                        &lds.x_st[t & 1][0][0], 0, (kb + 1) * 32, XLDK);
+                    ((kb & 1) ? a1A : a1B) = lds_load_a_frag(
+                        &lds.x_st[t & 1][0][0], 16, (kb + 1) * 32, XLDK);
+                }
 #pragma unroll
                 for (int s = 0; s < 3; ++s) {
-                    const int col0 = dir * G3 + (wid + 8 * s) * 16;
-                    const bf16x8 w = global_load_a_frag(w_ih_p, col0,
-                                                        kb * 32, KP);
-                    acc[s][0] = mfma16x16x32(a0, w, acc[s][0]);
-                    acc[s][1] = mfma16x16x32(a1, w, acc[s][1]);
+                    acc[s][0] = mfma16x16x32(a0, wc[s], acc[s][0]);
+                    acc[s][1] = mfma16x16x32(a1, wc[s], acc[s][1]);
                 }
             }
             // epilogue: + b_ih, scalar bf16 stores into the xg workspace
@@ -152,7 +174,7 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gru_layer_fwd_kernel(
                     for (int i = 0; i < 4; ++i) {
                         const int row = mt * 16 + lrow * 4 + i;
                         dst[(size_t)row * 2 * G3 + col] =
-                            f2bf(acc[s][mt][i] + bih_reg[s][i]);
+                            f2bf(acc[s][mt][i] + bih_reg[s]);
                     }
             }
             __syncthreads();
@@ -375,17 +397,17 @@ void gru_layer_fwd(const void* xg, const void* u, const float* bhh, void* hseq,
     dim3 grid(B / MB, 2);
     dim3 block(WAVES * 64);
     if (cache)
-        hipLaunchKernelGGL((gru_layer_fwd_kernel<true, false>), grid, block, 0,
+        hipLaunchKernelGGL((gru_layer_fwd_kernel<true, 0>), grid, block, 0,
                            stream,
                            static_cast<const bf16*>(xg), static_cast<const bf16*>(u),
                            bhh, static_cast<bf16*>(hseq), static_cast<bf16*>(cache),
-                           T, B, dbg, nullptr, nullptr, nullptr, 0, 0);
+                           T, B, dbg, nullptr, nullptr, nullptr, 0);
     else
-        hipLaunchKernelGGL((gru_layer_fwd_kernel<false, false>), grid, block, 0,
+        hipLaunchKernelGGL((gru_layer_fwd_kernel<false, 0>), grid, block, 0,
                            stream,
                            static_cast<const bf16*>(xg), static_cast<const bf16*>(u),
                            bhh, static_cast<bf16*>(hseq), nullptr, T, B, dbg,
-                           nullptr, nullptr, nullptr, 0, 0);
+                           nullptr, nullptr, nullptr, 0);
 }
 
 // serving variant: xg computed in-kernel from (x, W_ih, b_ih); `xg` is a
@@ -396,13 +418,26 @@ void gru_layer_fwd_fused(const void* x, const void* w_ih_p, const void* b_ih,
                          hipStream_t stream) {
     dim3 grid(B / MB, 2);
     dim3 block(WAVES * 64);
-    hipLaunchKernelGGL((gru_layer_fwd_kernel<false, true>), grid, block, 0,
-                       stream, static_cast<bf16*>(xg_ws),
-                       static_cast<const bf16*>(u), static_cast<const float*>(bhh),
-                       static_cast<bf16*>(hseq), nullptr, T, B, 0u,
-                       static_cast<const bf16*>(x),
-                       static_cast<const bf16*>(w_ih_p),
-                       static_cast<const bf16*>(b_ih), IN, KP);
+    if (KP == 512)
+        hipLaunchKernelGGL((gru_layer_fwd_kernel<false, 512>), grid, block, 0,
+                           stream, static_cast<bf16*>(xg_ws),
+                           static_cast<const bf16*>(u),
+                           static_cast<const float*>(bhh),
+                           static_cast<bf16*>(hseq), nullptr, T, B, 0u,
+                           static_cast<const bf16*>(x),
+                           static_cast<const bf16*>(w_ih_p),
+                           static_cast<const bf16*>(b_ih), IN);
+    else if (KP == 256)
+        hipLaunchKernelGGL((gru_layer_fwd_kernel<false, 256>), grid, block, 0,
+                           stream, static_cast<bf16*>(xg_ws),
+                           static_cast<const bf16*>(u),
+                           static_cast<const float*>(bhh),
+                           static_cast<bf16*>(hseq), nullptr, T, B, 0u,
+                           static_cast<const bf16*>(x),
+                           static_cast<const bf16*>(w_ih_p),
+                           static_cast<const bf16*>(b_ih), IN);
+    else
+        (void)0;  // unsupported KP: caller validates (bindings.cpp)
 }
 
 // ---------------------------------------------------------------------------
@@ -434,12 +469,16 @@ __global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
                                      // dir-major so each dir slice is a
                                      // contiguous (T*B, 384) GEMM operand
     int T, int B) {
+    // ONE barrier per step (was two): dhg is double-buffered so the next
+    // step's gate writes never touch the tile the current step's dh GEMM
+    // and output stores still read, and the dg_st staging tile is gone —
+    // dxg/dhg stores read straight from lds.dhg (dxr/dxz/dhgn blocks) plus
+    // per-lane register stores for the dxn column block.
     __shared__ struct {
         bf16 cache_st[2][MB][4 * H];  // double-buffered staged cache[t]
         bf16 dhin_st[2][MB][H];       // double-buffered staged dhin[t]
         bf16 hprev_st[2][MB][H];      // double-buffered staged h_{t-1}
-        bf16 dhg[MB][G3 + 8];         // A-operand of the dh GEMM
-        bf16 dg_st[MB][4 * H];        // staged output tile
+        bf16 dhg[2][MB][G3 + 8];      // A-operand of the dh GEMM (2 buffers)
     } lds;
 
     const int dir = blockIdx.y;
@@ -523,6 +562,7 @@ __global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
         // the stores — same restructure as the forward's gate phase (the
         // interleaved per-element form serialized the read latencies)
         float dhp_part[2][4];
+        float dxn8[2][4];  // kept in registers for the direct dxg store
         {
             const int j = j0 + lcol;
             float dh8[2][4], hp8[2][4];
@@ -551,21 +591,18 @@ __global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
                     const float dn = dh * (1.0f - z);
                     const float dz = dh * (hp8[mt][i] - n);
                     const float dan = dn * (1.0f - n * n);
-                    const float dxn = dan;
                     const float dhgn = dan * r;
                     const float dxr = dan * hgn * r * (1.0f - r);
                     const float dxz = dz * z * (1.0f - z);
                     dhp_part[mt][i] = dh * z;
-                    lds.dhg[row][0 * H + j] = f2bf(dxr);
-                    lds.dhg[row][1 * H + j] = f2bf(dxz);
-                    lds.dhg[row][2 * H + j] = f2bf(dhgn);
-                    lds.dg_st[row][0 * H + j] = f2bf(dxr);
-                    lds.dg_st[row][1 * H + j] = f2bf(dxz);
-                    lds.dg_st[row][2 * H + j] = f2bf(dxn);
-                    lds.dg_st[row][3 * H + j] = f2bf(dhgn);
+                    dxn8[mt][i] = dan;
+                    lds.dhg[curp][row][0 * H + j] = f2bf(dxr);
+                    lds.dhg[curp][row][1 * H + j] = f2bf(dxz);
+                    lds.dhg[curp][row][2 * H + j] = f2bf(dhgn);
                 }
         }
-        __syncthreads();  // dhg/dg_st and stage[curp^1] published
+        __syncthreads();  // dhg[curp] and stage[curp^1] published (the ONLY
+                          // barrier per step; next step writes dhg[curp^1])
 
         // ---- dh_prev = dh*z + dhg · U  (24 MFMA per wave) -----------------
         f32x4 acc[2];
@@ -575,7 +612,8 @@ __global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
         for (int kb = 0; kb < 12; ++kb) {
 #pragma unroll
             for (int mt = 0; mt < 2; ++mt) {
-                bf16x8 a = lds_load_a_frag(&lds.dhg[0][0], mt * 16, kb * 32, G3 + 8);
+                bf16x8 a = lds_load_a_frag(&lds.dhg[curp][0][0], mt * 16,
+                                           kb * 32, G3 + 8);
                 acc[mt] = mfma16x16x32(a, ufrag[kb], acc[mt]);
             }
         }
@@ -584,35 +622,41 @@ __global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
 #pragma unroll
             for (int i = 0; i < 4; ++i) dhc[mt][i] = dhp_part[mt][i] + acc[mt][i];
 
-        // ---- store dg tile -------------------------------------------------
         // ---- store dxg + dhg tiles (GEMM-ready layouts, no host cats) -----
+        // dxg = [dxr dxz dxn]: the dxr/dxz blocks copy wide from lds.dhg,
+        // dxn comes from this lane's registers (scalar stores)
         {
             bf16* dst = dxg + (((size_t)t * B + b0) * 2 + dir) * G3;
+            const int row = tid / 32, col = (tid % 32) * 8;  // 2H block
+#pragma unroll
+            for (int q = 0; q < 2; ++q)
+                *reinterpret_cast<bf16x8*>(
+                    dst + (size_t)(row + q * 16) * 2 * G3 + col) =
+                    *reinterpret_cast<const bf16x8*>(
+                        &lds.dhg[curp][row + q * 16][col]);
+            const int j = j0 + lcol;
+#pragma unroll
+            for (int mt = 0; mt < 2; ++mt)
+#pragma unroll
+                for (int i = 0; i < 4; ++i) {
+                    const int row2 = mt * 16 + lrow * 4 + i;
+                    dst[(size_t)row2 * 2 * G3 + 2 * H + j] =
+                        f2bf(dxn8[mt][i]);
+                }
+        }
+        {
+            // dhg tensor = [dxr dxz dhgn] — exactly lds.dhg's live columns
+            bf16* dst = dhg + (((size_t)dir * T + t) * B + b0) * G3;
             const int row = tid / 64, col = (tid % 64) * 8;
             if (col < G3) {
 #pragma unroll
                 for (int q = 0; q < 4; ++q)
                     *reinterpret_cast<bf16x8*>(
-                        dst + (size_t)(row + q * 8) * 2 * G3 + col) =
-                        *reinterpret_cast<const bf16x8*>(&lds.dg_st[row + q * 8][col]);
+                        dst + (size_t)(row + q * 8) * G3 + col) =
+                        *reinterpret_cast<const bf16x8*>(
+                            &lds.dhg[curp][row + q * 8][col]);
             }
         }
-        {
-            // dhg = [dxr dxz dhgn]: columns [0, 2H) then the dhgn block
-            bf16* dst = dhg + (((size_t)dir * T + t) * B + b0) * G3;
-            const int row = tid / 64, col = (tid % 64) * 8;
-#pragma unroll
-            for (int q = 0; q < 4; ++q) {
-                const int r2 = row + q * 8;
-                if (col < 2 * H)
-                    *reinterpret_cast<bf16x8*>(dst + (size_t)r2 * G3 + col) =
-                        *reinterpret_cast<const bf16x8*>(&lds.dg_st[r2][col]);
-                else if (col < 3 * H)
-                    *reinterpret_cast<bf16x8*>(dst + (size_t)r2 * G3 + col) =
-                        *reinterpret_cast<const bf16x8*>(&lds.dg_st[r2][col + H]);
-            }
-        }
-        __syncthreads();  // dhg/dg_st/stage[curp] reads done before reuse
     };
 
     int sidx = 0;

@@ -29,3 +29,24 @@ for train in (False, True):
     run(train, 8,  f"[{tag}] no xg staging")
     run(train, 16, f"[{tag}] no MFMA")
     run(train, 31, f"[{tag}] shell (barriers+LDS h only)")
+
+# fused (in-kernel xg GEMM) vs split timing per layer shape
+from roko_amd.model import RokoModel
+from roko_amd.ops import forward as fwd
+m = RokoModel().cuda().eval()
+w = fwd._bf16_weights(m)
+for l, IN in ((0, 500), (1, 256), (2, 256)):
+    x = (torch.randn(T, B, IN, device="cuda") * 0.3).to(torch.bfloat16)
+    def split():
+        xg2 = torch.addmm(w[f"b_ih{l}"], x.reshape(T * B, -1), w[f"w_ih_t{l}"]).view(T, B, 2, 384)
+        ext.gru_layer_fwd(xg2.contiguous(), w[f"u{l}"], w[f"bhh{l}"], False)
+    def fused():
+        ext.gru_layer_fused(x, w[f"w_ih_p{l}"], w[f"b_ih{l}"], w[f"u{l}"], w[f"bhh{l}"])
+    for name, fn in (("split", split), ("fused", fused)):
+        for _ in range(3): fn()
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(20): fn()
+        torch.cuda.synchronize()
+        us = (time.perf_counter() - t0) / 20 * 1e6
+        print(f"layer{l} IN={IN} {name:6s} {us:8.1f} us")
