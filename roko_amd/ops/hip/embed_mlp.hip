@@ -521,6 +521,15 @@ __global__ __launch_bounds__(WAVES3 * 64, 1) void embed_mlp_fwd3_kernel(
     bf16* t2 = lds.t2[wid];
     const uint8_t* win = ids + (size_t)b * R * W;
 
+    // loop-invariant operand fragments, register-resident for all columns
+    bf16x8 e_bv[4], w2_av[4];
+#pragma unroll
+    for (int nt = 0; nt < 4; ++nt)
+        e_bv[nt] = lds_load_b_frag_t(&lds.e_t[0][0], nt * 16, 0, 32);
+#pragma unroll
+    for (int kb = 0; kb < 4; ++kb)
+        w2_av[kb] = lds_load_a_frag(&lds.w2_lds[0][0], 0, kb * 32, LD1);
+
     // wave w handles columns w, w+WAVES3, ... — each fully privately
     for (int w = wid; w < W; w += WAVES3) {
         // ---- scatter: 200 one-hots (ids read straight through L1) ---------
@@ -538,35 +547,39 @@ __global__ __launch_bounds__(WAVES3 * 64, 1) void embed_mlp_fwd3_kernel(
 
         // ---- G1: A = W1 · Hot  (7 m-tiles x 7 k-steps, 49 MFMA) -----------
         // 7 independent accumulator chains; B-frag per k shared across m.
-        // Software-pipelined one k-step ahead: the L2 A-fragment loads for
-        // kb+1 issue before kb's MFMAs, so their ~200-cycle L2 latency hides
-        // under 7 MFMAs + the LDS B read instead of stalling each chain.
+        // THREE-deep k-window: at one wave per SIMD the register budget is
+        // 512/lane, so ~21 L2 A-fragments stay in flight — ~2 k-steps
+        // (~240 MFMA cycles) of cover for the ~200-cycle L2 latency (the
+        // 1-deep window measured neutral: 7 MFMAs of cover was not enough).
         {
             f32x4 acc[7];
 #pragma unroll
             for (int mt = 0; mt < 7; ++mt) acc[mt] = f32x4{0.f, 0.f, 0.f, 0.f};
-            bf16x8 afA[7], afB[7];
-            bf16x8 bfA, bfB;
+            bf16x8 af[3][7];
+            bf16x8 bf3[3];
 #pragma unroll
-            for (int mt = 0; mt < 7; ++mt)
-                afA[mt] = global_load_a_frag(w1g, mt * 16, 0, KP_LD);
-            bfA = lds_load_b_frag_t(&hot[0][0], 0, 0, KP_LD);
+            for (int p = 0; p < 2; ++p) {
+#pragma unroll
+                for (int mt = 0; mt < 7; ++mt)
+                    af[p][mt] = global_load_a_frag(w1g, mt * 16, p * 32,
+                                                   KP_LD);
+                bf3[p] = lds_load_b_frag_t(&hot[0][0], 0, p * 32, KP_LD);
+            }
 #pragma unroll
             for (int kb = 0; kb < 7; ++kb) {
-                bf16x8(&cur)[7] = (kb & 1) ? afB : afA;
-                bf16x8(&nxt)[7] = (kb & 1) ? afA : afB;
-                const bf16x8 bf_ = (kb & 1) ? bfB : bfA;
-                if (kb + 1 < 7) {
+                const int cur = kb % 3, nxt = (kb + 2) % 3;
+                if (kb + 2 < 7) {
 #pragma unroll
                     for (int mt = 0; mt < 7; ++mt)
-                        nxt[mt] = global_load_a_frag(w1g, mt * 16,
-                                                     (kb + 1) * 32, KP_LD);
-                    ((kb & 1) ? bfA : bfB) = lds_load_b_frag_t(
-                        &hot[0][0], 0, (kb + 1) * 32, KP_LD);
+                        af[nxt][mt] = global_load_a_frag(w1g, mt * 16,
+                                                         (kb + 2) * 32,
+                                                         KP_LD);
+                    bf3[nxt] = lds_load_b_frag_t(&hot[0][0], 0,
+                                                 (kb + 2) * 32, KP_LD);
                 }
 #pragma unroll
                 for (int mt = 0; mt < 7; ++mt)
-                    acc[mt] = mfma16x16x32(cur[mt], bf_, acc[mt]);
+                    acc[mt] = mfma16x16x32(af[cur][mt], bf3[cur], acc[mt]);
             }
             // epilogue: a[f][c] (c = lcol < 12 live, 12..15 zero products)
 #pragma unroll
@@ -587,12 +600,9 @@ __global__ __launch_bounds__(WAVES3 * 64, 1) void embed_mlp_fwd3_kernel(
             for (int mt = 0; mt < 7; ++mt) {
                 const bf16x8 af = lds_load_a_frag(&at[0][0], mt * 16, 0, 40);
 #pragma unroll
-                for (int nt = 0; nt < 4; ++nt) {
-                    const bf16x8 bf_ = lds_load_b_frag_t(&lds.e_t[0][0],
-                                                         nt * 16, 0, 32);
-                    acc[mt][nt] = mfma16x16x32(af, bf_,
+                for (int nt = 0; nt < 4; ++nt)
+                    acc[mt][nt] = mfma16x16x32(af, e_bv[nt],
                                                f32x4{0.f, 0.f, 0.f, 0.f});
-                }
             }
 #pragma unroll
             for (int mt = 0; mt < 7; ++mt)
@@ -623,12 +633,11 @@ __global__ __launch_bounds__(WAVES3 * 64, 1) void embed_mlp_fwd3_kernel(
 #pragma unroll
                 for (int kb = 0; kb < 4; kb += 2) {
                     acc0 = mfma16x16x32(
-                        lds_load_a_frag(&lds.w2_lds[0][0], 0, kb * 32, LD1),
+                        w2_av[kb],
                         lds_load_b_frag_t(&t1[0][0], nt * 16, kb * 32, LD1),
                         acc0);
                     acc1 = mfma16x16x32(
-                        lds_load_a_frag(&lds.w2_lds[0][0], 0, (kb + 1) * 32,
-                                        LD1),
+                        w2_av[kb + 1],
                         lds_load_b_frag_t(&t1[0][0], nt * 16, (kb + 1) * 32,
                                           LD1),
                         acc1);

@@ -79,7 +79,6 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gru_layer_fwd_kernel(
     __shared__ struct {
         bf16 h[2][MB][HPAD];         // double-buffered hidden-state mirror
         bf16 xgb[2][MB][G3];         // double-buffered step gate inputs
-        bf16 cache_st[TRAIN ? 2 : 1][TRAIN ? MB : 1][TRAIN ? 4 * H : 1];
         bf16 x_st[FUSEXG ? 2 : 1][FUSEXG ? MB : 1][XLDK];
     } lds;
 
@@ -307,10 +306,14 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gru_layer_fwd_kernel(
                     hreg[mt][i] = hnew;
                     lds.h[curp ^ 1][row][j] = f2bf(hnew);
                     if constexpr (TRAIN) {
-                        bf16x4 pk = {f2bf(hnew), f2bf(hnew), f2bf(hnew),
-                                     f2bf(hnew)};
-                        *reinterpret_cast<bf16x4*>(
-                            &lds.cache_st[curp ^ 1][row][4 * j]) = pk;
+                        if (!(dbg & 2u)) {
+                            bf16x4 pk = {f2bf(hnew), f2bf(hnew), f2bf(hnew),
+                                         f2bf(hnew)};
+                            *reinterpret_cast<bf16x4*>(
+                                cache +
+                                (((size_t)t * B + b0 + row) * 2 + dir) * 4 * H +
+                                4 * j) = pk;
+                        }
                     }
                 }
         } else {
@@ -351,10 +354,17 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gru_layer_fwd_kernel(
                     hreg[mt][i] = hnew;
                     lds.h[curp ^ 1][row][j] = f2bf(hnew);
                     if constexpr (TRAIN) {
-                        bf16x4 pk = {f2bf(r8[mt][i]), f2bf(z8[mt][i]),
-                                     f2bf(n8[mt][i]), f2bf(hg8[mt][i])};
-                        *reinterpret_cast<bf16x4*>(
-                            &lds.cache_st[curp ^ 1][row][4 * j]) = pk;
+                        // cache stores straight from registers (8-byte,
+                        // 128 B per 16-lane group): no LDS staging tile,
+                        // no post-barrier round-trip
+                        if (!(dbg & 2u)) {
+                            bf16x4 pk = {f2bf(r8[mt][i]), f2bf(z8[mt][i]),
+                                         f2bf(n8[mt][i]), f2bf(hg8[mt][i])};
+                            *reinterpret_cast<bf16x4*>(
+                                cache +
+                                (((size_t)t * B + b0 + row) * 2 + dir) * 4 * H +
+                                4 * j) = pk;
+                        }
                     }
                 }
         }
@@ -368,18 +378,6 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gru_layer_fwd_kernel(
             const int row = e / H, col = e % H;
             *reinterpret_cast<bf16x8*>(dst + (size_t)row * 2 * H + col) =
                 *reinterpret_cast<const bf16x8*>(&lds.h[curp ^ 1][row][col]);
-        }
-        if constexpr (TRAIN) {
-            if (!(dbg & 2u)) {
-                bf16* dst = cache + (((size_t)t * B + b0) * 2 + dir) * 4 * H;
-#pragma unroll
-                for (int p = 0; p < 4; ++p) {
-                    const int e = (p * WAVES * 64 + tid) * 8;
-                    const int row = e / (4 * H), col = e % (4 * H);
-                    *reinterpret_cast<bf16x8*>(dst + (size_t)row * 2 * 4 * H + col) =
-                        *reinterpret_cast<const bf16x8*>(&lds.cache_st[curp ^ 1][row][col]);
-                }
-            }
         }
     };
 
