@@ -55,7 +55,7 @@ void front_bwd(const uint8_t* ids, const void* dseq, const void* w1,
                const void* emb, float* dw1, float* db1, float* dw2, float* db2,
                const void* w1t_g, float* de, int B, uint32_t seed, float keep,
                hipStream_t stream, uint32_t phase_mask,
-               const uint32_t* seed_ptr);
+               const uint32_t* seed_ptr, const void* w1g);
 void gemm_bias(const void* A, const void* B, const float* bias, void* C,
                int M, int N, int K, hipStream_t stream);
 int atb_splitk_nslices(int K);
@@ -390,16 +390,19 @@ std::vector<torch::Tensor> front_bwd(torch::Tensor ids, torch::Tensor dseq,
     auto dw2 = torch::zeros({10, 100}, opt);
     auto db2 = torch::zeros({10}, opt);
     auto de = torch::zeros({12, 50}, opt);
-    // zero-padded W1^T image: the merged dm phase's L2-read B-fragments
+    // zero-padded W1 images: W1^T (208,128) is LDS-staged for the dm GEMM,
+    // W1 (112,232) feeds the G1 recompute's L2 A-fragments
     auto w1t_g = torch::zeros({208, 128}, w1.options());
     w1t_g.slice(0, 0, 200).slice(1, 0, 100).copy_(w1.t());
+    auto w1g = torch::zeros({112, 232}, w1.options());
+    w1g.slice(0, 0, 100).slice(1, 0, 200).copy_(w1);
     rk::front_bwd(ids.data_ptr<uint8_t>(), dseq.data_ptr(), w1.data_ptr(),
                   b1.data_ptr<float>(), w2.data_ptr(), b2.data_ptr<float>(),
                   emb.data_ptr(), dw1.data_ptr<float>(), db1.data_ptr<float>(),
                   dw2.data_ptr<float>(), db2.data_ptr<float>(),
                   w1t_g.data_ptr(), de.data_ptr<float>(), B, (uint32_t)seed,
                   (float)keep, cur_stream(), (uint32_t)phase_mask,
-                  seed_ptr_of(seed_buf));
+                  seed_ptr_of(seed_buf), w1g.data_ptr());
     return {de, dw1, db1, dw2, db2};
 }
 

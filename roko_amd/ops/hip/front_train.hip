@@ -294,30 +294,36 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
     float* __restrict__ db1,  // (F1)
     float* __restrict__ dw2,  // (F2, F1)
     float* __restrict__ db2,  // (F2)
-    const bf16* __restrict__ w1t_g,  // (208, 128) zero-padded W1^T in L2:
-                              // the dm GEMM's B-operand (a second transposed
-                              // W1 LDS copy would not fit; scalar W1 reads
-                              // made the dm phase HALF this kernel's time
-                              // in v1 — profiles/front_bwd_phases_r01)
+    const bf16* __restrict__ w1t_g,  // (208, 128) zero-padded W1^T image:
+                              // source of the LDS w1t_t stage
     float* __restrict__ de,   // (12, E) pre-zeroed, atomic-accumulated
     int B, uint32_t seed, float keep, uint32_t phase_mask,
-    const uint32_t* __restrict__ seed_ptr) {
+    const uint32_t* __restrict__ seed_ptr,
+    const bf16* __restrict__ w1g) {  // (112, 232) zero-padded W1 in L2:
+                              // the G1 recompute's A-fragments (frees the
+                              // 52 KB [f][r] LDS copy for w1t_t)
     // phase_mask: timing-experiment switch (default 0x1F = all phases).
     // bit0 G1 recompute, bit1 G3+dt2, bit2 dt1, bit3 dW1/dW2, bit4 dm/de.
     __shared__ struct {
-        bf16 w1t[MP][KP_LD];  // [f][r]: G1 A-operand; dm B via 8-scalar reads
-                              // (dm's K pad rows 112..127 read past this
-                              // array into the union below — safe: the
-                              // matching A-operand columns are zero)
+        // W1^T staged [r][f] ONCE at kernel start: the dm GEMM's B-operand
+        // as conflict-free b128 reads. The former [f][r] W1 copy is gone —
+        // the G1 recompute reads its A-fragments straight from the L2-
+        // resident padded W1 image (w1g), the v2/v3 eval-front pattern.
+        // Round-1's dm phase read W1^T from L2 per tile with no pipelining
+        // cover: the dm/de phases were 332 us of the 736 us kernel.
+        bf16 w1t_t[R + 8][136];
         union {
             bf16 m_t[EP][KP_LD];  // [e][r]: G1 B-operand (dead after G1)
             struct {              // live from the G3 phase on
-                bf16 dt2_je[16][72];   // [j][e]: dW2 A-operand
-                bf16 dt2_ej[EP][40];   // [e][j]: dt1 A-operand
+                bf16 je[16][72];  // [j][e]: dW2 A-operand
+                bf16 ej[EP][40];  // [e][j]: dt1 A-operand
             } g;
             bf16 dmm[EP][KP_LD];  // [e][r]: masked dm — the de GEMM's
                                   // B-operand (dead dt2 region, same size)
         } u1;
+        bf16 m_rt[R + 8][72];     // [r][e]: dW1 B-operand (b128 fragments;
+                                  // the k-major scalar read from m_t
+                                  // measured 4-way bank-conflicted: +82 us)
         union {
             bf16 t1_t[EP][136];   // [e][f]: G3 B-operand (dead after G3)
             // 128 rows: the dm A-fragment reads k = f up to 127; rows
@@ -326,7 +332,6 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
             // 0 x uninitialized-LDS-NaN = NaN — docs/KERNELS.md lesson 9)
             bf16 dt1_fe[128][72]; // [f][e]: dW1 + dm A-operand
         } u2;
-        bf16 m_rt[R + 8][72];    // [r][e]: dW1 B-operand
         union {
             bf16 t1_fe[MP][72];     // [f][e]: dW2 B-operand + relu/drop mask
             // hot aliases t1_fe's first 3712 elements: t1_fe is dead after
@@ -335,11 +340,9 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
             // never fragment-read (k stops at 63)
             bf16 hot_t[16][KP_LD];  // [c][r]: de GEMM A-operand
         } u4;
-        bf16 w2_lds[12][136];    // [j][f]: G3 A-operand. Only 12 rows: the
-                                 // A-fragment's row 12..15 reads run past
-                                 // the array (garbage), feeding accumulator
-                                 // rows j >= 10 that every epilogue ignores
-        bf16 w2t_t[MP][40];      // [f][j]: dt1^T B-operand
+        bf16 w2_lds[32][136];    // [j][f]: G3 A-operand + dt1 B-operand
+                                 // (k-major fragments read j up to 31:
+                                 // rows 10..31 stay zero)
         bf16 dseq_st[OUT];
         bf16 emb_s[12][E];
         bf16 b1s[MP];
@@ -360,24 +363,25 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
     const int w_end = (blockIdx.y + 1 == CSPLIT) ? W : w_begin + W / CSPLIT;
 
     // ---- one-time staging + zero ------------------------------------------
-    for (int e = tid; e < MP * KP_LD; e += 512) (&lds.w1t[0][0])[e] = f2bf(0.f);
     for (int e = tid; e < EP * KP_LD; e += 512) (&lds.u1.m_t[0][0])[e] = f2bf(0.f);
     for (int e = tid; e < EP * 136; e += 512) (&lds.u2.t1_t[0][0])[e] = f2bf(0.f);
-    for (int e = tid; e < (R + 8) * 72; e += 512) (&lds.m_rt[0][0])[e] = f2bf(0.f);
     for (int e = tid; e < 16 * 72; e += 512)
         lds.u2.dt1_fe[112 + e / 72][e % 72] = f2bf(0.f);
+    for (int e = tid; e < (R + 8) * 72; e += 512) (&lds.m_rt[0][0])[e] = f2bf(0.f);
     for (int e = tid; e < MP * 72; e += 512) (&lds.u4.t1_fe[0][0])[e] = f2bf(0.f);
-    for (int e = tid; e < 12 * 136; e += 512) (&lds.w2_lds[0][0])[e] = f2bf(0.f);
-    for (int e = tid; e < MP * 40; e += 512) (&lds.w2t_t[0][0])[e] = f2bf(0.f);
+    for (int e = tid; e < 32 * 136; e += 512) (&lds.w2_lds[0][0])[e] = f2bf(0.f);
     for (int e = tid; e < MP; e += 512) lds.b1s[e] = f2bf(0.f);
     for (int e = tid; e < 16; e += 512) lds.b2s[e] = f2bf(0.f);
     __syncthreads();
-    for (int e = tid; e < F1 * R; e += 512) lds.w1t[e / R][e % R] = w1[e];
-    for (int e = tid; e < 12 * E; e += 512) lds.emb_s[e / E][e % E] = emb[e];
-    for (int e = tid; e < F2 * F1; e += 512) {
-        lds.w2_lds[e / F1][e % F1] = w2[e];
-        lds.w2t_t[e % F1][e / F1] = w2[e];
+    // W1^T [r][f] from the padded L2 image (rows 200..207 / cols 100..127
+    // are zero in the source; stride pad cols 128..135 are never read)
+    for (int p = tid; p < (R + 8) * 128 / 8; p += 512) {
+        const int r = p / 16, f = (p % 16) * 8;
+        *reinterpret_cast<bf16x8*>(&lds.w1t_t[r][f]) =
+            *reinterpret_cast<const bf16x8*>(w1t_g + (size_t)r * 128 + f);
     }
+    for (int e = tid; e < 12 * E; e += 512) lds.emb_s[e / E][e % E] = emb[e];
+    for (int e = tid; e < F2 * F1; e += 512) lds.w2_lds[e / F1][e % F1] = w2[e];
     for (int e = tid; e < F1; e += 512) lds.b1s[e] = f2bf(b1[e]);
     for (int e = tid; e < F2; e += 512) lds.b2s[e] = f2bf(b2[e]);
 
@@ -459,11 +463,27 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
             if (tile < 28) {
                 const int mt = tile >> 2, nt = tile & 3;
                 f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+                // A-fragments from L2: batch-issue 4, then 3 more under the
+                // first MFMAs (un-pipelined per-kb loads exposed ~200 cyc
+                // of L2 latency per k-step: +74 us on this phase)
+                bf16x8 ag0[4], ag1[3];
 #pragma unroll
-                for (int kb = 0; kb < 7; ++kb) {
-                    bf16x8 a = lds_load_a_frag(&lds.w1t[0][0], mt * 16, kb * 32, KP_LD);
-                    bf16x8 bb = lds_load_b_frag_t(&lds.u1.m_t[0][0], nt * 16, kb * 32, KP_LD);
-                    acc = mfma16x16x32(a, bb, acc);
+                for (int q = 0; q < 4; ++q)
+                    ag0[q] = global_load_a_frag(w1g, mt * 16, q * 32, KP_LD);
+#pragma unroll
+                for (int kb = 0; kb < 4; ++kb) {
+                    if (kb < 3)
+                        ag1[kb] = global_load_a_frag(w1g, mt * 16,
+                                                     (kb + 4) * 32, KP_LD);
+                    bf16x8 bb = lds_load_b_frag_t(&lds.u1.m_t[0][0], nt * 16,
+                                                  kb * 32, KP_LD);
+                    acc = mfma16x16x32(ag0[kb], bb, acc);
+                }
+#pragma unroll
+                for (int kb = 0; kb < 3; ++kb) {
+                    bf16x8 bb = lds_load_b_frag_t(&lds.u1.m_t[0][0], nt * 16,
+                                                  (kb + 4) * 32, KP_LD);
+                    acc = mfma16x16x32(ag1[kb], bb, acc);
                 }
                 const int e = nt * 16 + lcol;
 #pragma unroll
@@ -510,8 +530,8 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
                     if (t2pre > 0.f && keep_half(h, 0, thresh16))
                         g = bf2f(lds.dseq_st[e * F2 + j]) * inv_keep;
                 }
-                lds.u1.g.dt2_je[j][e] = f2bf(g);
-                lds.u1.g.dt2_ej[e][j] = f2bf(g);
+                lds.u1.g.je[j][e] = f2bf(g);
+                lds.u1.g.ej[e][j] = f2bf(g);
                 db2acc[i] += g;  // per-lane partial; reduced once at the end
             }
         } else if (wid >= 4 && (phase_mask & 2u)) {
@@ -519,7 +539,7 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
             // dt1 GEMM's K loop; aliased memory holds stale m values)
             for (int z = tid - 256; z < EP * 2; z += 256) {
                 const int e = z >> 1, c = 16 + (z & 1) * 8;
-                *reinterpret_cast<bf16x8*>(&lds.u1.g.dt2_ej[e][c]) = bf16x8{};
+                *reinterpret_cast<bf16x8*>(&lds.u1.g.ej[e][c]) = bf16x8{};
             }
         }
         __syncthreads();
@@ -532,8 +552,10 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
             const int tile = wid + s * 8;
             if (tile < 28) {
                 const int emt = tile & 3, fnt = tile >> 2;
-                bf16x8 a = lds_load_a_frag(&lds.u1.g.dt2_ej[0][0], emt * 16, 0, 40);
-                bf16x8 bb = lds_load_b_frag_t(&lds.w2t_t[0][0], fnt * 16, 0, 40);
+                bf16x8 a = lds_load_a_frag(&lds.u1.g.ej[0][0], emt * 16, 0, 40);
+                // W2 read k-major from w2_lds (8-scalar; rows 10..31 zero)
+                bf16x8 bb = lds_load_b_frag_km(&lds.w2_lds[0][0], 0,
+                                               fnt * 16, 136);
                 f32x4 acc = mfma16x16x32(a, bb, f32x4{0.f, 0.f, 0.f, 0.f});
                 const int f = fnt * 16 + lcol;
                 const int e0 = emt * 16 + lrow * 4;
@@ -561,7 +583,7 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
             const int nt = wid;
 #pragma unroll
             for (int kb = 0; kb < 2; ++kb) {
-                bf16x8 a = lds_load_a_frag(&lds.u1.g.dt2_je[0][0], 0, kb * 32, 72);
+                bf16x8 a = lds_load_a_frag(&lds.u1.g.je[0][0], 0, kb * 32, 72);
                 bf16x8 bb = lds_load_b_frag_t(&lds.u4.t1_fe[0][0], nt * 16, kb * 32, 72);
                 dw2acc = mfma16x16x32(a, bb, dw2acc);
             }
@@ -593,7 +615,9 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
             if (tid < R) lds.u4.hot_t[lds.col_ids[tid]][tid] = f2bf(1.0f);
 
             // dm^T[e][r] = dt1^T · W1: A from dt1_fe ([f][e], transposed
-            // fragment reads), B = W1^T straight from L2
+            // fragment reads), B = W1^T from the LDS stage (b128 reads —
+            // round 1 read it per-tile from L2 with no latency cover and
+            // the dm/de phases were 45% of this kernel)
 #pragma clang loop unroll(disable)
             for (int s = 0; s < 7; ++s) {
                 const int tile = wid + s * 8;
@@ -605,8 +629,8 @@ __global__ __launch_bounds__(512, 2) void front_bwd_kernel(
                     for (int kb = 0; kb < 4; ++kb) {
                         bf16x8 a = lds_load_a_frag_t(&lds.u2.dt1_fe[0][0],
                                                      emt * 16, kb * 32, 72);
-                        bf16x8 bb = global_load_a_frag(w1t_g, rnt * 16,
-                                                       kb * 32, 128);
+                        bf16x8 bb = lds_load_b_frag_t(&lds.w1t_t[0][0],
+                                                      rnt * 16, kb * 32, 136);
                         acc = mfma16x16x32(a, bb, acc);
                     }
 #pragma unroll
@@ -881,7 +905,7 @@ void front_bwd(const uint8_t* ids, const void* dseq, const void* w1,
                const void* emb, float* dw1, float* db1, float* dw2, float* db2,
                const void* w1t_g, float* de, int B, uint32_t seed, float keep,
                hipStream_t stream, uint32_t phase_mask,
-               const uint32_t* seed_ptr) {
+               const uint32_t* seed_ptr, const void* w1g) {
     hipLaunchKernelGGL(front::front_bwd_kernel,
                        dim3(B, front::CSPLIT), dim3(512), 0, stream,
                        ids, static_cast<const bf16*>(dseq),
@@ -889,7 +913,7 @@ void front_bwd(const uint8_t* ids, const void* dseq, const void* w1,
                        static_cast<const bf16*>(w2), b2,
                        static_cast<const bf16*>(emb), dw1, db1, dw2, db2,
                        static_cast<const bf16*>(w1t_g), de, B, seed, keep,
-                       phase_mask, seed_ptr);
+                       phase_mask, seed_ptr, static_cast<const bf16*>(w1g));
 }
 
 }  // namespace rk
