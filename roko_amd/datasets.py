@@ -105,18 +105,24 @@ class InferenceDataset(Dataset):
         self.path = path
         f = RkwFile(path)
         self.group_meta = f.groups
-        all_groups = range(len(f.groups)) if groups is None else groups
-        self.index: List[Tuple[int, int]] = []
-        for gi in all_groups:
-            for j in range(f.groups[gi]["size"]):
-                self.index.append((gi, j))
+        self.groups = (list(range(len(f.groups))) if groups is None
+                       else list(groups))
+        # cumulative window counts instead of a per-window (gi, j) list:
+        # a whole-genome file has ~1e8 windows and a materialised index
+        # would cost tens of GB of host RAM (BASELINE config 5)
+        self._cum = np.cumsum(
+            [0] + [int(f.groups[gi]["size"]) for gi in self.groups])
         self._lazy = _LazyFiles([path])
 
     def __len__(self) -> int:
-        return len(self.index)
+        return int(self._cum[-1])
 
     def __getitem__(self, idx: int):
-        gi, j = self.index[idx]
+        if idx < 0 or idx >= len(self):
+            raise IndexError(idx)
+        k = int(np.searchsorted(self._cum, idx, side="right") - 1)
+        gi = self.groups[k]
+        j = idx - int(self._cum[k])
         f = self._lazy.get(0)
         _, _, ex, _ = f.group_arrays(gi)
         x = torch.from_numpy(np.array(ex[j]))

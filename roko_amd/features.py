@@ -177,25 +177,36 @@ def run(
 
         n_errors = 0
 
-        def consume(result):
+        def consume(result, job):
             nonlocal n_windows, n_errors
             if result is None:
                 return
             if result[0] == "__error__":
-                n_errors += 1
-                log(f"WARNING: region {result[1]} failed: {result[2]} (skipped)")
-                return
+                # one in-parent retry before giving the region up: transient
+                # faults (an fs hiccup, a worker killed mid-region) should
+                # not cost coverage (SURVEY.md §5.3 — the reference dies on
+                # the first worker exception and has no retry at all)
+                log(f"WARNING: region {result[1]} failed: {result[2]} "
+                    "(retrying once)")
+                result = func(job)
+                if result is None:
+                    return
+                if result[0] == "__error__":
+                    n_errors += 1
+                    log(f"WARNING: region {result[1]} failed twice: "
+                        f"{result[2]} (skipped)")
+                    return
             contig, start, end, positions, examples, labs = result
             writer.store(contig, start, end, positions, examples, labs)
             n_windows += len(positions)
 
         if workers <= 1:
             for job in jobs:
-                consume(func(job))
+                consume(func(job), job)
         else:
             with Pool(processes=workers) as pool:
-                for result in pool.imap(func, jobs):
-                    consume(result)
+                for result, job in zip(pool.imap(func, jobs), jobs):
+                    consume(result, job)
     dt = time.time() - t0
     log(f"wrote {n_windows} windows to {out_path} in {dt:.1f}s "
         f"({n_windows / max(dt, 1e-9):.0f} windows/s)"

@@ -142,3 +142,29 @@ def test_streaming_votes_match_batch_and_stay_bounded():
     keys_b, counts_b = accumulate_votes(pos, preds)
     assert np.array_equal(keys_s, keys_b)
     assert np.array_equal(counts_s, counts_b)
+
+
+def test_features_region_retry(tiny_assembly, tmp_path, monkeypatch):
+    """A transiently failing region is retried once in the parent and still
+    contributes windows (failure-recovery contract)."""
+    import roko_amd.features as F
+
+    calls = {"n": 0}
+    real = F._features_for_region
+
+    def flaky(bam, contig, start, end, cfg):
+        calls["n"] += 1
+        if calls["n"] == 1:
+            raise RuntimeError("injected transient fault")
+        return real(bam, contig, start, end, cfg)
+
+    monkeypatch.setattr(F, "_features_for_region", flaky)
+    msgs = []
+    out = str(tmp_path / "retry.rkw")
+    n = F.run(tiny_assembly["draft_fasta"], tiny_assembly["reads_bam"], out,
+              workers=1, cfg=F.FeatureConfig(region_size=2000,
+                                             region_overlap=300),
+              log=lambda *a: msgs.append(" ".join(str(x) for x in a)))
+    assert n > 0
+    assert any("retrying once" in m for m in msgs)
+    assert not any("failed twice" in m for m in msgs)
