@@ -533,8 +533,10 @@ class WGroup:
         self.children[name] = g
         return g
 
-    def create_dataset(self, name: str, data: np.ndarray, **kw) -> "WDataset":
-        d = WDataset(name, np.ascontiguousarray(data))
+    def create_dataset(self, name: str, data: np.ndarray,
+                       chunks: Optional[Tuple[int, ...]] = None,
+                       **kw) -> "WDataset":
+        d = WDataset(name, np.ascontiguousarray(data), chunks=chunks)
         self.children[name] = d
         return d
 
@@ -607,21 +609,100 @@ class WGroup:
 
 
 class WDataset:
-    def __init__(self, name: str, data: np.ndarray):
+    def __init__(self, name: str, data: np.ndarray,
+                 chunks: Optional[Tuple[int, ...]] = None):
         if data.dtype.kind not in "iufS":
             raise NotImplementedError(f"dataset dtype {data.dtype}")
+        if chunks is not None:
+            if len(chunks) != data.ndim:
+                raise ValueError("chunks rank mismatch")
+            if any(c < 1 for c in chunks):
+                raise ValueError("chunk dims must be >= 1")
         self.name = name
         self.data = data
+        self.chunks = tuple(chunks) if chunks else None
         self.attrs: Dict[str, object] = {}
 
+    # chunk B-tree v1 (type 1): superblock v0 implies indexed-storage K=32,
+    # so nodes hold up to 64 entries and readers fetch full-capacity images
+    _CHUNK_BT_CAP = 64
+
+    def _emit_chunked(self, w: _W) -> bytes:
+        """Write chunk data + B-tree; returns the layout message body."""
+        d, ck = self.data, self.chunks
+        nd = d.ndim
+        keysize = 8 + 8 * (nd + 1)
+        node_size = 24 + (2 * self._CHUNK_BT_CAP + 1) * keysize \
+            + 2 * self._CHUNK_BT_CAP * 8
+        grid = [range(0, d.shape[i], ck[i]) for i in range(nd)]
+        import itertools
+        entries = []  # (offsets tuple, addr, nbytes)
+        for offs in itertools.product(*grid):
+            sl = tuple(slice(o, min(o + ck[i], d.shape[i]))
+                       for i, o in enumerate(offs))
+            block = np.zeros(ck, d.dtype)
+            csl = tuple(slice(0, sl[i].stop - sl[i].start) for i in range(nd))
+            block[csl] = d[sl]
+            raw = block.tobytes()
+            entries.append((offs, w.alloc(raw), len(raw)))
+
+        def key(offs) -> bytes:
+            k = struct.pack("<II", 0, 0)  # size+mask only meaningful pre-child
+            for o in offs:
+                k += struct.pack("<Q", o)
+            return k + struct.pack("<Q", 0)
+
+        def entry_key(e) -> bytes:
+            k = struct.pack("<II", e[2], 0)
+            for o in e[0]:
+                k += struct.pack("<Q", o)
+            return k + struct.pack("<Q", 0)
+
+        end_offs = tuple(
+            (d.shape[i] + ck[i] - 1) // ck[i] * ck[i] for i in range(nd))
+
+        def emit_nodes(level: int, kids):
+            """kids: list of (first_offs, addr, first_key_bytes)."""
+            out = []
+            for i in range(0, len(kids), self._CHUNK_BT_CAP):
+                grp = kids[i:i + self._CHUNK_BT_CAP]
+                node = b"TREE" + struct.pack("<BBHQQ", 1, level, len(grp),
+                                             UNDEF, UNDEF)
+                for first_offs, addr, kb in grp:
+                    node += kb + struct.pack("<Q", addr)
+                # final key: next sibling's first offsets or one-past-end
+                j = i + len(grp)
+                nxt = kids[j][2] if j < len(kids) else key(end_offs)
+                node += nxt
+                node += b"\x00" * (node_size - len(node))
+                out.append((grp[0][0], w.alloc(node), grp[0][2]))
+            return out
+
+        layer = [(e[0], e[1], entry_key(e)) for e in entries]
+        level = 0
+        while True:
+            layer = emit_nodes(level, layer)
+            if len(layer) == 1:
+                btree_addr = layer[0][1]
+                break
+            level += 1
+        body = struct.pack("<BBBQ", 3, 2, nd + 1, btree_addr)
+        for c in ck:
+            body += struct.pack("<I", c)
+        body += struct.pack("<I", d.dtype.itemsize)
+        return body
+
     def _emit(self, w: _W) -> int:
-        raw_addr = w.alloc(self.data.tobytes())
+        if self.chunks:
+            layout = self._emit_chunked(w)
+        else:
+            raw_addr = w.alloc(self.data.tobytes())
+            layout = struct.pack("<BBQQ", 3, 1, raw_addr, self.data.nbytes)
         msgs = [
             _Msg(0x0001, _ds_msg(self.data.shape)),
             _Msg(0x0003, _dt_msg(self.data.dtype)),
             _Msg(0x0005, struct.pack("<BBBB", 2, 2, 2, 0)),  # fill: undefined
-            _Msg(0x0008, struct.pack("<BBQQ", 3, 1, raw_addr,
-                                     self.data.nbytes)),
+            _Msg(0x0008, layout),
         ]
         for an, av in self.attrs.items():
             msgs.append(_attr_msg(an, av))

@@ -83,7 +83,8 @@ def rkw_to_hdf5(rkw_path: str, h5_path: str) -> int:
             grp["positions"] = np.asarray(pos).astype(np.int64)
             if lab is not None:
                 grp["labels"] = np.asarray(lab).astype(np.int64)
-            grp.create_dataset("examples", np.asarray(ex).astype(np.uint8))
+            grp.create_dataset("examples", np.asarray(ex).astype(np.uint8),
+                               chunks=(1, ex.shape[1], ex.shape[2]))
             grp.attrs["contig"] = g["contig"]
             grp.attrs["size"] = int(g["size"])
             n_total += int(g["size"])
