@@ -31,7 +31,7 @@ void gru_layer_fwd_fused(const void* x, const void* w_ih_p, const void* b_ih,
                          hipStream_t stream);
 void gru_layer_bwd(const void* cache, const void* hseq, const void* dhin,
                    const void* ut, void* dxg, void* dhg, int T, int B,
-                   hipStream_t stream);
+                   hipStream_t stream, uint32_t dbg);
 void ce_fwd_bwd(const float* logits, const int64_t* target, float* dlogits,
                 float* loss_sum, int64_t n, hipStream_t stream);
 void adam_step(float* p, const float* g, float* m, float* v, int64_t n,
@@ -216,7 +216,8 @@ std::vector<torch::Tensor> gru_layer_fwd(torch::Tensor xg, torch::Tensor u,
 // BPTT sequential backward -> dxg (T,B,2,384) [dxr dxz dxn] and
 // dhg (2,T,B,384) [dxr dxz dhgn] in GEMM-ready layouts
 std::vector<torch::Tensor> gru_layer_bwd(torch::Tensor cache, torch::Tensor hseq,
-                                         torch::Tensor dhin, torch::Tensor ut) {
+                                         torch::Tensor dhin, torch::Tensor ut,
+                                         int64_t dbg = 0) {
     check(cache, torch::kBFloat16, "cache");
     check(hseq, torch::kBFloat16, "hseq");
     check(dhin, torch::kBFloat16, "dhin");
@@ -231,7 +232,7 @@ std::vector<torch::Tensor> gru_layer_bwd(torch::Tensor cache, torch::Tensor hseq
     auto dhg = torch::empty({2, T, B, 384}, cache.options());
     rk::gru_layer_bwd(cache.data_ptr(), hseq.data_ptr(), dhin.data_ptr(),
                       ut.data_ptr(), dxg.data_ptr(), dhg.data_ptr(), T, B,
-                      cur_stream());
+                      cur_stream(), (uint32_t)dbg);
     return {dxg, dhg};
 }
 
@@ -669,7 +670,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           py::arg("bhh"), py::arg("train") = false, py::arg("dbg") = 0);
     m.def("gru_layer_fused", &gru_layer_fused, py::arg("x"), py::arg("w_ih_p"),
           py::arg("b_ih"), py::arg("u"), py::arg("bhh"));
-    m.def("gru_layer_bwd", &gru_layer_bwd);
+    m.def("gru_layer_bwd", &gru_layer_bwd, py::arg("cache"), py::arg("hseq"),
+          py::arg("dhin"), py::arg("ut"), py::arg("dbg") = 0);
     m.def("ce_fwd_bwd", &ce_fwd_bwd);
     m.def("adam_step", &adam_step, py::arg("p"), py::arg("g"),
           py::arg("m"), py::arg("v"), py::arg("lr"), py::arg("beta1"),

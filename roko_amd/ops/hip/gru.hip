@@ -466,7 +466,9 @@ __global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
     bf16* __restrict__ dhg,          // (2, T, B, 3H) out: [dxr dxz dhgn]
                                      // dir-major so each dir slice is a
                                      // contiguous (T*B, 384) GEMM operand
-    int T, int B) {
+    int T, int B,
+    uint32_t dbg) {  // timing bisection: 1 no global stores, 2 no gate
+                     // VALU, 4 no MFMA, 8 no staging
     // ONE barrier per step (was two): dhg is double-buffered so the next
     // step's gate writes never touch the tile the current step's dh GEMM
     // and output stores still read, and the dg_st staging tile is gone —
@@ -511,6 +513,7 @@ __global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
     // register-staged loads for one step: cache (4 chunks) + dhin + hprev
     auto issue_loads = [&](int sidx, bf16x8 (&rc)[BWCH_C], bf16x8& rdh,
                            bf16x8& rhp) {
+        if (dbg & 8u) return;
         const int t = t_of(sidx);
         const int tp = (dir == 0) ? t - 1 : t + 1;
         const bf16* src = cache + (((size_t)t * B + b0) * 2 + dir) * 4 * H;
@@ -576,6 +579,20 @@ __global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
                         &lds.cache_st[curp][row][4 * j]);
                     hp8[mt][i] = bf2f(lds.hprev_st[curp][row][j]);
                 }
+            if (dbg & 2u) {  // timing: gate math stripped
+#pragma unroll
+                for (int mt = 0; mt < 2; ++mt)
+#pragma unroll
+                    for (int i = 0; i < 4; ++i) {
+                        const int row = mt * 16 + lrow * 4 + i;
+                        const float v = dh8[mt][i] + hp8[mt][i];
+                        dhp_part[mt][i] = v;
+                        dxn8[mt][i] = v;
+                        lds.dhg[curp][row][0 * H + j] = f2bf(v);
+                        lds.dhg[curp][row][1 * H + j] = f2bf(v);
+                        lds.dhg[curp][row][2 * H + j] = f2bf(v);
+                    }
+            } else
 #pragma unroll
             for (int mt = 0; mt < 2; ++mt)
 #pragma unroll
@@ -606,13 +623,15 @@ __global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
         f32x4 acc[2];
 #pragma unroll
         for (int mt = 0; mt < 2; ++mt) acc[mt] = f32x4{0.f, 0.f, 0.f, 0.f};
+        if (!(dbg & 4u)) {
 #pragma unroll
-        for (int kb = 0; kb < 12; ++kb) {
+            for (int kb = 0; kb < 12; ++kb) {
 #pragma unroll
-            for (int mt = 0; mt < 2; ++mt) {
-                bf16x8 a = lds_load_a_frag(&lds.dhg[curp][0][0], mt * 16,
-                                           kb * 32, G3 + 8);
-                acc[mt] = mfma16x16x32(a, ufrag[kb], acc[mt]);
+                for (int mt = 0; mt < 2; ++mt) {
+                    bf16x8 a = lds_load_a_frag(&lds.dhg[curp][0][0], mt * 16,
+                                               kb * 32, G3 + 8);
+                    acc[mt] = mfma16x16x32(a, ufrag[kb], acc[mt]);
+                }
             }
         }
 #pragma unroll
@@ -623,7 +642,7 @@ __global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
         // ---- store dxg + dhg tiles (GEMM-ready layouts, no host cats) -----
         // dxg = [dxr dxz dxn]: the dxr/dxz blocks copy wide from lds.dhg,
         // dxn comes from this lane's registers (scalar stores)
-        {
+        if (!(dbg & 1u)) {
             bf16* dst = dxg + (((size_t)t * B + b0) * 2 + dir) * G3;
             const int row = tid / 32, col = (tid % 32) * 8;  // 2H block
 #pragma unroll
@@ -642,7 +661,7 @@ __global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
                         f2bf(dxn8[mt][i]);
                 }
         }
-        {
+        if (!(dbg & 1u)) {
             // dhg tensor = [dxr dxz dhgn] — exactly lds.dhg's live columns
             bf16* dst = dhg + (((size_t)dir * T + t) * B + b0) * G3;
             const int row = tid / 64, col = (tid % 64) * 8;
@@ -667,13 +686,14 @@ __global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
 
 void gru_layer_bwd(const void* cache, const void* hseq, const void* dhin,
                    const void* ut, void* dxg, void* dhg, int T, int B,
-                   hipStream_t stream) {
+                   hipStream_t stream, uint32_t dbg) {
     dim3 grid(B / MB, 2);
     dim3 block(BW_WAVES * 64);
     hipLaunchKernelGGL(gru_layer_bwd_kernel, grid, block, 0, stream,
                        static_cast<const bf16*>(cache), static_cast<const bf16*>(hseq),
                        static_cast<const bf16*>(dhin), static_cast<const bf16*>(ut),
-                       static_cast<bf16*>(dxg), static_cast<bf16*>(dhg), T, B);
+                       static_cast<bf16*>(dxg), static_cast<bf16*>(dhg), T, B,
+                       dbg);
 }
 
 }  // namespace rk
