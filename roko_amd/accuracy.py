@@ -55,6 +55,52 @@ def seq_stats(query: str, truth: str, band: int = 0) -> Dict[str, float]:
     }
 
 
+def main(argv=None) -> None:
+    """CLI: score an assembly (and optionally the pre-polish draft) against
+    a truth FASTA — the in-repo equivalent of pomoxis assess_assembly.
+
+    Usage: python -m roko_amd.accuracy polished.fasta truth.fasta
+               [--draft draft.fasta] [--band N]
+    Contigs are matched by name; the summary aggregates over matched pairs.
+    """
+    import argparse
+
+    from .io.fasta import read_fasta
+
+    p = argparse.ArgumentParser(description=main.__doc__)
+    p.add_argument("assembly", help="polished (or any) assembly FASTA")
+    p.add_argument("truth", help="truth FASTA (contig names must match)")
+    p.add_argument("--draft", default=None,
+                   help="pre-polish draft FASTA: also report error reduction")
+    p.add_argument("--band", type=int, default=0,
+                   help="alignment band (0 = auto, grows on overflow)")
+    a = p.parse_args(argv)
+
+    asm = dict(read_fasta(a.assembly))
+    tru = dict(read_fasta(a.truth))
+    dra = dict(read_fasta(a.draft)) if a.draft else {}
+    names = sorted(set(asm) & set(tru))
+    if not names:
+        raise SystemExit("no contig names shared between assembly and truth")
+    tot_e = tot_n = 0
+    for name in names:
+        st = seq_stats(asm[name], tru[name], a.band)
+        line = (f"{name}: err {st['total_error']:.4%} "
+                f"(mm {st['mismatch']:.4%} ins {st['insertion']:.4%} "
+                f"del {st['deletion']:.4%}) Q{st['qscore']:.2f}")
+        if name in dra:
+            d = seq_stats(dra[name], tru[name], a.band)
+            red = (1 - st["total_error"] / d["total_error"]
+                   if d["total_error"] > 0 else 0.0)
+            line += f"  [draft err {d['total_error']:.4%}, reduction {red:.1%}]"
+        print(line)
+        tot_e += st["edit_distance"]
+        tot_n += len(tru[name])
+    err = tot_e / max(tot_n, 1)
+    q = (-10.0 * math.log10(err)) if err > 0 else float("inf")
+    print(f"TOTAL: err {err:.4%} Q{q:.2f} over {len(names)} contig(s)")
+
+
 def assess_polishing(draft: str, polished: str, truth: str,
                      band: int = 0) -> Dict[str, object]:
     """Compare draft-vs-truth and polished-vs-truth error; the headline
@@ -65,3 +111,7 @@ def assess_polishing(draft: str, polished: str, truth: str,
     red = (1.0 - p["total_error"] / d["total_error"]
            if d["total_error"] > 0 else 0.0)
     return {"draft": d, "polished": p, "error_reduction": red}
+
+
+if __name__ == "__main__":
+    main()

@@ -113,3 +113,19 @@ def test_polishing_beats_draft(tiny_assembly, tmp_path):
     print(f"accuracy gate: draft_err={res['draft']['total_error']:.4%} "
           f"polished_err={res['polished']['total_error']:.4%} "
           f"reduction={res['error_reduction']:.3f}")
+
+
+def test_accuracy_cli(tmp_path, capsys):
+    from roko_amd.accuracy import main as acc_main
+    from roko_amd.io.fasta import write_fasta
+
+    truth = "ACGTACGTAC" * 50
+    draft = truth[:200] + "T" + truth[201:]  # one substitution
+    write_fasta(str(tmp_path / "t.fa"), [("c1", truth)])
+    write_fasta(str(tmp_path / "a.fa"), [("c1", truth)])  # perfect assembly
+    write_fasta(str(tmp_path / "d.fa"), [("c1", draft)])
+    acc_main([str(tmp_path / "a.fa"), str(tmp_path / "t.fa"),
+              "--draft", str(tmp_path / "d.fa")])
+    out = capsys.readouterr().out
+    assert "TOTAL: err 0.0000%" in out
+    assert "reduction 100.0%" in out

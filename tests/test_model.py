@@ -78,3 +78,19 @@ def test_param_count_close_to_reference():
     # (identical layer dims, asserted above) is 1,099,731
     n = sum(p.numel() for p in RokoModel().parameters())
     assert 1_000_000 < n < 1_200_000
+
+
+def test_select_fused_path_logic(monkeypatch):
+    """The train CLI's GPU-default stepper selection."""
+    import torch
+
+    import roko_amd.train as T
+    from roko_amd.config import TrainConfig
+
+    monkeypatch.setattr("roko_amd.ops.train.train_step_available", lambda: True)
+    cpu, gpu = torch.device("cpu"), torch.device("cuda", 0)
+    assert not T._select_fused_path(cpu, TrainConfig())
+    assert T._select_fused_path(gpu, TrainConfig(batch_size=128))
+    assert not T._select_fused_path(gpu, TrainConfig(batch_size=100))  # %32
+    monkeypatch.setenv("ROKO_TRAIN_PATH", "autograd")
+    assert not T._select_fused_path(gpu, TrainConfig(batch_size=128))
