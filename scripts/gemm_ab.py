@@ -46,3 +46,18 @@ for label, M, N, K in shapes:
     rel = (got - ref).norm() / ref.norm()
     print(f"{label:28s} addmm {t_lt:7.1f}  mm {t_mm:7.1f}  "
           f"gemm_bias {t_ours:7.1f} us  rel {rel.item():.4f}")
+
+# train-path layout question: addmm with a TRANSPOSED-VIEW B picks hipBLASLt
+# MT16x16x256 (24 us) while the contiguous layout gets MT256x160 (16 us)
+print("\n-- B-operand layout (train xg GEMM shapes) --")
+for label, M, N, K in shapes[:2]:
+    A = (torch.randn(M, K, device="cuda") * 0.3).to(torch.bfloat16)
+    W = (torch.randn(N, K, device="cuda") * 0.3).to(torch.bfloat16)  # (768,K)
+    bias = torch.randn(N, device="cuda", dtype=torch.bfloat16)
+    out = torch.empty(M, N, device="cuda", dtype=torch.bfloat16)
+    t_view = bench(lambda: torch.addmm(bias, A, W.t(), out=out))
+    def with_copy():
+        Wt = W.t().contiguous()
+        torch.addmm(bias, A, Wt, out=out)
+    t_copy = bench(with_copy)
+    print(f"{label:28s} B=view {t_view:7.1f}  B=copy(+transpose) {t_copy:7.1f} us")
