@@ -48,8 +48,21 @@ bool Bgzf::load_block(uint64_t coffset) {
         phys_eof_ = true;
         return false;
     }
-    if (got < 18 || hdr[0] != 0x1f || hdr[1] != 0x8b || hdr[2] != 8 || !(hdr[3] & 4))
+    if (got < 18 || hdr[0] != 0x1f || hdr[1] != 0x8b || hdr[2] != 8 || !(hdr[3] & 4)) {
+        // recognise the common wrong-format cases and say so (this reader
+        // supports coordinate-sorted BAM+BAI only; CRAM/SAM are documented
+        // out of scope — convert with `samtools view -b`)
+        if (got >= 4 && std::memcmp(hdr, "CRAM", 4) == 0)
+            throw std::runtime_error(
+                "input is a CRAM file; this framework reads BAM only — "
+                "convert with `samtools view -b -o out.bam in.cram`");
+        if (got >= 3 && hdr[0] == '@' &&
+            (hdr[1] == 'H' || hdr[1] == 'S' || hdr[1] == 'R' || hdr[1] == 'P'))
+            throw std::runtime_error(
+                "input looks like uncompressed SAM text; this framework "
+                "reads BAM only — convert with `samtools view -b`");
         throw std::runtime_error("corrupt BGZF block header");
+    }
     uint16_t xlen = rd_u16(hdr + 10);
     // Find the BC subfield carrying BSIZE. The fixed 18-byte read already
     // includes the first 6 bytes of the extra field (the common case where

@@ -78,3 +78,22 @@ def test_multi_ref(tmp_path):
     write_bam(path, [("chrA", 100), ("chrB", 100)], recs)
     assert [r[0] for r in _pileup.fetch_records(path, "chrA", 0, 100)] == ["a"]
     assert [r[0] for r in _pileup.fetch_records(path, "chrB", 0, 100)] == ["b"]
+
+
+def test_cram_and_sam_inputs_rejected_with_clear_message(tmp_path):
+    """BAM+BAI only is a documented scope cut (README); the reader must say
+    what the input was and how to convert, not 'corrupt BGZF'."""
+    import pytest
+
+    from roko_amd.ops import pileup_ext
+
+    px = pileup_ext()
+    cram = tmp_path / "x.cram"
+    cram.write_bytes(b"CRAM\x03\x00" + b"\x00" * 64)
+    with pytest.raises(Exception, match="CRAM.*samtools view"):
+        px.bam_references(str(cram))
+
+    sam = tmp_path / "x.sam"
+    sam.write_text("@HD\tVN:1.6\n@SQ\tSN:c1\tLN:100\n")
+    with pytest.raises(Exception, match="SAM text.*samtools view"):
+        px.bam_references(str(sam))
