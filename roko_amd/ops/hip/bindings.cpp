@@ -22,7 +22,8 @@ void embed_mlp_fwd2(const uint8_t* ids, const void* w1g, const float* b1,
                     void* out, int B, hipStream_t stream);
 void embed_mlp_fwd3(const uint8_t* ids, const void* w1g, const float* b1,
                     const void* w2, const float* b2, const void* emb,
-                    void* out, int B, hipStream_t stream);
+                    void* out, int B, hipStream_t stream,
+                    unsigned long long* timing);
 void gru_layer_fwd(const void* xg, const void* u, const float* bhh, void* hseq,
                    void* cache, int T, int B, hipStream_t stream, uint32_t dbg);
 void gru_layer_fwd_fused(const void* x, const void* w_ih_p, const void* b_ih,
@@ -141,7 +142,8 @@ torch::Tensor embed_mlp_fwd2(torch::Tensor ids, torch::Tensor w1g,
 // same contract as embed_mlp_fwd2, wave-private-column kernel (v3)
 torch::Tensor embed_mlp_fwd3(torch::Tensor ids, torch::Tensor w1g,
                              torch::Tensor b1, torch::Tensor w2,
-                             torch::Tensor b2, torch::Tensor emb) {
+                             torch::Tensor b2, torch::Tensor emb,
+                             c10::optional<torch::Tensor> timing = c10::nullopt) {
     check(ids, torch::kUInt8, "ids");
     check(w1g, torch::kBFloat16, "w1g");
     check(b1, torch::kFloat32, "b1");
@@ -156,7 +158,10 @@ torch::Tensor embed_mlp_fwd3(torch::Tensor ids, torch::Tensor w1g,
     rk::embed_mlp_fwd3(ids.data_ptr<uint8_t>(), w1g.data_ptr(),
                        b1.data_ptr<float>(), w2.data_ptr(),
                        b2.data_ptr<float>(), emb.data_ptr(), out.data_ptr(),
-                       B, cur_stream());
+                       B, cur_stream(),
+                       timing ? reinterpret_cast<unsigned long long*>(
+                                    timing->data_ptr<int64_t>())
+                              : nullptr);
     return out;
 }
 
@@ -570,12 +575,16 @@ struct ServeSlot {
             at::cuda::CUDAStreamGuard guard(stream);
             hipStream_t s = stream.stream();
             if (n > 0) x_buf.narrow(0, 0, n).copy_(x.narrow(0, 0, n), true);
-            if (w1gt.defined())
-                (use_v3 ? rk::embed_mlp_fwd3 : rk::embed_mlp_fwd2)(
-                    x_buf.data_ptr<uint8_t>(), w1gt.data_ptr(),
-                    b1.data_ptr<float>(), w2.data_ptr(),
-                    b2.data_ptr<float>(), emb.data_ptr(),
-                    seq.data_ptr(), B, s);
+            if (w1gt.defined() && use_v3)
+                rk::embed_mlp_fwd3(x_buf.data_ptr<uint8_t>(), w1gt.data_ptr(),
+                                   b1.data_ptr<float>(), w2.data_ptr(),
+                                   b2.data_ptr<float>(), emb.data_ptr(),
+                                   seq.data_ptr(), B, s, nullptr);
+            else if (w1gt.defined())
+                rk::embed_mlp_fwd2(x_buf.data_ptr<uint8_t>(), w1gt.data_ptr(),
+                                   b1.data_ptr<float>(), w2.data_ptr(),
+                                   b2.data_ptr<float>(), emb.data_ptr(),
+                                   seq.data_ptr(), B, s);
             else
                 rk::embed_mlp_fwd(x_buf.data_ptr<uint8_t>(), w1.data_ptr(),
                                   b1.data_ptr<float>(), w2.data_ptr(),
@@ -662,7 +671,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.doc() = "roko-mi355x CDNA4 kernels (gfx950)";
     m.def("mfma_probe", &mfma_probe);
     m.def("embed_mlp_fwd2", &embed_mlp_fwd2);
-    m.def("embed_mlp_fwd3", &embed_mlp_fwd3);
+    m.def("embed_mlp_fwd3", &embed_mlp_fwd3, py::arg("ids"), py::arg("w1g"),
+          py::arg("b1"), py::arg("w2"), py::arg("b2"), py::arg("emb"),
+          py::arg("timing") = c10::nullopt);
     m.def("embed_mlp_fwd", &embed_mlp_fwd, py::arg("ids"), py::arg("w1"),
           py::arg("b1"), py::arg("w2"), py::arg("b2"), py::arg("emb"),
           py::arg("dbg") = 0, py::arg("timing") = c10::nullopt);

@@ -473,7 +473,9 @@ __global__ __launch_bounds__(WAVES3 * 64, 1) void embed_mlp_fwd3_kernel(
     const float* __restrict__ b2,     // (F2)
     const bf16* __restrict__ emb,     // (12, E)
     bf16* __restrict__ out,           // (W, B, OUT)
-    int B) {
+    int B,
+    unsigned long long* __restrict__ timing) {  // optional (6): per-phase
+                                                // cycle sums from lane 0
     __shared__ struct {
         // shared, read-only after staging
         bf16 e_t[64][32];        // G2 B-operand [e][k=c] (zero-padded)
@@ -521,7 +523,8 @@ __global__ __launch_bounds__(WAVES3 * 64, 1) void embed_mlp_fwd3_kernel(
     bf16* t2 = lds.t2[wid];
     const uint8_t* win = ids + (size_t)b * R * W;
 
-    // loop-invariant operand fragments, register-resident for all columns
+    // loop-invariant operand fragments + biases, register-resident for all
+    // columns (the per-column scalar b1s/b2s LDS reads were ~112/col)
     bf16x8 e_bv[4], w2_av[4];
 #pragma unroll
     for (int nt = 0; nt < 4; ++nt)
@@ -529,9 +532,27 @@ __global__ __launch_bounds__(WAVES3 * 64, 1) void embed_mlp_fwd3_kernel(
 #pragma unroll
     for (int kb = 0; kb < 4; ++kb)
         w2_av[kb] = lds_load_a_frag(&lds.w2_lds[0][0], 0, kb * 32, LD1);
+    float b1r[7][4], b2r[4];
+#pragma unroll
+    for (int mt = 0; mt < 7; ++mt)
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+            const int f = mt * 16 + lrow * 4 + i;
+            b1r[mt][i] = (f < F1) ? lds.b1s[f] : 0.f;
+        }
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+        b2r[i] = lds.b2s[lrow * 4 + i];
 
+    unsigned long long tacc[6] = {0, 0, 0, 0, 0, 0};
+#define V3_T0 unsigned long long tp0 = (timing && tid == 0) \
+        ? __builtin_amdgcn_s_memtime() : 0
+#define V3_T1(i) if (timing && tid == 0) { \
+        unsigned long long tn = __builtin_amdgcn_s_memtime(); \
+        tacc[i] += tn - tp0; tp0 = tn; }
     // wave w handles columns w, w+WAVES3, ... — each fully privately
     for (int w = wid; w < W; w += WAVES3) {
+        V3_T0;
         // ---- scatter: 200 one-hots (ids read straight through L1) ---------
         // lane r and r+64... handle reads r, r+64, r+128 (200 = 3*64 + 8)
         uint8_t myid[4];
@@ -544,6 +565,7 @@ __global__ __launch_bounds__(WAVES3 * 64, 1) void embed_mlp_fwd3_kernel(
 #pragma unroll
         for (int q = 0; q < 4; ++q)
             if (q < nr) hot[myid[q]][lane + q * 64] = f2bf(1.0f);
+        V3_T1(0);
 
         // ---- G1: A = W1 · Hot  (7 m-tiles x 7 k-steps, 49 MFMA) -----------
         // 7 independent accumulator chains; B-frag per k shared across m.
@@ -592,6 +614,7 @@ __global__ __launch_bounds__(WAVES3 * 64, 1) void embed_mlp_fwd3_kernel(
             for (int q = 0; q < 4; ++q)
                 if (q < nr) hot[myid[q]][lane + q * 64] = f2bf(0.0f);
         }
+        V3_T1(1);
 
         // ---- G2: t1 = relu(A · E + b1)  (7m x 4n single-K, 28 MFMA) -------
         {
@@ -614,9 +637,7 @@ __global__ __launch_bounds__(WAVES3 * 64, 1) void embed_mlp_fwd3_kernel(
                         bf16x4 pk;
 #pragma unroll
                         for (int i = 0; i < 4; ++i) {
-                            const int f = mt * 16 + lrow * 4 + i;
-                            const float v =
-                                acc[mt][nt][i] + (f < F1 ? lds.b1s[f] : 0.f);
+                            const float v = acc[mt][nt][i] + b1r[mt][i];
                             pk[i] = f2bf(fmaxf(v, 0.f));
                         }
                         *reinterpret_cast<bf16x4*>(
@@ -625,6 +646,7 @@ __global__ __launch_bounds__(WAVES3 * 64, 1) void embed_mlp_fwd3_kernel(
                 }
         }
 
+        V3_T1(2);
         // ---- G3: t2 = relu(W2 · t1 + b2)  (4 n-tiles x K=112, 16 MFMA) ----
         {
 #pragma unroll
@@ -648,14 +670,14 @@ __global__ __launch_bounds__(WAVES3 * 64, 1) void embed_mlp_fwd3_kernel(
 #pragma unroll
                     for (int i = 0; i < 4; ++i)
                         if (j + i < F2) {
-                            const float v =
-                                acc0[i] + acc1[i] + lds.b2s[j + i];
+                            const float v = acc0[i] + acc1[i] + b2r[i];
                             t2[e * F2 + j + i] = f2bf(fmaxf(v, 0.f));
                         }
                 }
             }
         }
 
+        V3_T1(3);
         // ---- coalesced store (wave-wide b128) -----------------------------
         {
             bf16* dst = out + ((size_t)w * B + b) * OUT;
@@ -666,20 +688,27 @@ __global__ __launch_bounds__(WAVES3 * 64, 1) void embed_mlp_fwd3_kernel(
             else if (e8 < OUT)
                 for (int q = e8; q < OUT; ++q) dst[q] = t2[q];
         }
+        V3_T1(4);
     }
+    if (timing && tid == 0)
+#pragma unroll
+        for (int i = 0; i < 5; ++i) atomicAdd(&timing[i], tacc[i]);
+#undef V3_T0
+#undef V3_T1
 }
 
 }  // namespace v3
 
 void embed_mlp_fwd3(const uint8_t* ids, const void* w1g, const float* b1,
                     const void* w2, const float* b2, const void* emb,
-                    void* out, int B, hipStream_t stream) {
+                    void* out, int B, hipStream_t stream,
+                    unsigned long long* timing) {
     hipLaunchKernelGGL(v3::embed_mlp_fwd3_kernel, dim3(B),
                        dim3(v3::WAVES3 * 64), 0, stream, ids,
                        static_cast<const bf16*>(w1g), b1,
                        static_cast<const bf16*>(w2), b2,
                        static_cast<const bf16*>(emb), static_cast<bf16*>(out),
-                       B);
+                       B, timing);
 }
 
 void embed_mlp_fwd2(const uint8_t* ids, const void* w1g, const float* b1,
