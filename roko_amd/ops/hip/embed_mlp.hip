@@ -619,14 +619,16 @@ __global__ __launch_bounds__(WAVES3 * 64, 1) void embed_mlp_fwd3_kernel(
         // ---- G2: t1 = relu(A · E + b1)  (7m x 4n single-K, 28 MFMA) -------
         {
             f32x4 acc[7][4];
+            bf16x8 av[7];
 #pragma unroll
-            for (int mt = 0; mt < 7; ++mt) {
-                const bf16x8 af = lds_load_a_frag(&at[0][0], mt * 16, 0, 40);
+            for (int mt = 0; mt < 7; ++mt)
+                av[mt] = lds_load_a_frag(&at[0][0], mt * 16, 0, 40);
+#pragma unroll
+            for (int mt = 0; mt < 7; ++mt)
 #pragma unroll
                 for (int nt = 0; nt < 4; ++nt)
-                    acc[mt][nt] = mfma16x16x32(af, e_bv[nt],
+                    acc[mt][nt] = mfma16x16x32(av[mt], e_bv[nt],
                                                f32x4{0.f, 0.f, 0.f, 0.f});
-            }
 #pragma unroll
             for (int mt = 0; mt < 7; ++mt)
 #pragma unroll
