@@ -379,6 +379,9 @@ class H5File:
             data = fh.read()
         if data[:8] != _SIG:
             raise ValueError(f"{path}: not an HDF5 file")
+        if len(data) < 96:
+            raise ValueError(f"{path}: truncated HDF5 file "
+                             f"({len(data)} bytes, superblock needs 96)")
         self.b = _Buf(data)
         ver = self.b.u8(8)
         if ver not in (0, 1):

@@ -185,3 +185,14 @@ def test_chunked_dataset_roundtrip(tmp_path):
     assert np.array_equal(np.asarray(f["g"]["a"]), a)
     assert np.array_equal(np.asarray(f["g"]["b"]), b)
     assert np.array_equal(np.asarray(f["g"]["c"]), a[:3])
+
+
+def test_reader_rejects_garbage(tmp_path):
+    bad = tmp_path / "x.hdf5"
+    bad.write_bytes(b"not an hdf5 file at all")
+    with pytest.raises(ValueError, match="not an HDF5"):
+        H5File(str(bad))
+    trunc = tmp_path / "t.hdf5"
+    trunc.write_bytes(b"\x89HDF\r\n\x1a\n\x00\x00")
+    with pytest.raises(ValueError, match="truncated"):
+        H5File(str(trunc))
