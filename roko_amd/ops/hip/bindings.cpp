@@ -59,6 +59,8 @@ void front_bwd(const uint8_t* ids, const void* dseq, const void* w1,
                const uint32_t* seed_ptr, const void* w1g);
 void gemm_bias(const void* A, const void* B, const float* bias, void* C,
                int M, int N, int K, hipStream_t stream);
+void xg_gemm(const void* A, const void* B, const void* bias, void* C, int M,
+             int KP, hipStream_t stream);
 int atb_splitk_nslices(int K);
 void atb_splitk(const void* A, const void* B, float* ws, float* C, int M,
                 int N, int K, hipStream_t stream);
@@ -371,6 +373,21 @@ torch::Tensor front_fwd(torch::Tensor ids, torch::Tensor w1, torch::Tensor b1,
                   out.data_ptr(), B, (uint32_t)seed, (float)keep, cur_stream(),
                   seed_ptr_of(seed_buf));
     return out;
+}
+
+// serving xg projection: A (M, 256|512) bf16 x Bt (768, KP) + bias -> (M, 768)
+torch::Tensor xg_gemm(torch::Tensor A, torch::Tensor Bt, torch::Tensor bias) {
+    check(A, torch::kBFloat16, "A");
+    check(Bt, torch::kBFloat16, "Bt");
+    check(bias, torch::kBFloat16, "bias");
+    const int M = A.size(0), KP = A.size(1);
+    TORCH_CHECK(KP == 256 || KP == 512, "A must be (M, 256|512) K-padded");
+    TORCH_CHECK(Bt.size(0) == 768 && Bt.size(1) == KP, "Bt must be (768, KP)");
+    TORCH_CHECK(M % 256 == 0, "M must be a multiple of 256");
+    auto C = torch::empty({M, 768}, A.options());
+    rk::xg_gemm(A.data_ptr(), Bt.data_ptr(), bias.data_ptr(), C.data_ptr(),
+                M, KP, cur_stream());
+    return C;
 }
 
 // fused train front bwd: -> (de, dw1, db1, dw2, db2) fp32
@@ -701,6 +718,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("atb_splitk", &atb_splitk);
     m.def("gemm_bias", &gemm_bias, py::arg("A"), py::arg("B"),
           py::arg("bias") = c10::nullopt);
+    m.def("xg_gemm", &xg_gemm);
     m.def("front_de_timed", &front_de_timed, py::arg("ids"), py::arg("dt1g"),
           py::arg("w1"), py::arg("seed"), py::arg("keep"), py::arg("dbg") = 0);
     m.def("front_bwd", &front_bwd, py::arg("ids"), py::arg("dseq"),

@@ -319,6 +319,134 @@ __global__ __launch_bounds__(256) void slice_sum_kernel(
 
 }  // namespace gemmatb
 
+// ---------------------------------------------------------------------------
+// Serving xg projection GEMM:  C (M, 768) = A (M, KP) · B (KP, 768) + bias
+// ---------------------------------------------------------------------------
+// hipBLASLt runs this shape (M = T*B = 11520, N = 768, K <= 512) at
+// ~543 TF/s on ~225 workgroups — ~105 CU·us of aggregate chip time per
+// served window, the single largest serving cost after the front
+// (profiles/infer_r02_v3_kernel_stats.csv). This kernel trades LDS staging
+// for an LDS-FREE, BARRIER-FREE design shaped for the pipelined server:
+//   * grid = (M/384) x 6: 180 workgroups — under one full wave of the
+//     256-CU chip, so concurrent serving chains fill the leftover CUs;
+//   * each wave owns a PRIVATE 48x128 C tile (3 m-subtiles x 8 n-subtiles,
+//     24 accumulators in AGPR space) and reads its A/B fragments straight
+//     from global (A: aligned b128 rows — the caller pads K to 512/256;
+//     B: L2-resident (KP,768) slice reused by the 30 m-workgroups of its
+//     n-range), software-pipelined one k-step ahead;
+//   * no __syncthreads anywhere: nothing is shared between waves.
+namespace xggemm {
+
+// MEASURED NEGATIVE (kept for the record, not wired into serving): both
+// tilings lose to hipBLASLt — 48x128/wave: 47.9 us vs addmm 25.6; 32x64:
+// 68.9 us. Without LDS staging every wave re-reads its whole B slice from
+// L2 (no cross-wave operand reuse), and the compiler's vmcnt scoreboard
+// still inserts ~44 full drains per kernel under 180+ live fragment
+// registers. The library's LDS-staged tiles win this shape; serving keeps
+// torch.addmm (see docs/KERNELS.md lesson 13-15).
+constexpr int XN = 768;
+constexpr int WM = 32;    // rows per wave (2 m-subtiles)
+constexpr int BN = 64;    // cols per workgroup (4 n-subtiles per wave)
+constexpr int WAVES = 8;  // 8 waves x 32 rows = 256 rows per workgroup
+
+template <int KP>
+__global__ __launch_bounds__(WAVES * 64, 2) void xg_gemm_kernel(
+    const bf16* __restrict__ A,     // (M, KP) row-major, 16-B-aligned rows
+    const bf16* __restrict__ Bt,    // (768, KP) row-major = B^T (the
+                                    // serving cache's padded W_ih image)
+    const bf16* __restrict__ bias,  // (768)
+    bf16* __restrict__ C,           // (M, 768)
+    int M) {
+    constexpr int KB = KP / 32;
+    const int tid = threadIdx.x;
+    const int wid = tid >> 6;
+    const int lane = tid & 63;
+    const int lrow = lane >> 4;
+    const int lcol = lane & 15;
+    const int m0 = blockIdx.x * (WAVES * WM) + wid * WM;
+    const int n0 = blockIdx.y * BN;
+
+    f32x4 acc[2][4];
+#pragma unroll
+    for (int mt = 0; mt < 2; ++mt)
+#pragma unroll
+        for (int nt = 0; nt < 4; ++nt) acc[mt][nt] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+    // fragment loaders (all global; A rows aligned because KP % 8 == 0)
+    auto a_frag = [&](int mt, int kb) {
+        return global_load_a_frag(A + (size_t)m0 * KP, mt * 16, kb * 32, KP);
+    };
+    auto b_frag = [&](int nt, int kb) {
+        // B^T (768, KP) row-major is the [col][k] layout: the B-fragment
+        // load is the A-fragment address pattern — aligned b128 per lane
+        return global_load_a_frag(Bt, n0 + nt * 16, kb * 32, KP);
+    };
+
+    // two explicit register sets, alternated by an unrolled two-body loop:
+    // selecting the sets through references compiled to 44 full vmcnt(0)
+    // drains (the compiler loses per-load scoreboard tracking — the same
+    // failure gru.hip's stage rotation hit, docs/KERNELS.md lesson 2)
+    bf16x8 afA[2], afB[2], bfA[4], bfB[4];
+    auto load_set = [&](int kb, bf16x8(&af)[2], bf16x8(&bf)[4]) {
+#pragma unroll
+        for (int mt = 0; mt < 2; ++mt) af[mt] = a_frag(mt, kb);
+#pragma unroll
+        for (int nt = 0; nt < 4; ++nt) bf[nt] = b_frag(nt, kb);
+    };
+    auto mfma_set = [&](bf16x8(&af)[2], bf16x8(&bf)[4]) {
+#pragma unroll
+        for (int mt = 0; mt < 2; ++mt)
+#pragma unroll
+            for (int nt = 0; nt < 4; ++nt)
+                acc[mt][nt] = mfma16x16x32(af[mt], bf[nt], acc[mt][nt]);
+    };
+    load_set(0, afA, bfA);
+#pragma unroll
+    for (int kb = 0; kb < KB; kb += 2) {
+        if (kb + 1 < KB) load_set(kb + 1, afB, bfB);
+        mfma_set(afA, bfA);
+        if (kb + 2 < KB) load_set(kb + 2, afA, bfA);
+        if (kb + 1 < KB) mfma_set(afB, bfB);
+    }
+
+    // epilogue: + bias, scalar bf16 stores (16-lane 32-B runs; the C rows
+    // this wave owns are private, no staging needed)
+    float br[4];
+#pragma unroll
+    for (int nt = 0; nt < 4; ++nt) br[nt] = bf2f(bias[n0 + nt * 16 + lcol]);
+#pragma unroll
+    for (int mt = 0; mt < 2; ++mt)
+#pragma unroll
+        for (int nt = 0; nt < 4; ++nt)
+#pragma unroll
+            for (int i = 0; i < 4; ++i) {
+                const int row = m0 + mt * 16 + lrow * 4 + i;
+                if (row < M)
+                    C[(size_t)row * XN + n0 + nt * 16 + lcol] =
+                        f2bf(acc[mt][nt][i] + br[nt]);
+            }
+}
+
+}  // namespace xggemm
+
+void xg_gemm(const void* A, const void* B, const void* bias, void* C, int M,
+             int KP, hipStream_t stream) {
+    dim3 grid(M / (xggemm::WAVES * xggemm::WM), xggemm::XN / xggemm::BN);
+    dim3 block(xggemm::WAVES * 64);
+    if (KP == 512)
+        hipLaunchKernelGGL(xggemm::xg_gemm_kernel<512>, grid, block, 0,
+                           stream, static_cast<const bf16*>(A),
+                           static_cast<const bf16*>(B),
+                           static_cast<const bf16*>(bias),
+                           static_cast<bf16*>(C), M);
+    else if (KP == 256)
+        hipLaunchKernelGGL(xggemm::xg_gemm_kernel<256>, grid, block, 0,
+                           stream, static_cast<const bf16*>(A),
+                           static_cast<const bf16*>(B),
+                           static_cast<const bf16*>(bias),
+                           static_cast<bf16*>(C), M);
+}
+
 int atb_splitk_nslices(int K) {
     return (K + gemmatb::KSLICE - 1) / gemmatb::KSLICE;
 }

@@ -61,3 +61,22 @@ for label, M, N, K in shapes[:2]:
         torch.addmm(bias, A, Wt, out=out)
     t_copy = bench(with_copy)
     print(f"{label:28s} B=view {t_view:7.1f}  B=copy(+transpose) {t_copy:7.1f} us")
+
+# custom LDS-free xg_gemm vs hipBLASLt at the padded serving shapes
+print("\n-- xg_gemm (custom) --")
+for KP, Kr in ((512, 500), (256, 256)):
+    M = 90 * 128
+    A = torch.zeros(M, KP, device="cuda", dtype=torch.bfloat16)
+    A[:, :Kr] = (torch.randn(M, Kr, device="cuda") * 0.3).to(torch.bfloat16)
+    Bt = (torch.randn(768, KP, device="cuda") * 0.3).to(torch.bfloat16)
+    Bt[:, Kr:] = 0
+    bias = (torch.randn(768, device="cuda") * 0.3).to(torch.bfloat16)
+    out = torch.empty(M, 768, device="cuda", dtype=torch.bfloat16)
+    t_lt = bench(lambda: torch.addmm(bias, A, Bt.t(), out=out))
+    t_x = bench(lambda: ext.xg_gemm(A, Bt, bias))
+    ref = torch.addmm(bias.float(), A.float(), Bt.float().t())
+    got = ext.xg_gemm(A, Bt, bias).float()
+    rel = (got - ref).norm() / ref.norm()
+    gf = 2.0 * M * 768 * Kr
+    print(f"KP={KP}: addmm {t_lt:6.1f} us ({gf/t_lt/1e6:.0f} TF/s)   "
+          f"xg_gemm {t_x:6.1f} us ({gf/t_x/1e6:.0f} TF/s)  rel {rel:.4f}")
