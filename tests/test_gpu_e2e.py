@@ -126,3 +126,45 @@ def test_training_loss_decreases_on_real_windows(tmp_path):
     opt = FusedAdam(list(model.parameters()), lr=2e-3)
     losses = [float(fused_train_step(model, x, y, opt)) for _ in range(60)]
     assert losses[-1] < losses[0] * 0.6, (losses[0], losses[-1])
+
+
+@requires_gpu
+def test_fused_train_resume_on_gpu(tmp_path):
+    """Resume with the GPU-default FusedAdam stepper: moments and step
+    counter must round-trip through the sidecar (the CPU resume test only
+    covers the autograd/torch-Adam path)."""
+    import numpy as np
+
+    from roko_amd import config as C
+    from roko_amd.config import TrainConfig
+    from roko_amd.rkdata import RkwWriter
+    from roko_amd.train import train
+
+    path = str(tmp_path / "t.rkw")
+    rng = np.random.default_rng(0)
+    w = RkwWriter(path, inference=False)
+    n = 256
+    P = np.zeros((n, C.WINDOW_COLS, 2), dtype=np.int32)
+    P[..., 0] = np.arange(C.WINDOW_COLS)[None, :]
+    w.store("c1", 0, C.WINDOW_COLS, P,
+            rng.integers(0, 12, (n, C.WINDOW_ROWS, C.WINDOW_COLS), dtype=np.uint8),
+            rng.integers(0, 5, (n, C.WINDOW_COLS), dtype=np.uint8))
+    w.write_contigs([("c1", "A" * 200)])
+    w.close()
+
+    out = str(tmp_path / "ckpt")
+    logs = []
+    cfg = TrainConfig(batch_size=32, epochs=1, seed=3, in_memory=True)
+    train(path, out, cfg=cfg, log=logs.append)
+    assert any("fused HIP step" in str(m) for m in logs), logs
+    assert os.path.exists(os.path.join(out, "train_state.pt"))
+
+    st = torch.load(os.path.join(out, "train_state.pt"), map_location="cpu",
+                    weights_only=False)
+    assert st["opt"]["kind"] == "fused_adam"
+    assert st["opt"]["step_count"] == n // 32
+
+    cfg2 = TrainConfig(batch_size=32, epochs=2, seed=3, in_memory=True,
+                       resume=True)
+    _, h2 = train(path, out, cfg=cfg2, log=lambda *a, **k: None)
+    assert len(h2) == 1 and h2[0]["epoch"] == 2
