@@ -101,13 +101,14 @@ def bench_inference(args, rank, world, device):
 
 def bench_train(args, rank, world, device):
     from roko_amd.ops.train import (FusedAdam, GraphedTrainStep,
-                                    fused_train_step, train_step_available)
+                                    fused_param_order, fused_train_step,
+                                    train_step_available)
 
     if device.type != "cuda" or not train_step_available():
         raise SystemExit("train bench requires the fused HIP train step on GPU")
     torch.manual_seed(0)
     model = RokoModel().to(device).train()
-    opt = FusedAdam(list(model.parameters()), lr=C.LR)
+    opt = FusedAdam(fused_param_order(model), lr=C.LR)
     g = torch.Generator().manual_seed(99 + rank)
     x = torch.randint(0, C.NUM_BASE_IDS, (args.batch, C.WINDOW_ROWS, C.WINDOW_COLS),
                       generator=g, dtype=torch.uint8).to(device)

@@ -220,13 +220,15 @@ def train(
     model = RokoModel().to(device)
     use_fused = _select_fused_path(device, cfg)
     if use_fused:
-        from .ops.train import FusedAdam
+        from .ops.train import FusedAdam, fused_param_order
 
         # DP sync for the fused path is FusedAdam.allreduce_grads (one flat
         # all-reduce per step) — no GradReducer hooks (the fused backward
-        # assigns grads outside autograd's accumulate hooks)
+        # assigns grads outside autograd's accumulate hooks). The param
+        # order makes the GRU direction pairs flat-adjacent so the per-step
+        # weight packs are zero-copy views (ops/train.py fused_param_order).
         _broadcast_initial_state(model)
-        opt = FusedAdam(list(model.parameters()), lr=cfg.lr)
+        opt = FusedAdam(fused_param_order(model), lr=cfg.lr)
         reducer = None
     else:
         reducer = GradReducer(list(model.parameters()), cfg.bucket_bytes)
