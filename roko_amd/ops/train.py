@@ -168,10 +168,27 @@ def _packed_gru(model, l: int):
     u = _adjacent_view(uf, ur, (2, 3 * H, H))
     bhh = _adjacent_view(bhf, bhr, (2, 3 * H))
     if all(t is not None for t in (w_ih, b_ih, u, bhh)):
+        # packed bf16 views from the FusedAdam flat mirror, when present
+        # (same layout, same adjacency)
+        mf = getattr(wf, "_rk_bf16", None)
+        if mf is not None:
+            w_ih._rk_bf16 = _adjacent_view(
+                mf, wr._rk_bf16, (6 * H, wf.shape[1]))
+            b_ih._rk_bf16 = _adjacent_view(bf_._rk_bf16, br._rk_bf16,
+                                           (6 * H,))
+            u._rk_bf16 = _adjacent_view(uf._rk_bf16, ur._rk_bf16,
+                                        (2, 3 * H, H))
         cache[l] = (w_ih, b_ih, u, bhh)
         return cache[l]
     return (torch.cat([wf, wr], 0), torch.cat([bf_, br]),
             torch.stack([uf, ur]), torch.stack([bhf, bhr]))
+
+
+def _bf(t: torch.Tensor) -> torch.Tensor:
+    """bf16 image of a weight tensor: the FusedAdam flat-mirror view when
+    present (refreshed once per step), else a fresh cast."""
+    m = getattr(t, "_rk_bf16", None)
+    return m if m is not None else t.detach().to(torch.bfloat16)
 
 
 def _accum_pref_grads(prefs, dw_ih, db_ih, du, dbhh) -> None:
@@ -202,13 +219,13 @@ class GruLayerFn(torch.autograd.Function):
         ext = _ext()
         T, B, _ = x_seq.shape
         x_bf = x_seq.to(torch.bfloat16)
-        w_ih_bf = w_ih.detach().to(torch.bfloat16)
+        w_ih_bf = _bf(w_ih)
         xg = torch.addmm(
-            b_ih_all.detach().to(torch.bfloat16),
+            _bf(b_ih_all),
             x_bf.reshape(T * B, -1),
             w_ih_bf.t(),
         ).view(T, B, 2, 384).contiguous()
-        u_bf = u.detach().to(torch.bfloat16).contiguous()
+        u_bf = _bf(u).contiguous()
         bhh_f = bhh.detach().float().contiguous()
         hseq, cache = ext.gru_layer_fwd(xg, u_bf, bhh_f, True)
         ctx.save_for_backward(x_bf, w_ih_bf, u_bf, hseq, cache)
@@ -345,6 +362,18 @@ class FusedAdam:
         for p in self.params:
             p.grad = None
 
+    def refresh_bf16(self):
+        """One flat fp32->bf16 cast of every parameter; per-param bf16
+        views hang off the params as ``_rk_bf16`` (train_forward uses them
+        instead of ~26 small per-tensor .to(bfloat16) kernels per step).
+        Call after step() / before the next forward."""
+        if not hasattr(self, "flat_bf"):
+            self.flat_bf = torch.empty_like(self.flat_p,
+                                            dtype=torch.bfloat16)
+            for p, off in zip(self.params, self.offs):
+                p._rk_bf16 = self.flat_bf[off : off + p.numel()].view(p.shape)
+        self.flat_bf.copy_(self.flat_p)
+
     def state_dict(self) -> dict:
         """Resume sidecar payload (roko_amd.train CheckpointManager)."""
         return {
@@ -454,11 +483,11 @@ class FrontFn(torch.autograd.Function):
         # host RNG (a host randint would freeze the mask into the graph)
         seed = 0 if seed_buf is not None else int(
             torch.randint(0, 2**31 - 1, (1,)).item())
-        w1b = w1.detach().to(torch.bfloat16).contiguous()
+        w1b = _bf(w1).contiguous()
         b1f = b1.detach().float().contiguous()
-        w2b = w2.detach().to(torch.bfloat16).contiguous()
+        w2b = _bf(w2).contiguous()
         b2f = b2.detach().float().contiguous()
-        embb = emb_w.detach().to(torch.bfloat16).contiguous()
+        embb = _bf(emb_w).contiguous()
         out = ext.front_fwd(ids_u8, w1b, b1f, w2b, b2f, embb, seed, keep,
                             seed_buf)
         ctx.save_for_backward(ids_u8, w1b, b1f, w2b, b2f, embb)
@@ -558,6 +587,8 @@ def fused_train_step(model, x, y, opt: Optional[FusedAdam] = None,
         raise ValueError(
             "fused_train_step does not support GradReducer; DP sync runs "
             "through FusedAdam.allreduce_grads (pass opt=FusedAdam(...))")
+    if opt is not None and opt.on_gpu:
+        opt.refresh_bf16()
     logits = train_forward(model, x)
     loss = fused_cross_entropy(logits, y)
     if opt is not None:
@@ -685,6 +716,7 @@ def dual_stream_train_step(model, x, y, opt: FusedAdam, streams=None):
     s1.wait_stream(cur)
 
     opt.zero_grad()
+    opt.refresh_bf16()
     with torch.cuda.stream(s0):
         logits0 = train_forward(model, x[:h])
         loss0 = fused_cross_entropy(logits0, y[:h])
