@@ -299,6 +299,61 @@ class GruLayerFn(torch.autograd.Function):
         return dx, dw_ih, db_ih, du, dbhh, None
 
 
+class HeadFn(torch.autograd.Function):
+    """fc4 head through the fused GEMV kernel (fp32 logits straight from
+    bf16 activations) with a bf16 backward: replaces the fp32 aten linear
+    chain (~150 us/step of skinny fp32 GEMMs plus an 11.8 MB (T,B,256)
+    fp32 activation copy — profiles/train_r02_final_kernel_stats.csv).
+    Reference op: rnn_model.py:59."""
+
+    @staticmethod
+    def forward(ctx, seq, w4, b4):
+        ext = _ext()
+        w4b = _bf(w4).contiguous()
+        (logits,) = ext.head_fwd(seq.contiguous(), w4b,
+                                 b4.detach().float().contiguous(),
+                                 True, False)
+        ctx.save_for_backward(seq, w4b)
+        ctx.prefs = (w4, b4)
+        return logits  # (B, T, 5) fp32
+
+    @staticmethod
+    def backward(ctx, dlogits):
+        ext = _ext()
+        seq, w4b = ctx.saved_tensors
+        T, B, _ = seq.shape
+        dl = (dlogits.permute(1, 0, 2).reshape(T * B, C.NUM_CLASSES)
+              .to(torch.bfloat16).contiguous())
+        dseq = dl.mm(w4b).view(T, B, 2 * C.HIDDEN_SIZE)
+
+        def wgrads():
+            dw4 = ext.atb_splitk(dl,
+                                 seq.reshape(T * B, 2 * C.HIDDEN_SIZE)
+                                 .contiguous())
+            ones = dl.new_ones(1, T * B)
+            db4 = ones.mm(dl).squeeze(0).float()
+            return dw4, db4
+
+        w4, b4 = ctx.prefs
+        if _defer_active():
+            cur = torch.cuda.current_stream()
+            side = _side_stream(dseq.device)
+            side.wait_stream(cur)
+            with torch.cuda.stream(side):
+                dw4, db4 = wgrads()
+            for t in (dl, seq):
+                t.record_stream(side)
+            _PENDING.extend([(w4, dw4), (b4, db4)])
+            return dseq, None, None
+        dw4, db4 = wgrads()
+        for p, g_ in ((w4, dw4), (b4, db4)):
+            if p.grad is None:
+                p.grad = g_.contiguous()
+            else:
+                p.grad.add_(g_)
+        return dseq, None, None
+
+
 class FusedCrossEntropy(torch.autograd.Function):
     """Mean CE over (N, 5) logits: fwd loss and dlogits in one kernel pass."""
 
@@ -572,7 +627,7 @@ def train_forward(model, x: torch.Tensor, seed_buf=None) -> torch.Tensor:
             getattr(g, f"bias_hh_l{l}"), getattr(g, f"bias_hh_l{l}_reverse"),
         )
         seq = GruLayerFn.apply(seq, w_ih, b_ih, u, bhh, prefs)
-    logits = model.fc4(seq.float()).transpose(0, 1)  # (B, T, 5)
+    logits = HeadFn.apply(seq, model.fc4.weight, model.fc4.bias)  # (B, T, 5)
     return logits
 
 
