@@ -51,3 +51,15 @@ def test_hw_queue_default_set_on_import():
     )
     assert out.returncode == 0, out.stderr[-500:]
     assert out.stdout.strip() == "20"
+
+
+def test_bench_gpus_flag_validated():
+    """--gpus N without a matching torchrun WORLD_SIZE must exit loudly
+    (round-1 flagged the flag as cosmetic)."""
+    out = subprocess.run(
+        [sys.executable, os.path.join(ROOT, "bench.py"), "--gpus", "4",
+         "--steps", "1", "--warmup", "0"],
+        cwd=ROOT, capture_output=True, text=True, timeout=300,
+    )
+    assert out.returncode != 0
+    assert "WORLD_SIZE" in out.stderr

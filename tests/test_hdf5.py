@@ -196,3 +196,18 @@ def test_reader_rejects_garbage(tmp_path):
     trunc.write_bytes(b"\x89HDF\r\n\x1a\n\x00\x00")
     with pytest.raises(ValueError, match="truncated"):
         H5File(str(trunc))
+
+
+def test_hdf5_compat_cli(tmp_path, capsys):
+    from roko_amd.io.hdf5_compat import main as compat_main
+
+    h5 = str(tmp_path / "a.hdf5")
+    _reference_layout_file(h5)
+    rkw = str(tmp_path / "a.rkw")
+    compat_main([h5, rkw])
+    assert "converted 6 windows" in capsys.readouterr().out
+    h52 = str(tmp_path / "b.hdf5")
+    compat_main([rkw, h52])
+    assert "converted 6 windows" in capsys.readouterr().out
+    f = H5File(h52)
+    assert any(k != "contigs" for k in f.keys())

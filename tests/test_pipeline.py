@@ -168,3 +168,32 @@ def test_features_region_retry(tiny_assembly, tmp_path, monkeypatch):
     assert n > 0
     assert any("retrying once" in m for m in msgs)
     assert not any("failed twice" in m for m in msgs)
+
+
+def test_streaming_votes_multi_contig():
+    import numpy as np
+
+    from roko_amd import config as C
+    from roko_amd.inference import StreamingVotes, accumulate_votes
+
+    rng = np.random.default_rng(11)
+    sv = StreamingVotes(chunk_windows=5)
+    per = {}
+    for contig in ("a", "b", "c"):
+        W = C.WINDOW_COLS
+        n = int(rng.integers(3, 20))
+        pos = np.zeros((n, W, 2), dtype=np.int64)
+        pos[..., 0] = rng.integers(0, 40, (n, W))
+        preds = rng.integers(0, C.NUM_CLASSES, (n, W)).astype(np.uint8)
+        per[contig] = (pos, preds)
+    # interleave adds across contigs
+    order = [(c, i) for c, (p, _) in per.items() for i in range(len(p))]
+    rng.shuffle(order)
+    for c, i in order:
+        pos, preds = per[c]
+        sv.add(c, pos[i], preds[i])
+    tables = sv.finalize()
+    for c, (pos, preds) in per.items():
+        k, cnt = accumulate_votes(pos, preds)
+        assert np.array_equal(tables[c][0], k)
+        assert np.array_equal(tables[c][1], cnt)
