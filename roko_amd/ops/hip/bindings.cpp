@@ -61,6 +61,9 @@ void gemm_bias(const void* A, const void* B, const float* bias, void* C,
                int M, int N, int K, hipStream_t stream);
 void xg_gemm(const void* A, const void* B, const void* bias, void* C, int M,
              int KP, hipStream_t stream);
+void xg_gemm2(const void* A, const void* Bt, const void* bias, void* C,
+              int M, int KREAL, int KP, hipStream_t stream,
+              unsigned long long* timing = nullptr);
 int atb_splitk_nslices(int K);
 void atb_splitk(const void* A, const void* B, float* ws, float* C, int M,
                 int N, int K, hipStream_t stream);
@@ -390,6 +393,28 @@ torch::Tensor xg_gemm(torch::Tensor A, torch::Tensor Bt, torch::Tensor bias) {
     return C;
 }
 
+// xg projection, LDS-staged specialized kernel: A (M, 500|512|256) bf16
+// UNPADDED x Bt (768, KP) K-padded + bias -> (M, 768)
+torch::Tensor xg_gemm2(torch::Tensor A, torch::Tensor Bt, torch::Tensor bias,
+                       c10::optional<torch::Tensor> timing) {
+    check(A, torch::kBFloat16, "A");
+    check(Bt, torch::kBFloat16, "Bt");
+    check(bias, torch::kBFloat16, "bias");
+    const int M = A.size(0), KREAL = A.size(1), KP = Bt.size(1);
+    TORCH_CHECK((KREAL == 500 && KP == 512) || (KREAL == 512 && KP == 512) ||
+                    (KREAL == 256 && KP == 256),
+                "unsupported (KREAL, KP): ", KREAL, " ", KP);
+    TORCH_CHECK(Bt.size(0) == 768, "Bt must be (768, KP)");
+    TORCH_CHECK(M % 256 == 0, "M must be a multiple of 256");
+    auto C = torch::empty({M, 768}, A.options());
+    unsigned long long* tptr = nullptr;
+    if (timing.has_value())
+        tptr = reinterpret_cast<unsigned long long*>(timing->data_ptr());
+    rk::xg_gemm2(A.data_ptr(), Bt.data_ptr(), bias.data_ptr(), C.data_ptr(),
+                 M, KREAL, KP, cur_stream(), tptr);
+    return C;
+}
+
 // fused train front bwd: -> (de, dw1, db1, dw2, db2) fp32
 std::vector<torch::Tensor> front_bwd(torch::Tensor ids, torch::Tensor dseq,
                                      torch::Tensor w1, torch::Tensor b1,
@@ -522,6 +547,11 @@ struct ServeSlot {
         const char* f = getenv("ROKO_XGFOLD");
         return f && std::string(f) == "1";
     }();
+    // specialized xg GEMM kernel (ROKO_XG2=0 to fall back to hipBLASLt)
+    bool use_xg2 = [] {
+        const char* f = getenv("ROKO_XG2");
+        return !(f && std::string(f) == "0");
+    }();
 
     ServeSlot(py::dict w, int B_, torch::Tensor host_out_)
         : B(B_), stream(at::cuda::getStreamFromPool(/*high_priority=*/false)) {
@@ -624,9 +654,22 @@ struct ServeSlot {
                     in_dim = 256;
                 }
             } else {
+                // xg projection: the specialized LDS-staged kernel halves
+                // the aggregate CU-time vs hipBLASLt on these shapes
+                // (~50 vs ~105 CU*us/window, r2 notes) — and serving is
+                // CU-time bound, so it wins even at higher wall latency.
+                const bool xg2_ok =
+                    use_xg2 && !w_ih_p.empty() && ((90 * B) % 256) == 0;
                 for (int l = 0; l < 3; ++l) {
-                    at::addmm_out(xg2d, b_ih[l], l == 0 ? seq2d : hseq2d,
-                                  w_ih_t[l]);
+                    if (xg2_ok)
+                        rk::xg_gemm2((l == 0 ? seq : hseq).data_ptr(),
+                                     w_ih_p[l].data_ptr(), b_ih[l].data_ptr(),
+                                     xg.data_ptr(), 90 * B,
+                                     l == 0 ? 500 : 256,
+                                     (int)w_ih_p[l].size(1), s);
+                    else
+                        at::addmm_out(xg2d, b_ih[l], l == 0 ? seq2d : hseq2d,
+                                      w_ih_t[l]);
                     rk::gru_layer_fwd(xg.data_ptr(), u[l].data_ptr(),
                                       bhh[l].data_ptr<float>(), hseq.data_ptr(),
                                       nullptr, 90, B, s, 0);
@@ -719,6 +762,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("gemm_bias", &gemm_bias, py::arg("A"), py::arg("B"),
           py::arg("bias") = c10::nullopt);
     m.def("xg_gemm", &xg_gemm);
+    m.def("xg_gemm2", &xg_gemm2, py::arg("A"), py::arg("Bt"), py::arg("bias"),
+          py::arg("timing") = py::none());
     m.def("front_de_timed", &front_de_timed, py::arg("ids"), py::arg("dt1g"),
           py::arg("w1"), py::arg("seed"), py::arg("keep"), py::arg("dbg") = 0);
     m.def("front_bwd", &front_bwd, py::arg("ids"), py::arg("dseq"),

@@ -429,6 +429,187 @@ __global__ __launch_bounds__(WAVES * 64, 2) void xg_gemm_kernel(
 
 }  // namespace xggemm
 
+// ---------------------------------------------------------------------------
+// xg_gemm2: the LDS-staged, shape-specialized retry. gemm_bias (the generic
+// kernel) compiles its runtime-K loop into per-chunk exec-mask branches and
+// a vmcnt(0) drain after every staging load (ISA audit in the r2 notes) —
+// 119 us for this shape. Here K is a template constant (fully unrolled,
+// register-set double-buffering), staging is branchless b64/b128 with no
+// bounds masks (M % 256 == 0, N = 768 exactly), and B stages straight from
+// the (768, KP) padded W image whose [n][k] rows already ARE the
+// b-fragment layout (no transpose, no conflicts).
+namespace xg2 {
+
+constexpr int XN = 768;
+constexpr int BM = 256;   // C rows per workgroup (2 m-subtiles per wave)
+constexpr int BN = 128;   // C cols per workgroup (8 n-subtiles)
+constexpr int LD = 40;    // LDS k-slice stride (32 + 8 pad)
+constexpr int WAVES = 8;
+
+// phase timing (wave 0, workgroup 0 only): cycles in [commit+issue, mfma,
+// barrier, epilogue] accumulated into timing[0..3] when non-null
+#define XG2_T0 unsigned long long tp0 = (timing && tid == 0) \
+        ? __builtin_amdgcn_s_memtime() : 0
+#define XG2_T1(i) if (timing && tid == 0) { \
+        unsigned long long tn = __builtin_amdgcn_s_memtime(); \
+        tacc[i] += tn - tp0; tp0 = tn; }
+
+template <int KP, int KREAL>
+__global__ __launch_bounds__(WAVES * 64, 1) void xg_gemm2_kernel(
+    const bf16* __restrict__ A,     // (M, KREAL) row-major (8-B aligned rows)
+    const bf16* __restrict__ Bt,    // (768, KP) row-major = B^T, zero-padded
+    const bf16* __restrict__ bias,  // (768)
+    bf16* __restrict__ C,           // (M, 768)
+    int M, unsigned long long* timing = nullptr) {
+    unsigned long long tacc[4] = {0, 0, 0, 0};
+    // 64-wide k-slices: half the barrier count of the 32-wide version
+    // (phase timing showed ~480 cyc/iter of barrier wave-skew), two
+    // 16-MFMA bursts per slice.
+    constexpr int BK = 64;
+    constexpr int KB = KP / BK;
+    constexpr int LDK = BK + 8;
+    __shared__ struct {
+        bf16 a[2][BM][LDK];   // [row][k]
+        bf16 b[2][BN][LDK];   // [n][k]
+    } lds;
+
+    const int tid = threadIdx.x;
+    const int wid = tid >> 6;
+    const int lane = tid & 63;
+    const int lrow = lane >> 4;
+    const int lcol = lane & 15;
+    const int m0 = blockIdx.x * BM;
+    const int n0 = blockIdx.y * BN;
+
+    f32x4 acc[2][8];
+#pragma unroll
+    for (int mt = 0; mt < 2; ++mt)
+#pragma unroll
+        for (int nt = 0; nt < 8; ++nt) acc[mt][nt] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+    // staging, branchless: A 256x64 = 32 el/thread as 8 b64 (8-B row
+    // alignment holds for KREAL % 4 == 0), B 128x64 = 16 el/thread as
+    // 2 b128 (KP % 8 == 0). Two register sets; the loop runs the MFMA
+    // burst FIRST and commits after it, so the commit's vmcnt wait sits
+    // a full ~1100-cycle MFMA burst after the loads were issued — the
+    // 3-set variant bought the same cover with 24 more registers.
+    uint64_t ra[2][8];
+    bf16x8 rb[2][2];
+    const int a_row = tid >> 1, a_c0 = (tid & 1) * 32;     // 2 threads/row
+    const int b_row = tid >> 2, b_c0 = (tid & 3) * 16;     // 4 threads/row
+    auto issue = [&](int kq, uint64_t(&va)[8], bf16x8(&vb)[2]) {
+        const int k0 = kq * BK;
+#pragma unroll
+        for (int q = 0; q < 8; ++q) {
+            const int col = k0 + a_c0 + q * 4;
+            va[q] = (KREAL == KP || col + 4 <= KREAL)
+                        ? *reinterpret_cast<const uint64_t*>(
+                              A + (size_t)(m0 + a_row) * KREAL + col)
+                        : 0u;
+        }
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+            vb[j] = *reinterpret_cast<const bf16x8*>(
+                Bt + (size_t)(n0 + b_row) * KP + k0 + b_c0 + j * 8);
+    };
+    auto commit = [&](int buf, uint64_t(&va)[8], bf16x8(&vb)[2]) {
+#pragma unroll
+        for (int q = 0; q < 8; ++q)
+            *reinterpret_cast<uint64_t*>(&lds.a[buf][a_row][a_c0 + q * 4]) =
+                va[q];
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+            *reinterpret_cast<bf16x8*>(&lds.b[buf][b_row][b_c0 + j * 8]) =
+                vb[j];
+    };
+    auto mfma_step = [&](int buf) {
+        // two k-halves; within each, batch the fragment reads before the
+        // MFMA burst so the lgkmcnt waits overlap
+#pragma unroll
+        for (int kh = 0; kh < 2; ++kh) {
+            bf16x8 af[2], bfr[8];
+#pragma unroll
+            for (int mt = 0; mt < 2; ++mt)
+                af[mt] = lds_load_a_frag(&lds.a[buf][0][0],
+                                         (wid * 2 + mt) * 16, kh * 32, LDK);
+#pragma unroll
+            for (int nt = 0; nt < 8; ++nt)
+                bfr[nt] = lds_load_b_frag_t(&lds.b[buf][0][0], nt * 16,
+                                            kh * 32, LDK);
+#pragma unroll
+            for (int mt = 0; mt < 2; ++mt)
+#pragma unroll
+                for (int nt = 0; nt < 8; ++nt)
+                    acc[mt][nt] = mfma16x16x32(af[mt], bfr[nt], acc[mt][nt]);
+        }
+    };
+
+    issue(0, ra[0], rb[0]);
+    commit(0, ra[0], rb[0]);
+    if (KB > 1) issue(1, ra[1], rb[1]);
+    __syncthreads();
+    XG2_T0;
+#pragma unroll
+    for (int kq = 0; kq < KB; ++kq) {
+        mfma_step(kq & 1);
+        XG2_T1(1);
+        if (kq + 1 < KB) commit((kq + 1) & 1, ra[(kq + 1) & 1],
+                                rb[(kq + 1) & 1]);
+        if (kq + 2 < KB) issue(kq + 2, ra[kq & 1], rb[kq & 1]);
+        XG2_T1(0);
+        __syncthreads();
+        XG2_T1(2);
+    }
+
+    // epilogue: bias + direct stores from the accumulators. A quarter-wave
+    // writes 16 consecutive bf16 (one 32-B chunk) per fragment row; no LDS
+    // round-trip (the staged version cost ~4k cycles of exposed latency).
+    float br[8];
+#pragma unroll
+    for (int nt = 0; nt < 8; ++nt)
+        br[nt] = bf2f(bias[n0 + nt * 16 + lcol]);
+#pragma unroll
+    for (int mt = 0; mt < 2; ++mt) {
+        const size_t rbase = (size_t)(m0 + (wid * 2 + mt) * 16 + lrow * 4);
+#pragma unroll
+        for (int nt = 0; nt < 8; ++nt)
+#pragma unroll
+            for (int i = 0; i < 4; ++i)
+                C[(rbase + i) * XN + n0 + nt * 16 + lcol] =
+                    f2bf(acc[mt][nt][i] + br[nt]);
+    }
+    XG2_T1(3);
+    if (timing && tid == 0 && blockIdx.x == 0 && blockIdx.y == 0)
+        for (int i = 0; i < 4; ++i) timing[i] = tacc[i];
+}
+
+}  // namespace xg2
+
+void xg_gemm2(const void* A, const void* Bt, const void* bias, void* C,
+              int M, int KREAL, int KP, hipStream_t stream,
+              unsigned long long* timing) {
+    dim3 grid(M / xg2::BM, xg2::XN / xg2::BN);
+    dim3 block(xg2::WAVES * 64);
+    if (KP == 512 && KREAL == 500)
+        hipLaunchKernelGGL((xg2::xg_gemm2_kernel<512, 500>), grid, block, 0,
+                           stream, static_cast<const bf16*>(A),
+                           static_cast<const bf16*>(Bt),
+                           static_cast<const bf16*>(bias),
+                           static_cast<bf16*>(C), M, timing);
+    else if (KP == 512 && KREAL == 512)
+        hipLaunchKernelGGL((xg2::xg_gemm2_kernel<512, 512>), grid, block, 0,
+                           stream, static_cast<const bf16*>(A),
+                           static_cast<const bf16*>(Bt),
+                           static_cast<const bf16*>(bias),
+                           static_cast<bf16*>(C), M, timing);
+    else if (KP == 256 && KREAL == 256)
+        hipLaunchKernelGGL((xg2::xg_gemm2_kernel<256, 256>), grid, block, 0,
+                           stream, static_cast<const bf16*>(A),
+                           static_cast<const bf16*>(Bt),
+                           static_cast<const bf16*>(bias),
+                           static_cast<bf16*>(C), M, timing);
+}
+
 void xg_gemm(const void* A, const void* B, const void* bias, void* C, int M,
              int KP, hipStream_t stream) {
     dim3 grid(M / (xggemm::WAVES * xggemm::WM), xggemm::XN / xggemm::BN);

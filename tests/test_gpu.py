@@ -511,3 +511,22 @@ def test_deferred_weight_grads_match_inline():
         tol = 1e-4 + 1e-3 * ref.abs().max()
         assert (ref - got).abs().max() <= tol, (
             n, float((ref - got).abs().max()))
+
+
+def test_xg_gemm2_numerics():
+    """xg_gemm2 (the serving xg projection kernel) vs fp32 torch on both
+    live shape families: l0 (K=500 padded to 512) and l1/l2 (K=256)."""
+    from roko_amd.ops import _hip_ops as ext
+
+    torch.manual_seed(0)
+    M = 90 * 128
+    for kreal, kp in [(500, 512), (256, 256)]:
+        A = torch.randn(M, kreal, device="cuda").bfloat16().contiguous()
+        W = (torch.randn(768, kreal, device="cuda") * 0.05).bfloat16()
+        bias = (torch.randn(768, device="cuda") * 0.1).bfloat16()
+        Wp = torch.zeros(768, kp, dtype=torch.bfloat16, device="cuda")
+        Wp[:, :kreal] = W
+        ref = torch.addmm(bias.float(), A.float(), W.t().float())
+        out = ext.xg_gemm2(A, Wp.contiguous(), bias)
+        rel = ((out.float() - ref).abs().max() / ref.abs().max()).item()
+        assert rel < 2e-2, (kreal, rel)
