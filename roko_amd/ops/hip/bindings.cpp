@@ -552,6 +552,23 @@ struct ServeSlot {
         const char* f = getenv("ROKO_XG2");
         return !(f && std::string(f) == "0");
     }();
+    // raw hipGraph per slot (ROKO_GSLOT=1 to enable): when the forward is
+    // torch-free (xg2 or fold path — no hipBLASLt workspace allocs), the
+    // whole kernel sequence is captured once and replayed with ONE
+    // hipGraphLaunch per batch instead of ~8 kernel enqueues. MEASURED
+    // NEGATIVE, default off: raw hipGraphLaunch costs MORE host time than
+    // the 8 direct enqueues on ROCm 7 (26.6 vs 27.7 M bases/s at b=128 —
+    // the round-1 "~15 us/node replay" cost is in the runtime, not in
+    // torch's wrapper). The effective host-bound lever is batch size
+    // (serve_soak sweep: b=128 26.6 -> b=512 38.1 M bases/s).
+    bool use_gslot = [] {
+        const char* f = getenv("ROKO_GSLOT");
+        return f && std::string(f) == "1";
+    }();
+    bool xg2_path = false;   // set in ctor: raw (capture-legal) layer path
+    hipGraph_t slot_graph = nullptr;
+    hipGraphExec_t slot_gexec = nullptr;
+    int n_runs = 0;
 
     ServeSlot(py::dict w, int B_, torch::Tensor host_out_)
         : B(B_), stream(at::cuda::getStreamFromPool(/*high_priority=*/false)) {
@@ -576,7 +593,9 @@ struct ServeSlot {
         if (w_ih_p.size() != 3) {
             w_ih_p.clear();
             use_fold = false;
+            use_xg2 = false;
         }
+        xg2_path = use_xg2 && !w_ih_p.empty() && ((90 * B_) % 256) == 0;
         TORCH_CHECK(B % 32 == 0, "serving batch must be a multiple of 32");
         TORCH_CHECK(host_out_.is_pinned() && host_out_.scalar_type() == t::kUInt8
                         && host_out_.size(0) == B && host_out_.size(1) == 90,
@@ -596,8 +615,79 @@ struct ServeSlot {
         (void)hipEventCreateWithFlags(&ev_done, hipEventDisableTiming);
     }
     ~ServeSlot() {
+        if (slot_gexec) (void)hipGraphExecDestroy(slot_gexec);
+        if (slot_graph) (void)hipGraphDestroy(slot_graph);
         if (ev_in) (void)hipEventDestroy(ev_in);
         if (ev_done) (void)hipEventDestroy(ev_done);
+    }
+
+    // the torch-free model body (front -> 3x(xg + GRU) -> head): every call
+    // is a raw kernel enqueue on `s`, so the sequence is hipGraph
+    // capture-legal. Callers guarantee use_fold || xg2_path.
+    void enqueue_model(hipStream_t s) {
+        if (w1gt.defined() && use_v3)
+            rk::embed_mlp_fwd3(x_buf.data_ptr<uint8_t>(), w1gt.data_ptr(),
+                               b1.data_ptr<float>(), w2.data_ptr(),
+                               b2.data_ptr<float>(), emb.data_ptr(),
+                               seq.data_ptr(), B, s, nullptr);
+        else if (w1gt.defined())
+            rk::embed_mlp_fwd2(x_buf.data_ptr<uint8_t>(), w1gt.data_ptr(),
+                               b1.data_ptr<float>(), w2.data_ptr(),
+                               b2.data_ptr<float>(), emb.data_ptr(),
+                               seq.data_ptr(), B, s);
+        else
+            rk::embed_mlp_fwd(x_buf.data_ptr<uint8_t>(), w1.data_ptr(),
+                              b1.data_ptr<float>(), w2.data_ptr(),
+                              b2.data_ptr<float>(), emb.data_ptr(),
+                              seq.data_ptr(), B, s, 0, nullptr);
+        if (use_fold) {
+            const void* xin = seq.data_ptr();
+            int in_dim = 500;
+            for (int l = 0; l < 3; ++l) {
+                torch::Tensor& out_t = (l % 2) ? hseq2 : hseq;
+                rk::gru_layer_fwd_fused(
+                    xin, w_ih_p[l].data_ptr(), b_ih[l].data_ptr(),
+                    u[l].data_ptr(), bhh[l].data_ptr<float>(), xg.data_ptr(),
+                    out_t.data_ptr(), 90, B, in_dim,
+                    (int)w_ih_p[l].size(1), s);
+                xin = out_t.data_ptr();
+                in_dim = 256;
+            }
+        } else {
+            for (int l = 0; l < 3; ++l) {
+                rk::xg_gemm2((l == 0 ? seq : hseq).data_ptr(),
+                             w_ih_p[l].data_ptr(), b_ih[l].data_ptr(),
+                             xg.data_ptr(), 90 * B, l == 0 ? 500 : 256,
+                             (int)w_ih_p[l].size(1), s);
+                rk::gru_layer_fwd(xg.data_ptr(), u[l].data_ptr(),
+                                  bhh[l].data_ptr<float>(), hseq.data_ptr(),
+                                  nullptr, 90, B, s, 0);
+            }
+        }
+        rk::head_fwd(hseq.data_ptr(), w4.data_ptr(), b4.data_ptr<float>(),
+                     nullptr, amax.data_ptr<uint8_t>(), 90, B, s);
+    }
+
+    void try_capture(hipStream_t s) {
+        hipError_t e =
+            hipStreamBeginCapture(s, hipStreamCaptureModeThreadLocal);
+        if (e != hipSuccess) { use_gslot = false; return; }
+        enqueue_model(s);
+        (void)hipMemcpyAsync(host_out.data_ptr(), amax.data_ptr(),
+                             (size_t)B * 90, hipMemcpyDeviceToHost, s);
+        e = hipStreamEndCapture(s, &slot_graph);
+        if (e != hipSuccess || !slot_graph) {
+            use_gslot = false;
+            slot_graph = nullptr;
+            return;
+        }
+        e = hipGraphInstantiate(&slot_gexec, slot_graph, nullptr, nullptr, 0);
+        if (e != hipSuccess) {
+            (void)hipGraphDestroy(slot_graph);
+            slot_graph = nullptr;
+            slot_gexec = nullptr;
+            use_gslot = false;
+        }
     }
 
     void run(torch::Tensor x, int64_t n) {
@@ -618,7 +708,27 @@ struct ServeSlot {
         if (x.is_cuda())
             c10::cuda::CUDACachingAllocator::recordStream(
                 x.storage().data_ptr(), stream);
-        {
+        if (use_fold || xg2_path) {
+            // torch-free path: raw copies + kernels, optionally one
+            // hipGraphLaunch replacing the per-batch kernel enqueues
+            hipStream_t s = stream.stream();
+            if (n > 0)
+                (void)hipMemcpyAsync(x_buf.data_ptr(), x.data_ptr(),
+                                     (size_t)n * 200 * 90, hipMemcpyDefault,
+                                     s);
+            // the first run executes eagerly (warms caches); the second
+            // captures + launches; later runs just launch
+            if (use_gslot && !slot_gexec && n_runs >= 1) try_capture(s);
+            if (slot_gexec) {
+                (void)hipGraphLaunch(slot_gexec, s);
+            } else {
+                enqueue_model(s);
+                (void)hipMemcpyAsync(host_out.data_ptr(), amax.data_ptr(),
+                                     (size_t)B * 90, hipMemcpyDeviceToHost,
+                                     s);
+            }
+            ++n_runs;
+        } else {
             at::cuda::CUDAStreamGuard guard(stream);
             hipStream_t s = stream.stream();
             if (n > 0) x_buf.narrow(0, 0, n).copy_(x.narrow(0, 0, n), true);
@@ -637,43 +747,12 @@ struct ServeSlot {
                                   b1.data_ptr<float>(), w2.data_ptr(),
                                   b2.data_ptr<float>(), emb.data_ptr(),
                                   seq.data_ptr(), B, s, 0, nullptr);
-            if (use_fold) {
-                // xg GEMM folded into the GRU kernel; layers ping-pong
-                // between hseq buffers (phase 1 of layer l reads the other
-                // dir's workgroup output, so in-place would race)
-                const void* xin = seq.data_ptr();
-                int in_dim = 500;
-                for (int l = 0; l < 3; ++l) {
-                    torch::Tensor& out_t = (l % 2) ? hseq2 : hseq;
-                    rk::gru_layer_fwd_fused(
-                        xin, w_ih_p[l].data_ptr(), b_ih[l].data_ptr(),
-                        u[l].data_ptr(), bhh[l].data_ptr<float>(),
-                        xg.data_ptr(), out_t.data_ptr(), 90, B, in_dim,
-                        (int)w_ih_p[l].size(1), s);
-                    xin = out_t.data_ptr();
-                    in_dim = 256;
-                }
-            } else {
-                // xg projection: the specialized LDS-staged kernel halves
-                // the aggregate CU-time vs hipBLASLt on these shapes
-                // (~50 vs ~105 CU*us/window, r2 notes) — and serving is
-                // CU-time bound, so it wins even at higher wall latency.
-                const bool xg2_ok =
-                    use_xg2 && !w_ih_p.empty() && ((90 * B) % 256) == 0;
-                for (int l = 0; l < 3; ++l) {
-                    if (xg2_ok)
-                        rk::xg_gemm2((l == 0 ? seq : hseq).data_ptr(),
-                                     w_ih_p[l].data_ptr(), b_ih[l].data_ptr(),
-                                     xg.data_ptr(), 90 * B,
-                                     l == 0 ? 500 : 256,
-                                     (int)w_ih_p[l].size(1), s);
-                    else
-                        at::addmm_out(xg2d, b_ih[l], l == 0 ? seq2d : hseq2d,
-                                      w_ih_t[l]);
-                    rk::gru_layer_fwd(xg.data_ptr(), u[l].data_ptr(),
-                                      bhh[l].data_ptr<float>(), hseq.data_ptr(),
-                                      nullptr, 90, B, s, 0);
-                }
+            for (int l = 0; l < 3; ++l) {
+                at::addmm_out(xg2d, b_ih[l], l == 0 ? seq2d : hseq2d,
+                              w_ih_t[l]);
+                rk::gru_layer_fwd(xg.data_ptr(), u[l].data_ptr(),
+                                  bhh[l].data_ptr<float>(), hseq.data_ptr(),
+                                  nullptr, 90, B, s, 0);
             }
             rk::head_fwd(hseq.data_ptr(), w4.data_ptr(), b4.data_ptr<float>(),
                          nullptr, amax.data_ptr<uint8_t>(), 90, B, s);

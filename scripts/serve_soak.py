@@ -9,12 +9,15 @@ from roko_amd.model import RokoModel
 from roko_amd.ops.forward import InferencePipeline
 
 secs = float(sys.argv[1]) if len(sys.argv) > 1 else 60.0
+batch = int(sys.argv[2]) if len(sys.argv) > 2 else 128
+depth = int(sys.argv[3]) if len(sys.argv) > 3 else 48
 torch.manual_seed(0)
 model = RokoModel().cuda().eval()
 ops.require()
-pipe = InferencePipeline(model, 128, depth=48)
+pipe = InferencePipeline(model, batch, depth=depth)
 g = torch.Generator().manual_seed(7)
-xs = [torch.randint(0, 12, (128, 200, 90), generator=g, dtype=torch.uint8).cuda()
+xs = [torch.randint(0, 12, (batch, 200, 90), generator=g,
+                    dtype=torch.uint8).cuda()
       for _ in range(64)]
 for i in range(100):
     pipe.submit(xs[i % 64], copy_out=False)
@@ -24,9 +27,9 @@ windows = 0
 t0 = time.perf_counter()
 marks = []
 while True:
-    for _ in range(2000):
+    for _ in range(max(256000 // batch, 200)):
         pipe.submit(xs[windows % 64], copy_out=False)
-        windows += 128
+        windows += batch
     torch.cuda.synchronize()
     el = time.perf_counter() - t0
     marks.append(windows * 30 / el)
