@@ -63,3 +63,28 @@ def test_bench_gpus_flag_validated():
     )
     assert out.returncode != 0
     assert "WORLD_SIZE" in out.stderr
+
+
+def test_bench_world2_gloo():
+    """The driver launches bench.py under torchrun with --nproc-per-node N;
+    rehearse that exact shape at N=2 on CPU (gloo): rendezvous, per-rank
+    work, MAX-reduce of elapsed, and rank-0-only JSON output."""
+    env = dict(os.environ)
+    env.pop("RANK", None), env.pop("WORLD_SIZE", None)
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29871", os.path.join(ROOT, "bench.py"),
+         "--gpus", "2", "--steps", "2", "--warmup", "1",
+         "--batch", "4", "--depth", "2"],
+        cwd=ROOT, capture_output=True, text=True, timeout=900, env=env,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [l for l in out.stdout.strip().splitlines() if l.startswith("{")]
+    # rank 0 prints once per record; no duplicate records from rank 1
+    recs = [json.loads(l) for l in lines]
+    assert len([r for r in recs if r["metric"] == "inference_bases_per_sec"]) == 1
+    r = recs[-1]
+    assert r["n_gpus"] == 2
+    assert r["config"]["parallelism"] == "dp2"
+    assert r["config"]["global_batch"] == 8
