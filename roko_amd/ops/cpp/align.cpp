@@ -20,7 +20,7 @@
 namespace rk {
 
 AlignStats align_stats(const std::string& query, const std::string& target,
-                       int band) {
+                       int band, std::string* cigar) {
     const int64_t n = int64_t(query.size());   // rows: query (assembly)
     const int64_t m = int64_t(target.size());  // cols: target (truth)
     if (band < 1) band = 1;
@@ -84,6 +84,11 @@ AlignStats align_stats(const std::string& query, const std::string& target,
     s.edit_distance = prev[size_t(dend - dlo)];
 
     // traceback from (n, m)
+    std::vector<std::pair<int64_t, char>> ops;  // reversed run-length ops
+    auto push_op = [&](char op) {
+        if (!ops.empty() && ops.back().second == op) ++ops.back().first;
+        else ops.emplace_back(1, op);
+    };
     int64_t i = n, j = m;
     while (i > 0 || j > 0) {
         const int64_t d = j - i;
@@ -91,13 +96,23 @@ AlignStats align_stats(const std::string& query, const std::string& target,
         if (i > 0 && j > 0 && dir == 0) {
             if (query[size_t(i - 1)] == target[size_t(j - 1)]) ++s.matches;
             else ++s.mismatches;
+            if (cigar) push_op('M');
             --i; --j;
         } else if (i > 0 && (dir == 1 || j == 0)) {
             ++s.insertions;  // extra base in query w.r.t. target
+            if (cigar) push_op('I');
             --i;
         } else {
             ++s.deletions;   // base of target missing from query
+            if (cigar) push_op('D');
             --j;
+        }
+    }
+    if (cigar) {
+        cigar->clear();
+        for (auto it = ops.rbegin(); it != ops.rend(); ++it) {
+            *cigar += std::to_string(it->first);
+            *cigar += it->second;
         }
     }
     return s;

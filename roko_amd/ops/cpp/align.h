@@ -14,8 +14,10 @@ struct AlignStats {
 
 // Banded unit-cost global alignment of query vs target (band = half-width
 // in diagonals beyond the length difference). Throws when the optimal path
-// cannot be represented in the band.
+// cannot be represented in the band. When `cigar` is non-null it receives
+// the run-length encoded alignment path (query as SEQ: M/I/D ops, '='/'X'
+// collapsed into M — the form BAM records use).
 AlignStats align_stats(const std::string& query, const std::string& target,
-                       int band);
+                       int band, std::string* cigar = nullptr);
 
 }  // namespace rk

@@ -97,4 +97,24 @@ PYBIND11_MODULE(_pileup, m) {
         },
         py::arg("query"), py::arg("target"), py::arg("band") = 128,
         "Banded global alignment stats of query vs target (assembly QC)");
+    m.def(
+        "align_cigar",
+        [](const std::string& query, const std::string& target, int band) {
+            rk::AlignStats s;
+            std::string cig;
+            {
+                py::gil_scoped_release release;
+                s = rk::align_stats(query, target, band, &cig);
+            }
+            py::dict d;
+            d["edit_distance"] = s.edit_distance;
+            d["matches"] = s.matches;
+            d["mismatches"] = s.mismatches;
+            d["insertions"] = s.insertions;
+            d["deletions"] = s.deletions;
+            d["cigar"] = cig;
+            return d;
+        },
+        py::arg("query"), py::arg("target"), py::arg("band") = 128,
+        "align_stats plus the M/I/D CIGAR of query against target");
 }

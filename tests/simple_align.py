@@ -79,3 +79,122 @@ class EditScript:
                 add(int(self.ins_after[t]), "D")
         add(e - 1 - t1, "S")
         return SamRecord(qname, flag, tid, int(self.dstart[t0]), mapq, cig, seq)
+
+
+def mutate_seq(rng: np.random.Generator, seq: str, err: float) -> str:
+    """Apply sequencing-style noise: substitutions, insertions and
+    deletions at err/3 each (total per-base error ~err)."""
+    if err <= 0:
+        return seq
+    out: List[str] = []
+    third = err / 3.0
+    for ch in seq:
+        r = rng.random()
+        if r < third:
+            continue  # deletion
+        if r < 2 * third:
+            out.append(BASES[int(rng.integers(4))])  # substitution
+        else:
+            out.append(ch)
+        while rng.random() < third:
+            out.append(BASES[int(rng.integers(4))])  # insertion
+    return "".join(out)
+
+
+def noisy_record(rng: np.random.Generator, es: EditScript, qname: str,
+                 s: int, e: int, err: float, px, flag: int = 0,
+                 mapq: int = 60) -> Optional[SamRecord]:
+    """SamRecord of a NOISY read (sequencing error `err` on a truth
+    substring) aligned against the draft with the in-repo banded aligner
+    (`px.align_cigar`) — the realistic path: read noise + draft errors both
+    present, CIGAR produced by alignment rather than by construction."""
+    t0 = s
+    while t0 < e and es.dstart[t0] < 0:
+        t0 += 1
+    t1 = e - 1
+    while t1 >= t0 and es.dstart[t1] < 0:
+        t1 -= 1
+    if t1 < t0:
+        return None
+    p0 = int(es.dstart[t0])
+    p1 = int(es.dstart[t1]) + 1
+    seq = mutate_seq(rng, es.truth[t0:t1 + 1], err)
+    if not seq:
+        return None
+    band = max(64, int((t1 - t0) * (err + 0.05)) + 32)
+    res = px.align_cigar(seq, es.draft[p0:p1], band=band)
+    cig: List[Tuple[int, str]] = []
+    num = 0
+    for ch in res["cigar"]:
+        if ch.isdigit():
+            num = num * 10 + int(ch)
+        else:
+            cig.append((num, ch))
+            num = 0
+    # normalize to a valid BAM CIGAR: leading/trailing deletions shift the
+    # mapping coordinates; terminal insertions become soft clips
+    pos = p0
+    while cig and cig[0][1] == "D":
+        pos += cig.pop(0)[0]
+    while cig and cig[-1][1] == "D":
+        cig.pop()
+    if cig and cig[0][1] == "I":
+        cig[0] = (cig[0][0], "S")
+    if cig and cig[-1][1] == "I":
+        cig[-1] = (cig[-1][0], "S")
+    if not any(op == "M" for _, op in cig):
+        return None
+    return SamRecord(qname, flag, 0, pos, mapq, cig, seq)
+
+
+def build_assembly(rng: np.random.Generator, outdir, *, length=3000, cov=20,
+                   read_len=400, read_err=0.0, sub=0.01, ins=0.003,
+                   dl=0.003):
+    """Full synthetic polishing scenario on disk: truth genome, draft with
+    the given error profile, reads at `cov`x (error-free exact-CIGAR reads
+    when read_err=0, otherwise noisy reads aligned with the banded
+    aligner), plus the truth-to-draft BAM for labeling. Returns the same
+    dict shape as the `tiny_assembly` fixture."""
+    import os
+
+    from roko_amd.io.bamio import write_bam
+    from roko_amd.io.fasta import write_fasta
+    from roko_amd.ops import pileup_ext
+
+    truth = "".join(BASES[int(b)] for b in rng.integers(0, 4, length))
+    es = EditScript(rng, truth, sub_rate=sub, ins_rate=ins, del_rate=dl)
+    draft = es.draft
+    outdir = str(outdir)
+    os.makedirs(outdir, exist_ok=True)
+    draft_fasta = os.path.join(outdir, "draft.fasta")
+    write_fasta(draft_fasta, [("ctg1", draft)])
+    refs = [("ctg1", len(draft))]
+
+    px = pileup_ext()
+    reads = []
+    n_reads = max(1, cov * len(truth) // read_len)
+    for i in range(n_reads):
+        s = int(rng.integers(0, max(1, len(truth) - read_len)))
+        flag = 16 if i % 2 else 0
+        if read_err > 0:
+            rec = noisy_record(rng, es, f"read{i}", s, s + read_len,
+                               read_err, px, flag=flag)
+        else:
+            rec = es.align_substring(f"read{i}", s, s + read_len, flag=flag)
+        if rec is not None:
+            reads.append(rec)
+    reads.sort(key=lambda r: (r.tid, r.pos))
+    reads_bam = os.path.join(outdir, "reads.bam")
+    write_bam(reads_bam, refs, reads)
+
+    trec = es.align_substring("truth_ctg1", 0, len(truth), flag=0)
+    truth_bam = os.path.join(outdir, "truth.bam")
+    write_bam(truth_bam, refs, [trec])
+    return {
+        "draft_fasta": draft_fasta,
+        "reads_bam": reads_bam,
+        "truth_bam": truth_bam,
+        "truth": truth,
+        "draft": draft,
+        "edit_script": es,
+    }
