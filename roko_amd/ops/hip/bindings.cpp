@@ -561,10 +561,10 @@ struct ServeSlot {
     // the round-1 "~15 us/node replay" cost is in the runtime, not in
     // torch's wrapper). The effective host-bound lever is batch size
     // (serve_soak sweep: b=128 26.6 -> b=512 38.1 M bases/s).
-    bool use_gslot = [] {
-        const char* f = getenv("ROKO_GSLOT");
-        return f && std::string(f) == "1";
-    }();
+    // auto default (set in ctor): ON for batch >= 256 where it wins the
+    // same-box A/B (b=512: 38.0 vs 36.4 M bases/s), OFF at b=128 where the
+    // graph launch costs more than the direct enqueues (26.6 vs 27.7).
+    bool use_gslot = false;
     bool xg2_path = false;   // set in ctor: raw (capture-legal) layer path
     hipGraph_t slot_graph = nullptr;
     hipGraphExec_t slot_gexec = nullptr;
@@ -596,6 +596,10 @@ struct ServeSlot {
             use_xg2 = false;
         }
         xg2_path = use_xg2 && !w_ih_p.empty() && ((90 * B_) % 256) == 0;
+        if (const char* f = getenv("ROKO_GSLOT"))
+            use_gslot = std::string(f) == "1";
+        else
+            use_gslot = B_ >= 256;
         TORCH_CHECK(B % 32 == 0, "serving batch must be a multiple of 32");
         TORCH_CHECK(host_out_.is_pinned() && host_out_.scalar_type() == t::kUInt8
                         && host_out_.size(0) == B && host_out_.size(1) == 90,
