@@ -32,6 +32,7 @@ def _ext():
 
 _DEFER = False
 _PENDING: list = []
+_PENDING_FNS: list = []   # (ready_event, closure) run at drain time
 _SIDE = None       # per-layer weight-grad GEMMs
 _SIDE2 = None      # the front recompute backward (starts once the last GRU
                    # layer's dx exists; by then the main stream is idle)
@@ -74,6 +75,7 @@ class deferred_weight_grads:
         # discard stashed grads from a step that died mid-backward — they
         # would otherwise be attached to the params on the next drain
         _PENDING.clear()
+        _PENDING_FNS.clear()
         self._prev = _DEFER
         _DEFER = True
         return self
@@ -87,9 +89,32 @@ def _defer_active() -> bool:
     return _DEFER and not torch.cuda.is_current_stream_capturing()
 
 
+def _run_pending_fns():
+    """Enqueue the stashed weight-grad closures on the side stream.
+
+    The closures are HOST-deferred, not just stream-deferred: enqueueing
+    the ~15 aten/kernel calls of one layer's weight grads costs ~250 us of
+    host time, and doing it inside GruLayerFn.backward stalled the NEXT
+    layer's BPTT launch (kernel-trace: 250-370 us main-stream gaps between
+    gru_layer_bwd kernels). FrontFn.backward calls this right after
+    enqueueing front_bwd, so both the host cost and the side-stream GPU
+    work hide under the ~700 us front_bwd kernel; drain_deferred_grads
+    keeps a fallback call for paths without a FrontFn."""
+    if not _PENDING_FNS:
+        return
+    side = _side_stream(torch.cuda.current_device())
+    for ev, fn in _PENDING_FNS:
+        side.wait_event(ev)
+        with torch.cuda.stream(side):
+            fn()
+    _PENDING_FNS.clear()
+
+
 def drain_deferred_grads():
-    """Join the side streams and attach the stashed gradients to params."""
+    """Run any remaining weight-grad closures, join the side streams, and
+    attach the stashed gradients to params."""
     global _PENDING
+    _run_pending_fns()
     if not _PENDING:
         return
     cur = torch.cuda.current_stream()
@@ -273,21 +298,29 @@ class GruLayerFn(torch.autograd.Function):
             return du, dw_ih, dbhh, db_ih
 
         if _defer_active() and ctx.prefs is not None:
-            cur = torch.cuda.current_stream()
-            side = _side_stream(dx.device)
-            side.wait_stream(cur)
-            with torch.cuda.stream(side):
-                du, dw_ih, dbhh, db_ih = weight_grads()
-            for t in (hseq, dhg, dxg, x_bf):
-                t.record_stream(side)
-            wf, wr, bf_, br, uf, ur, bhf, bhr = ctx.prefs
+            # host-defer the whole weight-grad section: enqueueing it here
+            # (~15 calls, ~250 us host) stalls the next layer's BPTT launch.
+            # The closure runs on the side stream at drain_deferred_grads(),
+            # ordered after this point by the event.
+            ev = torch.cuda.Event()
+            ev.record(torch.cuda.current_stream())
+            prefs = ctx.prefs
             G3 = 3 * H
-            _PENDING.extend([
-                (wf, dw_ih[:G3]), (wr, dw_ih[G3:]),
-                (bf_, db_ih[:G3]), (br, db_ih[G3:]),
-                (uf, du[0]), (ur, du[1]),
-                (bhf, dbhh[0]), (bhr, dbhh[1]),
-            ])
+
+            def deferred():
+                du, dw_ih, dbhh, db_ih = weight_grads()
+                side = torch.cuda.current_stream()
+                for t in (hseq, dhg, dxg, x_bf):
+                    t.record_stream(side)
+                wf, wr, bf_, br, uf, ur, bhf, bhr = prefs
+                _PENDING.extend([
+                    (wf, dw_ih[:G3]), (wr, dw_ih[G3:]),
+                    (bf_, db_ih[:G3]), (br, db_ih[G3:]),
+                    (uf, du[0]), (ur, du[1]),
+                    (bhf, dbhh[0]), (bhr, dbhh[1]),
+                ])
+
+            _PENDING_FNS.append((ev, deferred))
             return dx, None, None, None, None, None
 
         du, dw_ih, dbhh, db_ih = weight_grads()
@@ -336,14 +369,17 @@ class HeadFn(torch.autograd.Function):
 
         w4, b4 = ctx.prefs
         if _defer_active():
-            cur = torch.cuda.current_stream()
-            side = _side_stream(dseq.device)
-            side.wait_stream(cur)
-            with torch.cuda.stream(side):
+            ev = torch.cuda.Event()
+            ev.record(torch.cuda.current_stream())
+
+            def deferred():
                 dw4, db4 = wgrads()
-            for t in (dl, seq):
-                t.record_stream(side)
-            _PENDING.extend([(w4, dw4), (b4, db4)])
+                side = torch.cuda.current_stream()
+                for t in (dl, seq):
+                    t.record_stream(side)
+                _PENDING.extend([(w4, dw4), (b4, db4)])
+
+            _PENDING_FNS.append((ev, deferred))
             return dseq, None, None
         dw4, db4 = wgrads()
         for p, g_ in ((w4, dw4), (b4, db4)):
@@ -557,7 +593,13 @@ class FrontFn(torch.autograd.Function):
         if _defer_active():
             # the front produces ONLY weight gradients (it is the first
             # layer): the whole recompute backward runs on the side stream,
-            # overlapping the GRU BPTT kernels still on the main stream
+            # overlapping the GRU BPTT kernels still on the main stream.
+            # Enqueue the stashed GRU/head weight-grad closures FIRST:
+            # front_bwd fills the whole chip, so side-stream kernels cannot
+            # co-run with it (kernel trace: a 12 us slice_sum stretched to
+            # 685 us under it) — their only overlap window is the 8-WG BPTT
+            # region still draining on the main stream.
+            _run_pending_fns()
             cur = torch.cuda.current_stream()
             side = _side2_stream(dseq.device)
             side.wait_stream(cur)
