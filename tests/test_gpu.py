@@ -530,3 +530,24 @@ def test_xg_gemm2_numerics():
         out = ext.xg_gemm2(A, Wp.contiguous(), bias)
         rel = ((out.float() - ref).abs().max() / ref.abs().max()).item()
         assert rel < 2e-2, (kreal, rel)
+
+
+def test_pipeline_parity_large_batch():
+    """Large serving batches auto-enable the C++ hipGraph slot path
+    (B >= 256); outputs must stay bit-exact vs the eager forward."""
+    from roko_amd.ops.forward import InferencePipeline, roko_argmax
+
+    model = RokoModel().cuda().eval()
+    g = torch.Generator().manual_seed(11)
+    for b in (256, 512):
+        pipe = InferencePipeline(model, b, depth=3)
+        x = torch.randint(0, 12, (b, 200, 90), generator=g,
+                          dtype=torch.uint8).cuda()
+        # several submits so the graph capture (2nd run) and replay paths
+        # are both exercised on fresh inputs
+        for _ in range(3):
+            t = pipe.submit(x)
+        got = t()
+        ref = roko_argmax(model, x).cpu()
+        assert torch.equal(got, ref), (b, (got != ref).float().mean())
+        del pipe
