@@ -119,3 +119,58 @@ def test_train_resume_roundtrip(tmp_path, small_train_rkw=None):
     m2, h2 = train(path, out, cfg=cfg2, log=lambda *a, **k: None)
     # resumed run starts at epoch 2 -> exactly one more epoch of history
     assert len(h2) == 1 and h2[0]["epoch"] == 2
+
+
+def test_multi_contig_end_to_end(rng, tmp_path):
+    """Two contigs in one draft/BAM: features must window both, inference
+    must vote/stitch each independently and emit both polished sequences
+    (reference behavior: per-contig groups, inference.py stitching)."""
+    import numpy as np
+    import torch as _torch
+
+    from roko_amd import features as F
+    from roko_amd.inference import infer
+    from roko_amd.io.bamio import write_bam
+    from roko_amd.io.fasta import write_fasta
+    from tests.simple_align import EditScript
+
+    refs, reads, fasta_entries = [], [], []
+    truths = {}
+    for tid, name in enumerate(("ctgA", "ctgB")):
+        truth = "".join("ACGT"[int(b)] for b in rng.integers(0, 4, 1500))
+        es = EditScript(rng, truth, sub_rate=0.01, ins_rate=0.003,
+                        del_rate=0.003)
+        truths[name] = (truth, es)
+        fasta_entries.append((name, es.draft))
+        refs.append((name, len(es.draft)))
+        for i in range(1500 * 12 // 300):
+            s = int(rng.integers(0, 1200))
+            rec = es.align_substring(f"{name}_r{i}", s, s + 300,
+                                     flag=16 if i % 2 else 0, tid=tid)
+            if rec is not None:
+                reads.append(rec)
+    reads.sort(key=lambda r: (r.tid, r.pos))
+    draft_fasta = str(tmp_path / "draft.fasta")
+    write_fasta(draft_fasta, fasta_entries)
+    bam = str(tmp_path / "reads.bam")
+    write_bam(bam, refs, reads)
+
+    infer_rkw = str(tmp_path / "infer.rkw")
+    F.run(draft_fasta, bam, infer_rkw, workers=1,
+          cfg=F.FeatureConfig(region_size=1000, region_overlap=300),
+          log=lambda *a: None)
+
+    from roko_amd.model import RokoModel
+    model = RokoModel()
+    ckpt = str(tmp_path / "m.pth")
+    _torch.save(model.state_dict(), ckpt)
+    seqs = infer(infer_rkw, ckpt, str(tmp_path / "out.fasta"),
+                 batch_size=16, log=lambda *a: None)
+    assert set(seqs) == {"ctgA", "ctgB"}
+    for name, s in seqs.items():
+        assert 500 < len(s) < 3000, (name, len(s))
+        assert set(s) <= set("ACGT")
+    # the FASTA on disk round-trips both contigs
+    from roko_amd.io.fasta import read_fasta
+    back = dict(read_fasta(str(tmp_path / "out.fasta")))
+    assert set(back) == {"ctgA", "ctgB"}
