@@ -30,6 +30,9 @@
 //   h' = (1 - z) * n + z * h
 
 #include <cstdint>
+#include <cstdlib>
+#include <string>
+#include <type_traits>
 
 #include "common.h"
 
@@ -58,8 +61,12 @@ constexpr int XLD = XKP + 8;    // x-stage LDS row stride (bank pad)
 // fraction of the aggregate CU time; the per-chain latency rises (~+60 us),
 // which deep pipelining absorbs. Not used by training: there the step's
 // LATENCY is the metric and 225 workgroups beat 8.
-template <bool TRAIN, int FUSEKP = 0>  // FUSEKP: 0 = xg precomputed;
-                                       // 256/512 = in-kernel xg GEMM width
+template <bool TRAIN, int FUSEKP = 0, int MBT = MB>
+// FUSEKP: 0 = xg precomputed; 256/512 = in-kernel xg GEMM width.
+// MBT: batch rows per workgroup. 32 is the default; 16 (eval only) doubles
+// the workgroup count and HALVES each thread's gate-chain count — the gate
+// phase sits at the 2-wave VALU-issue floor, so per-step time drops with
+// per-thread instruction count, and b=128 serving gets 16 WGs of supply.
 __global__ __launch_bounds__(WAVES * 64, 2) void gru_layer_fwd_kernel(
     const bf16* __restrict__ xg,   // (T, B, 2, 3H)  W_ih·x + b_ih
                                    // (FUSEXG: workspace this kernel fills)
@@ -75,15 +82,20 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gru_layer_fwd_kernel(
     const bf16* __restrict__ b_ih = nullptr,   // (768)
     int IN = 0) {
     constexpr bool FUSEXG = FUSEKP > 0;
+    static_assert(!FUSEXG || MBT == 32, "fused xg path is MB=32 only");
+    constexpr int MT = MBT / 16;         // 16-row m-subtiles per thread
+    constexpr int CW = (MBT == 32) ? 8 : 4;  // staging chunk width (bf16)
+    constexpr int XCHT = MBT * G3 / (WAVES * 64 * CW);  // xg chunks/thread
+    using xvec = std::conditional_t<CW == 8, bf16x8, bf16x4>;
     constexpr int XLDK = FUSEXG ? FUSEKP + 8 : 1;
     __shared__ struct {
-        bf16 h[2][MB][HPAD];         // double-buffered hidden-state mirror
-        bf16 xgb[2][MB][G3];         // double-buffered step gate inputs
-        bf16 x_st[FUSEXG ? 2 : 1][FUSEXG ? MB : 1][XLDK];
+        bf16 h[2][MBT][HPAD];        // double-buffered hidden-state mirror
+        bf16 xgb[2][MBT][G3];        // double-buffered step gate inputs
+        bf16 x_st[FUSEXG ? 2 : 1][FUSEXG ? MBT : 1][XLDK];
     } lds;
 
     const int dir = blockIdx.y;
-    const int b0 = blockIdx.x * MB;
+    const int b0 = blockIdx.x * MBT;
     const int tid = threadIdx.x;
     const int wid = tid >> 6;
     const int lane = tid & 63;
@@ -108,7 +120,7 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gru_layer_fwd_kernel(
         auto stage_x = [&](int t, int buf) {
             const bf16* src = x + ((size_t)t * B + b0) * IN;
 #pragma unroll
-            for (int p = 0; p < MB * FUSEKP / 4 / (WAVES * 64); ++p) {
+            for (int p = 0; p < MBT * FUSEKP / 4 / (WAVES * 64); ++p) {
                 const int c = p * WAVES * 64 + tid;
                 const int row = (c * 4) / FUSEKP, col = (c * 4) % FUSEKP;
                 uint64_t v = 0;  // 4 bf16 (zero pads cols IN..KP)
@@ -204,12 +216,12 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gru_layer_fwd_kernel(
     for (int g = 0; g < 3; ++g) bhh_reg[g] = bhh[dir * G3 + g * H + j0 + lcol];
 
     // ---- zero hidden state -------------------------------------------------
-    float hreg[2][4];  // [mt][i] fp32 master copy, fragment-shaped
+    float hreg[MT][4];  // [mt][i] fp32 master copy, fragment-shaped
 #pragma unroll
-    for (int mt = 0; mt < 2; ++mt)
+    for (int mt = 0; mt < MT; ++mt)
 #pragma unroll
         for (int i = 0; i < 4; ++i) hreg[mt][i] = 0.0f;
-    for (int e = tid; e < MB * HPAD; e += WAVES * 64) lds.h[0][0][e] = f2bf(0.0f);
+    for (int e = tid; e < MBT * HPAD; e += WAVES * 64) lds.h[0][0][e] = f2bf(0.0f);
 
     const int stp = (dir == 0) ? 1 : -1;
     const int tfirst = (dir == 0) ? 0 : T - 1;
@@ -221,22 +233,22 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gru_layer_fwd_kernel(
     {
         const bf16* src = xg_src(tfirst);
 #pragma unroll
-        for (int p = 0; p < XCH; ++p) {
-            int e = (p * WAVES * 64 + tid) * 8;
+        for (int p = 0; p < XCHT; ++p) {
+            int e = (p * WAVES * 64 + tid) * CW;
             int row = e / G3, col = e % G3;
-            *reinterpret_cast<bf16x8*>(&lds.xgb[0][row][col]) =
-                *reinterpret_cast<const bf16x8*>(src + (size_t)row * 2 * G3 + col);
+            *reinterpret_cast<xvec*>(&lds.xgb[0][row][col]) =
+                *reinterpret_cast<const xvec*>(src + (size_t)row * 2 * G3 + col);
         }
     }
-    bf16x8 stageA[XCH], stageB[XCH];
+    xvec stageA[XCHT], stageB[XCHT];
     if (T > 1) {
         const bf16* src = xg_src(tfirst + stp);
 #pragma unroll
-        for (int p = 0; p < XCH; ++p) {
-            int e = (p * WAVES * 64 + tid) * 8;
+        for (int p = 0; p < XCHT; ++p) {
+            int e = (p * WAVES * 64 + tid) * CW;
             int row = e / G3, col = e % G3;
             stageA[p] =
-                *reinterpret_cast<const bf16x8*>(src + (size_t)row * 2 * G3 + col);
+                *reinterpret_cast<const xvec*>(src + (size_t)row * 2 * G3 + col);
         }
     }
     __syncthreads();
@@ -247,41 +259,41 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gru_layer_fwd_kernel(
     // write into LDS only touches registers loaded a FULL step earlier (the
     // v3 kernel's stage rotation copy forced an end-of-step vmcnt(0) on
     // loads ~200 cycles old — profiles/pmc_gru_r01.txt's 36% SQ_WAIT_ANY).
-    auto body = [&](int ti, int curp, bf16x8 (&st_wr)[XCH],
-                    bf16x8 (&st_ld)[XCH]) {
+    auto body = [&](int ti, int curp, xvec (&st_wr)[XCHT],
+                    xvec (&st_ld)[XCHT]) {
         const int t = tfirst + stp * ti;
         // issue xg[t+2] loads FIRST (they retire next step), then write the
         // staged xg[t+1] (loaded last step) into the back LDS buffer
         if (ti + 2 < T && !(dbg & 8u)) {
             const bf16* src = xg_src(t + 2 * stp);
 #pragma unroll
-            for (int p = 0; p < XCH; ++p) {
-                int e = (p * WAVES * 64 + tid) * 8;
+            for (int p = 0; p < XCHT; ++p) {
+                int e = (p * WAVES * 64 + tid) * CW;
                 int row = e / G3, col = e % G3;
                 st_ld[p] =
-                    *reinterpret_cast<const bf16x8*>(src + (size_t)row * 2 * G3 + col);
+                    *reinterpret_cast<const xvec*>(src + (size_t)row * 2 * G3 + col);
             }
         }
         if (ti + 1 < T && !(dbg & 8u)) {
 #pragma unroll
-            for (int p = 0; p < XCH; ++p) {
-                int e = (p * WAVES * 64 + tid) * 8;
+            for (int p = 0; p < XCHT; ++p) {
+                int e = (p * WAVES * 64 + tid) * CW;
                 int row = e / G3, col = e % G3;
-                *reinterpret_cast<bf16x8*>(&lds.xgb[curp ^ 1][row][col]) = st_wr[p];
+                *reinterpret_cast<xvec*>(&lds.xgb[curp ^ 1][row][col]) = st_wr[p];
             }
         }
 
         // gates_h = h · U^T  (24 MFMA per wave, A-frags shared 3-ways)
-        f32x4 acc[2][3];
+        f32x4 acc[MT][3];
 #pragma unroll
-        for (int mt = 0; mt < 2; ++mt)
+        for (int mt = 0; mt < MT; ++mt)
 #pragma unroll
             for (int g = 0; g < 3; ++g) acc[mt][g] = f32x4{0.f, 0.f, 0.f, 0.f};
         if (!(dbg & 16u)) {
 #pragma unroll
             for (int kb = 0; kb < 4; ++kb) {
 #pragma unroll
-                for (int mt = 0; mt < 2; ++mt) {
+                for (int mt = 0; mt < MT; ++mt) {
                     bf16x8 a = lds_load_a_frag(&lds.h[curp][0][0], mt * 16, kb * 32, HPAD);
 #pragma unroll
                     for (int g = 0; g < 3; ++g)
@@ -300,7 +312,7 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gru_layer_fwd_kernel(
         // phase alone was 55% of the kernel (scripts/gru_timing.py).
         if (dbg & 4u) {  // timing: gate math stripped (separate cold loop)
 #pragma unroll
-            for (int mt = 0; mt < 2; ++mt)
+            for (int mt = 0; mt < MT; ++mt)
 #pragma unroll
                 for (int i = 0; i < 4; ++i) {
                     const int row = mt * 16 + lrow * 4 + i;
@@ -322,9 +334,9 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gru_layer_fwd_kernel(
                 }
         } else {
             const int j = j0 + lcol;
-            float xr[2][4], xz[2][4], xn[2][4];
+            float xr[MT][4], xz[MT][4], xn[MT][4];
 #pragma unroll
-            for (int mt = 0; mt < 2; ++mt)
+            for (int mt = 0; mt < MT; ++mt)
 #pragma unroll
                 for (int i = 0; i < 4; ++i) {
                     const int row = mt * 16 + lrow * 4 + i;
@@ -332,9 +344,9 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gru_layer_fwd_kernel(
                     xz[mt][i] = bf2f(lds.xgb[curp][row][1 * H + j]);
                     xn[mt][i] = bf2f(lds.xgb[curp][row][2 * H + j]);
                 }
-            float r8[2][4], z8[2][4], n8[2][4], hg8[2][4];
+            float r8[MT][4], z8[MT][4], n8[MT][4], hg8[MT][4];
 #pragma unroll
-            for (int mt = 0; mt < 2; ++mt)
+            for (int mt = 0; mt < MT; ++mt)
 #pragma unroll
                 for (int i = 0; i < 4; ++i) {
                     hg8[mt][i] = acc[mt][2][i] + bhh_reg[2];
@@ -344,12 +356,12 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gru_layer_fwd_kernel(
                         sigmoidf_dev(xz[mt][i] + acc[mt][1][i] + bhh_reg[1]);
                 }
 #pragma unroll
-            for (int mt = 0; mt < 2; ++mt)
+            for (int mt = 0; mt < MT; ++mt)
 #pragma unroll
                 for (int i = 0; i < 4; ++i)
                     n8[mt][i] = tanhf_dev(xn[mt][i] + r8[mt][i] * hg8[mt][i]);
 #pragma unroll
-            for (int mt = 0; mt < 2; ++mt)
+            for (int mt = 0; mt < MT; ++mt)
 #pragma unroll
                 for (int i = 0; i < 4; ++i) {
                     const int row = mt * 16 + lrow * 4 + i;
@@ -378,10 +390,10 @@ __global__ __launch_bounds__(WAVES * 64, 2) void gru_layer_fwd_kernel(
         // overlaps the next step's MFMAs as plain VMEM traffic
         if (!(dbg & 1u)) {
             bf16* dst = hseq + (((size_t)t * B + b0) * 2 + dir) * H;
-            const int e = tid * 8;
+            const int e = tid * CW;
             const int row = e / H, col = e % H;
-            *reinterpret_cast<bf16x8*>(dst + (size_t)row * 2 * H + col) =
-                *reinterpret_cast<const bf16x8*>(&lds.h[curp ^ 1][row][col]);
+            *reinterpret_cast<xvec*>(dst + (size_t)row * 2 * H + col) =
+                *reinterpret_cast<const xvec*>(&lds.h[curp ^ 1][row][col]);
         }
     };
 
@@ -398,18 +410,52 @@ void gru_layer_fwd(const void* xg, const void* u, const float* bhh, void* hseq,
                    uint32_t dbg) {
     dim3 grid(B / MB, 2);
     dim3 block(WAVES * 64);
-    if (cache)
-        hipLaunchKernelGGL((gru_layer_fwd_kernel<true, 0>), grid, block, 0,
-                           stream,
-                           static_cast<const bf16*>(xg), static_cast<const bf16*>(u),
-                           bhh, static_cast<bf16*>(hseq), static_cast<bf16*>(cache),
-                           T, B, dbg, nullptr, nullptr, nullptr, 0);
-    else
-        hipLaunchKernelGGL((gru_layer_fwd_kernel<false, 0>), grid, block, 0,
-                           stream,
-                           static_cast<const bf16*>(xg), static_cast<const bf16*>(u),
-                           bhh, static_cast<bf16*>(hseq), nullptr, T, B, dbg,
-                           nullptr, nullptr, nullptr, 0);
+    // MBT=16 halves each thread's gate-chain count — the gate phase sits
+    // at the 2-wave VALU-issue floor, so the wall nearly halves (eval
+    // 175.6 -> 97.7 us measured) and the WG count doubles (serving
+    // packing). ROKO_GRU_MB=32 reverts both paths.
+    static const int mb_sel = [] {
+        const char* f = getenv("ROKO_GRU_MB");
+        return (f && std::string(f) == "32") ? 32 : 16;
+    }();
+    if (cache) {
+        if (mb_sel == 16 && (B % 16) == 0)
+            hipLaunchKernelGGL((gru_layer_fwd_kernel<true, 0, 16>),
+                               dim3(B / 16, 2), block, 0, stream,
+                               static_cast<const bf16*>(xg),
+                               static_cast<const bf16*>(u), bhh,
+                               static_cast<bf16*>(hseq),
+                               static_cast<bf16*>(cache), T, B, dbg, nullptr,
+                               nullptr, nullptr, 0);
+        else
+            hipLaunchKernelGGL((gru_layer_fwd_kernel<true, 0>), grid, block,
+                               0, stream, static_cast<const bf16*>(xg),
+                               static_cast<const bf16*>(u), bhh,
+                               static_cast<bf16*>(hseq),
+                               static_cast<bf16*>(cache), T, B, dbg, nullptr,
+                               nullptr, nullptr, 0);
+    }
+    else {
+        // eval tile choice: MBT=16 doubles the WG count and halves each
+        // thread's gate-chain count (ROKO_GRU_MB=32 reverts)
+        static const int mb_eval = [] {
+            const char* f = getenv("ROKO_GRU_MB");
+            return (f && std::string(f) == "32") ? 32 : 16;
+        }();
+        if (mb_eval == 16 && (B % 16) == 0)
+            hipLaunchKernelGGL((gru_layer_fwd_kernel<false, 0, 16>),
+                               dim3(B / 16, 2), block, 0, stream,
+                               static_cast<const bf16*>(xg),
+                               static_cast<const bf16*>(u), bhh,
+                               static_cast<bf16*>(hseq), nullptr, T, B, dbg,
+                               nullptr, nullptr, nullptr, 0);
+        else
+            hipLaunchKernelGGL((gru_layer_fwd_kernel<false, 0>), grid, block,
+                               0, stream, static_cast<const bf16*>(xg),
+                               static_cast<const bf16*>(u), bhh,
+                               static_cast<bf16*>(hseq), nullptr, T, B, dbg,
+                               nullptr, nullptr, nullptr, 0);
+    }
 }
 
 // serving variant: xg computed in-kernel from (x, W_ih, b_ih); `xg` is a
