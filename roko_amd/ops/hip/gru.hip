@@ -506,6 +506,7 @@ void gru_layer_fwd_fused(const void* x, const void* w_ih_p, const void* b_ih,
 constexpr int BW_WAVES = 8;  // dhg GEMM has K=384: 24 MFMA/wave at 8 waves
 constexpr int BWCH_C = 4;    // cache chunks per thread (32x512 / 512 / 8)
 
+template <int MBT = MB>
 __global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
     const bf16* __restrict__ cache,  // (T, B, 2, H, 4) [r z n hgn] packed
     const bf16* __restrict__ hseq,   // (T, B, 2, H)
@@ -524,15 +525,17 @@ __global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
     // and output stores still read, and the dg_st staging tile is gone —
     // dxg/dhg stores read straight from lds.dhg (dxr/dxz/dhgn blocks) plus
     // per-lane register stores for the dxn column block.
+    constexpr int MT = MBT / 16;               // 16-row m-subtiles
+    constexpr int BWCH = MBT * 4 * H / (BW_WAVES * 64 * 8);  // cache chunks
     __shared__ struct {
-        bf16 cache_st[2][MB][4 * H];  // double-buffered staged cache[t]
-        bf16 dhin_st[2][MB][H];       // double-buffered staged dhin[t]
-        bf16 hprev_st[2][MB][H];      // double-buffered staged h_{t-1}
-        bf16 dhg[2][MB][G3 + 8];      // A-operand of the dh GEMM (2 buffers)
+        bf16 cache_st[2][MBT][4 * H];  // double-buffered staged cache[t]
+        bf16 dhin_st[2][MBT][H];       // double-buffered staged dhin[t]
+        bf16 hprev_st[2][MBT][H];      // double-buffered staged h_{t-1}
+        bf16 dhg[2][MBT][G3 + 8];      // A-operand of the dh GEMM (2 bufs)
     } lds;
 
     const int dir = blockIdx.y;
-    const int b0 = blockIdx.x * MB;
+    const int b0 = blockIdx.x * MBT;
     const int tid = threadIdx.x;
     const int wid = tid >> 6;
     const int lane = tid & 63;
@@ -551,9 +554,9 @@ __global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
             ut + ((size_t)dir * H + col) * G3 + j);
     }
 
-    float dhc[2][4];  // dh carry, fragment-shaped
+    float dhc[MT][4];  // dh carry, fragment-shaped
 #pragma unroll
-    for (int mt = 0; mt < 2; ++mt)
+    for (int mt = 0; mt < MT; ++mt)
 #pragma unroll
         for (int i = 0; i < 4; ++i) dhc[mt][i] = 0.0f;
 
@@ -562,7 +565,7 @@ __global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
     auto t_of = [&](int sidx) { return t0 + stp * sidx; };
 
     // register-staged loads for one step: cache (4 chunks) + dhin + hprev
-    auto issue_loads = [&](int sidx, bf16x8 (&rc)[BWCH_C], bf16x8& rdh,
+    auto issue_loads = [&](int sidx, bf16x8 (&rc)[BWCH], bf16x8& rdh,
                            bf16x8& rhp) {
         if (dbg & 8u) return;
         const int t = t_of(sidx);
@@ -570,30 +573,34 @@ __global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
         const bf16* src = cache + (((size_t)t * B + b0) * 2 + dir) * 4 * H;
         const int row4 = tid / 64, col4 = (tid % 64) * 8;
 #pragma unroll
-        for (int q = 0; q < BWCH_C; ++q)
+        for (int q = 0; q < BWCH; ++q)
             rc[q] = *reinterpret_cast<const bf16x8*>(
                 src + (size_t)(row4 + q * 8) * 2 * 4 * H + col4);
         const int row = tid / 16, col = (tid % 16) * 8;
-        rdh = *reinterpret_cast<const bf16x8*>(
-            dhin + (((size_t)t * B + b0) * 2 + dir) * H + (size_t)row * 2 * H + col);
-        if (tp >= 0 && tp < T)
-            rhp = *reinterpret_cast<const bf16x8*>(
-                hseq + (((size_t)tp * B + b0) * 2 + dir) * H + (size_t)row * 2 * H + col);
-        else
-            rhp = bf16x8{};
+        rhp = bf16x8{};
+        if (row < MBT) {
+            rdh = *reinterpret_cast<const bf16x8*>(
+                dhin + (((size_t)t * B + b0) * 2 + dir) * H + (size_t)row * 2 * H + col);
+            if (tp >= 0 && tp < T)
+                rhp = *reinterpret_cast<const bf16x8*>(
+                    hseq + (((size_t)tp * B + b0) * 2 + dir) * H + (size_t)row * 2 * H + col);
+        } else
+            rdh = bf16x8{};
     };
-    auto write_stage = [&](int buf, bf16x8 (&rc)[BWCH_C], bf16x8& rdh,
+    auto write_stage = [&](int buf, bf16x8 (&rc)[BWCH], bf16x8& rdh,
                            bf16x8& rhp) {
         const int row4 = tid / 64, col4 = (tid % 64) * 8;
 #pragma unroll
-        for (int q = 0; q < BWCH_C; ++q)
+        for (int q = 0; q < BWCH; ++q)
             *reinterpret_cast<bf16x8*>(&lds.cache_st[buf][row4 + q * 8][col4]) = rc[q];
         const int row = tid / 16, col = (tid % 16) * 8;
-        *reinterpret_cast<bf16x8*>(&lds.dhin_st[buf][row][col]) = rdh;
-        *reinterpret_cast<bf16x8*>(&lds.hprev_st[buf][row][col]) = rhp;
+        if (row < MBT) {
+            *reinterpret_cast<bf16x8*>(&lds.dhin_st[buf][row][col]) = rdh;
+            *reinterpret_cast<bf16x8*>(&lds.hprev_st[buf][row][col]) = rhp;
+        }
     };
 
-    bf16x8 rcA[BWCH_C], rcB[BWCH_C];
+    bf16x8 rcA[BWCH], rcB[BWCH];
     bf16x8 rdhA, rhpA, rdhB, rhpB;
     // prologue: step 0 straight to LDS buffer 0; step 1 to registers A
     issue_loads(0, rcA, rdhA, rhpA);
@@ -601,8 +608,8 @@ __global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
     if (T > 1) issue_loads(1, rcA, rdhA, rhpA);
     __syncthreads();
 
-    auto body = [&](int sidx, int curp, bf16x8 (&rc_wr)[BWCH_C], bf16x8& rdh_wr,
-                    bf16x8& rhp_wr, bf16x8 (&rc_ld)[BWCH_C], bf16x8& rdh_ld,
+    auto body = [&](int sidx, int curp, bf16x8 (&rc_wr)[BWCH], bf16x8& rdh_wr,
+                    bf16x8& rhp_wr, bf16x8 (&rc_ld)[BWCH], bf16x8& rdh_ld,
                     bf16x8& rhp_ld) {
         const int t = t_of(sidx);
         // issue step s+2 loads FIRST, then write step s+1's staged registers
@@ -613,14 +620,14 @@ __global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
         // reads batched first, then 8 independent arithmetic chains, then
         // the stores — same restructure as the forward's gate phase (the
         // interleaved per-element form serialized the read latencies)
-        float dhp_part[2][4];
-        float dxn8[2][4];  // kept in registers for the direct dxg store
+        float dhp_part[MT][4];
+        float dxn8[MT][4];  // kept in registers for the direct dxg store
         {
             const int j = j0 + lcol;
-            float dh8[2][4], hp8[2][4];
-            bf16x4 pk8[2][4];
+            float dh8[MT][4], hp8[MT][4];
+            bf16x4 pk8[MT][4];
 #pragma unroll
-            for (int mt = 0; mt < 2; ++mt)
+            for (int mt = 0; mt < MT; ++mt)
 #pragma unroll
                 for (int i = 0; i < 4; ++i) {
                     const int row = mt * 16 + lrow * 4 + i;
@@ -632,7 +639,7 @@ __global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
                 }
             if (dbg & 2u) {  // timing: gate math stripped
 #pragma unroll
-                for (int mt = 0; mt < 2; ++mt)
+                for (int mt = 0; mt < MT; ++mt)
 #pragma unroll
                     for (int i = 0; i < 4; ++i) {
                         const int row = mt * 16 + lrow * 4 + i;
@@ -645,7 +652,7 @@ __global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
                     }
             } else
 #pragma unroll
-            for (int mt = 0; mt < 2; ++mt)
+            for (int mt = 0; mt < MT; ++mt)
 #pragma unroll
                 for (int i = 0; i < 4; ++i) {
                     const int row = mt * 16 + lrow * 4 + i;
@@ -671,14 +678,14 @@ __global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
                           // barrier per step; next step writes dhg[curp^1])
 
         // ---- dh_prev = dh*z + dhg · U  (24 MFMA per wave) -----------------
-        f32x4 acc[2];
+        f32x4 acc[MT];
 #pragma unroll
-        for (int mt = 0; mt < 2; ++mt) acc[mt] = f32x4{0.f, 0.f, 0.f, 0.f};
+        for (int mt = 0; mt < MT; ++mt) acc[mt] = f32x4{0.f, 0.f, 0.f, 0.f};
         if (!(dbg & 4u)) {
 #pragma unroll
             for (int kb = 0; kb < 12; ++kb) {
 #pragma unroll
-                for (int mt = 0; mt < 2; ++mt) {
+                for (int mt = 0; mt < MT; ++mt) {
                     bf16x8 a = lds_load_a_frag(&lds.dhg[curp][0][0], mt * 16,
                                                kb * 32, G3 + 8);
                     acc[mt] = mfma16x16x32(a, ufrag[kb], acc[mt]);
@@ -686,7 +693,7 @@ __global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
             }
         }
 #pragma unroll
-        for (int mt = 0; mt < 2; ++mt)
+        for (int mt = 0; mt < MT; ++mt)
 #pragma unroll
             for (int i = 0; i < 4; ++i) dhc[mt][i] = dhp_part[mt][i] + acc[mt][i];
 
@@ -697,14 +704,14 @@ __global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
             bf16* dst = dxg + (((size_t)t * B + b0) * 2 + dir) * G3;
             const int row = tid / 32, col = (tid % 32) * 8;  // 2H block
 #pragma unroll
-            for (int q = 0; q < 2; ++q)
+            for (int q = 0; q < MT; ++q)
                 *reinterpret_cast<bf16x8*>(
                     dst + (size_t)(row + q * 16) * 2 * G3 + col) =
                     *reinterpret_cast<const bf16x8*>(
                         &lds.dhg[curp][row + q * 16][col]);
             const int j = j0 + lcol;
 #pragma unroll
-            for (int mt = 0; mt < 2; ++mt)
+            for (int mt = 0; mt < MT; ++mt)
 #pragma unroll
                 for (int i = 0; i < 4; ++i) {
                     const int row2 = mt * 16 + lrow * 4 + i;
@@ -718,7 +725,7 @@ __global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
             const int row = tid / 64, col = (tid % 64) * 8;
             if (col < G3) {
 #pragma unroll
-                for (int q = 0; q < 4; ++q)
+                for (int q = 0; q < MBT / 8; ++q)
                     *reinterpret_cast<bf16x8*>(
                         dst + (size_t)(row + q * 8) * G3 + col) =
                         *reinterpret_cast<const bf16x8*>(
@@ -738,13 +745,27 @@ __global__ __launch_bounds__(BW_WAVES * 64, 2) void gru_layer_bwd_kernel(
 void gru_layer_bwd(const void* cache, const void* hseq, const void* dhin,
                    const void* ut, void* dxg, void* dhg, int T, int B,
                    hipStream_t stream, uint32_t dbg) {
-    dim3 grid(B / MB, 2);
     dim3 block(BW_WAVES * 64);
-    hipLaunchKernelGGL(gru_layer_bwd_kernel, grid, block, 0, stream,
-                       static_cast<const bf16*>(cache), static_cast<const bf16*>(hseq),
-                       static_cast<const bf16*>(dhin), static_cast<const bf16*>(ut),
-                       static_cast<bf16*>(dxg), static_cast<bf16*>(dhg), T, B,
-                       dbg);
+    static const int mb_sel = [] {
+        const char* f = getenv("ROKO_GRU_MB");
+        return (f && std::string(f) == "32") ? 32 : 16;
+    }();
+    if (mb_sel == 16 && (B % 16) == 0)
+        hipLaunchKernelGGL((gru_layer_bwd_kernel<16>), dim3(B / 16, 2), block,
+                           0, stream, static_cast<const bf16*>(cache),
+                           static_cast<const bf16*>(hseq),
+                           static_cast<const bf16*>(dhin),
+                           static_cast<const bf16*>(ut),
+                           static_cast<bf16*>(dxg), static_cast<bf16*>(dhg),
+                           T, B, dbg);
+    else
+        hipLaunchKernelGGL((gru_layer_bwd_kernel<32>), dim3(B / MB, 2), block,
+                           0, stream, static_cast<const bf16*>(cache),
+                           static_cast<const bf16*>(hseq),
+                           static_cast<const bf16*>(dhin),
+                           static_cast<const bf16*>(ut),
+                           static_cast<bf16*>(dxg), static_cast<bf16*>(dhg),
+                           T, B, dbg);
 }
 
 }  // namespace rk
