@@ -6,7 +6,10 @@
 //     done on the host side (hipBLASLt via torch.matmul) — this kernel gets
 //     the precomputed per-step gate inputs `xg`;
 //   * ONE kernel launch runs the full T-step recurrence of a layer: grid =
-//     (B/32 batch tiles) x (2 directions); each workgroup owns 32 batch rows
+//     (B/MBT batch tiles) x (2 directions); each workgroup owns MBT batch
+//     rows (default 16 — the gate phase is VALU-issue-floor bound, so the
+//     smallest MFMA-viable tile halves the per-step wall AND doubles the
+//     workgroup supply for serving packing; docs/KERNELS.md lesson 18)
 //     whose hidden state lives in fp32 registers with a bf16 mirror in LDS
 //     as the MFMA A-operand — no per-step launches, no grid-wide sync;
 //   * 8 waves per workgroup = two per SIMD: the matrix and VALU pipes are
