@@ -67,6 +67,11 @@ void xg_gemm2(const void* A, const void* Bt, const void* bias, void* C,
 int atb_splitk_nslices(int K);
 void atb_splitk(const void* A, const void* B, float* ws, float* C, int M,
                 int N, int K, hipStream_t stream);
+void atb_splitk_ld(const void* A, int lda, const void* B, int ldb,
+                   float* ws, float* C, int M, int N, int K,
+                   hipStream_t stream);
+void colsum_f32(const void* X, int ldx, int K, int N, float* out,
+                hipStream_t stream);
 void front_de(const uint8_t* ids, const void* dt1g, const void* w1t_g,
               float* de, int B, uint32_t seed, float keep, hipStream_t stream,
               unsigned long long* timing, uint32_t dbg,
@@ -413,6 +418,76 @@ torch::Tensor xg_gemm2(torch::Tensor A, torch::Tensor Bt, torch::Tensor bias,
     rk::xg_gemm2(A.data_ptr(), Bt.data_ptr(), bias.data_ptr(), C.data_ptr(),
                  M, KREAL, KP, cur_stream(), tptr);
     return C;
+}
+
+// GRU layer weight gradients in ONE binding call: dU (both dirs, via
+// strided-slice split-K AtB — the t-shifted h_prev pairing makes both
+// operands regular slices, no torch.cat), dW_ih, and the three bias
+// column-sums. The aten form of this section (cats, contiguous copies,
+// ones-GEMVs, stack/float) cost ~1.3 ms of HOST enqueue per train step
+// (kernel trace tr5) — this is 7 raw launches.
+void gru_wgrads(torch::Tensor dhg,   // (2, T, B, 384) bf16, dir-major
+                torch::Tensor dxg,   // (T*B, 768) bf16 view
+                torch::Tensor hseq,  // (T, B, 2, 128) bf16
+                torch::Tensor x,     // (T*B, IN) bf16
+                torch::Tensor ws_u,  // (S', 384, 128) f32 scratch
+                torch::Tensor ws_w,  // (S, 768, IN) f32 scratch
+                torch::Tensor du,    // (2, 384, 128) f32 out
+                torch::Tensor dw,    // (768, IN) f32 out
+                torch::Tensor dbhh,  // (2, 384) f32 out
+                torch::Tensor dbih)  // (768,) f32 out
+{
+    check(dhg, torch::kBFloat16, "dhg");
+    check(dxg, torch::kBFloat16, "dxg");
+    check(hseq, torch::kBFloat16, "hseq");
+    check(x, torch::kBFloat16, "x");
+    const int TB = dxg.size(0), IN = x.size(1);
+    const int T = hseq.size(0), B = hseq.size(1);
+    constexpr int H = 128, G3 = 384;
+    TORCH_CHECK(T * B == TB && dhg.size(1) == T && dhg.size(2) == B,
+                "shape mismatch");
+    const int Kp = TB - B;
+    TORCH_CHECK(ws_u.size(0) >= rk::atb_splitk_nslices(Kp) &&
+                    ws_w.size(0) >= rk::atb_splitk_nslices(TB) &&
+                    ws_w.size(2) >= IN,
+                "workspace too small");
+    hipStream_t s = cur_stream();
+    const auto* dhg_f = static_cast<const uint16_t*>(dhg.data_ptr());
+    const auto* dhg_r = dhg_f + (size_t)TB * G3;
+    const auto* hs = static_cast<const uint16_t*>(hseq.data_ptr());
+    float* du_p = du.data_ptr<float>();
+    float* dbhh_p = dbhh.data_ptr<float>();
+    // forward dir: dhg[t] x h[t-1] -> drop t=0 (h_prev = 0)
+    rk::atb_splitk_ld(dhg_f + (size_t)B * G3, G3, hs, 2 * H,
+                      ws_u.data_ptr<float>(), du_p, G3, H, Kp, s);
+    // reverse dir: dhg[t] x h[t+1] -> drop t=T-1
+    rk::atb_splitk_ld(dhg_r, G3, hs + (size_t)B * 2 * H + H, 2 * H,
+                      ws_u.data_ptr<float>(), du_p + (size_t)G3 * H, G3, H,
+                      Kp, s);
+    rk::atb_splitk_ld(dxg.data_ptr(), 2 * G3, x.data_ptr(), IN,
+                      ws_w.data_ptr<float>(), dw.data_ptr<float>(), 2 * G3,
+                      IN, TB, s);
+    rk::colsum_f32(dhg_f, G3, TB, G3, dbhh_p, s);
+    rk::colsum_f32(dhg_r, G3, TB, G3, dbhh_p + G3, s);
+    rk::colsum_f32(dxg.data_ptr(), 2 * G3, TB, 2 * G3,
+                   dbih.data_ptr<float>(), s);
+}
+
+// head weight gradients: dw4 = dl^T x seq, db4 = column sums of dl
+void head_wgrads(torch::Tensor dl,    // (T*B, 5) bf16
+                 torch::Tensor seq,   // (T*B, 256) bf16
+                 torch::Tensor ws,    // (S, 5, 256) f32 scratch
+                 torch::Tensor dw4,   // (5, 256) f32 out
+                 torch::Tensor db4)   // (5,) f32 out
+{
+    check(dl, torch::kBFloat16, "dl");
+    check(seq, torch::kBFloat16, "seq");
+    const int TB = dl.size(0), N = seq.size(1), M = dl.size(1);
+    hipStream_t s = cur_stream();
+    rk::atb_splitk_ld(dl.data_ptr(), M, seq.data_ptr(), N,
+                      ws.data_ptr<float>(), dw4.data_ptr<float>(), M, N, TB,
+                      s);
+    rk::colsum_f32(dl.data_ptr(), M, TB, M, db4.data_ptr<float>(), s);
 }
 
 // fused train front bwd: -> (de, dw1, db1, dw2, db2) fp32
@@ -845,6 +920,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("gemm_bias", &gemm_bias, py::arg("A"), py::arg("B"),
           py::arg("bias") = c10::nullopt);
     m.def("xg_gemm", &xg_gemm);
+    m.def("gru_wgrads", &gru_wgrads);
+    m.def("head_wgrads", &head_wgrads);
     m.def("xg_gemm2", &xg_gemm2, py::arg("A"), py::arg("Bt"), py::arg("bias"),
           py::arg("timing") = py::none());
     m.def("front_de_timed", &front_de_timed, py::arg("ids"), py::arg("dt1g"),

@@ -184,12 +184,12 @@ constexpr int LDM = BM + 8;  // [k][m] row stride (staged UNtransposed)
 constexpr int WAVES = 8;
 
 __global__ __launch_bounds__(WAVES * 64, 2) void atb_splitk_kernel(
-    const bf16* __restrict__ A,  // (K, M)
-    const bf16* __restrict__ B,  // (K, N)
+    const bf16* __restrict__ A,  // (K, M) rows strided by lda
+    const bf16* __restrict__ B,  // (K, N) rows strided by ldb
     float* __restrict__ ws,      // (n_slices, M, N) partials (plain stores —
                                  // the atomic-commit version spent ~50 us in
                                  // 45-way same-address contention on dW_ih)
-    int M, int N, int K) {
+    int M, int N, int K, int lda, int ldb) {
     // staged in the GLOBAL orientation [k][m] / [k][n]: the transposed
     // staging this replaced wrote 8 scalar u16 per thread at an 8-row lane
     // stride whose dword period gcd'd with the 64 banks to 32 — 8-way bank
@@ -221,8 +221,8 @@ __global__ __launch_bounds__(WAVES * 64, 2) void atb_splitk_kernel(
     // [m][k]; same for B. 128*64 elems each = 16/thread.
     // same alignment rule as gemm_bias_kernel: vector loads only when the
     // row stride keeps every row 16-byte aligned
-    const bool m_aligned = (M % 8) == 0;
-    const bool n_aligned = (N % 8) == 0;
+    const bool m_aligned = (lda % 8) == 0;
+    const bool n_aligned = (ldb % 8) == 0;
     auto stage = [&](int buf, int k0) {
 #pragma unroll
         for (int p = 0; p < 2; ++p) {
@@ -232,12 +232,12 @@ __global__ __launch_bounds__(WAVES * 64, 2) void atb_splitk_kernel(
             const int krow = k0 + kk;
             if (krow < K && m_aligned && m0 + m + 7 < M) {
                 *reinterpret_cast<bf16x8*>(v) = *reinterpret_cast<const bf16x8*>(
-                    A + (size_t)krow * M + m0 + m);
+                    A + (size_t)krow * lda + m0 + m);
             } else {
 #pragma unroll
                 for (int q = 0; q < 8; ++q)
                     v[q] = (krow < K && m0 + m + q < M)
-                               ? A[(size_t)krow * M + m0 + m + q] : f2bf(0.f);
+                               ? A[(size_t)krow * lda + m0 + m + q] : f2bf(0.f);
             }
             *reinterpret_cast<bf16x8*>(&lds.at[buf][kk][m]) =
                 *reinterpret_cast<const bf16x8*>(v);
@@ -250,12 +250,12 @@ __global__ __launch_bounds__(WAVES * 64, 2) void atb_splitk_kernel(
             const int krow = k0 + kk;
             if (krow < K && n_aligned && n0 + n + 7 < N) {
                 *reinterpret_cast<bf16x8*>(v) = *reinterpret_cast<const bf16x8*>(
-                    B + (size_t)krow * N + n0 + n);
+                    B + (size_t)krow * ldb + n0 + n);
             } else {
 #pragma unroll
                 for (int q = 0; q < 8; ++q)
                     v[q] = (krow < K && n0 + n + q < N)
-                               ? B[(size_t)krow * N + n0 + n + q] : f2bf(0.f);
+                               ? B[(size_t)krow * ldb + n0 + n + q] : f2bf(0.f);
             }
             *reinterpret_cast<bf16x8*>(&lds.bt[buf][kk][n]) =
                 *reinterpret_cast<const bf16x8*>(v);
@@ -632,21 +632,71 @@ int atb_splitk_nslices(int K) {
     return (K + gemmatb::KSLICE - 1) / gemmatb::KSLICE;
 }
 
-void atb_splitk(const void* A, const void* B, float* ws, float* C, int M,
-                int N, int K, hipStream_t stream) {
+void atb_splitk_ld(const void* A, int lda, const void* B, int ldb,
+                   float* ws, float* C, int M, int N, int K,
+                   hipStream_t stream) {
     const int S = atb_splitk_nslices(K);
     dim3 grid(S, (M + gemmatb::BM - 1) / gemmatb::BM,
               (N + gemmatb::BN - 1) / gemmatb::BN);
     hipLaunchKernelGGL(gemmatb::atb_splitk_kernel, grid,
                        dim3(gemmatb::WAVES * 64), 0, stream,
                        static_cast<const bf16*>(A), static_cast<const bf16*>(B),
-                       ws, M, N, K);
+                       ws, M, N, K, lda, ldb);
     const int64_t MN = (int64_t)M * N;
     // ceil over the 4-element stride: floor(MN/4) left the last MN%4
     // outputs unwritten for odd shapes
     hipLaunchKernelGGL(gemmatb::slice_sum_kernel,
                        dim3(((MN + 3) / 4 + 255) / 256), dim3(256), 0, stream,
                        ws, C, S, MN);
+}
+
+void atb_splitk(const void* A, const void* B, float* ws, float* C, int M,
+                int N, int K, hipStream_t stream) {
+    atb_splitk_ld(A, M, B, N, ws, C, M, N, K, stream);
+}
+
+// column sums: out[n] = sum_k X[k*ldx + n], bf16 in / fp32 out. Replaces
+// the ones-vector hipBLASLt GEMVs in the weight-grad path (each a separate
+// aten launch; the whole weight-grad section is now raw kernel calls —
+// the aten form cost ~1.3 ms of HOST enqueue per train step, tr5 trace).
+namespace colsum {
+
+__global__ __launch_bounds__(256, 4) void colsum_kernel(
+    const bf16* __restrict__ X, float* __restrict__ out, int ldx, int K,
+    int N, int ksplit) {
+    const int n = blockIdx.x * 256 + threadIdx.x;
+    if (n >= N) return;
+    const int k0 = blockIdx.y * ksplit;
+    const int k1 = min(K, k0 + ksplit);
+    // 8 independent accumulators with the loads batched per iteration:
+    // a single-accumulator runtime loop compiled to load->wait->add per
+    // element (~340 us measured); this form keeps 8 loads in flight
+    float acc[8] = {0.f};
+    int k = k0;
+    for (; k + 8 <= k1; k += 8) {
+        float v[8];
+#pragma unroll
+        for (int q = 0; q < 8; ++q)
+            v[q] = bf2f(X[(size_t)(k + q) * ldx + n]);
+#pragma unroll
+        for (int q = 0; q < 8; ++q) acc[q] += v[q];
+    }
+    for (; k < k1; ++k) acc[0] += bf2f(X[(size_t)k * ldx + n]);
+    const float total = ((acc[0] + acc[1]) + (acc[2] + acc[3])) +
+                        ((acc[4] + acc[5]) + (acc[6] + acc[7]));
+    atomicAdd(&out[n], total);
+}
+
+}  // namespace colsum
+
+void colsum_f32(const void* X, int ldx, int K, int N, float* out,
+                hipStream_t stream) {
+    (void)hipMemsetAsync(out, 0, sizeof(float) * N, stream);
+    const int KSP = 32;
+    const int ksplit = (K + KSP - 1) / KSP;
+    hipLaunchKernelGGL(colsum::colsum_kernel,
+                       dim3((N + 255) / 256, KSP), dim3(256), 0, stream,
+                       static_cast<const bf16*>(X), out, ldx, K, N, ksplit);
 }
 
 }  // namespace rk

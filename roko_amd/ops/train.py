@@ -37,6 +37,41 @@ _SIDE = None       # per-layer weight-grad GEMMs
 _SIDE2 = None      # the front recompute backward (starts once the last GRU
                    # layer's dx exists; by then the main stream is idle)
 
+# Cached scratch + output buffers for the raw weight-grad kernels
+# (ext.gru_wgrads / ext.head_wgrads — one pybind call replaces the ~18-op
+# aten section whose host enqueue cost ~1.3 ms/step, kernel trace tr5).
+# Outputs ping-pong over TWO sets per key so a grad attached at drain time
+# stays valid while the NEXT step's closure writes the other set (at most
+# 2 deferred backwards may share a drain).
+_WGRAD_BUFS: dict = {}
+
+
+def _wgrad_bufs(key, dev, *shapes):
+    """key -> {'pp': int, 'sets': [tuple(tensors), tuple(tensors)]}.
+    shapes: (name, *dims) per output tensor; returns the next set."""
+    ent = _WGRAD_BUFS.get(key)
+    if ent is None:
+        mk = lambda dims: torch.empty(*dims, device=dev, dtype=torch.float32)
+        ent = {"pp": 0, "sets": [tuple(mk(d) for d in shapes),
+                                 tuple(mk(d) for d in shapes)]}
+        _WGRAD_BUFS[key] = ent
+    ent["pp"] ^= 1
+    return ent["sets"][ent["pp"]]
+
+
+def _wgrad_ws(dev, slices, rows, cols):
+    """Shared fp32 split-K workspace, grown on demand (scratch only —
+    consumed within each gru_wgrads/head_wgrads call, safe to share
+    across layers on one stream)."""
+    key = ("ws", dev)
+    ws = _WGRAD_BUFS.get(key)
+    if (ws is None or ws.shape[0] < slices or ws.shape[1] < rows
+            or ws.shape[2] < cols):
+        ws = torch.empty(slices, rows, cols, device=dev,
+                         dtype=torch.float32)
+        _WGRAD_BUFS[key] = ws
+    return ws
+
 
 def _side_stream(dev) -> "torch.cuda.Stream":
     global _SIDE
@@ -299,16 +334,25 @@ class GruLayerFn(torch.autograd.Function):
 
         if _defer_active() and ctx.prefs is not None:
             # host-defer the whole weight-grad section: enqueueing it here
-            # (~15 calls, ~250 us host) stalls the next layer's BPTT launch.
-            # The closure runs on the side stream at drain_deferred_grads(),
-            # ordered after this point by the event.
+            # stalls the next layer's BPTT launch. The closure runs on the
+            # side stream at drain_deferred_grads(), ordered after this
+            # point by the event — and it is ONE pybind call into raw
+            # kernels (ext.gru_wgrads): the aten form (cats, contiguous
+            # copies, ones-GEMVs) cost ~1.3 ms/step of host enqueue.
             ev = torch.cuda.Event()
             ev.record(torch.cuda.current_stream())
             prefs = ctx.prefs
             G3 = 3 * H
+            IN = x_bf.shape[-1]
 
             def deferred():
-                du, dw_ih, dbhh, db_ih = weight_grads()
+                du, dw_ih, dbhh, db_ih = _wgrad_bufs(
+                    id(prefs[0]), dx.device,
+                    (2, G3, H), (2 * G3, IN), (2, G3), (2 * G3,))
+                ws = _wgrad_ws(dx.device, (TB + 255) // 256, 2 * G3,
+                               max(IN, 128))
+                ext.gru_wgrads(dhg, dxg_cat, hseq, x_bf.reshape(TB, IN),
+                               ws, ws, du, dw_ih, dbhh, db_ih)
                 side = torch.cuda.current_stream()
                 for t in (hseq, dhg, dxg, x_bf):
                     t.record_stream(side)
@@ -373,7 +417,14 @@ class HeadFn(torch.autograd.Function):
             ev.record(torch.cuda.current_stream())
 
             def deferred():
-                dw4, db4 = wgrads()
+                dw4, db4 = _wgrad_bufs(id(w4), dseq.device,
+                                       (C.NUM_CLASSES, 2 * C.HIDDEN_SIZE),
+                                       (C.NUM_CLASSES,))
+                ws = _wgrad_ws(dseq.device, (T * B + 255) // 256,
+                               2 * 3 * C.HIDDEN_SIZE,
+                               max(2 * C.HIDDEN_SIZE, 128))
+                ext.head_wgrads(dl, seq.reshape(T * B, 2 * C.HIDDEN_SIZE),
+                                ws, dw4, db4)
                 side = torch.cuda.current_stream()
                 for t in (dl, seq):
                     t.record_stream(side)
