@@ -71,7 +71,7 @@ void atb_splitk_ld(const void* A, int lda, const void* B, int ldb,
                    float* ws, float* C, int M, int N, int K,
                    hipStream_t stream);
 void colsum_f32(const void* X, int ldx, int K, int N, float* out,
-                hipStream_t stream);
+                float* part, hipStream_t stream);
 void front_de(const uint8_t* ids, const void* dt1g, const void* w1t_g,
               float* de, int B, uint32_t seed, float keep, hipStream_t stream,
               unsigned long long* timing, uint32_t dbg,
@@ -467,10 +467,11 @@ void gru_wgrads(torch::Tensor dhg,   // (2, T, B, 384) bf16, dir-major
     rk::atb_splitk_ld(dxg.data_ptr(), 2 * G3, x.data_ptr(), IN,
                       ws_w.data_ptr<float>(), dw.data_ptr<float>(), 2 * G3,
                       IN, TB, s);
-    rk::colsum_f32(dhg_f, G3, TB, G3, dbhh_p, s);
-    rk::colsum_f32(dhg_r, G3, TB, G3, dbhh_p + G3, s);
+    float* part = ws_u.data_ptr<float>();  // scratch; stream-ordered reuse
+    rk::colsum_f32(dhg_f, G3, TB, G3, dbhh_p, part, s);
+    rk::colsum_f32(dhg_r, G3, TB, G3, dbhh_p + G3, part, s);
     rk::colsum_f32(dxg.data_ptr(), 2 * G3, TB, 2 * G3,
-                   dbih.data_ptr<float>(), s);
+                   dbih.data_ptr<float>(), part, s);
 }
 
 // head weight gradients: dw4 = dl^T x seq, db4 = column sums of dl
@@ -487,7 +488,8 @@ void head_wgrads(torch::Tensor dl,    // (T*B, 5) bf16
     rk::atb_splitk_ld(dl.data_ptr(), M, seq.data_ptr(), N,
                       ws.data_ptr<float>(), dw4.data_ptr<float>(), M, N, TB,
                       s);
-    rk::colsum_f32(dl.data_ptr(), M, TB, M, db4.data_ptr<float>(), s);
+    rk::colsum_f32(dl.data_ptr(), M, TB, M, db4.data_ptr<float>(),
+                   ws.data_ptr<float>(), s);
 }
 
 // fused train front bwd: -> (de, dw1, db1, dw2, db2) fp32

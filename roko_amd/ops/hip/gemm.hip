@@ -682,21 +682,34 @@ __global__ __launch_bounds__(256, 4) void colsum_kernel(
         for (int q = 0; q < 8; ++q) acc[q] += v[q];
     }
     for (; k < k1; ++k) acc[0] += bf2f(X[(size_t)k * ldx + n]);
-    const float total = ((acc[0] + acc[1]) + (acc[2] + acc[3])) +
-                        ((acc[4] + acc[5]) + (acc[6] + acc[7]));
-    atomicAdd(&out[n], total);
+    // plain store of this k-split's partial (no memset, no atomics: a
+    // hipMemsetAsync enqueued just before front_bwd stretched to 659 us
+    // of CU starvation — tr6 trace)
+    out[(size_t)blockIdx.y * N + n] = ((acc[0] + acc[1]) + (acc[2] + acc[3])) +
+                                      ((acc[4] + acc[5]) + (acc[6] + acc[7]));
+}
+
+__global__ __launch_bounds__(256, 4) void colsum_reduce_kernel(
+    const float* __restrict__ part, float* __restrict__ out, int N,
+    int ksp) {
+    const int n = blockIdx.x * 256 + threadIdx.x;
+    if (n >= N) return;
+    float acc = 0.f;
+    for (int q = 0; q < ksp; ++q) acc += part[(size_t)q * N + n];
+    out[n] = acc;
 }
 
 }  // namespace colsum
 
 void colsum_f32(const void* X, int ldx, int K, int N, float* out,
-                hipStream_t stream) {
-    (void)hipMemsetAsync(out, 0, sizeof(float) * N, stream);
+                float* part, hipStream_t stream) {
     const int KSP = 32;
     const int ksplit = (K + KSP - 1) / KSP;
     hipLaunchKernelGGL(colsum::colsum_kernel,
                        dim3((N + 255) / 256, KSP), dim3(256), 0, stream,
-                       static_cast<const bf16*>(X), out, ldx, K, N, ksplit);
+                       static_cast<const bf16*>(X), part, ldx, K, N, ksplit);
+    hipLaunchKernelGGL(colsum::colsum_reduce_kernel, dim3((N + 255) / 256),
+                       dim3(256), 0, stream, part, out, N, KSP);
 }
 
 }  // namespace rk
