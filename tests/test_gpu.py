@@ -508,10 +508,11 @@ def test_deferred_weight_grads_match_inline():
     assert abs(l0 - l1) < 1e-5, (l0, l1)
     for n in g0:
         ref, got = g0[n].float(), g1[n].float()
-        # 2e-3 relative: the two paths reduce bf16 inputs in different
-        # orders (hipBLASLt GEMV vs split-K colsum), so bias sums differ
-        # at the bf16-ulp scale of the summands
-        tol = 1e-4 + 2e-3 * ref.abs().max()
+        # 4e-3 relative: the two paths reduce bf16 inputs in different
+        # orders (hipBLASLt GEMV vs split-K colsum), so 11520-term bias
+        # sums differ at the bf16-ulp scale of the summands; real layout
+        # bugs produce O(1) relative errors, far above this
+        tol = 2e-4 + 4e-3 * ref.abs().max()
         assert (ref - got).abs().max() <= tol, (
             n, float((ref - got).abs().max()))
 
