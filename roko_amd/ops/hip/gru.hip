@@ -16,8 +16,9 @@
 //     separate and arbitrate between co-resident waves, so one wave's
 //     dependent gate math overlaps its partner's MFMAs;
 //   * U = weight_hh^T fragments are loaded ONCE into registers (12 bf16x8
-//     per wave) and reused for all T steps; per-step work is 24
-//     v_mfma_f32_16x16x32_bf16 per wave + fused sigmoid/tanh gate math;
+//     per wave) and reused for all T steps; per-step work is 12 (MBT=16)
+//     or 24 (MBT=32) v_mfma_f32_16x16x32_bf16 per wave + fused
+//     sigmoid/tanh gate math;
 //   * the hidden-state mirror is double-buffered in LDS, so each step needs
 //     exactly ONE __syncthreads(): MFMAs read h[cur] while gate math writes
 //     h[cur^1]; the barrier publishes h[cur^1] for the next step;

@@ -94,3 +94,21 @@ def test_select_fused_path_logic(monkeypatch):
     assert not T._select_fused_path(gpu, TrainConfig(batch_size=100))  # %32
     monkeypatch.setenv("ROKO_TRAIN_PATH", "autograd")
     assert not T._select_fused_path(gpu, TrainConfig(batch_size=128))
+
+
+def test_wgrad_buffer_ping_pong():
+    """_wgrad_bufs must alternate between exactly two cached sets per key
+    (a grad attached at drain time stays valid while the NEXT step's
+    closure writes the other set)."""
+    from roko_amd.ops.train import _WGRAD_BUFS, _wgrad_bufs
+
+    _WGRAD_BUFS.clear()
+    a1 = _wgrad_bufs("k1", "cpu", (2, 3), (4,))
+    b1 = _wgrad_bufs("k1", "cpu", (2, 3), (4,))
+    a2 = _wgrad_bufs("k1", "cpu", (2, 3), (4,))
+    assert a1[0].shape == (2, 3) and a1[1].shape == (4,)
+    assert a1[0] is not b1[0]          # alternates
+    assert a1[0] is a2[0]              # ...between exactly two sets
+    other = _wgrad_bufs("k2", "cpu", (2, 3), (4,))
+    assert other[0] is not a1[0] and other[0] is not b1[0]
+    _WGRAD_BUFS.clear()
