@@ -133,11 +133,12 @@ class InferencePipeline:
     therefore overlap, so the engine keeps `depth` in-flight batches on
     per-slot HIP streams. Each slot's whole forward (front kernel -> 3x
     (xg GEMM + persistent GRU kernel) -> fused head+argmax -> pinned D2H)
-    is enqueued by ONE C++ call (`ext.ServeSlot.submit`) — host submit cost
-    is a first-order serving bound on ROCm (profiles/PERF_HISTORY.md). A
-    hipGraph backend (each slot captured once, replayed per step —
-    BASELINE.json config 4's "hipGraph-captured GRU steps") is kept as
-    ROKO_SERVE=graph; measured equal throughput at higher host cost.
+    is enqueued by ONE C++ call (`ext.ServeSlot.submit`); for batches
+    >= 256 the slot auto-captures its kernel sequence into a RAW hipGraph
+    (BASELINE.json config 4's "hipGraph-captured GRU steps") — a measured
+    win at large batch and a measured loss at b=128, both A/B'd
+    (profiles/PERF_HISTORY.md; ROKO_GSLOT overrides). The older
+    torch-level graph backend is kept as ROKO_SERVE=graph for A/B.
 
     Every submitted batch runs the full model; nothing is cached or skipped
     — pipelining only overlaps independent batches, as a serving deployment
