@@ -41,3 +41,37 @@ def test_features_worker_failure_is_fenced(tmp_path, tiny_assembly):
     )
     assert n == 0
     assert any("FAILED" in str(l) or "WARNING" in str(l) for l in logs)
+
+
+def test_meter_jsonl_reports(monkeypatch):
+    """Meter emits valid JSONL with totals + windowed rates and a final
+    record on close (the serving/train CLIs' observability channel)."""
+    import io
+    import json
+
+    from roko_amd.utils.metrics import Meter
+
+    buf = io.StringIO()
+    m = Meter("teststage", report_every=0.0, stream=buf, rank=3)
+    m.add(windows=128, bases=3840)
+    m.add(windows=64)
+    final = m.close()
+    lines = [json.loads(l) for l in buf.getvalue().strip().splitlines()]
+    assert len(lines) >= 2
+    assert all(r["stage"] == "teststage" and r["rank"] == 3 for r in lines)
+    assert lines[-1]["final"] is True
+    assert lines[-1]["windows_total"] == 192.0
+    assert lines[-1]["bases_total"] == 3840.0
+    assert final["windows_per_s_avg"] > 0
+
+
+def test_env_world_parsing(monkeypatch):
+    from roko_amd.parallel.ddp import env_world
+
+    for k in ("RANK", "LOCAL_RANK", "WORLD_SIZE"):
+        monkeypatch.delenv(k, raising=False)
+    assert env_world() == (0, 0, 1)
+    monkeypatch.setenv("RANK", "5")
+    monkeypatch.setenv("LOCAL_RANK", "1")
+    monkeypatch.setenv("WORLD_SIZE", "8")
+    assert env_world() == (5, 1, 8)
