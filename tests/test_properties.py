@@ -92,3 +92,19 @@ def test_align_stats_property(a, b):
     # symmetry of the distance (ins/del swap)
     s2 = pileup_ext().align_stats(b, a, band=max(len(a), len(b)))
     assert s2["edit_distance"] == s["edit_distance"]
+
+
+@given(st.text(alphabet="ACGT", min_size=1, max_size=80),
+       st.text(alphabet="ACGT", min_size=1, max_size=80))
+@settings(max_examples=40, deadline=None)
+def test_align_stats_metric_properties(a, b):
+    """Edit distance is symmetric, zero iff equal, and bounded by the
+    length difference below and max length above."""
+    from roko_amd.ops import pileup_ext
+
+    px = pileup_ext()
+    d_ab = px.align_stats(a, b, band=96)["edit_distance"]
+    d_ba = px.align_stats(b, a, band=96)["edit_distance"]
+    assert d_ab == d_ba
+    assert (d_ab == 0) == (a == b)
+    assert abs(len(a) - len(b)) <= d_ab <= max(len(a), len(b))
