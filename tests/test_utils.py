@@ -75,3 +75,18 @@ def test_env_world_parsing(monkeypatch):
     monkeypatch.setenv("LOCAL_RANK", "1")
     monkeypatch.setenv("WORLD_SIZE", "8")
     assert env_world() == (5, 1, 8)
+
+
+def test_fasta_roundtrip(tmp_path):
+    """FASTA writer/reader round-trip incl. line wrapping and multiple
+    contigs (the inference output path)."""
+    from roko_amd.io.fasta import read_fasta, write_fasta
+
+    entries = [("ctg1 description here", "ACGT" * 50),
+               ("ctg2", "A"), ("ctg3", "GATTACA" * 123)]
+    p = str(tmp_path / "x.fasta")
+    write_fasta(p, entries)
+    back = list(read_fasta(p))  # read_fasta is a generator
+    assert [n.split()[0] for n, _ in entries] == [n.split()[0]
+                                                 for n, _ in back]
+    assert [s for _, s in entries] == [s for _, s in back]
