@@ -90,3 +90,29 @@ def test_fasta_roundtrip(tmp_path):
     assert [n.split()[0] for n, _ in entries] == [n.split()[0]
                                                  for n, _ in back]
     assert [s for _, s in entries] == [s for _, s in back]
+
+
+def test_reference_contract_constants():
+    """The frozen interface contract with the reference:
+    window geometry, encodings, and model dims determine the .pth
+    checkpoint format and RKW/HDF5 schema — any drift silently breaks
+    reference interop, so pin them."""
+    from roko_amd import config as C
+
+    assert (C.WINDOW_ROWS, C.WINDOW_COLS, C.WINDOW_STRIDE) == (200, 90, 30)
+    assert C.MAX_INS == 3
+    assert (C.REGION_SIZE, C.REGION_OVERLAP) == (100_000, 300)
+    assert C.NUM_BASE_IDS == 12       # base 0-5 fwd, +6 reverse
+    assert C.NUM_CLASSES == 5         # ACGT*
+    assert (C.EMBED_DIM, C.IN_SIZE) == (50, 500)
+    assert (C.HIDDEN_SIZE, C.NUM_LAYERS) == (128, 3)
+
+    # checkpoint key map matches the reference module names exactly
+    from roko_amd.model import RokoModel
+    keys = set(RokoModel().state_dict().keys())
+    assert {"embedding.weight", "fc1.weight", "fc1.bias", "fc2.weight",
+            "fc2.bias", "fc4.weight", "fc4.bias"} <= keys
+    for l in range(3):
+        for side in ("", "_reverse"):
+            for part in ("weight_ih", "weight_hh", "bias_ih", "bias_hh"):
+                assert f"gru.{part}_l{l}{side}" in keys
